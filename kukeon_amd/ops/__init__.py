@@ -1,0 +1,82 @@
+"""Device-op dispatch for kukeon_amd.
+
+On a CUDA (ROCm) device every op is the hand-written gfx950 HIP kernel from
+``kukeon_amd._C`` — if the extension is missing there we raise instead of
+falling back, so a GPU run can never silently use an eager path.  On CPU the
+fp32 reference implementations run (tests, GPU-less control-plane hosts).
+"""
+from __future__ import annotations
+
+import torch
+
+from . import reference
+
+_C = None
+_IMPORT_ERROR: Exception | None = None
+try:
+    from kukeon_amd import _C  # type: ignore
+except Exception as e:  # pragma: no cover - exercised only on broken installs
+    _IMPORT_ERROR = e
+
+
+def native_available() -> bool:
+    return _C is not None
+
+
+def _native():
+    if _C is None:
+        raise RuntimeError(
+            "kukeon_amd._C (gfx950 HIP extension) is not built but an op was "
+            "called on a CUDA device. Run `python -m kukeon_amd.ops.build` "
+            f"(import error: {_IMPORT_ERROR})")
+    return _C
+
+
+def _impl(t: torch.Tensor):
+    return _native() if t.is_cuda else reference
+
+
+def rmsnorm(out, input, weight, eps: float) -> None:
+    _impl(input).rmsnorm(out, input, weight, eps)
+
+
+def fused_add_rmsnorm(input, residual, weight, eps: float) -> None:
+    _impl(input).fused_add_rmsnorm(input, residual, weight, eps)
+
+
+def silu_mul(out, gate_up) -> None:
+    _impl(gate_up).silu_mul(out, gate_up)
+
+
+def rope_kv_append(qkv, k_cache, v_cache, cos_sin, positions, slot_mapping,
+                   num_q_heads: int, num_kv_heads: int, head_dim: int) -> None:
+    _impl(qkv).rope_kv_append(qkv, k_cache, v_cache, cos_sin, positions,
+                              slot_mapping, num_q_heads, num_kv_heads,
+                              head_dim)
+
+
+def paged_attention(out, q, k_cache, v_cache, block_table, seq_lens,
+                    q_offset: int, num_splits: int, scale: float,
+                    tmp_out, tmp_ml) -> None:
+    _impl(q).paged_attention(out, q, k_cache, v_cache, block_table, seq_lens,
+                             q_offset, num_splits, scale, tmp_out, tmp_ml)
+
+
+def prefill_attention(out, q, k_cache, v_cache, block_table, seq_lens,
+                      q_starts, qb_seq, qb_start, q_offset: int,
+                      scale: float) -> None:
+    _impl(q).prefill_attention(out, q, k_cache, v_cache, block_table,
+                               seq_lens, q_starts, qb_seq, qb_start, q_offset,
+                               scale)
+
+
+def sample(tokens, logits, temps, top_k, top_p, seed, workspace) -> None:
+    _impl(logits).sample(tokens, logits, temps, top_k, top_p, seed, workspace)
+
+
+def moe_gather_tokens(out, input, row_map) -> None:
+    _impl(input).moe_gather_tokens(out, input, row_map)
+
+
+def moe_scatter_tokens(out, input, inv_map, weights, top_k: int) -> None:
+    _impl(input).moe_scatter_tokens(out, input, inv_map, weights, top_k)

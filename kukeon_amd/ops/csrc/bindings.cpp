@@ -1,0 +1,53 @@
+// Python bindings for the kukeon_amd gfx950 kernel library.
+#include <torch/extension.h>
+
+namespace kukeon {
+void rmsnorm(torch::Tensor out, torch::Tensor input, torch::Tensor weight,
+             double eps);
+void fused_add_rmsnorm(torch::Tensor input, torch::Tensor residual,
+                       torch::Tensor weight, double eps);
+void silu_mul(torch::Tensor out, torch::Tensor gate_up);
+void rope_kv_append(torch::Tensor qkv, torch::Tensor k_cache,
+                    torch::Tensor v_cache, torch::Tensor cos_sin,
+                    torch::Tensor positions, torch::Tensor slot_mapping,
+                    int64_t num_q_heads, int64_t num_kv_heads,
+                    int64_t head_dim);
+void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                     torch::Tensor v_cache, torch::Tensor block_table,
+                     torch::Tensor seq_lens, int64_t q_offset,
+                     int64_t num_splits, double scale, torch::Tensor tmp_out,
+                     torch::Tensor tmp_ml);
+void prefill_attention(torch::Tensor out, torch::Tensor q,
+                       torch::Tensor k_cache, torch::Tensor v_cache,
+                       torch::Tensor block_table, torch::Tensor seq_lens,
+                       torch::Tensor q_starts, torch::Tensor qb_seq,
+                       torch::Tensor qb_start, int64_t q_offset, double scale);
+void mfma_probe(torch::Tensor out, torch::Tensor a, torch::Tensor b);
+void sample(torch::Tensor tokens, torch::Tensor logits, torch::Tensor temps,
+            torch::Tensor top_k, torch::Tensor top_p, torch::Tensor seed,
+            torch::Tensor workspace);
+void moe_gather_tokens(torch::Tensor out, torch::Tensor input,
+                       torch::Tensor row_map);
+void moe_scatter_tokens(torch::Tensor out, torch::Tensor input,
+                        torch::Tensor row_map, torch::Tensor weights,
+                        int64_t top_k);
+}  // namespace kukeon
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &kukeon::rmsnorm, "RMSNorm (bf16, gfx950)");
+  m.def("fused_add_rmsnorm", &kukeon::fused_add_rmsnorm,
+        "residual += input; input = rmsnorm(residual)");
+  m.def("silu_mul", &kukeon::silu_mul, "out = silu(gate)*up from fused [T,2I]");
+  m.def("rope_kv_append", &kukeon::rope_kv_append,
+        "in-place NEOX RoPE on fused qkv + paged KV append");
+  m.def("paged_attention", &kukeon::paged_attention,
+        "decode paged attention (GQA, split-KV)");
+  m.def("prefill_attention", &kukeon::prefill_attention,
+        "varlen causal paged prefill attention (MFMA)");
+  m.def("mfma_probe", &kukeon::mfma_probe,
+        "32x32x16 bf16 MFMA fragment-layout probe");
+  m.def("sample", &kukeon::sample, "top-k/top-p/temperature sampling");
+  m.def("moe_gather_tokens", &kukeon::moe_gather_tokens, "MoE permute");
+  m.def("moe_scatter_tokens", &kukeon::moe_scatter_tokens,
+        "MoE unpermute + weighted combine");
+}

@@ -1,0 +1,230 @@
+"""GPU numerics tests: every gfx950 HIP kernel vs the plain-PyTorch fp32
+reference of the same op (tests/conftest registers the `gpu` marker)."""
+import pytest
+import torch
+
+import kukeon_amd.ops as ops
+from kukeon_amd.ops import reference
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():
+    pytest.skip("needs MI355X", allow_module_level=True)
+
+DEV = "cuda:0"
+
+
+def test_native_loaded():
+    # on a GPU box the HIP extension must be present — no silent fallback
+    assert ops.native_available()
+
+
+def test_mfma_probe_layout():
+    from kukeon_amd import _C
+    torch.manual_seed(0)
+    # asymmetric inputs so operand/output transposes can't pass (guide G9)
+    a = (torch.randn(32, 16) * torch.linspace(0.2, 2.0, 16)).bfloat16().to(DEV)
+    b = (torch.randn(16, 32) * torch.linspace(-1.5, 1.5, 32)).bfloat16().to(DEV)
+    out = torch.empty(32, 32, dtype=torch.float32, device=DEV)
+    _C.mfma_probe(out, a, b)
+    ref = a.float() @ b.float()
+    torch.testing.assert_close(out.cpu(), ref.cpu(), rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("T,H", [(1, 4096), (64, 4096), (7, 8192), (3, 5120)])
+def test_rmsnorm(T, H):
+    torch.manual_seed(1)
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    out = torch.empty_like(x)
+    ops.rmsnorm(out, x, w, 1e-5)
+    ref = torch.empty_like(x.cpu())
+    reference.rmsnorm(ref, x.cpu(), w.cpu(), 1e-5)
+    torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=2e-2,
+                               atol=2e-2)
+
+
+def test_fused_add_rmsnorm():
+    torch.manual_seed(2)
+    T, H = 33, 4096
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    res = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    x_ref, res_ref = x.cpu().clone(), res.cpu().clone()
+    ops.fused_add_rmsnorm(x, res, w, 1e-5)
+    reference.fused_add_rmsnorm(x_ref, res_ref, w.cpu(), 1e-5)
+    torch.testing.assert_close(res.cpu().float(), res_ref.float(), rtol=2e-2,
+                               atol=2e-2)
+    torch.testing.assert_close(x.cpu().float(), x_ref.float(), rtol=2e-2,
+                               atol=3e-2)
+
+
+def test_silu_mul():
+    torch.manual_seed(3)
+    T, I = 17, 14336
+    gu = torch.randn(T, 2 * I, dtype=torch.bfloat16, device=DEV)
+    out = torch.empty(T, I, dtype=torch.bfloat16, device=DEV)
+    ops.silu_mul(out, gu)
+    ref = torch.empty(T, I, dtype=torch.bfloat16)
+    reference.silu_mul(ref, gu.cpu())
+    torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=2e-2,
+                               atol=2e-2)
+
+
+def _make_cache(nb, hk, bs, d):
+    k = torch.randn(nb, hk, bs, d, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(nb, hk, bs, d, dtype=torch.bfloat16, device=DEV)
+    return k, v
+
+
+def test_rope_kv_append():
+    torch.manual_seed(4)
+    T, Hq, Hk, D, BS, NB = 9, 8, 2, 128, 16, 8
+    qkv = torch.randn(T, (Hq + 2 * Hk) * D, dtype=torch.bfloat16, device=DEV)
+    kc, vc = _make_cache(NB, Hk, BS, D)
+    maxpos = 64
+    inv = 1.0 / (500000.0 ** (torch.arange(0, D, 2).float() / D))
+    t = torch.arange(maxpos).float()
+    fr = torch.outer(t, inv)
+    cos_sin = torch.cat([fr.cos(), fr.sin()], dim=1).to(DEV)
+    pos = torch.randint(0, maxpos, (T,), dtype=torch.int32, device=DEV)
+    slots = torch.randperm(NB * BS)[:T].to(torch.int32).to(DEV)
+
+    qkv_ref = qkv.cpu().clone()
+    kc_ref, vc_ref = kc.cpu().clone(), vc.cpu().clone()
+    ops.rope_kv_append(qkv, kc, vc, cos_sin, pos, slots, Hq, Hk, D)
+    reference.rope_kv_append(qkv_ref, kc_ref, vc_ref, cos_sin.cpu(), pos.cpu(),
+                             slots.cpu(), Hq, Hk, D)
+    torch.testing.assert_close(qkv.cpu().float(), qkv_ref.float(), rtol=2e-2,
+                               atol=2e-2)
+    torch.testing.assert_close(kc.cpu().float(), kc_ref.float(), rtol=2e-2,
+                               atol=2e-2)
+    torch.testing.assert_close(vc.cpu().float(), vc_ref.float(), rtol=0, atol=0)
+
+
+@pytest.mark.parametrize("G,splits", [(4, 1), (4, 4), (8, 1), (8, 3)])
+def test_paged_attention(G, splits):
+    torch.manual_seed(5)
+    B, Hk, D, BS = 5, 2, 128, 16
+    Hq = G * Hk
+    ctxs = [1, 16, 17, 100, 250]
+    NB = sum((c + BS - 1) // BS for c in ctxs) + 4
+    kc, vc = _make_cache(NB, Hk, BS, D)
+    maxb = max((c + BS - 1) // BS for c in ctxs)
+    bt = torch.zeros(B, maxb, dtype=torch.int32, device=DEV)
+    nxt = 0
+    for b, c in enumerate(ctxs):
+        n = (c + BS - 1) // BS
+        bt[b, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    seq_lens = torch.tensor(ctxs, dtype=torch.int32, device=DEV)
+    q = torch.randn(B, Hq * D, dtype=torch.bfloat16, device=DEV)
+    out = torch.empty(B, Hq * D, dtype=torch.bfloat16, device=DEV)
+    scale = D ** -0.5
+    tmp_out = torch.zeros(B, Hq, splits, D, dtype=torch.float32, device=DEV)
+    tmp_ml = torch.zeros(B, Hq, splits, 2, dtype=torch.float32, device=DEV)
+    ops.paged_attention(out, q, kc, vc, bt, seq_lens, 0, splits, scale,
+                        tmp_out, tmp_ml)
+    ref = torch.empty(B, Hq * D, dtype=torch.bfloat16)
+    reference.paged_attention(ref, q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                              seq_lens.cpu(), 0, 1, scale)
+    torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=2e-2,
+                               atol=2e-2)
+
+
+def test_paged_attention_q_offset():
+    """q passed as a row of a fused qkv tensor with stride + offset."""
+    torch.manual_seed(6)
+    B, Hq, Hk, D, BS = 3, 8, 2, 128, 16
+    ctxs = [40, 8, 64]
+    NB = 12
+    kc, vc = _make_cache(NB, Hk, BS, D)
+    bt = torch.arange(NB, dtype=torch.int32, device=DEV).reshape(3, 4)
+    seq_lens = torch.tensor(ctxs, dtype=torch.int32, device=DEV)
+    qkv = torch.randn(B, (Hq + 2 * Hk) * D, dtype=torch.bfloat16, device=DEV)
+    out = torch.empty(B, Hq * D, dtype=torch.bfloat16, device=DEV)
+    t1 = torch.zeros(1, dtype=torch.float32, device=DEV)
+    ops.paged_attention(out, qkv, kc, vc, bt, seq_lens, 0, 1, D ** -0.5, t1, t1)
+    ref = torch.empty(B, Hq * D, dtype=torch.bfloat16)
+    reference.paged_attention(ref, qkv.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                              seq_lens.cpu(), 0, 1, D ** -0.5)
+    torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=2e-2,
+                               atol=2e-2)
+
+
+def test_sample_greedy_exact():
+    torch.manual_seed(7)
+    B, V = 8, 128256
+    logits = torch.randn(B, V, dtype=torch.bfloat16, device=DEV)
+    tokens = torch.zeros(B, dtype=torch.int32, device=DEV)
+    temps = torch.zeros(B, dtype=torch.float32, device=DEV)
+    tk = torch.zeros(B, dtype=torch.int32, device=DEV)
+    tp = torch.ones(B, dtype=torch.float32, device=DEV)
+    seed = torch.zeros(1, dtype=torch.int64, device=DEV)
+    ws = torch.zeros(B, 4, dtype=torch.float32, device=DEV)
+    ops.sample(tokens, logits, temps, tk, tp, seed, ws)
+    ref = logits.float().argmax(dim=-1).to(torch.int32)
+    assert torch.equal(tokens.cpu(), ref.cpu())
+
+
+def test_sample_topk_membership():
+    torch.manual_seed(8)
+    B, V, K = 4, 50000, 20
+    logits = torch.randn(B, V, dtype=torch.bfloat16, device=DEV) * 3
+    temps = torch.full((B,), 0.8, dtype=torch.float32, device=DEV)
+    tk = torch.full((B,), K, dtype=torch.int32, device=DEV)
+    tp = torch.ones(B, dtype=torch.float32, device=DEV)
+    seed = torch.zeros(1, dtype=torch.int64, device=DEV)
+    ws = torch.zeros(B, 4, dtype=torch.float32, device=DEV)
+    tokens = torch.zeros(B, dtype=torch.int32, device=DEV)
+    # top-k sets computed on bf16-rounded logits (ties at 24-bit granularity
+    # can admit a couple extra members — allow K + small slack)
+    topsets = [set(torch.topk(logits[b].float(), K + 4).indices.tolist())
+               for b in range(B)]
+    for _ in range(50):
+        ops.sample(tokens, logits, temps, tk, tp, seed, ws)
+        for b in range(B):
+            assert int(tokens[b]) in topsets[b]
+
+
+def test_sample_topp_membership():
+    torch.manual_seed(9)
+    B, V = 4, 50000
+    logits = (torch.randn(B, V) * 4).bfloat16().to(DEV)
+    temps = torch.ones(B, dtype=torch.float32, device=DEV)
+    tk = torch.zeros(B, dtype=torch.int32, device=DEV)
+    tp = torch.full((B,), 0.7, dtype=torch.float32, device=DEV)
+    seed = torch.zeros(1, dtype=torch.int64, device=DEV)
+    ws = torch.zeros(B, 4, dtype=torch.float32, device=DEV)
+    tokens = torch.zeros(B, dtype=torch.int32, device=DEV)
+    allowed = []
+    for b in range(B):
+        probs = torch.softmax(logits[b].float(), -1)
+        srt, idx = torch.sort(probs, descending=True)
+        cum = torch.cumsum(srt, 0)
+        cut = int(torch.searchsorted(cum, 0.7).clamp(max=V - 1))
+        allowed.append(set(idx[: cut + 8].tolist()))  # small slack for ties
+    for _ in range(50):
+        ops.sample(tokens, logits, temps, tk, tp, seed, ws)
+        for b in range(B):
+            assert int(tokens[b]) in allowed[b]
+
+
+def test_moe_gather_scatter():
+    torch.manual_seed(10)
+    T, H, K = 13, 4096, 2
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    row_map = torch.randint(0, T, (T * K,), dtype=torch.int32, device=DEV)
+    out = torch.empty(T * K, H, dtype=torch.bfloat16, device=DEV)
+    ops.moe_gather_tokens(out, x, row_map)
+    assert torch.equal(out.cpu(), x.cpu()[row_map.cpu().long()])
+
+    e = torch.randn(T * K, H, dtype=torch.bfloat16, device=DEV)
+    inv = torch.randperm(T * K, device=DEV).to(torch.int32).reshape(T, K)
+    w = torch.rand(T, K, dtype=torch.float32, device=DEV)
+    out2 = torch.empty(T, H, dtype=torch.bfloat16, device=DEV)
+    ops.moe_scatter_tokens(out2, e, inv, w, K)
+    ref = torch.empty(T, H, dtype=torch.bfloat16)
+    reference.moe_scatter_tokens(ref, e.cpu(), inv.cpu(), w.cpu(), K)
+    torch.testing.assert_close(out2.cpu().float(), ref.float(), rtol=2e-2,
+                               atol=2e-2)
