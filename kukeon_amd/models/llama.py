@@ -1,0 +1,165 @@
+"""Llama-family decoder (Llama-3-8B / 70B shapes) built on the kukeon_amd
+gfx950 op set: fused qkv/gate_up GEMMs (hipBLASLt via F.linear), fused
+RoPE+KV-append, paged GQA attention, fused add+RMSNorm and SwiGLU kernels.
+
+Tensor parallelism: q/k/v and gate/up are column-sharded (head-aligned),
+o_proj and down_proj row-sharded with one RCCL all-reduce each over xGMI;
+embed/lm_head are replicated so every rank samples identically with no
+collective in the sampling path.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from kukeon_amd import ops
+from kukeon_amd import parallel
+from kukeon_amd.engine.config import ModelConfig
+
+
+@dataclass
+class AttnMeta:
+    """Per-forward attention metadata (all tensors on the model device)."""
+    mode: str                       # "prefill" | "decode"
+    positions: torch.Tensor         # [T] int32
+    slot_mapping: torch.Tensor      # [T] int32
+    block_table: torch.Tensor       # [nseq, max_blocks] int32
+    seq_lens: torch.Tensor          # [nseq] int32
+    q_starts: Optional[torch.Tensor] = None   # [nseq] int32 (prefill)
+    qb_seq: Optional[torch.Tensor] = None     # [nqb] int32 (prefill grid)
+    qb_start: Optional[torch.Tensor] = None   # [nqb] int32
+    num_splits: int = 1
+    tmp_out: Optional[torch.Tensor] = None    # decode split-KV workspaces
+    tmp_ml: Optional[torch.Tensor] = None
+
+
+def _init_weight(shape, device, std=0.02, seed=None):
+    w = torch.empty(shape, dtype=torch.bfloat16, device=device)
+    w.normal_(0.0, std)
+    return nn.Parameter(w, requires_grad=False)
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: ModelConfig, layer_idx: int, device):
+        super().__init__()
+        tp = parallel.tp_size()
+        assert cfg.num_q_heads % tp == 0 and cfg.num_kv_heads % tp == 0, (
+            "TP degree must divide q and kv head counts")
+        self.layer_idx = layer_idx
+        self.hq = cfg.num_q_heads // tp
+        self.hk = cfg.num_kv_heads // tp
+        self.d = cfg.head_dim
+        self.scale = cfg.head_dim ** -0.5
+        qkv_out = (self.hq + 2 * self.hk) * self.d
+        self.qkv_w = _init_weight((qkv_out, cfg.hidden_size), device)
+        self.o_w = _init_weight((cfg.hidden_size, self.hq * self.d), device)
+
+    def forward(self, x: torch.Tensor, cos_sin: torch.Tensor,
+                kc: torch.Tensor, vc: torch.Tensor, meta: AttnMeta):
+        qkv = F.linear(x, self.qkv_w)
+        ops.rope_kv_append(qkv, kc, vc, cos_sin, meta.positions,
+                           meta.slot_mapping, self.hq, self.hk, self.d)
+        out = torch.empty(x.shape[0], self.hq * self.d, dtype=x.dtype,
+                          device=x.device)
+        if meta.mode == "decode":
+            ops.paged_attention(out, qkv, kc, vc, meta.block_table,
+                                meta.seq_lens, 0, meta.num_splits, self.scale,
+                                meta.tmp_out, meta.tmp_ml)
+        else:
+            ops.prefill_attention(out, qkv, kc, vc, meta.block_table,
+                                  meta.seq_lens, meta.q_starts, meta.qb_seq,
+                                  meta.qb_start, 0, self.scale)
+        o = F.linear(out, self.o_w)
+        return parallel.tp_all_reduce(o)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: ModelConfig, device):
+        super().__init__()
+        tp = parallel.tp_size()
+        assert cfg.intermediate_size % tp == 0
+        self.inter = cfg.intermediate_size // tp
+        self.gate_up_w = _init_weight((2 * self.inter, cfg.hidden_size), device)
+        self.down_w = _init_weight((cfg.hidden_size, self.inter), device)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        gu = F.linear(x, self.gate_up_w)
+        act = torch.empty(x.shape[0], self.inter, dtype=x.dtype,
+                          device=x.device)
+        ops.silu_mul(act, gu)
+        return parallel.tp_all_reduce(F.linear(act, self.down_w))
+
+
+class LlamaLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, layer_idx: int, device):
+        super().__init__()
+        self.input_norm = nn.Parameter(
+            torch.ones(cfg.hidden_size, dtype=torch.bfloat16, device=device),
+            requires_grad=False)
+        self.post_norm = nn.Parameter(
+            torch.ones(cfg.hidden_size, dtype=torch.bfloat16, device=device),
+            requires_grad=False)
+        self.attn = LlamaAttention(cfg, layer_idx, device)
+        self.mlp = LlamaMLP(cfg, device)
+        self.eps = cfg.rms_eps
+
+    def forward(self, x, residual, cos_sin, kc, vc, meta):
+        if residual is None:
+            residual = x
+            h = torch.empty_like(x)
+            ops.rmsnorm(h, x, self.input_norm, self.eps)
+        else:
+            ops.fused_add_rmsnorm(x, residual, self.input_norm, self.eps)
+            h = x
+        h = self.attn.forward(h, cos_sin, kc, vc, meta)
+        ops.fused_add_rmsnorm(h, residual, self.post_norm, self.eps)
+        h = self.mlp.forward(h)
+        return h, residual
+
+
+class LlamaModel(nn.Module):
+    """Random-init Llama decoder for the serving engine (bench contract:
+    synthetic data / random weights — no checkpoints in this environment)."""
+
+    def __init__(self, cfg: ModelConfig, device="cuda"):
+        super().__init__()
+        self.cfg = cfg
+        self.device_ = torch.device(device)
+        torch.manual_seed(42)
+        self.embed = _init_weight((cfg.vocab_size, cfg.hidden_size),
+                                  self.device_)
+        self.layers = nn.ModuleList(
+            [LlamaLayer(cfg, i, self.device_) for i in range(cfg.num_layers)])
+        self.final_norm = nn.Parameter(
+            torch.ones(cfg.hidden_size, dtype=torch.bfloat16,
+                       device=self.device_), requires_grad=False)
+        self.lm_head = _init_weight((cfg.vocab_size, cfg.hidden_size),
+                                    self.device_)
+        self.cos_sin = self._build_rope_table().to(self.device_)
+
+    def _build_rope_table(self) -> torch.Tensor:
+        cfg = self.cfg
+        half = cfg.head_dim // 2
+        inv = 1.0 / (cfg.rope_theta ** (
+            torch.arange(0, cfg.head_dim, 2, dtype=torch.float64) /
+            cfg.head_dim))
+        t = torch.arange(cfg.max_position, dtype=torch.float64)
+        fr = torch.outer(t, inv)
+        return torch.cat([fr.cos(), fr.sin()], dim=1).float()
+
+    def forward(self, input_ids: torch.Tensor, kv_k: torch.Tensor,
+                kv_v: torch.Tensor, meta: AttnMeta) -> torch.Tensor:
+        x = F.embedding(input_ids.long(), self.embed)
+        residual = None
+        for i, layer in enumerate(self.layers):
+            x, residual = layer.forward(x, residual, self.cos_sin, kv_k[i],
+                                        kv_v[i], meta)
+        ops.fused_add_rmsnorm(x, residual, self.final_norm, self.cfg.rms_eps)
+        return x
+
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        return F.linear(hidden, self.lm_head)

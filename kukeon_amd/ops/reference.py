@@ -121,10 +121,10 @@ def prefill_attention(out: torch.Tensor, q: torch.Tensor,
     Hq = out.shape[1] // D
     G = Hq // Hk
     nseq = seq_lens.shape[0]
-    row = 0
     for s in range(nseq):
         ctx = int(seq_lens[s])
-        start = int(q_starts[s])
+        start = int(q_starts[s, 0])
+        row = int(q_starts[s, 1])
         qlen = ctx - start
         if qlen <= 0:
             continue
@@ -142,7 +142,7 @@ def prefill_attention(out: torch.Tensor, q: torch.Tensor,
             p = torch.softmax(sc, dim=-1)
             o = p @ v[:, hk]
             out[row: row + qlen].view(qlen, Hq, D)[:, h] = o.to(out.dtype)
-        row += qlen
+        pass
 
 
 def sample(tokens: torch.Tensor, logits: torch.Tensor, temps: torch.Tensor,

@@ -1,0 +1,76 @@
+"""Distributed setup: one process per GPU, torch.distributed over RCCL
+(backend "nccl" IS RCCL on ROCm) for xGMI collectives; gloo on CPU hosts.
+
+The control plane needs no collectives (single host, unix sockets); this
+package serves the data plane only: tensor parallelism for large backing
+models and the data-parallel session-sharded serving bench.
+"""
+from __future__ import annotations
+
+import os
+
+import torch
+import torch.distributed as dist
+
+_TP_GROUP = None
+_TP_RANK = 0
+_TP_SIZE = 1
+
+
+def init_distributed(backend: str | None = None) -> tuple[int, int]:
+    """Initialize from torchrun env vars; returns (rank, world_size)."""
+    if dist.is_initialized():
+        return dist.get_rank(), dist.get_world_size()
+    if "RANK" not in os.environ:
+        return 0, 1
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29500")
+    dist.init_process_group(backend=backend)
+    rank = dist.get_rank()
+    if torch.cuda.is_available():
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+    return rank, dist.get_world_size()
+
+
+def init_tensor_parallel(tp_size: int) -> None:
+    """Carve TP groups out of the world (world must be a multiple)."""
+    global _TP_GROUP, _TP_RANK, _TP_SIZE
+    _TP_SIZE = tp_size
+    if tp_size == 1 or not dist.is_initialized():
+        _TP_GROUP, _TP_RANK = None, 0
+        return
+    world = dist.get_world_size()
+    rank = dist.get_rank()
+    assert world % tp_size == 0
+    for start in range(0, world, tp_size):
+        ranks = list(range(start, start + tp_size))
+        g = dist.new_group(ranks)
+        if rank in ranks:
+            _TP_GROUP = g
+            _TP_RANK = rank - start
+    _TP_SIZE = tp_size
+
+
+def tp_rank() -> int:
+    return _TP_RANK
+
+
+def tp_size() -> int:
+    return _TP_SIZE
+
+
+def tp_group():
+    return _TP_GROUP
+
+
+def tp_all_reduce(t: torch.Tensor) -> torch.Tensor:
+    if _TP_SIZE > 1:
+        dist.all_reduce(t, group=_TP_GROUP)
+    return t
+
+
+def barrier() -> None:
+    if dist.is_initialized():
+        dist.barrier()
