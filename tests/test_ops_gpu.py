@@ -228,3 +228,68 @@ def test_moe_gather_scatter():
     reference.moe_scatter_tokens(ref, e.cpu(), inv.cpu(), w.cpu(), K)
     torch.testing.assert_close(out2.cpu().float(), ref.float(), rtol=2e-2,
                                atol=2e-2)
+
+
+def _prefill_case(ctxs_starts, Hq, Hk, seed=11):
+    """ctxs_starts: list of (total_ctx, q_start). Builds caches+q and runs
+    both impls."""
+    torch.manual_seed(seed)
+    D, BS = 128, 16
+    nseq = len(ctxs_starts)
+    NB = sum((c + BS - 1) // BS for c, _ in ctxs_starts) + 2
+    kc, vc = _make_cache(NB, Hk, BS, D)
+    maxb = max((c + BS - 1) // BS for c, _ in ctxs_starts)
+    bt = torch.zeros(nseq, maxb, dtype=torch.int32, device=DEV)
+    nxt = 0
+    for s, (c, _) in enumerate(ctxs_starts):
+        n = (c + BS - 1) // BS
+        bt[s, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    seq_lens = torch.tensor([c for c, _ in ctxs_starts], dtype=torch.int32,
+                            device=DEV)
+    rows = []
+    qs_pairs, qb_seq, qb_start = [], [], []
+    row = 0
+    for s, (c, st) in enumerate(ctxs_starts):
+        qlen = c - st
+        qs_pairs.append((st, row))
+        for qb in range(0, qlen, 32):
+            qb_seq.append(s)
+            qb_start.append(qb)
+        row += qlen
+    T = row
+    q = torch.randn(T, Hq * D, dtype=torch.bfloat16, device=DEV)
+    q_starts = torch.tensor(qs_pairs, dtype=torch.int32, device=DEV)
+    t_qb_seq = torch.tensor(qb_seq, dtype=torch.int32, device=DEV)
+    t_qb_start = torch.tensor(qb_start, dtype=torch.int32, device=DEV)
+    out = torch.empty(T, Hq * D, dtype=torch.bfloat16, device=DEV)
+    scale = D ** -0.5
+    ops.prefill_attention(out, q, kc, vc, bt, seq_lens, q_starts, t_qb_seq,
+                          t_qb_start, 0, scale)
+    ref = torch.empty(T, Hq * D, dtype=torch.bfloat16)
+    reference.prefill_attention(ref, q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                                seq_lens.cpu(), q_starts.cpu(), None, None,
+                                0, scale)
+    torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=2.5e-2,
+                               atol=2.5e-2)
+
+
+def test_prefill_single_seq_aligned():
+    _prefill_case([(128, 0)], Hq=8, Hk=2)
+
+
+def test_prefill_ragged_tails():
+    _prefill_case([(33, 0), (100, 0), (17, 0), (160, 0)], Hq=8, Hk=2)
+
+
+def test_prefill_continuation():
+    # multi-turn: queries are the tail of an existing context
+    _prefill_case([(200, 150), (90, 64), (70, 69)], Hq=8, Hk=2)
+
+
+def test_prefill_gqa8():
+    _prefill_case([(77, 0), (130, 40)], Hq=8, Hk=1)
+
+
+def test_prefill_single_token_turns():
+    _prefill_case([(40, 39), (16, 15)], Hq=4, Hk=1)
