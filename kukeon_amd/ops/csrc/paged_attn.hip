@@ -141,8 +141,9 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
         psum += __shfl_xor(psum, 8, WAVE);
         psum += __shfl_xor(psum, 16, WAVE);
         psum += __shfl_xor(psum, 32, WAVE);
-        // p is replicated over the 4 lanes of each token group
-        l[g] = l[g] * alpha + psum * 0.25f;
+        // the xor-reduce over offsets {4,8,16,32} keeps low lane bits fixed,
+        // so each token group contributes exactly once
+        l[g] = l[g] * alpha + psum;
         m[g] = mn;
         o0[g] *= alpha;
         o1[g] *= alpha;
