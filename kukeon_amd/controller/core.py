@@ -1,0 +1,989 @@
+"""The controller: verb orchestration, desired-state apply, reconcile.
+
+The capability map of the reference's internal/controller + runner (SURVEY.md
+§2.3/2.4) re-built on the process-cell runtime: provisioning (cgroups,
+per-space subnets, egress policy), cell lifecycle with idempotent start and
+markCellFailed rollback, restart policies with backoff floor + retry cap,
+AutoDelete, Session lifetime enforcement (wallClock/idleTimeout + onEnd
+persist), OutOfSync re-derivation for blueprint/config-materialized cells,
+and GPU pinning via the amdgpu allocator.
+"""
+from __future__ import annotations
+
+import base64
+import contextlib
+import logging
+import shutil
+import time
+from dataclasses import dataclass, field
+from pathlib import Path
+from typing import Dict, List, Optional, Tuple
+
+from kukeon_amd.api import errors
+from kukeon_amd.api import v1beta1 as api
+from kukeon_amd.controller import blueprint as bpmod
+from kukeon_amd.controller import diff as diffmod
+from kukeon_amd.controller import naming, parser
+from kukeon_amd.controller.subnet import SubnetAllocator
+from kukeon_amd.netpolicy import Enforcer, NoopEnforcer, build_policy
+from kukeon_amd.runtime.cgroup import CgroupManager
+from kukeon_amd.runtime.devices import (GPUAllocator, device_paths_for,
+                                        visible_devices_env)
+from kukeon_amd.runtime.process import (ROOT_CONTAINER, ProcessRuntime,
+                                        Runtime)
+from kukeon_amd.state.store import METADATA_FILE, Store
+
+log = logging.getLogger("kukeon.controller")
+
+
+def now_iso(ts: Optional[float] = None) -> str:
+    return time.strftime("%Y-%m-%dT%H:%M:%SZ",
+                         time.gmtime(ts if ts is not None else time.time()))
+
+
+def parse_iso(text: str) -> float:
+    import calendar
+    return calendar.timegm(time.strptime(text, "%Y-%m-%dT%H:%M:%SZ"))
+
+
+@dataclass
+class ResourceResult:
+    kind: str
+    name: str
+    action: str   # created | updated | unchanged | failed | recreated
+    error: str = ""
+
+
+class Controller:
+    def __init__(self, run_path: str, runtime: Optional[Runtime] = None,
+                 cgroups: Optional[CgroupManager] = None,
+                 enforcer: Optional[Enforcer] = None,
+                 gpu_devices: Optional[List[int]] = None,
+                 now_fn=time.time):
+        self.run_path = Path(run_path)
+        self.store = Store(run_path)
+        self.cgroups = cgroups or CgroupManager(enabled=False)
+        self.runtime = runtime or ProcessRuntime(self.cgroups)
+        self.enforcer = enforcer or NoopEnforcer()
+        self.subnets = SubnetAllocator(self.store)
+        self.gpus = GPUAllocator(str(self.run_path / "gpus.json"),
+                                 devices=gpu_devices)
+        self.now = now_fn
+        self._restart_state: Dict[str, Tuple[float, int]] = {}
+
+    # ==================================================================
+    # bootstrap
+    # ==================================================================
+    def bootstrap(self) -> None:
+        (self.run_path / "data").mkdir(parents=True, exist_ok=True)
+        (self.run_path / "bin").mkdir(parents=True, exist_ok=True)
+        self.cgroups.create("")
+        for realm, space, stack in (
+            (naming.DEFAULT_REALM, naming.DEFAULT_SPACE, naming.DEFAULT_STACK),
+            (naming.SYSTEM_REALM, naming.SYSTEM_SPACE, naming.SYSTEM_STACK),
+        ):
+            self.ensure_realm(realm)
+            self.ensure_space(realm, space)
+            self.ensure_stack(realm, space, stack)
+
+    def ensure_realm(self, name: str) -> api.RealmDoc:
+        try:
+            return self.get_realm(name)
+        except errors.RealmNotFound:
+            doc = api.RealmDoc(metadata=api.Metadata(name=name))
+            return self.create_realm(doc)
+
+    def ensure_space(self, realm: str, name: str) -> api.SpaceDoc:
+        try:
+            return self.get_space(realm, name)
+        except errors.SpaceNotFound:
+            doc = api.SpaceDoc(metadata=api.Metadata(name=name),
+                               spec=api.SpaceSpec(realm_id=realm))
+            return self.create_space(doc)
+
+    def ensure_stack(self, realm: str, space: str, name: str) -> api.StackDoc:
+        try:
+            return self.get_stack(realm, space, name)
+        except errors.StackNotFound:
+            doc = api.StackDoc(metadata=api.Metadata(name=name),
+                               spec=api.StackSpec(realm_id=realm,
+                                                  space_id=space))
+            return self.create_stack(doc)
+
+    # ==================================================================
+    # realm / space / stack
+    # ==================================================================
+    def create_realm(self, doc: api.RealmDoc) -> api.RealmDoc:
+        naming.validate_name(doc.metadata.name, "realm")
+        path = self.store.realm_dir(doc.metadata.name) / METADATA_FILE
+        doc.status.state = api.STATE_READY
+        doc.status.cgroup_path = "/".join(
+            self.cgroups.create(doc.metadata.name)[:1])
+        self.store.create_exclusive(path, doc.to_dict())
+        return doc
+
+    def get_realm(self, name: str) -> api.RealmDoc:
+        data = self.store.read(self.store.realm_dir(name) / METADATA_FILE)
+        if data is None:
+            raise errors.RealmNotFound(name)
+        return api.RealmDoc.from_dict(data)
+
+    def list_realms(self) -> List[api.RealmDoc]:
+        return [self.get_realm(n)
+                for n in self.store.list_children(self.store.data_root)]
+
+    def delete_realm(self, name: str, cascade: bool = False) -> None:
+        self.get_realm(name)
+        spaces = self.store.list_children(self.store.realm_dir(name))
+        if spaces and not cascade:
+            raise errors.NotEmpty(f"realm {name} has spaces: {spaces}")
+        for s in spaces:
+            self.delete_space(name, s, cascade=True)
+        self.store.delete_tree(self.store.realm_dir(name))
+        self.cgroups.delete(name)
+
+    def create_space(self, doc: api.SpaceDoc) -> api.SpaceDoc:
+        naming.validate_name(doc.metadata.name, "space")
+        self.get_realm(doc.spec.realm_id)
+        path = (self.store.space_dir(doc.spec.realm_id, doc.metadata.name) /
+                METADATA_FILE)
+        doc.status.state = api.STATE_READY
+        self.cgroups.create(f"{doc.spec.realm_id}/{doc.metadata.name}")
+        doc.status.subnet = self.subnets.allocate(doc.spec.realm_id,
+                                                  doc.metadata.name)
+        self._apply_egress(doc)
+        self.store.create_exclusive(path, doc.to_dict())
+        return doc
+
+    def _apply_egress(self, doc: api.SpaceDoc) -> None:
+        egress = doc.spec.network.egress if doc.spec.network else None
+        policy = build_policy(egress)
+        subnet = doc.status.subnet or self.subnets.lookup(
+            doc.spec.realm_id, doc.metadata.name) or ""
+        if subnet:
+            self.enforcer.apply(doc.spec.realm_id, doc.metadata.name, subnet,
+                                policy)
+
+    def get_space(self, realm: str, name: str) -> api.SpaceDoc:
+        data = self.store.read(self.store.space_dir(realm, name) /
+                               METADATA_FILE)
+        if data is None:
+            raise errors.SpaceNotFound(f"{realm}/{name}")
+        return api.SpaceDoc.from_dict(data)
+
+    def list_spaces(self, realm: str) -> List[api.SpaceDoc]:
+        return [self.get_space(realm, n)
+                for n in self.store.list_children(self.store.realm_dir(realm))]
+
+    def delete_space(self, realm: str, name: str, cascade: bool = False) -> None:
+        self.get_space(realm, name)
+        stacks = self.store.list_children(self.store.space_dir(realm, name))
+        if stacks and not cascade:
+            raise errors.NotEmpty(f"space {name} has stacks: {stacks}")
+        for st in stacks:
+            self.delete_stack(realm, name, st, cascade=True)
+        self.enforcer.remove(realm, name)
+        self.subnets.release(realm, name)
+        self.store.delete_tree(self.store.space_dir(realm, name))
+        self.cgroups.delete(f"{realm}/{name}")
+
+    def create_stack(self, doc: api.StackDoc) -> api.StackDoc:
+        naming.validate_name(doc.metadata.name, "stack")
+        self.get_space(doc.spec.realm_id, doc.spec.space_id)
+        path = (self.store.stack_dir(doc.spec.realm_id, doc.spec.space_id,
+                                     doc.metadata.name) / METADATA_FILE)
+        doc.status.state = api.STATE_READY
+        self.cgroups.create(f"{doc.spec.realm_id}/{doc.spec.space_id}/"
+                            f"{doc.metadata.name}")
+        self.store.create_exclusive(path, doc.to_dict())
+        return doc
+
+    def get_stack(self, realm: str, space: str, name: str) -> api.StackDoc:
+        data = self.store.read(self.store.stack_dir(realm, space, name) /
+                               METADATA_FILE)
+        if data is None:
+            raise errors.StackNotFound(f"{realm}/{space}/{name}")
+        return api.StackDoc.from_dict(data)
+
+    def list_stacks(self, realm: str, space: str) -> List[api.StackDoc]:
+        return [self.get_stack(realm, space, n) for n in
+                self.store.list_children(self.store.space_dir(realm, space))]
+
+    def delete_stack(self, realm: str, space: str, name: str,
+                     cascade: bool = False) -> None:
+        self.get_stack(realm, space, name)
+        cells = self.store.list_children(
+            self.store.stack_dir(realm, space, name))
+        if cells and not cascade:
+            raise errors.NotEmpty(f"stack {name} has cells: {cells}")
+        for c in cells:
+            self.delete_cell(realm, space, name, c, force=True)
+        self.store.delete_tree(self.store.stack_dir(realm, space, name))
+        self.cgroups.delete(f"{realm}/{space}/{name}")
+
+    # ==================================================================
+    # cells
+    # ==================================================================
+    def _cell_path(self, realm, space, stack, name) -> Path:
+        return self.store.cell_dir(realm, space, stack, name) / METADATA_FILE
+
+    def _cell_cgroup(self, doc: api.CellDoc) -> str:
+        s = doc.spec
+        return f"{s.realm_id}/{s.space_id}/{s.stack_id}/{doc.metadata.name}"
+
+    def create_cell(self, doc: api.CellDoc) -> api.CellDoc:
+        parser.validate_document(doc)
+        self.get_stack(doc.spec.realm_id, doc.spec.space_id, doc.spec.stack_id)
+        doc.spec.id = doc.spec.id or doc.metadata.name
+        doc.spec.root_container_id = naming.root_container_id(
+            doc.spec.space_id, doc.spec.stack_id, doc.metadata.name)
+        for c in doc.spec.containers:
+            c.id = c.id or "main"
+            c.realm_id = doc.spec.realm_id
+            c.space_id = doc.spec.space_id
+            c.stack_id = doc.spec.stack_id
+            c.cell_id = doc.metadata.name
+        doc.status.state = api.STATE_PENDING
+        path = self._cell_path(doc.spec.realm_id, doc.spec.space_id,
+                               doc.spec.stack_id, doc.metadata.name)
+        self.store.create_exclusive(path, doc.to_dict())
+        return doc
+
+    def get_cell(self, realm, space, stack, name) -> api.CellDoc:
+        data = self.store.read(self._cell_path(realm, space, stack, name))
+        if data is None:
+            raise errors.CellNotFound(f"{realm}/{space}/{stack}/{name}")
+        return api.CellDoc.from_dict(data)
+
+    def list_cells(self, realm, space, stack) -> List[api.CellDoc]:
+        return [self.get_cell(realm, space, stack, n) for n in
+                self.store.list_children(
+                    self.store.stack_dir(realm, space, stack))]
+
+    def _persist_cell(self, doc: api.CellDoc) -> None:
+        path = self._cell_path(doc.spec.realm_id, doc.spec.space_id,
+                               doc.spec.stack_id, doc.metadata.name)
+        self.store.write_cas(path, doc.to_dict(), bump=False)
+
+    def _container_env(self, doc: api.CellDoc, c: api.ContainerSpec,
+                       gpu_ids: List[int]) -> List[str]:
+        env: Dict[str, str] = {}
+        space = self.get_space(doc.spec.realm_id, doc.spec.space_id)
+        defaults = (space.spec.defaults.container
+                    if space.spec.defaults else None)
+        if defaults and defaults.user:
+            env["KUKEON_DEFAULT_USER"] = defaults.user
+        for kv in c.env:
+            k, _, v = kv.partition("=")
+            env[k] = v
+        # secrets (env channel)
+        for s in c.secrets:
+            if not s.env:
+                continue
+            sec = self.get_secret(doc.spec.realm_id, doc.spec.space_id, s.name)
+            for k, v in sec.spec.data.items():
+                env[s.env if len(sec.spec.data) == 1 else k] = _b64maybe(v)
+        # git identity sugar
+        if c.git:
+            if c.git.name:
+                env["GIT_AUTHOR_NAME"] = c.git.name
+                env["GIT_COMMITTER_NAME"] = c.git.name
+            if c.git.email:
+                env["GIT_AUTHOR_EMAIL"] = c.git.email
+                env["GIT_COMMITTER_EMAIL"] = c.git.email
+        # CLI --env RuntimeEnv overlay (attachable container, collide-replace)
+        if c.attachable or len(doc.spec.containers) == 1:
+            for kv in doc.spec.runtime_env:
+                k, _, v = kv.partition("=")
+                env[k] = v
+        if gpu_ids:
+            for kv in visible_devices_env(gpu_ids):
+                k, _, v = kv.partition("=")
+                env[k] = v
+        env["KUKEON_CELL"] = doc.metadata.name
+        env["KUKEON_REALM"] = doc.spec.realm_id
+        env["KUKEON_SPACE"] = doc.spec.space_id
+        env["KUKEON_STACK"] = doc.spec.stack_id
+        return [f"{k}={v}" for k, v in env.items()]
+
+    def start_cell(self, realm, space, stack, name) -> api.CellDoc:
+        doc = self.get_cell(realm, space, stack, name)
+        cell_dir = self.store.cell_dir(realm, space, stack, name)
+        cg = self._cell_cgroup(doc)
+        # idempotency guard: all containers running -> no-op
+        if doc.status.state == api.STATE_READY and self._all_running(doc):
+            return doc
+        self.cgroups.create(cg)
+        started = []
+        try:
+            root_probe = self.runtime.probe(cell_dir / ROOT_CONTAINER)
+            if not root_probe.running:
+                self.runtime.start_root(cell_dir, cg)
+            gpu_total = sum(c.gpus for c in doc.spec.containers)
+            gpu_ids: List[int] = []
+            if gpu_total > 0:
+                gpu_ids = self.gpus.allocate(self._gpu_owner(doc), gpu_total)
+                self.cgroups.allow_devices(cg, device_paths_for(gpu_ids))
+            doc.status.containers = []
+            cursor = 0
+            for c in doc.spec.containers:
+                cdir = cell_dir / (c.id or "main")
+                mine = gpu_ids[cursor: cursor + c.gpus]
+                cursor += c.gpus
+                env = self._container_env(doc, c, mine)
+                cdoc = api.ContainerDoc(
+                    metadata=api.Metadata(name=c.id or "main"), spec=c)
+                self.store.write(cdir / METADATA_FILE, cdoc.to_dict())
+                probe = self.runtime.probe(cdir)
+                if not probe.running:
+                    self.runtime.start_container(cdir, c, env, cg)
+                started.append(cdir)
+                if c.resources:
+                    self.cgroups.set_memory_limit(
+                        cg, c.resources.memory_limit_bytes)
+                    self.cgroups.set_cpu_shares(cg, c.resources.cpu_shares)
+                st = api.ContainerStatus(state=api.STATE_READY,
+                                         started_at=now_iso(self.now()),
+                                         gpu_ids=mine)
+                st.pid = self.runtime.probe(cdir).pid
+                doc.status.containers.append(st)
+            doc.status.state = api.STATE_READY
+            doc.status.message = ""
+            doc.status.cgroup_path = f"kukeon/{cg}"
+            self._persist_cell(doc)
+            return doc
+        except Exception as e:
+            # markCellFailed rollback: kill whatever came up mid-provision
+            for cdir in started:
+                with contextlib.suppress(Exception):
+                    self.runtime.kill(cdir)
+            with contextlib.suppress(Exception):
+                self.runtime.kill(cell_dir / ROOT_CONTAINER)
+            doc.status.state = api.STATE_FAILED
+            doc.status.message = str(e)
+            with contextlib.suppress(Exception):
+                self._persist_cell(doc)
+            raise
+
+    def _gpu_owner(self, doc: api.CellDoc) -> str:
+        s = doc.spec
+        return f"cell:{s.realm_id}/{s.space_id}/{s.stack_id}/{doc.metadata.name}"
+
+    def _all_running(self, doc: api.CellDoc) -> bool:
+        cell_dir = self.store.cell_dir(doc.spec.realm_id, doc.spec.space_id,
+                                       doc.spec.stack_id, doc.metadata.name)
+        for c in doc.spec.containers:
+            if not self.runtime.probe(cell_dir / (c.id or "main")).running:
+                return False
+        return True
+
+    def stop_cell(self, realm, space, stack, name,
+                  grace_seconds: float = 10.0) -> api.CellDoc:
+        doc = self.get_cell(realm, space, stack, name)
+        cell_dir = self.store.cell_dir(realm, space, stack, name)
+        for c in reversed(doc.spec.containers):
+            self.runtime.stop(cell_dir / (c.id or "main"), grace_seconds)
+        self.runtime.stop(cell_dir / ROOT_CONTAINER, grace_seconds)
+        doc.status.state = api.STATE_STOPPED
+        self._persist_cell(doc)
+        return doc
+
+    def kill_cell(self, realm, space, stack, name) -> api.CellDoc:
+        doc = self.get_cell(realm, space, stack, name)
+        cell_dir = self.store.cell_dir(realm, space, stack, name)
+        for c in reversed(doc.spec.containers):
+            self.runtime.kill(cell_dir / (c.id or "main"))
+        self.runtime.kill(cell_dir / ROOT_CONTAINER)
+        self.cgroups.kill_all(self._cell_cgroup(doc))
+        doc.status.state = api.STATE_STOPPED
+        self._persist_cell(doc)
+        return doc
+
+    def restart_cell(self, realm, space, stack, name) -> api.CellDoc:
+        self.stop_cell(realm, space, stack, name)
+        return self.start_cell(realm, space, stack, name)
+
+    def delete_cell(self, realm, space, stack, name,
+                    force: bool = False) -> None:
+        try:
+            doc = self.get_cell(realm, space, stack, name)
+        except errors.CellNotFound:
+            if force:
+                self.store.delete_tree(
+                    self.store.cell_dir(realm, space, stack, name))
+                return
+            raise
+        if doc.status.state == api.STATE_READY and not force:
+            self.stop_cell(realm, space, stack, name)
+        elif force:
+            with contextlib.suppress(Exception):
+                self.kill_cell(realm, space, stack, name)
+        self.gpus.release(self._gpu_owner(doc))
+        self.store.delete_tree(self.store.cell_dir(realm, space, stack, name))
+        self.cgroups.delete(self._cell_cgroup(doc))
+
+    def purge_cell(self, realm, space, stack, name) -> None:
+        """Force residual-state removal even when metadata is damaged."""
+        with contextlib.suppress(Exception):
+            self.kill_cell(realm, space, stack, name)
+        self.gpus.release(f"cell:{realm}/{space}/{stack}/{name}")
+        self.store.delete_tree(self.store.cell_dir(realm, space, stack, name))
+        self.cgroups.delete(f"{realm}/{space}/{stack}/{name}")
+
+    def recreate_cell(self, doc: api.CellDoc) -> api.CellDoc:
+        was_running = False
+        try:
+            old = self.get_cell(doc.spec.realm_id, doc.spec.space_id,
+                                doc.spec.stack_id, doc.metadata.name)
+            was_running = old.status.state == api.STATE_READY
+            self.delete_cell(doc.spec.realm_id, doc.spec.space_id,
+                             doc.spec.stack_id, doc.metadata.name, force=True)
+        except errors.CellNotFound:
+            pass
+        created = self.create_cell(doc)
+        if was_running:
+            return self.start_cell(doc.spec.realm_id, doc.spec.space_id,
+                                   doc.spec.stack_id, doc.metadata.name)
+        return created
+
+    # ---- attach / logs ----------------------------------------------
+    def attach_path(self, realm, space, stack, name) -> str:
+        doc = self.get_cell(realm, space, stack, name)
+        target = None
+        for c in doc.spec.containers:
+            if c.attachable and not c.root:
+                target = c
+                break
+        if target is None:
+            raise errors.InvalidArgument(
+                f"cell {name} has no attachable container")
+        cdir = self.store.cell_dir(realm, space, stack, name) / (
+            target.id or "main")
+        probe = self.runtime.probe(cdir)
+        if not probe.running:
+            raise errors.NotReady(f"container {target.id} is not running")
+        sock = cdir / "tty" / "socket"
+        # SUN_PATH 107-byte guard: hand back a short symlink when needed
+        if len(str(sock)) > 100:
+            return str(self.store.socket_link(str(sock)))
+        return str(sock)
+
+    def log_path(self, realm, space, stack, name,
+                 container: str = "") -> str:
+        doc = self.get_cell(realm, space, stack, name)
+        if not container:
+            for c in doc.spec.containers:
+                if c.attachable:
+                    container = c.id or "main"
+                    break
+            else:
+                container = doc.spec.containers[0].id or "main"
+        cdir = self.store.cell_dir(realm, space, stack, name) / container
+        cap = cdir / "capture.log"
+        if cap.exists():
+            return str(cap)
+        return str(cdir / "shim.log")
+
+    # ==================================================================
+    # scoped documents: secrets / blueprints / configs / volumes
+    # ==================================================================
+    def put_secret(self, doc: api.SecretDoc) -> None:
+        p = self.store.scoped_doc_path(doc.spec.realm_id, doc.spec.space_id,
+                                       "secrets", doc.metadata.name)
+        self.store.write(p, doc.to_dict(), mode=0o600)
+
+    def get_secret(self, realm, space, name) -> api.SecretDoc:
+        data = self.store.read(
+            self.store.scoped_doc_path(realm, space, "secrets", name))
+        if data is None:
+            raise errors.SecretNotFound(f"{realm}/{space}/{name}")
+        return api.SecretDoc.from_dict(data)
+
+    def list_secrets(self, realm, space) -> List[str]:
+        return self.store.list_scoped_docs(
+            self.store.space_dir(realm, space) / "secrets")
+
+    def delete_secret(self, realm, space, name) -> None:
+        if not self.store.delete(
+                self.store.scoped_doc_path(realm, space, "secrets", name)):
+            raise errors.SecretNotFound(name)
+
+    def put_blueprint(self, doc: api.CellBlueprintDoc) -> None:
+        p = self.store.scoped_doc_path(doc.spec.realm_id, doc.spec.space_id,
+                                       "blueprints", doc.metadata.name)
+        self.store.write(p, doc.to_dict())
+
+    def get_blueprint(self, realm, space, name) -> api.CellBlueprintDoc:
+        data = self.store.read(
+            self.store.scoped_doc_path(realm, space, "blueprints", name))
+        if data is None:
+            raise errors.BlueprintNotFound(f"{realm}/{space}/{name}")
+        return api.CellBlueprintDoc.from_dict(data)
+
+    def list_blueprints(self, realm, space) -> List[str]:
+        return self.store.list_scoped_docs(
+            self.store.space_dir(realm, space) / "blueprints")
+
+    def delete_blueprint(self, realm, space, name) -> None:
+        if not self.store.delete(
+                self.store.scoped_doc_path(realm, space, "blueprints", name)):
+            raise errors.BlueprintNotFound(name)
+
+    def put_config(self, doc: api.CellConfigDoc) -> None:
+        p = self.store.scoped_doc_path(doc.spec.realm_id, doc.spec.space_id,
+                                       "configs", doc.metadata.name)
+        self.store.write(p, doc.to_dict())
+
+    def get_config(self, realm, space, name) -> api.CellConfigDoc:
+        data = self.store.read(
+            self.store.scoped_doc_path(realm, space, "configs", name))
+        if data is None:
+            raise errors.ConfigNotFound(f"{realm}/{space}/{name}")
+        return api.CellConfigDoc.from_dict(data)
+
+    def list_configs(self, realm, space) -> List[str]:
+        return self.store.list_scoped_docs(
+            self.store.space_dir(realm, space) / "configs")
+
+    def delete_config(self, realm, space, name) -> None:
+        if not self.store.delete(
+                self.store.scoped_doc_path(realm, space, "configs", name)):
+            raise errors.ConfigNotFound(name)
+
+    def put_volume(self, doc: api.VolumeDoc) -> api.VolumeDoc:
+        d = self.store.volume_data_dir(doc.spec.realm_id, doc.spec.space_id,
+                                       doc.metadata.name)
+        d.mkdir(parents=True, exist_ok=True)
+        doc.status.state = api.STATE_READY
+        doc.status.path = str(d)
+        p = self.store.scoped_doc_path(doc.spec.realm_id, doc.spec.space_id,
+                                       "volume-meta", doc.metadata.name)
+        self.store.write(p, doc.to_dict())
+        return doc
+
+    def get_volume(self, realm, space, name) -> api.VolumeDoc:
+        data = self.store.read(
+            self.store.scoped_doc_path(realm, space, "volume-meta", name))
+        if data is None:
+            raise errors.VolumeNotFound(f"{realm}/{space}/{name}")
+        return api.VolumeDoc.from_dict(data)
+
+    def delete_volume(self, realm, space, name) -> None:
+        self.get_volume(realm, space, name)
+        self.store.delete(
+            self.store.scoped_doc_path(realm, space, "volume-meta", name))
+        self.store.delete_tree(self.store.volume_data_dir(realm, space, name))
+
+    # ==================================================================
+    # run (blueprint/config materialization)
+    # ==================================================================
+    def run_from_blueprint(self, realm, space, stack, blueprint_name,
+                           params: Dict[str, str],
+                           env: Optional[List[str]] = None,
+                           name: Optional[str] = None) -> api.CellDoc:
+        bp = self.get_blueprint(realm, space, blueprint_name)
+        taken = set(self.store.list_children(
+            self.store.stack_dir(realm, space, stack)))
+        cname = name or naming.generate_cell_name(
+            bp.spec.name_prefix or blueprint_name, taken)
+        cell = bpmod.materialize(bp, cname, params, env_overlay=env)
+        cell.spec.realm_id = cell.spec.realm_id or realm
+        cell.spec.space_id = cell.spec.space_id or space
+        cell.spec.stack_id = cell.spec.stack_id or stack
+        self.create_cell(cell)
+        return self.start_cell(realm, space, stack, cname)
+
+    def run_from_config(self, realm, space, stack, config_name,
+                        params: Dict[str, str],
+                        name: Optional[str] = None) -> api.CellDoc:
+        cfg = self.get_config(realm, space, config_name)
+        bp = self.get_blueprint(realm, space, cfg.spec.blueprint)
+        taken = set(self.store.list_children(
+            self.store.stack_dir(realm, space, stack)))
+        cname = name or naming.generate_cell_name(
+            cfg.spec.name_prefix or config_name, taken)
+        cell = bpmod.materialize_from_config(cfg, bp, cname, params)
+        cell.spec.realm_id = cell.spec.realm_id or realm
+        cell.spec.space_id = cell.spec.space_id or space
+        cell.spec.stack_id = cell.spec.stack_id or stack
+        self.create_cell(cell)
+        return self.start_cell(realm, space, stack, cname)
+
+    # ==================================================================
+    # sessions
+    # ==================================================================
+    def create_session(self, doc: api.SessionDoc) -> api.SessionDoc:
+        parser.validate_document(doc)
+        realm = doc.spec.realm_id or naming.DEFAULT_REALM
+        space = doc.spec.space_id or naming.DEFAULT_SPACE
+        doc.spec.realm_id, doc.spec.space_id = realm, space
+        self.ensure_stack(realm, space, doc.spec.stack_id)
+        if doc.spec.gpus > 0:
+            ids = self.gpus.allocate(f"session:{doc.metadata.name}",
+                                     doc.spec.gpus)
+            doc.status.gpu_ids = ids
+        doc.status.state = api.STATE_RUNNING
+        doc.status.started_at = now_iso(self.now())
+        if doc.spec.lifetime and doc.spec.lifetime.wall_clock:
+            secs = parser.parse_duration(doc.spec.lifetime.wall_clock)
+            doc.status.deadline = now_iso(self.now() + secs)
+        doc.status.last_activity_at = doc.status.started_at
+        path = self.store.session_path(realm, space, doc.spec.stack_id,
+                                       doc.metadata.name)
+        self.store.create_exclusive(path, doc.to_dict())
+        return doc
+
+    def get_session(self, realm, space, stack, name) -> api.SessionDoc:
+        data = self.store.read(
+            self.store.session_path(realm, space, stack, name))
+        if data is None:
+            raise errors.SessionNotFound(f"{realm}/{space}/{stack}/{name}")
+        return api.SessionDoc.from_dict(data)
+
+    def list_sessions(self) -> List[api.SessionDoc]:
+        root = self.run_path / "sessions"
+        out = []
+        for p in sorted(root.glob("*/*/*/*.json")):
+            data = self.store.read(p)
+            if data:
+                out.append(api.SessionDoc.from_dict(data))
+        return out
+
+    def close_session(self, realm, space, stack, name,
+                      state: str = api.STATE_COMPLETED) -> api.SessionDoc:
+        doc = self.get_session(realm, space, stack, name)
+        if doc.status.state in (api.STATE_COMPLETED, api.STATE_TERMINATED):
+            return doc
+        # persist declared outputs before teardown
+        persist_dir = self.run_path / "persist" / name
+        if doc.spec.on_end:
+            for p in doc.spec.on_end.persist:
+                src = Path(p.volume)
+                if src.is_absolute():
+                    # cell-relative volumes: look under the cell dirs
+                    src = Path(str(src))
+                if src.exists():
+                    dst = persist_dir / src.name
+                    dst.parent.mkdir(parents=True, exist_ok=True)
+                    with contextlib.suppress(Exception):
+                        if src.is_dir():
+                            shutil.copytree(src, dst, dirs_exist_ok=True)
+                        else:
+                            shutil.copy2(src, dst)
+        # destroy the stack's cells (ephemeral by default)
+        with contextlib.suppress(errors.StackNotFound):
+            for cell in self.store.list_children(
+                    self.store.stack_dir(realm, space, stack)):
+                with contextlib.suppress(Exception):
+                    self.delete_cell(realm, space, stack, cell, force=True)
+        self.gpus.release(f"session:{name}")
+        doc.status.state = state
+        doc.status.ended_at = now_iso(self.now())
+        doc.status.gpu_ids = []
+        self.store.write(self.store.session_path(realm, space, stack, name),
+                         doc.to_dict())
+        return doc
+
+    def delete_session(self, realm, space, stack, name) -> None:
+        self.close_session(realm, space, stack, name, api.STATE_TERMINATED)
+        self.store.delete(self.store.session_path(realm, space, stack, name))
+
+    def touch_session(self, realm, space, stack, name) -> None:
+        doc = self.get_session(realm, space, stack, name)
+        doc.status.last_activity_at = now_iso(self.now())
+        self.store.write(self.store.session_path(realm, space, stack, name),
+                         doc.to_dict())
+
+    # ==================================================================
+    # apply
+    # ==================================================================
+    def apply_documents(self, text: str,
+                        team: str = "") -> List[ResourceResult]:
+        docs = parser.parse_documents(text, validate=False)
+        order = {k: i for i, k in enumerate(api.APPLY_ORDER)}
+        docs.sort(key=lambda d: order.get(d.kind, 99))
+        results = []
+        for doc in docs:
+            if team:
+                doc.metadata.labels[api.LABEL_TEAM] = team
+            try:
+                parser.validate_document(doc)
+                results.append(self._apply_one(doc))
+            except Exception as e:
+                results.append(ResourceResult(doc.kind, doc.metadata.name,
+                                              "failed", str(e)))
+        return results
+
+    def _apply_one(self, doc) -> ResourceResult:
+        kind, name = doc.kind, doc.metadata.name
+        if kind == api.KIND_REALM:
+            try:
+                self.get_realm(name)
+                return ResourceResult(kind, name, "unchanged")
+            except errors.RealmNotFound:
+                self.create_realm(doc)
+                return ResourceResult(kind, name, "created")
+        if kind == api.KIND_SPACE:
+            try:
+                cur = self.get_space(doc.spec.realm_id, name)
+            except errors.SpaceNotFound:
+                self.create_space(doc)
+                return ResourceResult(kind, name, "created")
+            d = diffmod.diff_space(doc, cur)
+            if d.change_type == diffmod.ChangeType.NONE:
+                return ResourceResult(kind, name, "unchanged")
+            cur.spec = doc.spec
+            self._apply_egress(cur)
+            self.store.write_cas(
+                self.store.space_dir(doc.spec.realm_id, name) / METADATA_FILE,
+                cur.to_dict())
+            return ResourceResult(kind, name, "updated")
+        if kind == api.KIND_STACK:
+            try:
+                self.get_stack(doc.spec.realm_id, doc.spec.space_id, name)
+                return ResourceResult(kind, name, "unchanged")
+            except errors.StackNotFound:
+                self.create_stack(doc)
+                return ResourceResult(kind, name, "created")
+        if kind == api.KIND_SECRET:
+            self.put_secret(doc)
+            return ResourceResult(kind, name, "updated")
+        if kind == api.KIND_VOLUME:
+            self.put_volume(doc)
+            return ResourceResult(kind, name, "updated")
+        if kind == api.KIND_CELL_BLUEPRINT:
+            self.put_blueprint(doc)
+            return ResourceResult(kind, name, "updated")
+        if kind == api.KIND_CELL_CONFIG:
+            self.put_config(doc)
+            return ResourceResult(kind, name, "updated")
+        if kind == api.KIND_SESSION:
+            try:
+                self.get_session(doc.spec.realm_id or naming.DEFAULT_REALM,
+                                 doc.spec.space_id or naming.DEFAULT_SPACE,
+                                 doc.spec.stack_id, name)
+                return ResourceResult(kind, name, "unchanged")
+            except errors.SessionNotFound:
+                self.create_session(doc)
+                return ResourceResult(kind, name, "created")
+        if kind == api.KIND_CELL:
+            try:
+                cur = self.get_cell(doc.spec.realm_id, doc.spec.space_id,
+                                    doc.spec.stack_id, name)
+            except errors.CellNotFound:
+                self.create_cell(doc)
+                return ResourceResult(kind, name, "created")
+            d = diffmod.diff_cell(doc, cur)
+            ct = d.change_type
+            if ct == diffmod.ChangeType.NONE:
+                return ResourceResult(kind, name, "unchanged")
+            if ct == diffmod.ChangeType.BREAKING:
+                self.recreate_cell(doc)
+                return ResourceResult(kind, name, "recreated")
+            cur.spec = doc.spec
+            self.store.write_cas(
+                self._cell_path(doc.spec.realm_id, doc.spec.space_id,
+                                doc.spec.stack_id, name), cur.to_dict())
+            return ResourceResult(kind, name, "updated")
+        raise errors.ValidationError(f"kind {kind} is not applyable")
+
+    # ==================================================================
+    # reconcile (the control-plane hot loop)
+    # ==================================================================
+    def reconcile_cells(self) -> int:
+        """One pass over every cell: status re-derivation, restart policy,
+        AutoDelete. Returns cells visited."""
+        visited = 0
+        for realm in self.store.list_children(self.store.data_root):
+            for space in self.store.list_children(self.store.realm_dir(realm)):
+                for stack in self.store.list_children(
+                        self.store.space_dir(realm, space)):
+                    for cell in self.store.list_children(
+                            self.store.stack_dir(realm, space, stack)):
+                        with contextlib.suppress(errors.CellNotFound):
+                            self.reconcile_cell(realm, space, stack, cell)
+                            visited += 1
+        return visited
+
+    def reconcile_cell(self, realm, space, stack, name) -> api.CellDoc:
+        doc = self.get_cell(realm, space, stack, name)
+        if doc.status.state in (api.STATE_PENDING, api.STATE_STOPPED,
+                                api.STATE_FAILED):
+            return doc
+        cell_dir = self.store.cell_dir(realm, space, stack, name)
+        now = self.now()
+        running = 0
+        clean_exits = 0
+        crashed_terminal = 0
+        restarting = 0
+        statuses = []
+        for c in doc.spec.containers:
+            cdir = cell_dir / (c.id or "main")
+            probe = self.runtime.probe(cdir)
+            key = str(cdir)
+            last, count = self._restart_state.get(key, (0.0, 0))
+            st = api.ContainerStatus(state=probe.state, pid=probe.pid,
+                                     exit_code=probe.exit_code,
+                                     started_at=probe.started_at,
+                                     finished_at=probe.finished_at,
+                                     restart_count=count)
+            if probe.running:
+                running += 1
+                if count and probe.started_at and \
+                        now - last > max(60.0, 2 * self._backoff(c)):
+                    # stable again: reset the crash-loop counter
+                    self._restart_state[key] = (last, 0)
+            elif probe.exit_code is not None or probe.exists:
+                rc = probe.exit_code
+                should = self._should_restart(c, rc, count)
+                if should and now - last >= self._backoff(c):
+                    env_gpus = [s.gpu_ids for s in doc.status.containers
+                                if s.gpu_ids]
+                    env = self._container_env(
+                        doc, c, env_gpus[0] if env_gpus else [])
+                    with contextlib.suppress(Exception):
+                        self.runtime.start_container(
+                            cdir, c, env, self._cell_cgroup(doc))
+                        self._restart_state[key] = (now, count + 1)
+                        st.restart_count = count + 1
+                        st.state = api.STATE_READY
+                        restarting += 1
+                elif should:
+                    restarting += 1  # waiting out the backoff floor
+                elif rc == 0:
+                    clean_exits += 1
+                else:
+                    crashed_terminal += 1
+            statuses.append(st)
+        doc.status.containers = statuses
+        n = len(doc.spec.containers)
+        if running == n:
+            doc.status.state = api.STATE_READY
+        elif clean_exits == n:
+            doc.status.state = api.STATE_EXITED
+        elif crashed_terminal > 0 and restarting == 0 and running == 0:
+            doc.status.state = api.STATE_ERROR
+        elif crashed_terminal > 0 and restarting == 0:
+            doc.status.state = api.STATE_ERROR
+        elif running + restarting + clean_exits == n and (restarting or
+                                                          clean_exits):
+            doc.status.state = api.STATE_DEGRADED
+        self._reconcile_outofsync(doc)
+        doc.status.observed_generation = doc.metadata.generation
+        self._persist_cell(doc)
+        if doc.spec.auto_delete and doc.status.state in \
+                api.TERMINAL_CELL_STATES:
+            self.delete_cell(realm, space, stack, name, force=True)
+        return doc
+
+    @staticmethod
+    def _backoff(c: api.ContainerSpec) -> float:
+        if c.restart_backoff_seconds is not None:
+            return float(c.restart_backoff_seconds)
+        return float(api.DEFAULT_RESTART_BACKOFF_SECONDS)
+
+    @staticmethod
+    def _should_restart(c: api.ContainerSpec, rc: Optional[int],
+                        count: int) -> bool:
+        if c.root:
+            return False
+        if c.restart_policy == api.RESTART_ALWAYS:
+            return True
+        if c.restart_policy == api.RESTART_ON_FAILURE:
+            if rc is not None and rc == 0:
+                return False
+            cap = (c.restart_max_retries
+                   if c.restart_max_retries is not None
+                   else api.DEFAULT_RESTART_MAX_RETRIES)
+            return count < cap
+        return False
+
+    def _reconcile_outofsync(self, doc: api.CellDoc) -> None:
+        """Config-lineage drift detection (reference reconcile_outofsync)."""
+        prov = doc.spec.provenance
+        if prov is None:
+            return
+        try:
+            if prov.binding_kind == "config":
+                cfg = self.get_config(doc.spec.realm_id, doc.spec.space_id,
+                                      prov.binding_ref)
+                bp = self.get_blueprint(doc.spec.realm_id, doc.spec.space_id,
+                                        cfg.spec.blueprint)
+                want = bpmod.materialize_from_config(
+                    cfg, bp, doc.metadata.name, dict(prov.params))
+            else:
+                bp = self.get_blueprint(doc.spec.realm_id, doc.spec.space_id,
+                                        prov.binding_ref)
+                want = bpmod.materialize(bp, doc.metadata.name,
+                                         dict(prov.params),
+                                         env_overlay=list(prov.env))
+            want.spec.realm_id = doc.spec.realm_id
+            want.spec.space_id = doc.spec.space_id
+            want.spec.stack_id = doc.spec.stack_id
+            for c in want.spec.containers:
+                c.id = c.id or "main"
+                c.realm_id, c.space_id = doc.spec.realm_id, doc.spec.space_id
+                c.stack_id, c.cell_id = doc.spec.stack_id, doc.metadata.name
+            d = diffmod.diff_cell(want, doc)
+            if d.change_type == diffmod.ChangeType.NONE:
+                doc.status.out_of_sync = False
+                doc.status.out_of_sync_reason = ""
+                doc.status.out_of_sync_error = ""
+            else:
+                doc.status.out_of_sync = True
+                doc.status.out_of_sync_reason = (
+                    f"{d.change_type.value}: " + ", ".join(d.paths[:5]))
+        except errors.NotFound as e:
+            doc.status.out_of_sync = True
+            doc.status.out_of_sync_error = str(e)
+
+    def reconcile_sessions(self) -> int:
+        visited = 0
+        now = self.now()
+        for doc in self.list_sessions():
+            visited += 1
+            if doc.status.state != api.STATE_RUNNING:
+                continue
+            realm, space, stack = (doc.spec.realm_id, doc.spec.space_id,
+                                   doc.spec.stack_id)
+            if doc.status.deadline and now >= parse_iso(doc.status.deadline):
+                self.close_session(realm, space, stack, doc.metadata.name,
+                                   api.STATE_TERMINATED)
+                continue
+            if doc.spec.lifetime and doc.spec.lifetime.idle_timeout:
+                idle = parser.parse_duration(doc.spec.lifetime.idle_timeout)
+                last = self._session_activity(doc)
+                if last and now - last >= idle:
+                    self.close_session(realm, space, stack,
+                                       doc.metadata.name,
+                                       api.STATE_TERMINATED)
+        return visited
+
+    def _session_activity(self, doc: api.SessionDoc) -> Optional[float]:
+        """Latest activity: shim activity files of the session's stack."""
+        latest = (parse_iso(doc.status.last_activity_at)
+                  if doc.status.last_activity_at else None)
+        stack_dir = self.store.stack_dir(doc.spec.realm_id, doc.spec.space_id,
+                                         doc.spec.stack_id)
+        for act in stack_dir.glob("*/*/activity"):
+            with contextlib.suppress(OSError):
+                m = act.stat().st_mtime
+                latest = m if latest is None else max(latest, m)
+        return latest
+
+    def reconcile_space_networks(self) -> int:
+        visited = 0
+        for realm in self.store.list_children(self.store.data_root):
+            for space in self.store.list_children(self.store.realm_dir(realm)):
+                with contextlib.suppress(errors.SpaceNotFound):
+                    self._apply_egress(self.get_space(realm, space))
+                    visited += 1
+        return visited
+
+
+def _b64maybe(v: str) -> str:
+    """Secret values may be base64 (K8s style); fall back to plain."""
+    try:
+        return base64.b64decode(v, validate=True).decode()
+    except Exception:
+        return v

@@ -1,0 +1,275 @@
+"""Container shim: the kuketty/kukepause analog for process cells.
+
+One shim per container (the runtime spawns `python -m kukeon_amd.tty.shim
+--dir <container_dir>` in its own session/process group). It:
+
+* reads the spawn spec (`spawn.json`) the runner rendered — argv, env, cwd,
+  attachable flag, capture path — and records `runtime.json` (shim pid +
+  start time, workload pid) for liveness probes that survive daemon restarts,
+* supervises the workload and writes `status.json` {exitCode, finishedAt}
+  durably on exit (the containerd-shim role: exit codes outlive the daemon),
+* for attachable containers: owns the PTY master, serves the attach socket
+  at <dir>/tty/socket (multi-client byte pump, capture file for `kuke log`,
+  activity timestamps for Session idleTimeout), runs the optional tty init
+  script first,
+* `--pause` mode is the cell root: a signal-wait PID holding the cell's
+  process group open (kukepause analog; exits 0 on SIGTERM/SIGINT).
+"""
+from __future__ import annotations
+
+import argparse
+import fcntl
+import json
+import os
+import pty
+import select
+import signal
+import socket
+import sys
+import time
+from pathlib import Path
+
+
+def now_iso() -> str:
+    return time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+
+
+def write_json(path: Path, data: dict) -> None:
+    tmp = path.with_suffix(".tmp")
+    tmp.write_text(json.dumps(data))
+    os.replace(tmp, path)
+
+
+def run_pause() -> int:
+    """Root-container pause: reap children, exit 0 on TERM/INT."""
+    stop = {"flag": False}
+
+    def on_term(signum, frame):
+        stop["flag"] = True
+
+    def on_chld(signum, frame):
+        try:
+            while os.waitpid(-1, os.WNOHANG)[0] > 0:
+                pass
+        except ChildProcessError:
+            pass
+
+    signal.signal(signal.SIGTERM, on_term)
+    signal.signal(signal.SIGINT, on_term)
+    signal.signal(signal.SIGCHLD, on_chld)
+    while not stop["flag"]:
+        signal.pause()
+    return 0
+
+
+class Shim:
+    def __init__(self, cdir: Path):
+        self.dir = cdir
+        self.spec = json.loads((cdir / "spawn.json").read_text())
+        self.child_pid = 0
+
+    def record_runtime(self, workload_pid: int) -> None:
+        from kukeon_amd.runtime.proc import proc_starttime
+        write_json(self.dir / "runtime.json", {
+            "shimPid": os.getpid(),
+            "shimStarttime": proc_starttime(os.getpid()),
+            "workloadPid": workload_pid,
+            "startedAt": now_iso(),
+        })
+
+    def record_status(self, rc: int) -> None:
+        write_json(self.dir / "status.json",
+                   {"exitCode": rc, "finishedAt": now_iso()})
+
+    def child_env(self) -> dict:
+        env = dict(os.environ)
+        for kv in self.spec.get("env", []):
+            k, _, v = kv.partition("=")
+            env[k] = v
+        env.setdefault("TERM", "xterm-256color")
+        return env
+
+    # ------------------------------------------------------------------
+    def run_plain(self) -> int:
+        pid = os.fork()
+        if pid == 0:
+            self._exec_child()
+        self.child_pid = pid
+        self.record_runtime(pid)
+        _, status = os.waitpid(pid, 0)
+        rc = os.waitstatus_to_exitcode(status)
+        self.record_status(rc if rc >= 0 else 128 - rc)
+        return 0
+
+    def _exec_child(self):
+        spec = self.spec
+        if spec.get("cwd"):
+            try:
+                os.chdir(spec["cwd"])
+            except OSError:
+                pass
+        argv = spec["argv"]
+        try:
+            os.execvpe(argv[0], argv, self.child_env())
+        except OSError as e:
+            sys.stderr.write(f"exec {argv[0]}: {e}\n")
+            os._exit(127)
+
+    # ------------------------------------------------------------------
+    def run_attachable(self) -> int:
+        tty_dir = self.dir / "tty"
+        tty_dir.mkdir(parents=True, exist_ok=True)
+        sock_path = tty_dir / "socket"
+        try:
+            sock_path.unlink()
+        except FileNotFoundError:
+            pass
+        srv = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        srv.bind(str(sock_path))
+        os.chmod(sock_path, 0o660)
+        srv.listen(8)
+        srv.setblocking(False)
+
+        pid, master = pty.fork()
+        if pid == 0:
+            # child: on the PTY slave, its own session
+            spec = self.spec
+            if spec.get("cwd"):
+                try:
+                    os.chdir(spec["cwd"])
+                except OSError:
+                    pass
+            init = spec.get("init_script")
+            argv = list(spec["argv"])
+            if init:
+                argv = ["/bin/sh", "-c", f"{init}\nexec \"$@\"", "sh"] + argv
+            try:
+                os.execvpe(argv[0], argv, self.child_env())
+            except OSError as e:
+                sys.stderr.write(f"exec {argv[0]}: {e}\n")
+                os._exit(127)
+
+        self.child_pid = pid
+        self.record_runtime(pid)
+        capture = open(self.dir / "capture.log", "ab", buffering=0)
+        activity = self.dir / "activity"
+        activity.touch()
+        clients = []
+        rc = 0
+        fcntl.fcntl(master, fcntl.F_SETFL,
+                    fcntl.fcntl(master, fcntl.F_GETFL) | os.O_NONBLOCK)
+        try:
+            while True:
+                rl = [srv, master] + clients
+                try:
+                    rd, _, _ = select.select(rl, [], [], 0.5)
+                except InterruptedError:
+                    rd = []
+                for r in rd:
+                    if r is srv:
+                        try:
+                            c, _ = srv.accept()
+                            c.setblocking(False)
+                            self._handle_handshake(c)
+                            clients.append(c)
+                        except OSError:
+                            pass
+                    elif r is master:
+                        try:
+                            data = os.read(master, 65536)
+                        except (OSError, IOError):
+                            data = b""
+                        if not data:
+                            raise EOFError
+                        capture.write(data)
+                        dead = []
+                        for c in clients:
+                            try:
+                                c.sendall(data)
+                            except OSError:
+                                dead.append(c)
+                        for c in dead:
+                            clients.remove(c)
+                            c.close()
+                    else:
+                        try:
+                            data = r.recv(65536)
+                        except OSError:
+                            data = b""
+                        if not data:
+                            clients.remove(r)
+                            r.close()
+                            continue
+                        activity.touch()
+                        self._handle_client_data(r, master, data)
+                # reap
+                done, status = os.waitpid(pid, os.WNOHANG)
+                if done == pid:
+                    rc = os.waitstatus_to_exitcode(status)
+                    break
+        except EOFError:
+            _, status = os.waitpid(pid, 0)
+            rc = os.waitstatus_to_exitcode(status)
+        finally:
+            self.record_status(rc if rc >= 0 else 128 - rc)
+            for c in clients:
+                c.close()
+            srv.close()
+            capture.close()
+        return 0
+
+    def _handle_handshake(self, c: socket.socket) -> None:
+        hello = json.dumps({"ok": True, "proto": "kukeon-tty/1",
+                            "pid": self.child_pid}) + "\n"
+        try:
+            c.sendall(hello.encode())
+        except OSError:
+            pass
+
+    def _handle_client_data(self, c, master: int, data: bytes) -> None:
+        # in-band resize escape: \x00R{"rows":..,"cols":..}\n ; raw otherwise
+        if data.startswith(b"\x00R"):
+            nl = data.find(b"\n")
+            if nl > 0:
+                try:
+                    import struct
+                    import termios
+                    ws = json.loads(data[2:nl])
+                    fcntl.ioctl(master, termios.TIOCSWINSZ,
+                                struct.pack("HHHH", ws["rows"], ws["cols"],
+                                            0, 0))
+                except (ValueError, OSError, KeyError):
+                    pass
+                data = data[nl + 1:]
+        if data:
+            try:
+                os.write(master, data)
+            except OSError:
+                pass
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--dir")
+    ap.add_argument("--pause", action="store_true")
+    args = ap.parse_args()
+    if args.pause:
+        return run_pause()
+    shim = Shim(Path(args.dir))
+
+    def on_term(signum, frame):
+        # forward to the workload; the wait loop observes the exit
+        if shim.child_pid > 0:
+            try:
+                os.kill(shim.child_pid, signal.SIGTERM)
+            except ProcessLookupError:
+                pass
+
+    signal.signal(signal.SIGTERM, on_term)
+    if shim.spec.get("attachable"):
+        return shim.run_attachable()
+    return shim.run_plain()
+
+
+if __name__ == "__main__":
+    sys.exit(main())

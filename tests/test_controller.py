@@ -1,0 +1,298 @@
+"""Controller unit tests against the FakeRuntime (reference test strategy:
+fakes at every layer boundary, frozen clocks via injected now_fn)."""
+import pytest
+
+from kukeon_amd.api import errors
+from kukeon_amd.api import v1beta1 as api
+from kukeon_amd.controller.core import Controller
+from kukeon_amd.controller import naming
+from kukeon_amd.runtime.process import ROOT_CONTAINER, FakeRuntime
+
+
+class Clock:
+    def __init__(self, t=1000.0):
+        self.t = t
+
+    def __call__(self):
+        return self.t
+
+
+@pytest.fixture
+def ctl(tmp_path):
+    c = Controller(str(tmp_path / "run"), runtime=FakeRuntime(),
+                   gpu_devices=[0, 1, 2, 3], now_fn=Clock())
+    c.bootstrap()
+    return c
+
+
+def make_cell(name="c1", **kw):
+    spec = dict(realm_id="default", space_id="default", stack_id="default")
+    cspec = api.ContainerSpec(id="main", image="none",
+                              command="sleep", args=["100"], **kw)
+    return api.CellDoc(metadata=api.Metadata(name=name),
+                       spec=api.CellSpec(containers=[cspec], **spec))
+
+
+def test_bootstrap_creates_hierarchies(ctl):
+    assert {r.metadata.name for r in ctl.list_realms()} == {
+        "default", naming.SYSTEM_REALM}
+    ctl.get_stack("default", "default", "default")
+    ctl.get_stack(naming.SYSTEM_REALM, naming.SYSTEM_SPACE,
+                  naming.SYSTEM_STACK)
+
+
+def test_cell_lifecycle(ctl):
+    ctl.create_cell(make_cell())
+    doc = ctl.start_cell("default", "default", "default", "c1")
+    assert doc.status.state == api.STATE_READY
+    assert len(ctl.runtime.started) == 2  # root + main
+    got = ctl.get_cell("default", "default", "default", "c1")
+    assert got.status.state == api.STATE_READY
+    ctl.stop_cell("default", "default", "default", "c1")
+    got = ctl.get_cell("default", "default", "default", "c1")
+    assert got.status.state == api.STATE_STOPPED
+    ctl.delete_cell("default", "default", "default", "c1")
+    with pytest.raises(errors.CellNotFound):
+        ctl.get_cell("default", "default", "default", "c1")
+
+
+def test_cell_create_duplicate(ctl):
+    ctl.create_cell(make_cell())
+    with pytest.raises(errors.AlreadyExists):
+        ctl.create_cell(make_cell())
+
+
+def test_start_rollback_on_failure(ctl):
+    ctl.create_cell(make_cell())
+    cdir = ctl.store.cell_dir("default", "default", "default", "c1") / "main"
+    ctl.runtime.fail_on[str(cdir)] = RuntimeError("boom")
+    with pytest.raises(RuntimeError):
+        ctl.start_cell("default", "default", "default", "c1")
+    doc = ctl.get_cell("default", "default", "default", "c1")
+    assert doc.status.state == api.STATE_FAILED
+    assert "boom" in doc.status.message
+
+
+def test_restart_policy_on_failure_with_backoff(ctl):
+    cell = make_cell()
+    cell.spec.containers[0].restart_policy = "on-failure"
+    cell.spec.containers[0].restart_backoff_seconds = 30
+    cell.spec.containers[0].restart_max_retries = 2
+    ctl.create_cell(cell)
+    ctl.start_cell("default", "default", "default", "c1")
+    cdir = ctl.store.cell_dir("default", "default", "default", "c1") / "main"
+    ctl.runtime.mark_exited(cdir, 1)
+
+    # first restart fires immediately (no prior restart -> no floor)
+    ctl.reconcile_cell("default", "default", "default", "c1")
+    assert ctl.runtime.started.count(str(cdir)) == 2
+    # crash again: inside the 30s floor -> restart deferred
+    ctl.runtime.mark_exited(cdir, 1)
+    ctl.reconcile_cell("default", "default", "default", "c1")
+    assert ctl.runtime.started.count(str(cdir)) == 2
+    ctl.now.t += 31
+    ctl.reconcile_cell("default", "default", "default", "c1")
+    assert ctl.runtime.started.count(str(cdir)) == 3
+    # third crash -> cap of 2 retries reached -> terminal Error
+    ctl.runtime.mark_exited(cdir, 1)
+    ctl.now.t += 31
+    doc = ctl.reconcile_cell("default", "default", "default", "c1")
+    assert ctl.runtime.started.count(str(cdir)) == 3  # capped
+    assert doc.status.state == api.STATE_ERROR
+
+
+def test_clean_exit_is_exited_not_error(ctl):
+    ctl.create_cell(make_cell())
+    ctl.start_cell("default", "default", "default", "c1")
+    cdir = ctl.store.cell_dir("default", "default", "default", "c1") / "main"
+    ctl.runtime.mark_exited(cdir, 0)
+    doc = ctl.reconcile_cell("default", "default", "default", "c1")
+    assert doc.status.state == api.STATE_EXITED
+
+
+def test_autodelete_on_terminal(ctl):
+    cell = make_cell()
+    cell.spec.auto_delete = True
+    ctl.create_cell(cell)
+    ctl.start_cell("default", "default", "default", "c1")
+    cdir = ctl.store.cell_dir("default", "default", "default", "c1") / "main"
+    ctl.runtime.mark_exited(cdir, 0)
+    ctl.reconcile_cell("default", "default", "default", "c1")
+    with pytest.raises(errors.CellNotFound):
+        ctl.get_cell("default", "default", "default", "c1")
+
+
+def test_gpu_pinning_and_release(ctl):
+    cell = make_cell()
+    cell.spec.containers[0].gpus = 2
+    ctl.create_cell(cell)
+    doc = ctl.start_cell("default", "default", "default", "c1")
+    assert doc.status.containers[0].gpu_ids == [0, 1]
+    assert ctl.gpus.free == [2, 3]
+    ctl.delete_cell("default", "default", "default", "c1", force=True)
+    assert ctl.gpus.free == [0, 1, 2, 3]
+
+
+def test_gpu_exhaustion(ctl):
+    cell = make_cell()
+    cell.spec.containers[0].gpus = 5
+    ctl.create_cell(cell)
+    with pytest.raises(errors.GPUUnavailable):
+        ctl.start_cell("default", "default", "default", "c1")
+
+
+def test_apply_pipeline_and_diff(ctl):
+    yaml_text = """
+apiVersion: v1beta1
+kind: Realm
+metadata: {name: prod}
+---
+apiVersion: v1beta1
+kind: Space
+metadata: {name: web}
+spec:
+  realmId: prod
+  network:
+    egress:
+      default: deny
+      allow:
+        - cidr: 10.0.0.0/8
+---
+apiVersion: v1beta1
+kind: Stack
+metadata: {name: app}
+spec: {realmId: prod, spaceId: web}
+---
+apiVersion: v1beta1
+kind: Cell
+metadata: {name: worker}
+spec:
+  realmId: prod
+  spaceId: web
+  stackId: app
+  containers:
+    - id: main
+      image: none
+      command: sleep
+      args: ["60"]
+"""
+    results = ctl.apply_documents(yaml_text)
+    assert [(r.kind, r.action) for r in results] == [
+        ("Realm", "created"), ("Space", "created"), ("Stack", "created"),
+        ("Cell", "created")]
+    # re-apply: everything unchanged
+    results = ctl.apply_documents(yaml_text)
+    assert all(r.action == "unchanged" for r in results), [
+        (r.kind, r.action, r.error) for r in results]
+    # breaking change to the cell -> recreate
+    results = ctl.apply_documents(yaml_text.replace('args: ["60"]',
+                                                    'args: ["120"]'))
+    cellres = [r for r in results if r.kind == "Cell"][0]
+    assert cellres.action == "recreated"
+
+
+def test_apply_validation_failure_reported(ctl):
+    bad = """
+apiVersion: v1beta1
+kind: Cell
+metadata: {name: bad}
+spec:
+  realmId: default
+  spaceId: default
+  stackId: default
+  containers: []
+"""
+    res = ctl.apply_documents(bad)
+    assert res[0].action == "failed" and "non-empty" in res[0].error
+
+
+def test_blueprint_run_and_outofsync(ctl):
+    bp = api.CellBlueprintDoc(
+        metadata=api.Metadata(name="agent"),
+        spec=api.CellBlueprintSpec(
+            realm_id="default", space_id="default", name_prefix="agent",
+            params=[api.BlueprintParam(name="CMD", default="sleep")],
+            template={
+                "spec": {
+                    "realmId": "default", "spaceId": "default",
+                    "stackId": "default",
+                    "containers": [{"id": "main", "image": "none",
+                                    "command": "${CMD}", "args": ["5"]}],
+                }
+            }))
+    ctl.put_blueprint(bp)
+    doc = ctl.run_from_blueprint("default", "default", "default", "agent", {})
+    assert doc.metadata.name.startswith("agent-")
+    assert doc.status.state == api.STATE_READY
+    assert doc.spec.provenance.binding_kind == "blueprint"
+    # no drift yet
+    rec = ctl.reconcile_cell("default", "default", "default",
+                             doc.metadata.name)
+    assert not rec.status.out_of_sync
+    # mutate the blueprint -> reconcile flags OutOfSync
+    bp.spec.template["spec"]["containers"][0]["args"] = ["99"]
+    ctl.put_blueprint(bp)
+    rec = ctl.reconcile_cell("default", "default", "default",
+                             doc.metadata.name)
+    assert rec.status.out_of_sync
+    assert "Breaking" in rec.status.out_of_sync_reason
+
+
+def test_session_wallclock_deadline(ctl):
+    ses = api.SessionDoc(
+        metadata=api.Metadata(name="s1"),
+        spec=api.SessionSpec(stack_id="default", gpus=1,
+                             lifetime=api.SessionLifetime(wall_clock="30m")))
+    ctl.create_session(ses)
+    got = ctl.get_session("default", "default", "default", "s1")
+    assert got.status.state == api.STATE_RUNNING
+    assert got.status.gpu_ids == [0]
+    # a cell lives in the stack; the session teardown must remove it
+    ctl.create_cell(make_cell())
+    ctl.start_cell("default", "default", "default", "c1")
+    ctl.now.t += 29 * 60
+    ctl.reconcile_sessions()
+    assert ctl.get_session("default", "default", "default",
+                           "s1").status.state == api.STATE_RUNNING
+    ctl.now.t += 2 * 60
+    ctl.reconcile_sessions()
+    got = ctl.get_session("default", "default", "default", "s1")
+    assert got.status.state == api.STATE_TERMINATED
+    assert got.status.ended_at
+    assert ctl.gpus.free == [0, 1, 2, 3]
+    with pytest.raises(errors.CellNotFound):
+        ctl.get_cell("default", "default", "default", "c1")
+
+
+def test_secret_env_injection(ctl):
+    ctl.put_secret(api.SecretDoc(
+        metadata=api.Metadata(name="api-key"),
+        spec=api.SecretSpec(realm_id="default", space_id="default",
+                            data={"KEY": "sk-123"})))
+    cell = make_cell()
+    cell.spec.containers[0].secrets = [
+        api.ContainerSecret(name="api-key", env="ANTHROPIC_API_KEY")]
+    ctl.create_cell(cell)
+    env = ctl._container_env(
+        ctl.get_cell("default", "default", "default", "c1"),
+        cell.spec.containers[0], [])
+    assert "ANTHROPIC_API_KEY=sk-123" in env
+
+
+def test_scope_delete_cascade_guard(ctl):
+    ctl.create_cell(make_cell())
+    with pytest.raises(errors.NotEmpty):
+        ctl.delete_stack("default", "default", "default")
+    ctl.delete_stack("default", "default", "default", cascade=True)
+    with pytest.raises(errors.StackNotFound):
+        ctl.get_stack("default", "default", "default")
+
+
+def test_subnet_allocation_per_space(ctl):
+    ctl.create_space(api.SpaceDoc(metadata=api.Metadata(name="s2"),
+                                  spec=api.SpaceSpec(realm_id="default")))
+    subs = {ctl.subnets.lookup("default", s)
+            for s in ("default", "s2")}
+    assert len(subs) == 2
+    for s in subs:
+        assert s.startswith("10.88.") and s.endswith(".0/24")
