@@ -1,0 +1,572 @@
+"""kuke — the CLI. `python -m kukeon_amd.cli` or the bin/kuke wrapper.
+
+Verb surface (reference cmd/kuke parity): init, apply, run, attach, get,
+create, delete, purge, start, stop, kill, restart, log, status, doctor,
+session, daemon, version. Talks JSON-RPC to kukeond over the unix socket;
+`--local` (or a missing daemon socket during init) promotes verbs in-process
+over a Controller, like the reference's image path.
+"""
+from __future__ import annotations
+
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+import click
+import yaml
+
+from kukeon_amd.api import errors
+from kukeon_amd.api import v1beta1 as api
+from kukeon_amd.api.client import DEFAULT_SOCKET, LocalClient, UnixClient
+
+DEFAULT_RUN_PATH = os.environ.get("KUKEON_RUN_PATH", "/run/kukeon")
+
+
+def load_client_config() -> dict:
+    p = Path(os.environ.get("KUKE_CONFIG",
+                            os.path.expanduser("~/.kuke/kuke.yaml")))
+    if p.exists():
+        try:
+            doc = yaml.safe_load(p.read_text()) or {}
+            return doc.get("spec", doc)
+        except Exception:
+            return {}
+    return {}
+
+
+class Ctx:
+    def __init__(self, socket_path, run_path, local, realm, space, stack):
+        cfg = load_client_config()
+        self.socket_path = socket_path or cfg.get("socket") or os.environ.get(
+            "KUKEOND_SOCKET") or str(Path(run_path) / "kukeond.sock")
+        self.run_path = run_path
+        self.local = local
+        self.realm = realm or cfg.get("defaultRealm") or "default"
+        self.space = space or cfg.get("defaultSpace") or "default"
+        self.stack = stack or cfg.get("defaultStack") or "default"
+        self._client = None
+
+    @property
+    def client(self):
+        if self._client is None:
+            if self.local or not os.path.exists(self.socket_path):
+                self._client = LocalClient(self._controller())
+            else:
+                self._client = UnixClient(self.socket_path)
+        return self._client
+
+    def _controller(self):
+        from kukeon_amd.controller.core import Controller
+        from kukeon_amd.runtime.cgroup import CgroupManager
+        from kukeon_amd.netpolicy import make_enforcer
+        ctl = Controller(self.run_path, cgroups=CgroupManager(),
+                         enforcer=make_enforcer())
+        return ctl
+
+
+pass_ctx = click.make_pass_decorator(Ctx)
+
+
+@click.group()
+@click.option("--socket", "socket_path", default=None,
+              help="kukeond unix socket path")
+@click.option("--run-path", default=DEFAULT_RUN_PATH, show_default=True)
+@click.option("--local", is_flag=True, help="run verbs in-process")
+@click.option("--realm", default=None)
+@click.option("--space", default=None)
+@click.option("--stack", default=None)
+@click.pass_context
+def cli(ctx, socket_path, run_path, local, realm, space, stack):
+    ctx.obj = Ctx(socket_path, run_path, local, realm, space, stack)
+
+
+def _die(e: Exception):
+    click.echo(f"error: {e}", err=True)
+    sys.exit(1)
+
+
+@cli.command()
+@pass_ctx
+def init(ctx):
+    """Bootstrap the run path, hierarchies and cgroups."""
+    ctl = ctx._controller()
+    ctl.bootstrap()
+    from kukeon_amd.daemon.server import verify_or_write_instance
+    verify_or_write_instance(Path(ctx.run_path))
+    click.echo(f"initialized kukeon at {ctx.run_path} "
+               f"(cgroups: {ctl.cgroups.mode}, "
+               f"gpus: {ctl.gpus.devices or 'none'})")
+
+
+@cli.command()
+@click.option("-f", "--file", "files", multiple=True, required=True,
+              type=click.Path(exists=True, allow_dash=True))
+@click.option("--team", default="")
+@pass_ctx
+def apply(ctx, files, team):
+    """Apply declarative manifests (multi-doc YAML)."""
+    for fp in files:
+        text = (sys.stdin.read() if fp == "-" else
+                Path(fp).read_text())
+        try:
+            results = ctx.client.ApplyDocuments(yaml=text, team=team)
+        except errors.KukeonError as e:
+            _die(e)
+        rc = 0
+        for r in results:
+            line = f"{r['kind'].lower()}/{r['name']} {r['action']}"
+            if r["error"]:
+                line += f": {r['error']}"
+                rc = 1
+            click.echo(line)
+        if rc:
+            sys.exit(rc)
+
+
+@cli.command()
+@click.option("-f", "--file", "file_", type=click.Path(exists=True))
+@click.option("-b", "--blueprint", default=None)
+@click.option("-c", "--config", "config_", default=None)
+@click.option("-p", "--param", "params", multiple=True,
+              help="KEY=VALUE blueprint params")
+@click.option("--env", "env_", multiple=True, help="KEY=VALUE runtime env")
+@click.option("--name", default=None)
+@click.option("--rm", "auto_delete", is_flag=True)
+@click.option("--attach/--no-attach", default=True)
+@pass_ctx
+def run(ctx, file_, blueprint, config_, params, env_, name, auto_delete,
+        attach):
+    """Create (or reuse) a cell and attach to it."""
+    kv = dict(p.split("=", 1) for p in params)
+    try:
+        if file_:
+            raw = yaml.safe_load(Path(file_).read_text())
+            doc = api.CellDoc.from_dict(raw)
+            doc.spec.auto_delete = doc.spec.auto_delete or auto_delete
+            doc.spec.realm_id = doc.spec.realm_id or ctx.realm
+            doc.spec.space_id = doc.spec.space_id or ctx.space
+            doc.spec.stack_id = doc.spec.stack_id or ctx.stack
+            try:
+                cell = ctx.client.GetCell(
+                    realm=doc.spec.realm_id, space=doc.spec.space_id,
+                    stack=doc.spec.stack_id, name=doc.metadata.name)
+                click.echo(f"cell {doc.metadata.name} exists — starting")
+            except errors.NotFound:
+                ctx.client.CreateCell(doc=doc.to_dict(),
+                                      runtimeEnv=list(env_))
+            cell = ctx.client.StartCell(
+                realm=doc.spec.realm_id, space=doc.spec.space_id,
+                stack=doc.spec.stack_id, name=doc.metadata.name)
+        elif blueprint:
+            cell = ctx.client.RunFromBlueprint(
+                realm=ctx.realm, space=ctx.space, stack=ctx.stack,
+                blueprint=blueprint, params=kv, env=list(env_), name=name)
+        elif config_:
+            cell = ctx.client.RunFromConfig(
+                realm=ctx.realm, space=ctx.space, stack=ctx.stack,
+                config=config_, params=kv, name=name)
+        else:
+            _die(errors.InvalidArgument("one of -f / -b / -c is required"))
+    except errors.KukeonError as e:
+        _die(e)
+    cname = cell["metadata"]["name"]
+    click.echo(f"cell {cname}: {cell['status']['state']}")
+    has_attachable = any(c.get("attachable")
+                         for c in cell["spec"]["containers"])
+    if attach and has_attachable:
+        _attach_cell(ctx, cell["spec"]["realmId"], cell["spec"]["spaceId"],
+                     cell["spec"]["stackId"], cname)
+
+
+def _attach_cell(ctx, realm, space, stack, name, timeout=10.0):
+    from kukeon_amd.tty import attach as attach_mod
+    deadline = time.monotonic() + timeout
+    path = None
+    while time.monotonic() < deadline:
+        try:
+            path = ctx.client.AttachContainer(realm=realm, space=space,
+                                              stack=stack, name=name)
+            path = path["hostSocketPath"]
+            if attach_mod.ping(path):
+                break
+        except errors.KukeonError:
+            pass
+        time.sleep(0.2)
+    else:
+        _die(errors.AttachPingTimeout(f"cell {name} tty never became ready"))
+    click.echo("(attached — detach: Ctrl-] Ctrl-])", err=True)
+    rc = attach_mod.attach(path)
+    click.echo("", err=True)
+    sys.exit(rc if rc != 2 else 0)
+
+
+@cli.command()
+@click.argument("name")
+@pass_ctx
+def attach(ctx, name):
+    """Attach to a running cell's tty."""
+    _attach_cell(ctx, ctx.realm, ctx.space, ctx.stack, name)
+
+
+KIND_ALIASES = {
+    "realm": "Realm", "realms": "Realm", "space": "Space", "spaces": "Space",
+    "stack": "Stack", "stacks": "Stack", "cell": "Cell", "cells": "Cell",
+    "session": "Session", "sessions": "Session", "secret": "Secret",
+    "secrets": "Secret", "blueprint": "CellBlueprint",
+    "blueprints": "CellBlueprint", "config": "CellConfig",
+    "configs": "CellConfig", "volume": "Volume", "volumes": "Volume",
+}
+
+
+@cli.command()
+@click.argument("kind")
+@click.argument("name", required=False)
+@click.option("-o", "--output", default="table",
+              type=click.Choice(["table", "yaml", "json"]))
+@pass_ctx
+def get(ctx, kind, name, output):
+    """Get resources: kuke get cells | kuke get cell NAME -o yaml."""
+    k = KIND_ALIASES.get(kind.lower())
+    if k is None:
+        _die(errors.InvalidArgument(f"unknown kind {kind}"))
+    c, r, s, st = ctx.client, ctx.realm, ctx.space, ctx.stack
+    try:
+        if k == "Realm":
+            docs = [c.GetRealm(name=name)] if name else c.ListRealms()
+        elif k == "Space":
+            docs = [c.GetSpace(realm=r, name=name)] if name else \
+                c.ListSpaces(realm=r)
+        elif k == "Stack":
+            docs = [c.GetStack(realm=r, space=s, name=name)] if name else \
+                c.ListStacks(realm=r, space=s)
+        elif k == "Cell":
+            docs = [c.GetCell(realm=r, space=s, stack=st, name=name)] \
+                if name else c.ListCells(realm=r, space=s, stack=st)
+        elif k == "Session":
+            docs = [c.GetSession(realm=r, space=s, stack=st, name=name)] \
+                if name else c.ListSessions()
+        elif k == "Secret":
+            docs = ([c.GetSecret(realm=r, space=s, name=name)] if name else
+                    [{"metadata": {"name": n}, "kind": "Secret",
+                      "status": {"state": "-"}}
+                     for n in c.ListSecrets(realm=r, space=s)])
+        elif k == "CellBlueprint":
+            docs = ([c.GetBlueprint(realm=r, space=s, name=name)] if name
+                    else [{"metadata": {"name": n}, "kind": "CellBlueprint",
+                           "status": {"state": "-"}}
+                          for n in c.ListBlueprints(realm=r, space=s)])
+        elif k == "CellConfig":
+            docs = ([c.GetConfig(realm=r, space=s, name=name)] if name else
+                    [{"metadata": {"name": n}, "kind": "CellConfig",
+                      "status": {"state": "-"}}
+                     for n in c.ListConfigs(realm=r, space=s)])
+        elif k == "Volume":
+            docs = [c.GetVolume(realm=r, space=s, name=name)]
+        else:
+            docs = []
+    except errors.KukeonError as e:
+        _die(e)
+    if output == "yaml":
+        click.echo(yaml.safe_dump_all(docs, sort_keys=False).rstrip())
+    elif output == "json":
+        click.echo(json.dumps(docs, indent=2))
+    else:
+        click.echo(f"{'NAME':<28} {'KIND':<14} {'STATE':<10}")
+        for d in docs:
+            click.echo(f"{d['metadata']['name']:<28} "
+                       f"{d.get('kind', k):<14} "
+                       f"{d.get('status', {}).get('state', '-'):<10}")
+
+
+def _scope_verb(ctx, verb, kind, name, cascade=False, force=False):
+    c, r, s, st = ctx.client, ctx.realm, ctx.space, ctx.stack
+    k = KIND_ALIASES.get(kind.lower())
+    try:
+        if verb == "delete":
+            if k == "Realm":
+                c.DeleteRealm(name=name, cascade=cascade)
+            elif k == "Space":
+                c.DeleteSpace(realm=r, name=name, cascade=cascade)
+            elif k == "Stack":
+                c.DeleteStack(realm=r, space=s, name=name, cascade=cascade)
+            elif k == "Cell":
+                c.DeleteCell(realm=r, space=s, stack=st, name=name,
+                             force=force)
+            elif k == "Session":
+                c.DeleteSession(realm=r, space=s, stack=st, name=name)
+            elif k == "Secret":
+                c.DeleteSecret(realm=r, space=s, name=name)
+            elif k == "CellBlueprint":
+                c.DeleteBlueprint(realm=r, space=s, name=name)
+            elif k == "CellConfig":
+                c.DeleteConfig(realm=r, space=s, name=name)
+            elif k == "Volume":
+                c.DeleteVolume(realm=r, space=s, name=name)
+            else:
+                _die(errors.InvalidArgument(f"cannot delete kind {kind}"))
+    except errors.KukeonError as e:
+        _die(e)
+    click.echo(f"{kind}/{name} deleted")
+
+
+@cli.command()
+@click.argument("kind")
+@click.argument("name")
+@click.option("--cascade", is_flag=True)
+@click.option("--force", is_flag=True)
+@pass_ctx
+def delete(ctx, kind, name, cascade, force):
+    """Delete a resource (graceful; --cascade for scopes)."""
+    _scope_verb(ctx, "delete", kind, name, cascade, force)
+
+
+@cli.command()
+@click.argument("kind")
+@click.argument("name")
+@pass_ctx
+def purge(ctx, kind, name):
+    """Force residual-state removal of a cell."""
+    if KIND_ALIASES.get(kind.lower()) != "Cell":
+        _die(errors.InvalidArgument("purge supports cells"))
+    try:
+        ctx.client.PurgeCell(realm=ctx.realm, space=ctx.space,
+                             stack=ctx.stack, name=name)
+    except errors.KukeonError as e:
+        _die(e)
+    click.echo(f"cell/{name} purged")
+
+
+def _lifecycle(ctx, method, name):
+    try:
+        cell = ctx.client.call(method, realm=ctx.realm, space=ctx.space,
+                               stack=ctx.stack, name=name)
+        click.echo(f"cell {name}: {cell['status']['state']}")
+    except errors.KukeonError as e:
+        _die(e)
+
+
+@cli.command()
+@click.argument("name")
+@pass_ctx
+def start(ctx, name):
+    """Start a cell."""
+    _lifecycle(ctx, "StartCell", name)
+
+
+@cli.command()
+@click.argument("name")
+@pass_ctx
+def stop(ctx, name):
+    """Gracefully stop a cell (SIGTERM, 10s, SIGKILL)."""
+    _lifecycle(ctx, "StopCell", name)
+
+
+@cli.command()
+@click.argument("name")
+@pass_ctx
+def kill(ctx, name):
+    """Kill a cell immediately."""
+    _lifecycle(ctx, "KillCell", name)
+
+
+@cli.command()
+@click.argument("name")
+@pass_ctx
+def restart(ctx, name):
+    """Restart a cell."""
+    _lifecycle(ctx, "RestartCell", name)
+
+
+@cli.command()
+@click.argument("name")
+@click.option("--container", default="")
+@click.option("-f", "--follow", is_flag=True)
+@click.option("-n", "--lines", default=100)
+@pass_ctx
+def log(ctx, name, container, follow, lines):
+    """Show a cell's capture/log file (path resolved by the daemon; bytes
+    never cross the RPC)."""
+    try:
+        res = ctx.client.LogPath(realm=ctx.realm, space=ctx.space,
+                                 stack=ctx.stack, name=name,
+                                 container=container)
+    except errors.KukeonError as e:
+        _die(e)
+    path = Path(res["path"])
+    if not path.exists():
+        _die(errors.NotFound(f"no log at {path}"))
+    data = path.read_bytes().splitlines()[-lines:]
+    for line in data:
+        click.echo(line.decode("utf-8", "replace"))
+    if follow:
+        with open(path, "rb") as f:
+            f.seek(0, 2)
+            try:
+                while True:
+                    chunk = f.read(65536)
+                    if chunk:
+                        sys.stdout.buffer.write(chunk)
+                        sys.stdout.flush()
+                    else:
+                        time.sleep(0.25)
+            except KeyboardInterrupt:
+                pass
+
+
+@cli.command()
+@pass_ctx
+def status(ctx):
+    """Daemon/host/state checks."""
+    checks = []
+    sock = Path(ctx.socket_path)
+    checks.append(("daemon socket", sock.exists(), str(sock)))
+    if sock.exists():
+        try:
+            st = ctx.client.Status()
+            checks.append(("daemon rpc", True, f"pid {st['pid']}"))
+            checks.append(("cgroups", st["cgroupMode"] != "none",
+                           st["cgroupMode"]))
+            g = st["gpus"]
+            checks.append(("gpus", True,
+                           f"{len(g['devices'])} devices, "
+                           f"{len(g['free'])} free"))
+        except Exception as e:
+            checks.append(("daemon rpc", False, str(e)))
+    data = Path(ctx.run_path) / "data"
+    checks.append(("state tree", data.is_dir(), str(data)))
+    ok = True
+    for name, good, detail in checks:
+        mark = "ok" if good else "FAIL"
+        ok &= good
+        click.echo(f"[{mark:>4}] {name:<16} {detail}")
+    sys.exit(0 if ok else 1)
+
+
+@cli.command()
+@click.argument("what", default="all")
+@pass_ctx
+def doctor(ctx, what):
+    """Preflight checks (cgroups, gpus, tooling)."""
+    from kukeon_amd.runtime.cgroup import CgroupManager
+    from kukeon_amd.runtime.devices import discover_gpus
+    import shutil as sh
+    cg = CgroupManager()
+    click.echo(f"cgroups: mode={cg.mode} "
+               f"controllers={','.join(cg.available_controllers()) or '-'}")
+    gpus = discover_gpus()
+    click.echo(f"amdgpu: {len(gpus)} device(s) {gpus}")
+    click.echo(f"kfd: {'present' if os.path.exists('/dev/kfd') else 'absent'}")
+    click.echo(f"iptables: {'present' if sh.which('iptables') else 'absent'}")
+    try:
+        import kukeon_amd.ops as ops
+        click.echo(f"hip extension: "
+                   f"{'loaded' if ops.native_available() else 'MISSING'}")
+    except Exception as e:
+        click.echo(f"hip extension: error ({e})")
+
+
+@cli.group()
+def session():
+    """Session lifecycle verbs."""
+
+
+@session.command("create")
+@click.argument("name")
+@click.option("--task", default="")
+@click.option("--owner", default="")
+@click.option("--gpus", default=0)
+@click.option("--wall-clock", default="")
+@click.option("--idle-timeout", default="")
+@pass_ctx
+def session_create(ctx, name, task, owner, gpus, wall_clock, idle_timeout):
+    doc = api.SessionDoc(
+        metadata=api.Metadata(name=name),
+        spec=api.SessionSpec(
+            realm_id=ctx.realm, space_id=ctx.space, stack_id=ctx.stack,
+            owner=owner, task=task, gpus=gpus,
+            lifetime=api.SessionLifetime(wall_clock=wall_clock,
+                                         idle_timeout=idle_timeout)
+            if (wall_clock or idle_timeout) else None))
+    try:
+        res = ctx.client.CreateSession(doc=doc.to_dict())
+    except errors.KukeonError as e:
+        _die(e)
+    click.echo(f"session {name}: {res['status']['state']} "
+               f"gpus={res['status'].get('gpuIds', [])} "
+               f"deadline={res['status'].get('deadline', '-')}")
+
+
+@session.command("close")
+@click.argument("name")
+@pass_ctx
+def session_close(ctx, name):
+    try:
+        res = ctx.client.CloseSession(realm=ctx.realm, space=ctx.space,
+                                      stack=ctx.stack, name=name)
+    except errors.KukeonError as e:
+        _die(e)
+    click.echo(f"session {name}: {res['status']['state']}")
+
+
+@cli.group()
+def daemon():
+    """Daemon management."""
+
+
+@daemon.command("serve")
+@click.option("--reconcile-interval", default=30.0)
+@click.option("--foreground/--no-foreground", default=True)
+@pass_ctx
+def daemon_serve(ctx, reconcile_interval, foreground):
+    import logging
+    logging.basicConfig(
+        level=logging.INFO,
+        format="%(asctime)s %(levelname)s %(name)s %(message)s")
+    ctl = ctx._controller()
+    ctl.bootstrap()
+    from kukeon_amd.daemon.server import Server
+    srv = Server(ctl, ctx.socket_path, reconcile_interval)
+    srv.start()
+    if foreground:
+        srv.wait()
+
+
+@daemon.command("stop")
+@pass_ctx
+def daemon_stop(ctx):
+    try:
+        ctx.client.DaemonStop()
+        click.echo("daemon stopping")
+    except errors.KukeonError as e:
+        _die(e)
+
+
+@daemon.command("status")
+@pass_ctx
+def daemon_status(ctx):
+    try:
+        st = ctx.client.Status()
+        click.echo(json.dumps(st, indent=2))
+    except Exception as e:
+        _die(errors.KukeonError(f"daemon unreachable: {e}"))
+
+
+@cli.command()
+def version():
+    import kukeon_amd
+    click.echo(f"kuke {kukeon_amd.__version__} (MI355X-native)")
+
+
+def main():
+    prog = os.path.basename(sys.argv[0])
+    if prog == "kukeond":
+        sys.argv.insert(1, "daemon")
+        if len(sys.argv) == 2:
+            sys.argv.append("serve")
+    cli(prog_name="kuke")
+
+
+if __name__ == "__main__":
+    main()

@@ -8,15 +8,24 @@ import time
 from typing import Optional
 
 
-def proc_starttime(pid: int) -> Optional[int]:
-    """Kernel start time (clock ticks) of pid, None if gone."""
+def _stat_fields(pid: int) -> Optional[list]:
     try:
         with open(f"/proc/{pid}/stat", "rb") as f:
             data = f.read().decode("utf-8", "replace")
-        # field 22, but comm may contain spaces/parens: split after last ')'
-        rest = data.rsplit(")", 1)[1].split()
-        return int(rest[19])
+        # comm may contain spaces/parens: split after the last ')'
+        return data.rsplit(")", 1)[1].split()
     except (FileNotFoundError, ProcessLookupError, IndexError, ValueError):
+        return None
+
+
+def proc_starttime(pid: int) -> Optional[int]:
+    """Kernel start time (clock ticks) of pid; None if gone or zombie."""
+    rest = _stat_fields(pid)
+    if rest is None or rest[0] == "Z":  # zombies are dead for our purposes
+        return None
+    try:
+        return int(rest[19])
+    except (IndexError, ValueError):
         return None
 
 

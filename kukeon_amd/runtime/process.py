@@ -77,18 +77,29 @@ class Runtime:
 class ProcessRuntime(Runtime):
     def __init__(self, cgroups: Optional[CgroupManager] = None):
         self.cgroups = cgroups or CgroupManager(enabled=False)
+        self._children: List[subprocess.Popen] = []
+
+    def _reap(self) -> None:
+        """Collect exited shim children so they don't linger as zombies."""
+        self._children = [p for p in self._children if p.poll() is None]
 
     # ------------------------------------------------------------------
     def _spawn_shim(self, cdir: Path, args: List[str],
                     cgroup_rel: str) -> int:
         cdir.mkdir(parents=True, exist_ok=True)
         logf = open(cdir / "shim.log", "ab")
+        env = dict(os.environ)
+        pkg_root = str(Path(__file__).resolve().parent.parent.parent)
+        env["PYTHONPATH"] = pkg_root + (
+            ":" + env["PYTHONPATH"] if env.get("PYTHONPATH") else "")
         p = subprocess.Popen(
             [sys.executable, "-m", "kukeon_amd.tty.shim"] + args,
             stdin=subprocess.DEVNULL, stdout=logf, stderr=logf,
-            start_new_session=True, close_fds=True,
+            start_new_session=True, close_fds=True, env=env,
             cwd=str(cdir))
         logf.close()
+        self._children.append(p)
+        self._reap()
         if cgroup_rel:
             self.cgroups.attach(cgroup_rel, p.pid)
         return p.pid
@@ -122,9 +133,15 @@ class ProcessRuntime(Runtime):
         _write_json(cdir / "spawn.json", spawn)
         pid = self._spawn_shim(cdir, ["--dir", str(cdir)], cgroup_rel)
         # wait briefly for the shim to record the workload
-        for _ in range(100):
+        for _ in range(300):
             if (cdir / "runtime.json").exists():
                 break
+            if not proc.alive(pid) and not (cdir / "status.json").exists():
+                log_tail = ""
+                with contextlib.suppress(OSError):
+                    log_tail = (cdir / "shim.log").read_text()[-500:]
+                raise RuntimeError(
+                    f"container shim died at startup: {log_tail}")
             time.sleep(0.01)
         return pid
 
@@ -140,6 +157,7 @@ class ProcessRuntime(Runtime):
 
     # ------------------------------------------------------------------
     def probe(self, cdir: Path) -> ContainerProbe:
+        self._reap()
         rt = _read_json(cdir / "runtime.json")
         status = _read_json(cdir / "status.json")
         if rt is None and status is None:
@@ -159,6 +177,7 @@ class ProcessRuntime(Runtime):
         return p
 
     def stop(self, cdir: Path, grace_seconds: float = 10.0) -> None:
+        self._reap()
         rt = _read_json(cdir / "runtime.json")
         if not rt:
             return

@@ -125,7 +125,14 @@ class Shim:
         except FileNotFoundError:
             pass
         srv = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
-        srv.bind(str(sock_path))
+        # bind via a relative path: the absolute one can exceed SUN_PATH
+        # (107 bytes); clients reach it through the daemon's short symlink
+        cwd = os.getcwd()
+        try:
+            os.chdir(tty_dir)
+            srv.bind("socket")
+        finally:
+            os.chdir(cwd)
         os.chmod(sock_path, 0o660)
         srv.listen(8)
         srv.setblocking(False)
