@@ -56,30 +56,27 @@ def main() -> None:
         device = "cpu"
 
     cfg = MODEL_PRESETS[args.model]()
+    tp_size = world if args.model == "llama-3-70b" else 1
+    # MoE decode runs eager: per-expert GEMM sizes are data-dependent.
+    # TP ranks share a sampling seed (identical draws, no collective needed).
     ecfg = EngineConfig(max_model_len=args.max_model_len,
                         max_sessions=max(args.sessions, 8),
-                        use_graphs=not args.no_graphs and device != "cpu",
-                        seed=1234 + rank)
+                        use_graphs=(not args.no_graphs and device != "cpu"
+                                    and not cfg.is_moe),
+                        seed=1234 + rank // tp_size)
     if device == "cpu":
         ecfg.num_kv_blocks = (args.sessions *
                               (args.ctx_cap // ecfg.block_size + 2) + 64)
 
-    if args.model == "llama-3-8b":
-        from kukeon_amd.models.llama import LlamaModel
-        model = LlamaModel(cfg, device=device)
-    elif args.model == "tiny-llama":
-        from kukeon_amd.models.llama import LlamaModel
-        model = LlamaModel(cfg, device=device)
-    elif args.model == "mixtral-8x7b":
-        from kukeon_amd.models.mixtral import MixtralModel
-        model = MixtralModel(cfg, device=device)
-    elif args.model == "llama-3-70b":
+    if args.model == "llama-3-70b":
         parallel.init_tensor_parallel(world)
         ecfg.tp_size = world
+    if cfg.is_moe:
+        from kukeon_amd.models.mixtral import MixtralModel
+        model = MixtralModel(cfg, device=device)
+    else:
         from kukeon_amd.models.llama import LlamaModel
         model = LlamaModel(cfg, device=device)
-    else:
-        raise SystemExit(f"unknown model {args.model}")
 
     engine = LLMEngine(model, cfg, ecfg, device=device)
     tp = ecfg.tp_size

@@ -52,6 +52,13 @@ def mixtral_8x7b() -> ModelConfig:
                        rope_theta=1e6, num_experts=8, top_k_experts=2)
 
 
+def tiny_mixtral() -> ModelConfig:
+    return ModelConfig(name="tiny-mixtral", hidden_size=256, num_layers=2,
+                       num_q_heads=2, num_kv_heads=2, head_dim=128,
+                       intermediate_size=256, vocab_size=512,
+                       max_position=512, num_experts=4, top_k_experts=2)
+
+
 def tiny_llama() -> ModelConfig:
     """CPU-test sized model (same head_dim=128 the kernels require)."""
     return ModelConfig(name="tiny-llama", hidden_size=256, num_layers=2,
@@ -65,6 +72,7 @@ MODEL_PRESETS = {
     "llama-3-70b": llama3_70b,
     "mixtral-8x7b": mixtral_8x7b,
     "tiny-llama": tiny_llama,
+    "tiny-mixtral": tiny_mixtral,
 }
 
 
