@@ -971,6 +971,41 @@ class Controller:
                 latest = m if latest is None else max(latest, m)
         return latest
 
+    def provision_modelhub_cell(self, model: str = "llama-3-8b",
+                                gpus: int = 1, socket_path: str = "",
+                                extra_args: Optional[List[str]] = None,
+                                start: bool = True) -> api.CellDoc:
+        """Provision the inference server as a system-realm cell (the same
+        pattern the reference uses for kukeond itself)."""
+        import sys as _sys
+        self.ensure_realm(naming.SYSTEM_REALM)
+        self.ensure_space(naming.SYSTEM_REALM, naming.SYSTEM_SPACE)
+        self.ensure_stack(naming.SYSTEM_REALM, naming.SYSTEM_SPACE,
+                          naming.SYSTEM_STACK)
+        sock = socket_path or str(self.run_path / "modelhub.sock")
+        args = ["-m", "kukeon_amd.serve.server", "--model", model,
+                "--socket", sock] + list(extra_args or [])
+        doc = api.CellDoc(
+            metadata=api.Metadata(name="modelhub"),
+            spec=api.CellSpec(
+                realm_id=naming.SYSTEM_REALM, space_id=naming.SYSTEM_SPACE,
+                stack_id=naming.SYSTEM_STACK,
+                containers=[api.ContainerSpec(
+                    id="modelhub", image="kukeon.internal/modelhub",
+                    command=_sys.executable, args=args, gpus=gpus,
+                    restart_policy=api.RESTART_ALWAYS)]))
+        try:
+            self.create_cell(doc)
+        except errors.AlreadyExists:
+            pass
+        if start:
+            return self.start_cell(naming.SYSTEM_REALM, naming.SYSTEM_SPACE,
+                                   naming.SYSTEM_STACK, "modelhub")
+        return doc
+
+    def modelhub_socket(self) -> str:
+        return str(self.run_path / "modelhub.sock")
+
     def reconcile_space_networks(self) -> int:
         visited = 0
         for realm in self.store.list_children(self.store.data_root):

@@ -1,0 +1,265 @@
+"""The modelhub server: the agent-backing inference service.
+
+Runs as a system-realm cell (provisioned like kukeond itself — SURVEY.md
+§2.9 integration contract): one process per GPU serving the engine over a
+unix socket. Agent sessions hold named server-side KV contexts that stay
+resident in HBM across turns.
+
+Wire protocol (newline JSON, one connection per client, requests may
+interleave across clients — the engine continuously batches them):
+  {"id": N, "method": "generate", "params": {"session": "s1",
+      "tokens": [...], "max_new_tokens": 128, "temperature": 0.7,
+      "top_k": 50, "top_p": 0.9}}
+  -> {"id": N, "result": {"tokens": [...], "context_len": M}}
+  other methods: ping, stats, release (free a session's KV).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import logging
+import os
+import queue
+import socket
+import socketserver
+import threading
+from dataclasses import dataclass
+from typing import Dict, List, Optional
+
+from kukeon_amd.engine.config import (EngineConfig, MODEL_PRESETS,
+                                      SamplingParams)
+from kukeon_amd.engine.engine import LLMEngine
+from kukeon_amd.engine.kv_cache import SequenceKV
+
+log = logging.getLogger("kukeon.modelhub")
+
+
+@dataclass
+class _Pending:
+    req_id: int
+    reply: "queue.Queue"
+    tokens: List[int]
+
+
+class ModelhubServer:
+    def __init__(self, model, cfg, ecfg: EngineConfig, socket_path: str,
+                 device: str = "cuda:0"):
+        self.engine = LLMEngine(model, cfg, ecfg, device=device)
+        self.socket_path = socket_path
+        self.sessions: Dict[str, SequenceKV] = {}
+        self._submit: "queue.Queue" = queue.Queue()
+        self._pending: Dict[int, _Pending] = {}
+        self._lock = threading.Lock()
+        self._stop = threading.Event()
+        self._srv = None
+        self._threads = []
+
+    # ---- engine loop --------------------------------------------------
+    def _engine_loop(self):
+        self.engine.capture_all()
+        while not self._stop.is_set():
+            # drain submissions
+            try:
+                while True:
+                    fn = self._submit.get(
+                        timeout=None if not self.engine.has_work() and
+                        self._submit.empty() else 0.0)
+                    if fn is None:
+                        return
+                    fn()
+            except queue.Empty:
+                pass
+            if not self.engine.has_work():
+                continue
+            outs = self.engine.step()
+            with self._lock:
+                for o in outs:
+                    p = self._pending.get(o.req_id)
+                    if p is None:
+                        continue
+                    p.tokens.extend(o.new_tokens)
+                    if o.finished:
+                        del self._pending[o.req_id]
+                        p.reply.put({"tokens": p.tokens})
+
+    # ---- request handling --------------------------------------------
+    def handle(self, method: str, params: dict) -> dict:
+        if method == "ping":
+            return {"ok": True, "pid": os.getpid()}
+        if method == "stats":
+            kv = self.engine.kv
+            return {
+                "sessions": len(self.sessions),
+                "running": self.engine.num_running,
+                "waiting": len(self.engine.waiting),
+                "kv_blocks_total": kv.num_blocks,
+                "kv_blocks_free": kv.allocator.num_free,
+            }
+        if method == "release":
+            name = params["session"]
+            done: "queue.Queue" = queue.Queue()
+
+            def _do_release():
+                kv = self.sessions.pop(name, None)
+                if kv is not None:
+                    self.engine.free_sequence(kv)
+                done.put({})
+            self._submit.put(_do_release)
+            return done.get(timeout=60)
+        if method == "generate":
+            return self._generate(params)
+        raise ValueError(f"unknown method {method}")
+
+    def _generate(self, params: dict) -> dict:
+        name = params["session"]
+        tokens = list(params["tokens"])
+        sp = SamplingParams(
+            temperature=float(params.get("temperature", 0.7)),
+            top_k=int(params.get("top_k", 50)),
+            top_p=float(params.get("top_p", 0.9)),
+            max_new_tokens=int(params.get("max_new_tokens", 128)))
+        reply: "queue.Queue" = queue.Queue()
+
+        def _do_submit():
+            kv = self.sessions.get(name)
+            if kv is None:
+                kv = SequenceKV(self.engine.ecfg.block_size)
+                self.sessions[name] = kv
+            try:
+                rid = self.engine.add_request(kv, tokens, sp)
+            except Exception as e:  # overlong context etc.
+                reply.put({"error": str(e)})
+                return
+            with self._lock:
+                self._pending[rid] = _Pending(rid, reply, [])
+        self._submit.put(_do_submit)
+        out = reply.get(timeout=600)
+        if "error" in out:
+            raise ValueError(out["error"])
+        kv = self.sessions.get(name)
+        out["context_len"] = kv.num_tokens if kv else 0
+        return out
+
+    # ---- server plumbing ----------------------------------------------
+    def start(self):
+        sp = self.socket_path
+        if os.path.exists(sp):
+            os.unlink(sp)
+        os.makedirs(os.path.dirname(sp) or ".", exist_ok=True)
+        hub = self
+
+        class Handler(socketserver.StreamRequestHandler):
+            def handle(self):
+                for line in self.rfile:
+                    line = line.strip()
+                    if not line:
+                        continue
+                    try:
+                        req = json.loads(line)
+                        result = hub.handle(req.get("method", ""),
+                                            req.get("params") or {})
+                        resp = {"id": req.get("id"), "result": result}
+                    except Exception as e:  # noqa: BLE001
+                        resp = {"id": req.get("id") if isinstance(req, dict)
+                                else None, "error": str(e)}
+                    try:
+                        self.wfile.write((json.dumps(resp) + "\n").encode())
+                        self.wfile.flush()
+                    except OSError:
+                        return
+
+        cwd = os.getcwd()
+        bind_path = sp
+        if len(sp) > 100:
+            os.chdir(os.path.dirname(sp))
+            bind_path = os.path.basename(sp)
+        try:
+            self._srv = socketserver.ThreadingUnixStreamServer(
+                bind_path, Handler)
+        finally:
+            os.chdir(cwd)
+        self._srv.daemon_threads = True
+        os.chmod(sp, 0o666)
+        t = threading.Thread(target=self._srv.serve_forever, daemon=True)
+        t.start()
+        self._threads.append(t)
+        te = threading.Thread(target=self._engine_loop, daemon=True)
+        te.start()
+        self._threads.append(te)
+        log.info("modelhub serving on %s", sp)
+
+    def stop(self):
+        self._stop.set()
+        self._submit.put(None)
+        if self._srv:
+            self._srv.shutdown()
+            self._srv.server_close()
+
+
+class ModelhubClient:
+    def __init__(self, socket_path: str, timeout: float = 600.0):
+        self.sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        self.sock.settimeout(timeout)
+        self.sock.connect(socket_path)
+        self._rf = self.sock.makefile("rb")
+        self._id = 0
+        self._lock = threading.Lock()
+
+    def call(self, method: str, **params):
+        with self._lock:
+            self._id += 1
+            self.sock.sendall((json.dumps(
+                {"id": self._id, "method": method, "params": params}) +
+                "\n").encode())
+            line = self._rf.readline()
+        if not line:
+            raise ConnectionError("modelhub closed the connection")
+        resp = json.loads(line)
+        if "error" in resp:
+            raise RuntimeError(resp["error"])
+        return resp["result"]
+
+    def generate(self, session: str, tokens: List[int], **kw):
+        return self.call("generate", session=session, tokens=tokens, **kw)
+
+    def close(self):
+        self.sock.close()
+
+
+def main(argv: Optional[List[str]] = None) -> None:
+    ap = argparse.ArgumentParser("kukeon-modelhub")
+    ap.add_argument("--model", default="llama-3-8b")
+    ap.add_argument("--socket", default="/run/kukeon/modelhub.sock")
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--max-model-len", type=int, default=4096)
+    ap.add_argument("--max-sessions", type=int, default=256)
+    ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--kv-blocks", type=int, default=0)
+    args = ap.parse_args(argv)
+    logging.basicConfig(level=logging.INFO)
+    import torch
+    device = args.device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+    cfg = MODEL_PRESETS[args.model]()
+    ecfg = EngineConfig(
+        max_model_len=args.max_model_len, max_sessions=args.max_sessions,
+        num_kv_blocks=args.kv_blocks or (512 if device == "cpu" else 0),
+        use_graphs=not args.no_graphs and device != "cpu" and not cfg.is_moe)
+    if cfg.is_moe:
+        from kukeon_amd.models.mixtral import MixtralModel
+        model = MixtralModel(cfg, device=device)
+    else:
+        from kukeon_amd.models.llama import LlamaModel
+        model = LlamaModel(cfg, device=device)
+    hub = ModelhubServer(model, cfg, ecfg, args.socket, device=device)
+    hub.start()
+    import signal
+    stop = threading.Event()
+    signal.signal(signal.SIGTERM, lambda *_: stop.set())
+    signal.signal(signal.SIGINT, lambda *_: stop.set())
+    while not stop.is_set():
+        stop.wait(1.0)
+    hub.stop()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,87 @@
+"""Modelhub server tests on CPU (tiny model): concurrent sessions over the
+unix socket, persistent multi-turn KV, release, and the system-cell
+provisioning path through the controller + process runtime."""
+import threading
+import time
+import uuid
+
+import pytest
+
+from kukeon_amd.engine.config import EngineConfig, tiny_llama
+from kukeon_amd.models.llama import LlamaModel
+from kukeon_amd.serve.server import ModelhubClient, ModelhubServer
+
+
+@pytest.fixture
+def hub():
+    sock = f"/tmp/mh-{uuid.uuid4().hex[:10]}.sock"
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=256, max_sessions=8, num_kv_blocks=256,
+                        use_graphs=False)
+    model = LlamaModel(cfg, device="cpu")
+    h = ModelhubServer(model, cfg, ecfg, sock, device="cpu")
+    h.start()
+    yield h, sock
+    h.stop()
+
+
+def test_generate_multi_turn_and_sessions(hub):
+    h, sock = hub
+    c = ModelhubClient(sock, timeout=120)
+    r1 = c.generate("alice", [1, 2, 3, 4], max_new_tokens=4, temperature=0.0)
+    assert len(r1["tokens"]) == 4
+    assert r1["context_len"] == 4 + 4 - 1  # final token pending
+    r2 = c.generate("alice", [9, 9], max_new_tokens=3, temperature=0.0)
+    assert r2["context_len"] == r1["context_len"] + 1 + 2 + 3 - 1
+    rb = c.generate("bob", [5, 6, 7], max_new_tokens=2, temperature=0.0)
+    assert len(rb["tokens"]) == 2
+    st = c.call("stats")
+    assert st["sessions"] == 2
+    c.call("release", session="alice")
+    assert c.call("stats")["sessions"] == 1
+    c.close()
+
+
+def test_concurrent_clients_batched(hub):
+    h, sock = hub
+    results = {}
+
+    def worker(name):
+        c = ModelhubClient(sock, timeout=120)
+        results[name] = c.generate(name, [ord(x) % 512 for x in name] * 3,
+                                   max_new_tokens=5, temperature=0.0)
+        c.close()
+
+    ts = [threading.Thread(target=worker, args=(f"s{i}",)) for i in range(4)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=120)
+    assert len(results) == 4
+    assert all(len(r["tokens"]) == 5 for r in results.values())
+
+
+def test_modelhub_as_system_cell(tmp_path):
+    from kukeon_amd.controller.core import Controller
+    from kukeon_amd.serve.server import ModelhubClient
+    ctl = Controller(str(tmp_path / "run"), gpu_devices=[])
+    ctl.bootstrap()
+    sock = f"/tmp/mhc-{uuid.uuid4().hex[:8]}.sock"
+    doc = ctl.provision_modelhub_cell(
+        model="tiny-llama", gpus=0, socket_path=sock,
+        extra_args=["--max-model-len", "128", "--kv-blocks", "64"])
+    assert doc.status.state == "Ready"
+    deadline = time.monotonic() + 60
+    client = None
+    while time.monotonic() < deadline:
+        try:
+            client = ModelhubClient(sock, timeout=60)
+            break
+        except OSError:
+            time.sleep(0.5)
+    assert client is not None, "modelhub cell never served its socket"
+    r = client.generate("agent-1", [1, 2, 3], max_new_tokens=3,
+                        temperature=0.0)
+    assert len(r["tokens"]) == 3
+    client.close()
+    ctl.kill_cell("kuke-system", "kukeon", "kukeon", "modelhub")
