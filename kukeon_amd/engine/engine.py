@@ -99,7 +99,8 @@ class LLMEngine:
         self.d_topp = torch.ones(self.Bmax, dtype=torch.float32, device=d)
         self.d_tokens = torch.zeros(self.Bmax, dtype=torch.int32, device=d)
         self.d_seed = torch.full((1,), ecfg.seed, dtype=torch.int64, device=d)
-        self.d_ws = torch.zeros(self.Bmax, 4, dtype=torch.float32, device=d)
+        self.d_ws = torch.zeros(self.Bmax, 528, dtype=torch.float32,
+                                device=d)  # sampling workspace rows
         self.max_splits = 16
         self.d_tmp_out = {}
         self.d_tmp_ml = {}

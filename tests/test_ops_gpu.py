@@ -161,7 +161,7 @@ def test_sample_greedy_exact():
     tk = torch.zeros(B, dtype=torch.int32, device=DEV)
     tp = torch.ones(B, dtype=torch.float32, device=DEV)
     seed = torch.zeros(1, dtype=torch.int64, device=DEV)
-    ws = torch.zeros(B, 4, dtype=torch.float32, device=DEV)
+    ws = torch.zeros(B, 528, dtype=torch.float32, device=DEV)
     ops.sample(tokens, logits, temps, tk, tp, seed, ws)
     ref = logits.float().argmax(dim=-1).to(torch.int32)
     assert torch.equal(tokens.cpu(), ref.cpu())
@@ -175,7 +175,7 @@ def test_sample_topk_membership():
     tk = torch.full((B,), K, dtype=torch.int32, device=DEV)
     tp = torch.ones(B, dtype=torch.float32, device=DEV)
     seed = torch.zeros(1, dtype=torch.int64, device=DEV)
-    ws = torch.zeros(B, 4, dtype=torch.float32, device=DEV)
+    ws = torch.zeros(B, 528, dtype=torch.float32, device=DEV)
     tokens = torch.zeros(B, dtype=torch.int32, device=DEV)
     # top-k sets computed on bf16-rounded logits (ties at 24-bit granularity
     # can admit a couple extra members — allow K + small slack)
@@ -195,7 +195,7 @@ def test_sample_topp_membership():
     tk = torch.zeros(B, dtype=torch.int32, device=DEV)
     tp = torch.full((B,), 0.7, dtype=torch.float32, device=DEV)
     seed = torch.zeros(1, dtype=torch.int64, device=DEV)
-    ws = torch.zeros(B, 4, dtype=torch.float32, device=DEV)
+    ws = torch.zeros(B, 528, dtype=torch.float32, device=DEV)
     tokens = torch.zeros(B, dtype=torch.int32, device=DEV)
     allowed = []
     for b in range(B):
