@@ -116,6 +116,14 @@ class LLMEngine:
         self.h_params = torch.zeros(self.Bmax, 3, dtype=torch.float32,
                                     pin_memory=pin)
         self._params_dirty = False
+        # numpy views of the pinned staging buffers: per-row scalar writes
+        # through torch cost ~5-10us each; through numpy they are ~50ns,
+        # and the views share the pinned memory so the H2D copies see them
+        self.n_ids = self.h_ids.numpy()
+        self.n_pos = self.h_pos.numpy()
+        self.n_slots = self.h_slots.numpy()
+        self.n_seq_lens = self.h_seq_lens.numpy()
+        self.n_bt = self.h_bt.numpy()
 
         self.graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self._graph_pool = None
@@ -259,8 +267,8 @@ class LLMEngine:
         if req.row >= 0:
             self._rows[req.row] = None
             self._free_rows.append(req.row)
-            self.h_slots[req.row] = -1
-            self.h_seq_lens[req.row] = 0
+            self.n_slots[req.row] = -1
+            self.n_seq_lens[req.row] = 0
             req.row = -1
             self.num_running -= 1
 
@@ -303,25 +311,24 @@ class LLMEngine:
         for row in range(nrows):
             r = self._rows[row]
             if r is None:
-                self.h_slots[row] = -1
-                self.h_seq_lens[row] = 0
+                self.n_slots[row] = -1
+                self.n_seq_lens[row] = 0
                 continue
             active.append(r)
-            self.h_ids[row] = r.output_tokens[-1]
-            self.h_pos[row] = r.kv.num_tokens
-            self.h_slots[row] = self._alloc_for(r, 1)[0]
-            self.h_seq_lens[row] = r.kv.num_tokens
+            self.n_ids[row] = r.output_tokens[-1]
+            self.n_pos[row] = r.kv.num_tokens
+            self.n_slots[row] = self._alloc_for(r, 1)[0]
+            self.n_seq_lens[row] = r.kv.num_tokens
             n = len(r.kv.blocks)
             if n > r.bt_written:
-                self.h_bt[row, r.bt_written: n] = torch.tensor(
-                    r.kv.blocks[r.bt_written:], dtype=torch.int32)
+                self.n_bt[row, r.bt_written: n] = r.kv.blocks[r.bt_written:]
                 r.bt_written = n
         use_graph = self.is_cuda and self.ecfg.use_graphs
         bucket = self._bucket(nrows) if use_graph else nrows
         nb = bucket
         for row in range(nrows, nb):
-            self.h_slots[row] = -1
-            self.h_seq_lens[row] = 0
+            self.n_slots[row] = -1
+            self.n_seq_lens[row] = 0
         self.d_ids[:nb].copy_(self.h_ids[:nb], non_blocking=True)
         self.d_pos[:nb].copy_(self.h_pos[:nb], non_blocking=True)
         self.d_slots[:nb].copy_(self.h_slots[:nb], non_blocking=True)

@@ -60,7 +60,7 @@ class LlamaAttention(nn.Module):
 
     def forward(self, x: torch.Tensor, cos_sin: torch.Tensor,
                 kc: torch.Tensor, vc: torch.Tensor, meta: AttnMeta):
-        qkv = F.linear(x, self.qkv_w)
+        qkv = ops.linear(x, self.qkv_w)
         ops.rope_kv_append(qkv, kc, vc, cos_sin, meta.positions,
                            meta.slot_mapping, self.hq, self.hk, self.d)
         out = torch.empty(x.shape[0], self.hq * self.d, dtype=x.dtype,
@@ -73,7 +73,7 @@ class LlamaAttention(nn.Module):
             ops.prefill_attention(out, qkv, kc, vc, meta.block_table,
                                   meta.seq_lens, meta.q_starts, meta.qb_seq,
                                   meta.qb_start, 0, self.scale)
-        o = F.linear(out, self.o_w)
+        o = ops.linear(out, self.o_w)
         return parallel.tp_all_reduce(o)
 
 
@@ -87,11 +87,11 @@ class LlamaMLP(nn.Module):
         self.down_w = _init_weight((cfg.hidden_size, self.inter), device)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        gu = F.linear(x, self.gate_up_w)
+        gu = ops.linear(x, self.gate_up_w)
         act = torch.empty(x.shape[0], self.inter, dtype=x.dtype,
                           device=x.device)
         ops.silu_mul(act, gu)
-        return parallel.tp_all_reduce(F.linear(act, self.down_w))
+        return parallel.tp_all_reduce(ops.linear(act, self.down_w))
 
 
 class LlamaLayer(nn.Module):
@@ -162,4 +162,4 @@ class LlamaModel(nn.Module):
         return x
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
-        return F.linear(hidden, self.lm_head)
+        return ops.linear(hidden, self.lm_head)

@@ -74,6 +74,30 @@ def sample(tokens, logits, temps, top_k, top_p, seed, workspace) -> None:
     _impl(logits).sample(tokens, logits, temps, top_k, top_p, seed, workspace)
 
 
+_SKINNY_WS = {}
+
+
+def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """F.linear with the weight-streaming skinny-GEMM kernel on the decode
+    path (CUDA, rows <= 64, aligned shapes); hipBLASLt otherwise."""
+    rows = x.shape[0]
+    if (x.is_cuda and x.dim() == 2 and rows <= 64
+            and w.shape[0] % 64 == 0 and w.shape[1] % 32 == 0
+            and x.dtype == torch.bfloat16):
+        N = w.shape[0]
+        out = torch.empty(rows, N, dtype=x.dtype, device=x.device)
+        key = (x.device.index or 0)
+        ws = _SKINNY_WS.get(key)
+        need = 64 * N
+        if ws is None or ws.numel() < need:
+            # grown only outside graph capture (engine warmup runs eager)
+            ws = torch.empty(need, dtype=torch.float32, device=x.device)
+            _SKINNY_WS[key] = ws
+        _native().skinny_gemm(out, x, w, ws)
+        return out
+    return torch.nn.functional.linear(x, w)
+
+
 def moe_gather_tokens(out, input, row_map) -> None:
     _impl(input).moe_gather_tokens(out, input, row_map)
 

@@ -26,6 +26,7 @@ SOURCES = [
     "prefill_attn.hip",
     "sampling.hip",
     "moe.hip",
+    "skinny_gemm.hip",
     "bindings.cpp",
 ]
 

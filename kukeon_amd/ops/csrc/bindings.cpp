@@ -23,6 +23,9 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
                        torch::Tensor q_starts, torch::Tensor qb_seq,
                        torch::Tensor qb_start, int64_t q_offset, double scale);
 void mfma_probe(torch::Tensor out, torch::Tensor a, torch::Tensor b);
+void mfma_probe16(torch::Tensor out, torch::Tensor a, torch::Tensor b);
+void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                 torch::Tensor ws);
 void sample(torch::Tensor tokens, torch::Tensor logits, torch::Tensor temps,
             torch::Tensor top_k, torch::Tensor top_p, torch::Tensor seed,
             torch::Tensor workspace);
@@ -46,6 +49,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "varlen causal paged prefill attention (MFMA)");
   m.def("mfma_probe", &kukeon::mfma_probe,
         "32x32x16 bf16 MFMA fragment-layout probe");
+  m.def("mfma_probe16", &kukeon::mfma_probe16,
+        "16x16x32 bf16 MFMA fragment-layout probe");
+  m.def("skinny_gemm", &kukeon::skinny_gemm,
+        "weight-streaming decode GEMM (M<=64)");
   m.def("sample", &kukeon::sample, "top-k/top-p/temperature sampling");
   m.def("moe_gather_tokens", &kukeon::moe_gather_tokens, "MoE permute");
   m.def("moe_scatter_tokens", &kukeon::moe_scatter_tokens,

@@ -80,25 +80,33 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
     m[g] = -INFINITY; l[g] = 0.f; o0[g] = 0.f; o1[g] = 0.f;
   }
 
+  // stage geometry: each thread owns one 16B chunk of K and V
+  const int stg_t = threadIdx.x / (D / 8);      // token 0..15
+  const int stg_d0 = (threadIdx.x % (D / 8)) * 8;
+  const int stg_dsw = (((stg_d0 / 32) ^ (stg_t & 3)) * 32) + (stg_d0 % 32);
+  uint4 kreg, vreg;
   if (blk_begin < blk_end) {
+    {  // prologue: fetch the first block into registers
+      const int pb0 = block_table[(long)b * max_blocks + blk_begin];
+      kreg = *reinterpret_cast<const uint4*>(
+          k_cache + (((long)pb0 * Hk + hk) * BS + stg_t) * D + stg_d0);
+      vreg = *reinterpret_cast<const uint4*>(
+          v_cache + (((long)pb0 * Hk + hk) * BS + stg_t) * D + stg_d0);
+    }
     for (int bi = blk_begin; bi < blk_end; ++bi) {
-      const int pblock = block_table[(long)b * max_blocks + bi];
-      const unsigned short* kg =
-          k_cache + (((long)pblock * Hk + hk) * BS) * D;
-      const unsigned short* vg =
-          v_cache + (((long)pblock * Hk + hk) * BS) * D;
-      // ---- cooperative stage: each thread one 16B chunk ----
-      {
-        const int t = threadIdx.x / (D / 8);      // token 0..15
-        const int d0 = (threadIdx.x % (D / 8)) * 8;
-        const int subw = d0 / 32;
-        const int dsw = ((subw ^ (t & 3)) * 32) + (d0 % 32);  // K swizzle
-        __syncthreads();
-        *reinterpret_cast<uint4*>(&kbuf[t * D + dsw]) =
-            *reinterpret_cast<const uint4*>(kg + t * D + d0);
-        *reinterpret_cast<uint4*>(&vbuf[t * D + d0]) =
-            *reinterpret_cast<const uint4*>(vg + t * D + d0);
-        __syncthreads();
+      // ---- write the prefetched block, then issue the next block's
+      // loads so their HBM latency hides under this block's compute
+      // (guide T14 async-stage split) ----
+      __syncthreads();
+      *reinterpret_cast<uint4*>(&kbuf[stg_t * D + stg_dsw]) = kreg;
+      *reinterpret_cast<uint4*>(&vbuf[stg_t * D + stg_d0]) = vreg;
+      __syncthreads();
+      if (bi + 1 < blk_end) {
+        const int pbn = block_table[(long)b * max_blocks + bi + 1];
+        kreg = *reinterpret_cast<const uint4*>(
+            k_cache + (((long)pbn * Hk + hk) * BS + stg_t) * D + stg_d0);
+        vreg = *reinterpret_cast<const uint4*>(
+            v_cache + (((long)pbn * Hk + hk) * BS + stg_t) * D + stg_d0);
       }
       if (!active) continue;
       // ---- scores ----

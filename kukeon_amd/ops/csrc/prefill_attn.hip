@@ -63,6 +63,40 @@ __global__ void mfma_probe_kernel(float* __restrict__ c,
   }
 }
 
+__global__ void mfma_probe16_kernel(float* __restrict__ c,
+                                    const unsigned short* __restrict__ a,
+                                    const unsigned short* __restrict__ b) {
+  typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4_t;
+  const int lane = threadIdx.x & 63;
+  const int g = lane >> 4;     // k group
+  const int rc = lane & 15;
+  unsigned int areg[4], breg[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int k0 = 8 * g + 2 * i;
+    areg[i] = (unsigned)a[rc * 32 + k0] | ((unsigned)a[rc * 32 + k0 + 1] << 16);
+    breg[i] = (unsigned)b[k0 * 16 + rc] | ((unsigned)b[(k0 + 1) * 16 + rc] << 16);
+  }
+  f32x4_t acc = {};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+      as_frag(areg[0], areg[1], areg[2], areg[3]),
+      as_frag(breg[0], breg[1], breg[2], breg[3]), acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) c[(4 * g + r) * 16 + rc] = acc[r];
+}
+
+void mfma_probe16(torch::Tensor out, torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.sizes() == torch::IntArrayRef({16, 32}));
+  TORCH_CHECK(b.sizes() == torch::IntArrayRef({32, 16}));
+  TORCH_CHECK(out.sizes() == torch::IntArrayRef({16, 16}));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  mfma_probe16_kernel<<<1, 64, 0, stream>>>(
+      out.data_ptr<float>(),
+      reinterpret_cast<const unsigned short*>(a.data_ptr()),
+      reinterpret_cast<const unsigned short*>(b.data_ptr()));
+  HIP_CHECK_KERNEL();
+}
+
 void mfma_probe(torch::Tensor out, torch::Tensor a, torch::Tensor b) {
   TORCH_CHECK(a.sizes() == torch::IntArrayRef({32, 16}));
   TORCH_CHECK(b.sizes() == torch::IntArrayRef({16, 32}));

@@ -58,6 +58,52 @@ __global__ void rmsnorm_kernel(unsigned short* __restrict__ out,
   }
 }
 
+// Small-batch (decode) variant: one WAVE per row, no cross-wave barrier —
+// at T<=128 the block-per-row kernel is launch/latency-bound (64 blocks on
+// 256 CUs with two __syncthreads); a wave-local reduce halves the latency.
+template <int ITERS, bool FUSED_ADD>
+__global__ void rmsnorm_wave_kernel(unsigned short* __restrict__ out,
+                                    unsigned short* __restrict__ input,
+                                    unsigned short* __restrict__ residual,
+                                    const unsigned short* __restrict__ weight,
+                                    float eps, int H, int T) {
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= T) return;
+  const int lane = threadIdx.x & 63;
+  unsigned short* in_row = input + (long)row * H;
+  unsigned short* res_row = FUSED_ADD ? residual + (long)row * H : nullptr;
+  unsigned short* out_row = out + (long)row * H;
+  float vals[ITERS][8];
+  float ss = 0.f;
+#pragma unroll
+  for (int it = 0; it < ITERS; ++it) {
+    const int base = (it * WAVE + lane) * 8;
+    bf16x8 v = load_bf16x8(in_row + base);
+    if (FUSED_ADD) {
+      bf16x8 r = load_bf16x8(res_row + base);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[it][j] = v.f(j) + r.f(j);
+      *reinterpret_cast<uint4*>(res_row + base) = pack_bf16x8(vals[it]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[it][j] = v.f(j);
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ss += vals[it][j] * vals[it][j];
+  }
+  ss = wave_sum(ss);
+  const float rs = rsqrtf(ss / (float)H + eps);
+#pragma unroll
+  for (int it = 0; it < ITERS; ++it) {
+    const int base = (it * WAVE + lane) * 8;
+    bf16x8 w = load_bf16x8(weight + base);
+    float o[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = vals[it][j] * rs * w.f(j);
+    *reinterpret_cast<uint4*>(out_row + base) = pack_bf16x8(o);
+  }
+}
+
 // Generic fallback for hidden sizes that don't match a template instance.
 template <bool FUSED_ADD>
 __global__ void rmsnorm_kernel_generic(unsigned short* __restrict__ out,
@@ -102,10 +148,17 @@ static void launch_rmsnorm(torch::Tensor& out, torch::Tensor& input,
   auto* res = residual ? reinterpret_cast<unsigned short*>(residual->data_ptr()) : nullptr;
   auto* w = reinterpret_cast<const unsigned short*>(weight.data_ptr());
 
-#define RMS_CASE(N)                                                          \
-  case N * 8 * 256:                                                          \
-    rmsnorm_kernel<N, FUSED_ADD><<<grid, block, 0, stream>>>(o, in, res, w,  \
-                                                             (float)eps, H); \
+#define RMS_CASE(N)                                                           \
+  case N * 8 * 256:                                                           \
+    if (T <= 128) {                                                           \
+      rmsnorm_wave_kernel<N * 4, FUSED_ADD>                                   \
+          <<<dim3((unsigned)((T + 3) / 4)), 256, 0, stream>>>(o, in, res, w,  \
+                                                              (float)eps, H,  \
+                                                              (int)T);        \
+    } else {                                                                  \
+      rmsnorm_kernel<N, FUSED_ADD><<<grid, block, 0, stream>>>(o, in, res, w, \
+                                                               (float)eps, H);\
+    }                                                                         \
     break;
   switch (H) {
     RMS_CASE(1)  // 2048

@@ -24,6 +24,10 @@ __global__ void rope_kv_kernel(
     const int* __restrict__ slot_mapping,    // [T] flat slot = block*BS + off
     int Hq, int Hk, int D, int BS) {
   const long t = blockIdx.x;
+  // grid.y splits the per-token chunk loops so small decode batches still
+  // put >=256 workgroups on the chip
+  const int yoff = blockIdx.y * blockDim.x + threadIdx.x;
+  const int ystride = gridDim.y * blockDim.x;
   const int half = D / 2;
   const long row_stride = (long)(Hq + 2 * Hk) * D;
   unsigned short* row = qkv + t * row_stride;
@@ -33,7 +37,7 @@ __global__ void rope_kv_kernel(
 
   // ---- rotate q (in place) ----
   const int q_chunks = Hq * (D / 16);  // 8 pairs per chunk
-  for (int c = threadIdx.x; c < q_chunks; c += blockDim.x) {
+  for (int c = yoff; c < q_chunks; c += ystride) {
     const int h = c / (D / 16);
     const int d0 = (c % (D / 16)) * 8;
     unsigned short* base = row + (long)h * D;
@@ -54,7 +58,7 @@ __global__ void rope_kv_kernel(
   const long cache_tok_base =
       slot >= 0 ? ((long)(slot / BS) * Hk * BS + (long)(slot % BS)) * D : 0;
   const int k_chunks = Hk * (D / 16);
-  for (int c = threadIdx.x; c < k_chunks; c += blockDim.x) {
+  for (int c = yoff; c < k_chunks; c += ystride) {
     const int h = c / (D / 16);
     const int d0 = (c % (D / 16)) * 8;
     unsigned short* base = row + (long)(Hq + h) * D;
@@ -80,7 +84,7 @@ __global__ void rope_kv_kernel(
   // ---- copy v to v_cache ----
   if (slot >= 0) {
     const int v_chunks = Hk * (D / 8);
-    for (int c = threadIdx.x; c < v_chunks; c += blockDim.x) {
+    for (int c = yoff; c < v_chunks; c += ystride) {
       const int h = c / (D / 8);
       const int d0 = (c % (D / 8)) * 8;
       const unsigned short* base = row + (long)(Hq + Hk + h) * D;
@@ -108,7 +112,8 @@ void rope_kv_append(torch::Tensor qkv, torch::Tensor k_cache,
   const int BS = (int)k_cache.size(2);
   TORCH_CHECK(k_cache.size(1) == num_kv_heads && k_cache.size(3) == D);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  rope_kv_kernel<<<dim3((unsigned)T), 256, 0, stream>>>(
+  const unsigned ysplit = T <= 128 ? 4 : 1;
+  rope_kv_kernel<<<dim3((unsigned)T, ysplit), 256, 0, stream>>>(
       reinterpret_cast<unsigned short*>(qkv.data_ptr()),
       reinterpret_cast<unsigned short*>(k_cache.data_ptr()),
       reinterpret_cast<unsigned short*>(v_cache.data_ptr()),

@@ -293,3 +293,30 @@ def test_prefill_gqa8():
 
 def test_prefill_single_token_turns():
     _prefill_case([(40, 39), (16, 15)], Hq=4, Hk=1)
+
+
+def test_mfma_probe16_layout():
+    from kukeon_amd import _C
+    torch.manual_seed(12)
+    a = (torch.randn(16, 32) * torch.linspace(0.3, 1.7, 32)).bfloat16().to(DEV)
+    b = (torch.randn(32, 16) * torch.linspace(-1.2, 1.2, 16)).bfloat16().to(DEV)
+    out = torch.empty(16, 16, dtype=torch.float32, device=DEV)
+    _C.mfma_probe16(out, a, b)
+    ref = a.float() @ b.float()
+    torch.testing.assert_close(out.cpu(), ref.cpu(), rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("M,N,K", [(1, 4096, 4096), (8, 6144, 4096),
+                                   (33, 4096, 14336), (64, 28672, 4096),
+                                   (64, 128256, 4096), (64, 4096, 4096)])
+def test_skinny_gemm(M, N, K):
+    from kukeon_amd import _C
+    torch.manual_seed(13)
+    x = (torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.5)
+    w = (torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05)
+    out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+    ws = torch.empty(64 * N, dtype=torch.float32, device=DEV)
+    _C.skinny_gemm(out, x, w, ws)
+    ref = (x.float() @ w.float().T)
+    torch.testing.assert_close(out.float().cpu(), ref.cpu(), rtol=3e-2,
+                               atol=3e-2)
