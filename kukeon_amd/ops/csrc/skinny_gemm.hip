@@ -1,15 +1,20 @@
 // Skinny decode GEMM for gfx950: out[M,N] = x[M,K] @ W[N,K]^T, M <= 64.
 //
-// Decode is weight-bandwidth-bound (the whole W streams through once per
-// step); hipBLASLt's skinny tiles measured only 36-50% of the HBM roofline
-// on the o/down/qkv shapes (profiles/r01*). This kernel streams W with one
-// 16 B load per lane per MFMA directly into fragments (no LDS round trip —
-// guide §5 "GEMV / M<=16 decode weights" row generalized to M<=64 via
-// mfma_f32_16x16x32_bf16), reads the L2-resident x straight into B-frags,
-// and split-Ks with f32 atomicAdd partials so every shape puts >=512
-// workgroups on the 256-CU chip.
-//
-// Grid: (N/64, SPLITK); 4 waves per WG, wave w owns N rows [n0+16w, +16).
+// Decode is weight-bandwidth-bound (the whole W streams through HBM once
+// per step); hipBLASLt's skinny tiles measured 36-54% of the roofline on
+// the o/down/qkv shapes (profiles/). Design learned from the pattern probe
+// (scripts/pattern_probe.hip: the 16-row-strided nt read reaches 5.6 TB/s,
+// so the access pattern is not the limit — duty cycle is):
+//   * x (tiny, L2-resident) is staged into LDS ONCE per kernel for the
+//     whole 512-deep k-slice (XOR-swizzled, conflict-free ds_read_b128
+//     B-fragments that consume no vmcnt slots),
+//   * W streams with nt 16B/lane loads in 8-deep batches, software-
+//     pipelined so batch b+1 is in flight while batch b feeds the MFMAs,
+//     and batch 0 is issued BEFORE the staging barrier (it has no LDS
+//     dependence) so even the prologue overlaps,
+//   * split-K (slice = 512) gives every shape >= 512 workgroups; partials
+//     combine through f32 atomicAdd + a cast pass.
+// Grid: (N/64, K/512); block = 4 waves, wave w owns N rows [64b+16w, +16).
 #include "common.h"
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
@@ -18,12 +23,22 @@ namespace kukeon {
 
 typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8_t;
 typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4_t;
+typedef __attribute__((__vector_size__(4 * sizeof(unsigned int)))) unsigned int u32x4_t;
 
+DEV_INLINE bf16x8_t frag_of(u32x4_t v) {
+  return __builtin_bit_cast(bf16x8_t, v);
+}
 struct uint4_s { unsigned int x[4]; };
 DEV_INLINE bf16x8_t frag_of(uint4 v) {
   uint4_s u{{v.x, v.y, v.z, v.w}};
   return __builtin_bit_cast(bf16x8_t, u);
 }
+DEV_INLINE u32x4_t nt_load16(const void* p) {
+  return __builtin_nontemporal_load(reinterpret_cast<const u32x4_t*>(p));
+}
+
+constexpr int KSLICE = 512;
+constexpr int U = 8;  // W k-chunks (of 32) per batch; 2 batches per slice
 
 __global__ void skinny_zero_kernel(float* __restrict__ ws, long n) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -34,6 +49,10 @@ __global__ void skinny_zero_kernel(float* __restrict__ ws, long n) {
   }
 }
 
+DEV_INLINE int xswz(int row, int byte_in_row) {
+  return row * (KSLICE * 2) + (byte_in_row ^ ((row & 15) << 4));
+}
+
 // MT = number of 16-row M subtiles (1 => M<=16, 4 => M<=64)
 template <int MT, bool SPLIT>
 __global__ __launch_bounds__(256) void skinny_gemm_kernel(
@@ -41,33 +60,75 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     float* __restrict__ ws,                // [M, N] f32  (SPLIT=true)
     const unsigned short* __restrict__ x,  // [M, K]
     const unsigned short* __restrict__ w,  // [N, K]
-    int M, int N, long K, int splitk) {
+    int M, int N, long K) {
+  __shared__ __align__(16) unsigned short xbuf[64 * KSLICE];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int n0 = blockIdx.x * 64 + wid * 16;     // this wave's 16 N rows
   const int row16 = lane & 15;                   // A row / B col
   const int kgrp = lane >> 4;                    // 0..3 -> k = 8*kgrp + j
-  const long kchunk = (K / 32 + gridDim.y - 1) / gridDim.y;
-  const long ks = (long)blockIdx.y * kchunk * 32;
-  const long ke = min(K, ks + kchunk * 32);
+  const long ks = (long)blockIdx.y * KSLICE;
+  const int klen = (int)min((long)KSLICE, K - ks);
+  const int nbatch = (klen + U * 32 - 1) / (U * 32);
 
-  const unsigned short* wrow = w + (long)(n0 + row16) * K;
+  const unsigned short* wrow = w + (long)(n0 + row16) * K + ks + 8 * kgrp;
+
+  // ---- batch 0 of W goes in flight before anything else ----
+  u32x4_t wa[U];
+#pragma unroll
+  for (int u = 0; u < U; ++u)
+    wa[u] = nt_load16(wrow + u * 32);
+
+  // ---- stage x[0..63][ks..ks+klen) into LDS (loads batched, then
+  // writes, so the 8 L2 round trips overlap) ----
+  {
+    const int xr = threadIdx.x >> 2;             // row 0..63
+    const int c0 = (threadIdx.x & 3) * 8;
+    const unsigned short* xs = x + (long)min(M - 1, xr) * K + ks;
+    uint4 v[KSLICE / 32];
+#pragma unroll
+    for (int i = 0; i < KSLICE / 32; ++i) {
+      const int c = c0 + i * 32;
+      v[i] = (c < klen) ? *reinterpret_cast<const uint4*>(xs + c)
+                        : uint4{0, 0, 0, 0};
+    }
+#pragma unroll
+    for (int i = 0; i < KSLICE / 32; ++i) {
+      const int c = c0 + i * 32;
+      *reinterpret_cast<uint4*>(reinterpret_cast<char*>(xbuf) +
+                                xswz(xr, c * 2)) = v[i];
+    }
+  }
+  __syncthreads();
+
   f32x4_t acc[MT];
 #pragma unroll
   for (int m = 0; m < MT; ++m) acc[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 
-  for (long k = ks; k < ke; k += 32) {
-    const uint4 wa =
-        *reinterpret_cast<const uint4*>(wrow + k + 8 * kgrp);
-    const bf16x8_t afrag = frag_of(wa);
+  u32x4_t wb[U];
+  for (int b = 0; b < nbatch; ++b) {
+    // issue the next batch while this one feeds the MFMAs
+    if (b + 1 < nbatch) {
 #pragma unroll
-    for (int m = 0; m < MT; ++m) {
-      const int xr = min(M - 1, m * 16 + row16);
-      const uint4 xb =
-          *reinterpret_cast<const uint4*>(x + (long)xr * K + k + 8 * kgrp);
-      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          afrag, frag_of(xb), acc[m], 0, 0, 0);
+      for (int u = 0; u < U; ++u)
+        wb[u] = nt_load16(wrow + (b + 1) * U * 32 + u * 32);
     }
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+      const int kc = b * U * 32 + u * 32 + 8 * kgrp;
+      if (kc >= klen) break;
+      const bf16x8_t afrag = frag_of(wa[u]);
+#pragma unroll
+      for (int m = 0; m < MT; ++m) {
+        const int xr = m * 16 + row16;
+        const uint4 xb = *reinterpret_cast<const uint4*>(
+            reinterpret_cast<const char*>(xbuf) + xswz(xr, kc * 2));
+        acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, frag_of(xb), acc[m], 0, 0, 0);
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < U; ++u) wa[u] = wb[u];
   }
 
   // C layout (16x16): lane -> col = lane&15 (the M index here),
@@ -113,10 +174,7 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int ntiles = N / 64;
-  int splitk = 1;
-  if (ntiles < 512) splitk = min(16, (512 + ntiles - 1) / ntiles);
-  // keep each slice >= 8 k-chunks so the split overhead stays small
-  splitk = max(1, min(splitk, (int)(K / 32 / 8)));
+  const int splitk = (int)((K + KSLICE - 1) / KSLICE);
   const int MT = (M + 15) / 16;
   dim3 grid(ntiles, splitk);
   auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
@@ -127,14 +185,14 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
 #define SK_LAUNCH(MT_)                                                       \
   if (splitk == 1) {                                                         \
     skinny_gemm_kernel<MT_, false><<<grid, 256, 0, stream>>>(                \
-        op, nullptr, xp, wp, M, N, K, splitk);                               \
+        op, nullptr, xp, wp, M, N, K);                                       \
   } else {                                                                   \
     float* wsp = ws.data_ptr<float>();                                       \
     TORCH_CHECK(ws.numel() >= total, "skinny_gemm workspace too small");     \
     skinny_zero_kernel<<<dim3((unsigned)((total / 4 + 255) / 256)), 256, 0,  \
                          stream>>>(wsp, total);                              \
     skinny_gemm_kernel<MT_, true><<<grid, 256, 0, stream>>>(                 \
-        nullptr, wsp, xp, wp, M, N, K, splitk);                              \
+        nullptr, wsp, xp, wp, M, N, K);                                      \
     skinny_cast_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256, 0,  \
                          stream>>>(op, wsp, total);                          \
   }
