@@ -24,6 +24,8 @@
 
 namespace kukeon {
 
+typedef __attribute__((__vector_size__(2 * sizeof(short)))) short bf16x2_t;
+
 template <int D, int BS, int GPW>
 __global__ __launch_bounds__(256) void paged_attn_kernel(
     unsigned short* __restrict__ out,     // [B, Hq*D] bf16 (splits==1)
@@ -55,8 +57,9 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
   __shared__ unsigned short kbuf[BS * D];
   __shared__ unsigned short vbuf[BS * D];
 
-  // ---- load Q for this wave's heads (scale folded in) ----
-  float qreg[GPW][32];
+  // ---- load Q for this wave's heads (packed bf16 pairs for v_dot2;
+  // scale is applied to the reduced score instead) ----
+  unsigned int qpk[GPW][16];
   const int h0 = wid * GPW;
 #pragma unroll
   for (int g = 0; g < GPW; ++g) {
@@ -66,9 +69,11 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
           q + (long)b * q_stride + (long)(hk * G + h) * D + sub * 32;
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
-        bf16x8 v = load_bf16x8(qp + c * 8);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) qreg[g][c * 8 + j] = v.f(j) * scale;
+        const uint4 v = *reinterpret_cast<const uint4*>(qp + c * 8);
+        qpk[g][c * 4 + 0] = v.x;
+        qpk[g][c * 4 + 1] = v.y;
+        qpk[g][c * 4 + 2] = v.z;
+        qpk[g][c * 4 + 3] = v.w;
       }
     }
   }
@@ -116,16 +121,20 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
 #pragma unroll
       for (int g = 0; g < GPW; ++g) s[g] = 0.f;
       {
+        // packed bf16 dot: one v_dot2_f32_bf16 per 2 elems (no unpack VALU)
         const unsigned short* kp = &kbuf[tok * D + ((sub ^ (tok & 3)) * 32)];
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
-          bf16x8 kv8 = load_bf16x8(kp + c * 8);
+          const uint4 kv = *reinterpret_cast<const uint4*>(kp + c * 8);
+          const unsigned int kw[4] = {kv.x, kv.y, kv.z, kv.w};
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const float kvf = kv8.f(j);
+          for (int j = 0; j < 4; ++j) {
 #pragma unroll
             for (int g = 0; g < GPW; ++g)
-              s[g] = fmaf(qreg[g][c * 8 + j], kvf, s[g]);
+              s[g] = __builtin_amdgcn_fdot2_f32_bf16(
+                  __builtin_bit_cast(bf16x2_t, kw[j]),
+                  __builtin_bit_cast(bf16x2_t, qpk[g][c * 4 + j]),
+                  s[g], false);
           }
         }
       }
@@ -133,6 +142,7 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
       for (int g = 0; g < GPW; ++g) {
         s[g] += __shfl_xor(s[g], 1, WAVE);
         s[g] += __shfl_xor(s[g], 2, WAVE);
+        s[g] *= scale;
         if (!valid) s[g] = -INFINITY;
         // every lane now has the score of its token (replicated x4);
         // reduce across the 16 token groups
