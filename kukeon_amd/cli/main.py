@@ -57,12 +57,12 @@ class Ctx:
                 self._client = UnixClient(self.socket_path)
         return self._client
 
-    def _controller(self):
+    def _controller(self, server_cfg=None):
         from kukeon_amd.controller.core import Controller
         from kukeon_amd.runtime.cgroup import CgroupManager
         from kukeon_amd.netpolicy import make_enforcer
         ctl = Controller(self.run_path, cgroups=CgroupManager(),
-                         enforcer=make_enforcer())
+                         enforcer=make_enforcer(), server_config=server_cfg)
         return ctl
 
 
@@ -511,20 +511,113 @@ def session_close(ctx, name):
 
 
 @cli.group()
+def team():
+    """Team distribution (kuke team init)."""
+
+
+@team.command("init")
+@click.option("-f", "--file", "file_", default="kuketeam.yaml",
+              type=click.Path(exists=True))
+@click.option("--teams-root", default=None)
+@click.option("--build/--no-build", "build_", default=True)
+@pass_ctx
+def team_init_cmd(ctx, file_, teams_root, build_):
+    from kukeon_amd.teams.pipeline import team_init
+    try:
+        res = team_init(ctx._controller(), file_, realm=ctx.realm,
+                        space=ctx.space, teams_root=teams_root,
+                        build_images=build_)
+    except errors.KukeonError as e:
+        _die(e)
+    click.echo(f"team {res['team']} (source {res['source']})")
+    for kind, name, action in res["applied"]:
+        click.echo(f"  {kind.lower()}/{name} {action}")
+    for ref in res["built"]:
+        click.echo(f"  image kukeon.internal/{ref} registered")
+    for p in res["pruned"]:
+        click.echo(f"  pruned {p}")
+
+
+@cli.group()
+def image():
+    """Image catalog verbs (always in-process, like the reference)."""
+
+
+@image.command("list")
+@pass_ctx
+def image_list(ctx):
+    for img in ctx._controller().list_images():
+        click.echo(img["name"])
+
+
+@image.command("get")
+@click.argument("name")
+@pass_ctx
+def image_get(ctx, name):
+    try:
+        click.echo(json.dumps(ctx._controller().get_image(name), indent=2))
+    except errors.KukeonError as e:
+        _die(e)
+
+
+@image.command("delete")
+@click.argument("name")
+@pass_ctx
+def image_delete(ctx, name):
+    try:
+        ctx._controller().delete_image(name)
+        click.echo(f"image {name} deleted")
+    except errors.KukeonError as e:
+        _die(e)
+
+
+@image.command("prune")
+@pass_ctx
+def image_prune(ctx):
+    for name in ctx._controller().prune_images():
+        click.echo(f"pruned {name}")
+
+
+@cli.command()
+@click.argument("name")
+@click.option("-f", "--file", "file_", required=True,
+              type=click.Path(exists=True),
+              help="image spec YAML (entrypoint/env/capabilities)")
+@pass_ctx
+def build(ctx, name, file_):
+    """Register an image from a build spec (kukebuild analog; OCI builds
+    are delegated to an external engine when one exists)."""
+    spec = yaml.safe_load(Path(file_).read_text()) or {}
+    ctx._controller().register_image(name, spec)
+    click.echo(f"image {name} registered")
+
+
+@cli.group()
 def daemon():
     """Daemon management."""
 
 
 @daemon.command("serve")
-@click.option("--reconcile-interval", default=30.0)
+@click.option("--reconcile-interval", default=None, type=float)
+@click.option("--configuration", default=None,
+              type=click.Path(exists=True),
+              help="ServerConfiguration YAML")
 @click.option("--foreground/--no-foreground", default=True)
 @pass_ctx
-def daemon_serve(ctx, reconcile_interval, foreground):
+def daemon_serve(ctx, reconcile_interval, configuration, foreground):
     import logging
     logging.basicConfig(
         level=logging.INFO,
         format="%(asctime)s %(levelname)s %(name)s %(message)s")
-    ctl = ctx._controller()
+    server_cfg = None
+    if configuration:
+        raw = yaml.safe_load(Path(configuration).read_text())
+        doc = api.ServerConfigurationDoc.from_dict(raw)
+        server_cfg = doc.spec
+    if reconcile_interval is None:
+        reconcile_interval = (server_cfg.reconcile_interval_seconds
+                              if server_cfg else 30.0)
+    ctl = ctx._controller(server_cfg)
     ctl.bootstrap()
     from kukeon_amd.daemon.server import Server
     srv = Server(ctl, ctx.socket_path, reconcile_interval)
@@ -551,6 +644,17 @@ def daemon_status(ctx):
         click.echo(json.dumps(st, indent=2))
     except Exception as e:
         _die(errors.KukeonError(f"daemon unreachable: {e}"))
+
+
+@cli.command()
+@pass_ctx
+def refresh(ctx):
+    """Re-derive all statuses from live runtime state."""
+    try:
+        res = ctx.client.RefreshAll()
+    except errors.KukeonError as e:
+        _die(e)
+    click.echo(f"refreshed: {res}")
 
 
 @cli.command()

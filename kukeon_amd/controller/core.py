@@ -26,6 +26,7 @@ from kukeon_amd.controller import diff as diffmod
 from kukeon_amd.controller import naming, parser
 from kukeon_amd.controller.subnet import SubnetAllocator
 from kukeon_amd.netpolicy import Enforcer, NoopEnforcer, build_policy
+from kukeon_amd.runtime import diskpressure
 from kukeon_amd.runtime.cgroup import CgroupManager
 from kukeon_amd.runtime.devices import (GPUAllocator, device_paths_for,
                                         visible_devices_env)
@@ -59,6 +60,7 @@ class Controller:
                  cgroups: Optional[CgroupManager] = None,
                  enforcer: Optional[Enforcer] = None,
                  gpu_devices: Optional[List[int]] = None,
+                 server_config: Optional[api.ServerConfigurationSpec] = None,
                  now_fn=time.time):
         self.run_path = Path(run_path)
         self.store = Store(run_path)
@@ -66,8 +68,15 @@ class Controller:
         self.runtime = runtime or ProcessRuntime(self.cgroups)
         self.enforcer = enforcer or NoopEnforcer()
         self.subnets = SubnetAllocator(self.store)
+        self.server_config = server_config or api.ServerConfigurationSpec()
+        if gpu_devices is None and self.server_config.gpu_devices:
+            gpu_devices = list(self.server_config.gpu_devices)
         self.gpus = GPUAllocator(str(self.run_path / "gpus.json"),
                                  devices=gpu_devices)
+        self.disk_guard = diskpressure.Guard(
+            str(self.run_path),
+            warn_percent=self.server_config.disk_pressure_warn_percent,
+            block_percent=self.server_config.disk_pressure_block_percent)
         self.now = now_fn
         self._restart_state: Dict[str, Tuple[float, int]] = {}
 
@@ -233,6 +242,7 @@ class Controller:
 
     def create_cell(self, doc: api.CellDoc) -> api.CellDoc:
         parser.validate_document(doc)
+        self.disk_guard.check(ignore=doc.spec.ignore_disk_pressure)
         self.get_stack(doc.spec.realm_id, doc.spec.space_id, doc.spec.stack_id)
         doc.spec.id = doc.spec.id or doc.metadata.name
         doc.spec.root_container_id = naming.root_container_id(
@@ -331,6 +341,7 @@ class Controller:
                 mine = gpu_ids[cursor: cursor + c.gpus]
                 cursor += c.gpus
                 env = self._container_env(doc, c, mine)
+                env += self._mount_volumes(doc, c, cdir)
                 cdoc = api.ContainerDoc(
                     metadata=api.Metadata(name=c.id or "main"), spec=c)
                 self.store.write(cdir / METADATA_FILE, cdoc.to_dict())
@@ -364,6 +375,39 @@ class Controller:
             with contextlib.suppress(Exception):
                 self._persist_cell(doc)
             raise
+
+    def _mount_volumes(self, doc: api.CellDoc, c: api.ContainerSpec,
+                       cdir: Path) -> List[str]:
+        """Process-cell volume exposure: named volumes (or host paths)
+        surface as KUKEON_VOLUME_<NAME> env + a symlink under the
+        container dir (no mount namespaces without a container engine)."""
+        env = []
+        mnt = cdir / "mnt"
+        for vm in c.volumes:
+            name = vm.name or Path(vm.target or vm.source or "vol").name
+            if vm.source:
+                src = Path(vm.source)
+            else:
+                vol = self.get_volume(doc.spec.realm_id, doc.spec.space_id,
+                                      name)
+                src = Path(vol.status.path)
+            mnt.mkdir(parents=True, exist_ok=True)
+            link = mnt / name
+            with contextlib.suppress(OSError):
+                if link.is_symlink() or link.exists():
+                    link.unlink()
+                link.symlink_to(src)
+            key = name.upper().replace("-", "_")
+            env.append(f"KUKEON_VOLUME_{key}={src}")
+        return env
+
+    def refresh_all(self) -> Dict[str, int]:
+        """Re-derive every status from live runtime state (kuke refresh)."""
+        return {
+            "cells": self.reconcile_cells(),
+            "sessions": self.reconcile_sessions(),
+            "spaces": self.reconcile_space_networks(),
+        }
 
     def _gpu_owner(self, doc: api.CellDoc) -> str:
         s = doc.spec
@@ -970,6 +1014,90 @@ class Controller:
                 m = act.stat().st_mtime
                 latest = m if latest is None else max(latest, m)
         return latest
+
+    # ==================================================================
+    # images (catalog of named runtime profiles; the containerd image
+    # store analog — no container engine in this runtime, so an image is
+    # a registered spec consumed by cells/teams rather than an OCI rootfs)
+    # ==================================================================
+    def register_image(self, name: str, spec: Dict) -> None:
+        safe = name.replace("/", "_")
+        p = self.run_path / "images" / f"{safe}.json"
+        self.store.write(p, {"name": name, "spec": spec,
+                             "registeredAt": now_iso(self.now())})
+
+    def get_image(self, name: str) -> Dict:
+        safe = name.replace("/", "_")
+        data = self.store.read(self.run_path / "images" / f"{safe}.json")
+        if data is None:
+            raise errors.NotFound(f"image {name}")
+        return data
+
+    def list_images(self) -> List[Dict]:
+        d = self.run_path / "images"
+        out = []
+        if d.is_dir():
+            for p in sorted(d.glob("*.json")):
+                data = self.store.read(p)
+                if data:
+                    out.append(data)
+        return out
+
+    def delete_image(self, name: str) -> None:
+        safe = name.replace("/", "_")
+        if not self.store.delete(self.run_path / "images" / f"{safe}.json"):
+            raise errors.NotFound(f"image {name}")
+
+    def prune_images(self) -> List[str]:
+        """Remove images not referenced by any cell spec."""
+        used = set()
+        for realm in self.store.list_children(self.store.data_root):
+            for space in self.store.list_children(self.store.realm_dir(realm)):
+                for stack in self.store.list_children(
+                        self.store.space_dir(realm, space)):
+                    for cell in self.store.list_children(
+                            self.store.stack_dir(realm, space, stack)):
+                        with contextlib.suppress(errors.CellNotFound):
+                            doc = self.get_cell(realm, space, stack, cell)
+                            for c in doc.spec.containers:
+                                used.add(c.image)
+        removed = []
+        for img in self.list_images():
+            if img["name"] not in used:
+                self.delete_image(img["name"])
+                removed.append(img["name"])
+        return removed
+
+    # ==================================================================
+    # team prune: delete team-labeled resources not in this apply's set
+    # ==================================================================
+    def prune_team(self, team: str, keep) -> List[str]:
+        pruned = []
+        for realm in self.store.list_children(self.store.data_root):
+            for space in self.store.list_children(self.store.realm_dir(realm)):
+                for kind_dir, kind in (("blueprints", api.KIND_CELL_BLUEPRINT),
+                                       ("configs", api.KIND_CELL_CONFIG)):
+                    for name in self.store.list_scoped_docs(
+                            self.store.space_dir(realm, space) / kind_dir):
+                        data = self.store.read(self.store.scoped_doc_path(
+                            realm, space, kind_dir, name))
+                        labels = (data or {}).get("metadata", {}).get(
+                            "labels", {})
+                        if labels.get(api.LABEL_TEAM) == team and                                 (kind, name) not in keep:
+                            self.store.delete(self.store.scoped_doc_path(
+                                realm, space, kind_dir, name))
+                            pruned.append(f"{kind}/{name}")
+                for stack in self.store.list_children(
+                        self.store.space_dir(realm, space)):
+                    for cell in self.store.list_children(
+                            self.store.stack_dir(realm, space, stack)):
+                        with contextlib.suppress(errors.CellNotFound):
+                            doc = self.get_cell(realm, space, stack, cell)
+                            if doc.metadata.labels.get(api.LABEL_TEAM) ==                                     team and (api.KIND_CELL, cell) not in keep:
+                                self.delete_cell(realm, space, stack, cell,
+                                                 force=True)
+                                pruned.append(f"Cell/{cell}")
+        return pruned
 
     def provision_modelhub_cell(self, model: str = "llama-3-8b",
                                 gpus: int = 1, socket_path: str = "",

@@ -282,6 +282,9 @@ class Service:
     def ReconcileSessions(self, p):
         return {"visited": self.ctl.reconcile_sessions()}
 
+    def RefreshAll(self, p):
+        return self.ctl.refresh_all()
+
 
 class Server:
     def __init__(self, ctl: Controller, socket_path: str,
