@@ -314,7 +314,24 @@ class Controller:
         env["KUKEON_REALM"] = doc.spec.realm_id
         env["KUKEON_SPACE"] = doc.spec.space_id
         env["KUKEON_STACK"] = doc.spec.stack_id
+        # a Running session over this stack wires its agent to the modelhub
+        ses = self._stack_session(doc.spec.realm_id, doc.spec.space_id,
+                                  doc.spec.stack_id)
+        if ses is not None:
+            env["KUKEON_SESSION"] = ses.metadata.name
+            if ses.spec.modelhub:
+                env["KUKEON_MODELHUB"] = ses.spec.modelhub
         return [f"{k}={v}" for k, v in env.items()]
+
+    def _stack_session(self, realm, space, stack) -> Optional[api.SessionDoc]:
+        d = self.run_path / "sessions" / realm / space / stack
+        if not d.is_dir():
+            return None
+        for p in sorted(d.glob("*.json")):
+            data = self.store.read(p)
+            if data and data.get("status", {}).get("state") ==                     api.STATE_RUNNING:
+                return api.SessionDoc.from_dict(data)
+        return None
 
     def start_cell(self, realm, space, stack, name) -> api.CellDoc:
         doc = self.get_cell(realm, space, stack, name)

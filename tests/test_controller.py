@@ -296,3 +296,25 @@ def test_subnet_allocation_per_space(ctl):
     assert len(subs) == 2
     for s in subs:
         assert s.startswith("10.88.") and s.endswith(".0/24")
+
+
+def test_session_modelhub_env_injection(ctl):
+    ses = api.SessionDoc(
+        metadata=api.Metadata(name="s-agent"),
+        spec=api.SessionSpec(stack_id="default",
+                             modelhub="/run/kukeon/modelhub.sock"))
+    ctl.create_session(ses)
+    cell = make_cell()
+    ctl.create_cell(cell)
+    env = ctl._container_env(
+        ctl.get_cell("default", "default", "default", "c1"),
+        cell.spec.containers[0], [])
+    assert "KUKEON_MODELHUB=/run/kukeon/modelhub.sock" in env
+    assert "KUKEON_SESSION=s-agent" in env
+    # after the session ends, new containers lose the wiring
+    ctl.close_session("default", "default", "default", "s-agent")
+    ctl.create_cell(make_cell("c2"))
+    env2 = ctl._container_env(
+        ctl.get_cell("default", "default", "default", "c2"),
+        cell.spec.containers[0], [])
+    assert not any(e.startswith("KUKEON_MODELHUB") for e in env2)
