@@ -41,6 +41,8 @@ class Request:
     finished: bool = False
     row: int = -1                       # sticky decode row
     bt_written: int = 0                 # block-table entries already staged
+    prefill_pos: int = 0                # prompt tokens already prefilled
+    preempted: int = 0
 
 
 @dataclass
@@ -149,6 +151,7 @@ class LLMEngine:
         kv.blocks = []
         kv.num_tokens = 0
         kv.pending_token = None
+        kv.history = []
 
     def has_work(self) -> bool:
         return bool(self.waiting) or self.num_running > 0
@@ -164,22 +167,38 @@ class LLMEngine:
         return []
 
     # ------------------------------------------------------------------
-    def _admit_prefill(self) -> List[Request]:
-        batch, tokens = [], 0
+    def _admit_prefill(self) -> List[Tuple[Request, int]]:
+        """-> [(request, chunk_len)]; a chunk smaller than the remaining
+        prompt keeps the request at the front of the queue (chunked
+        prefill for prompts beyond the per-step token budget)."""
+        batch: List[Tuple[Request, int]] = []
+        tokens = 0
         free = self.kv.allocator.num_free
         avail_rows = len(self._free_rows)
-        while self.waiting and avail_rows > 0:
+        budget = self.ecfg.max_prefill_tokens
+        while self.waiting and avail_rows > 0 and tokens < budget:
             req = self.waiting[0]
-            n = len(req.prompt_tokens)
-            if batch and tokens + n > self.ecfg.max_prefill_tokens:
+            remaining = len(req.prompt_tokens) - req.prefill_pos
+            chunk = min(remaining, budget - tokens)
+            if chunk <= 0:
                 break
-            need = req.kv.blocks_needed(n + req.sampling.max_new_tokens)
+            finishing = (req.prefill_pos + chunk) == len(req.prompt_tokens)
+            reserve = chunk + (req.sampling.max_new_tokens if finishing
+                               else 0)
+            need = req.kv.blocks_needed(reserve)
             if need > free:
+                if not batch:
+                    # can't make progress on the head request right now
+                    break
                 break
             free -= need
-            tokens += n
-            avail_rows -= 1
-            batch.append(self.waiting.pop(0))
+            tokens += chunk
+            if finishing:
+                avail_rows -= 1
+                batch.append((self.waiting.pop(0), chunk))
+            else:
+                batch.append((req, chunk))
+                break  # a partial chunk consumes the whole budget
         return batch
 
     def _alloc_for(self, req: Request, new_tokens: int) -> List[int]:
@@ -191,32 +210,35 @@ class LLMEngine:
         return slots
 
     # ------------------------------------------------------------------
-    def _run_prefill(self, batch: List[Request]) -> List[StepOutput]:
+    def _run_prefill(self, batch: List[Tuple[Request, int]]) -> List[StepOutput]:
         dev = self.device
         ids, pos, slots = [], [], []
         qs_pairs, qb_seq, qb_start = [], [], []
         row = 0
-        for s, req in enumerate(batch):
-            n = len(req.prompt_tokens)
+        for s, (req, chunk) in enumerate(batch):
+            toks = req.prompt_tokens[req.prefill_pos:
+                                     req.prefill_pos + chunk]
             start = req.kv.num_tokens
-            sl = self._alloc_for(req, n)
-            ids.extend(req.prompt_tokens)
-            pos.extend(range(start, start + n))
+            sl = self._alloc_for(req, chunk)
+            req.kv.history.extend(toks)
+            ids.extend(toks)
+            pos.extend(range(start, start + chunk))
             slots.extend(sl)
             qs_pairs.append((start, row))
-            for qb in range(0, n, 32):
+            for qb in range(0, chunk, 32):
                 qb_seq.append(s)
                 qb_start.append(qb)
-            row += n
+            row += chunk
+            req.prefill_pos += chunk
         t_ids = torch.tensor(ids, dtype=torch.int32, device=dev)
         t_pos = torch.tensor(pos, dtype=torch.int32, device=dev)
         t_slots = torch.tensor(slots, dtype=torch.int32, device=dev)
-        t_seq_lens = torch.tensor([r.kv.num_tokens for r in batch],
+        t_seq_lens = torch.tensor([r.kv.num_tokens for r, _ in batch],
                                   dtype=torch.int32, device=dev)
         t_qs = torch.tensor(qs_pairs, dtype=torch.int32, device=dev)
-        mbt = max(len(r.kv.blocks) for r in batch)
+        mbt = max(len(r.kv.blocks) for r, _ in batch)
         bt = torch.zeros(len(batch), mbt, dtype=torch.int32)
-        for i, r in enumerate(batch):
+        for i, (r, _) in enumerate(batch):
             bt[i, : len(r.kv.blocks)] = torch.tensor(r.kv.blocks,
                                                      dtype=torch.int32)
         meta = AttnMeta(mode="prefill", positions=t_pos, slot_mapping=t_slots,
@@ -227,15 +249,21 @@ class LLMEngine:
                         qb_start=torch.tensor(qb_start, dtype=torch.int32,
                                               device=dev))
         hidden = self.model.forward(t_ids, self.kv.k, self.kv.v, meta)
-        last_rows = []
-        acc = 0
-        for r in batch:
-            acc += len(r.prompt_tokens)
-            last_rows.append(acc - 1)
+        # sample only the requests whose prompt completed this pass
+        finishing = [(i, r) for i, (r, _) in enumerate(batch)
+                     if r.prefill_pos == len(r.prompt_tokens)]
+        if not finishing:
+            return []
+        last_rows, acc = [], 0
+        for i, (r, chunk) in enumerate(batch):
+            acc += chunk
+            if r.prefill_pos == len(r.prompt_tokens):
+                last_rows.append(acc - 1)
         logits = self.model.compute_logits(
             hidden[torch.tensor(last_rows, dtype=torch.long, device=dev)])
-        toks = self._sample_eager(batch, logits)
-        return [self._append_token(r, t) for r, t in zip(batch, toks)]
+        toks = self._sample_eager([r for _, r in finishing], logits)
+        return [self._append_token(r, t)
+                for (_, r), t in zip(finishing, toks)]
 
     def _sample_eager(self, reqs: List[Request], logits: torch.Tensor):
         B = len(reqs)
@@ -304,7 +332,44 @@ class LLMEngine:
                                              device=self.device)
         return self.d_tmp_out[key], self.d_tmp_ml[key]
 
+    def _preempt(self, req: Request) -> None:
+        """Evict a running request: free its KV and re-queue it for full
+        recompute (history + its un-processed last token become the new
+        prompt; already-emitted output_tokens are preserved)."""
+        log.warning("preempting request %d (KV exhausted)", req.req_id)
+        self._release_row(req)
+        history = list(req.kv.history)
+        last = req.output_tokens[-1] if req.output_tokens else None
+        self.kv.allocator.free(req.kv.blocks)
+        req.kv.blocks = []
+        req.kv.num_tokens = 0
+        req.kv.history = []
+        req.kv.pending_token = None
+        req.prompt_tokens = history + ([last] if last is not None else [])
+        req.prefill_pos = 0
+        req.preempted += 1
+        self.waiting.insert(0, req)
+
     def _run_decode(self) -> List[StepOutput]:
+        # alloc phase first: on KV exhaustion preempt from the back
+        # (recompute policy) and retry
+        for row in range(self.Bmax):
+            r = self._rows[row]
+            if r is None:
+                continue
+            try:
+                slot = self._alloc_for(r, 1)[0]
+            except MemoryError:
+                actives = [x for x in self._rows if x is not None]
+                victim = actives[-1] if actives[-1] is not r else r
+                self._preempt(victim)
+                if victim is r:
+                    continue
+                slot = self._alloc_for(r, 1)[0]
+            r._slot = slot
+            r.kv.history.append(r.output_tokens[-1])
+        if self.num_running == 0:
+            return []
         maxrow = max(i for i, r in enumerate(self._rows) if r is not None)
         nrows = maxrow + 1
         active = []
@@ -316,8 +381,8 @@ class LLMEngine:
                 continue
             active.append(r)
             self.n_ids[row] = r.output_tokens[-1]
-            self.n_pos[row] = r.kv.num_tokens
-            self.n_slots[row] = self._alloc_for(r, 1)[0]
+            self.n_pos[row] = r.kv.num_tokens - 1
+            self.n_slots[row] = r._slot
             self.n_seq_lens[row] = r.kv.num_tokens
             n = len(r.kv.blocks)
             if n > r.bt_written:

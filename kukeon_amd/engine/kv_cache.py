@@ -47,6 +47,9 @@ class SequenceKV:
         # last sampled token of the previous turn: sampled but never run
         # through the model, so the next turn's prefill must include it
         self.pending_token = None
+        # every token whose K/V lives in the cache, in order — the replay
+        # source for preemption-by-recompute
+        self.history = []
 
     def blocks_needed(self, new_tokens: int) -> int:
         total = self.num_tokens + new_tokens

@@ -140,3 +140,51 @@ def test_kv_exhaustion_defers_admission():
             if o.finished:
                 done.append(o.req_id)
     assert sorted(done) == [0, 1]
+
+
+def test_chunked_prefill_matches_naive():
+    torch.manual_seed(0)
+    engine, cfg, ecfg = make_engine(max_prefill_tokens=8)
+    prompt = [(i * 13 + 5) % cfg.vocab_size for i in range(37)]
+    sp = SamplingParams(temperature=0.0, max_new_tokens=4)
+    kv = SequenceKV(ecfg.block_size)
+    engine.add_request(kv, prompt, sp)
+    got, prefill_steps = [], 0
+    while engine.has_work():
+        outs = engine.step()
+        if not outs and engine.waiting:
+            prefill_steps += 1
+        for o in outs:
+            got.extend(o.new_tokens)
+    assert prefill_steps >= 4  # 37 tokens through an 8-token budget
+    assert got == naive_greedy(engine.model, cfg, prompt, 4)
+
+
+def test_preemption_recompute_matches_naive():
+    torch.manual_seed(0)
+    engine, cfg, ecfg = make_engine()
+    sp = SamplingParams(temperature=0.0, max_new_tokens=6)
+    kv1, kv2 = SequenceKV(16), SequenceKV(16)
+    p1 = [7, 3, 99, 140, 11, 42, 17, 23, 5, 81, 250, 33, 9, 1, 2, 4]
+    p2 = [4, 4, 8, 15, 16, 23, 42, 108]
+    engine.add_request(kv1, p1, sp)
+    engine.add_request(kv2, p2, sp)
+    # prefill both, then strangle the pool so the next block alloc fails
+    outs = {0: [], 1: []}
+    for o in engine.step():
+        outs[o.req_id].extend(o.new_tokens)
+    stolen = engine.kv.allocator.alloc(engine.kv.allocator.num_free)
+    preempted = False
+    for _ in range(40):
+        if not engine.has_work():
+            break
+        for o in engine.step():
+            outs[o.req_id].extend(o.new_tokens)
+        if engine.waiting and not preempted:
+            preempted = True
+            assert engine.waiting[0].preempted == 1
+            engine.kv.allocator.free(stolen)  # storage pressure clears
+            stolen = []
+    assert preempted, "expected a preemption under KV exhaustion"
+    assert outs[0] == naive_greedy(engine.model, cfg, p1, 6)
+    assert outs[1] == naive_greedy(engine.model, cfg, p2, 6)
