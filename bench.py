@@ -35,6 +35,7 @@ def main() -> None:
     ap.add_argument("--ctx-cap", type=int, default=3584)
     ap.add_argument("--max-model-len", type=int, default=4096)
     ap.add_argument("--no-graphs", action="store_true")
+    ap.add_argument("--decode-microbatch", type=int, default=32)
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
 
@@ -63,6 +64,7 @@ def main() -> None:
                         max_sessions=max(args.sessions, 8),
                         use_graphs=(not args.no_graphs and device != "cpu"
                                     and not cfg.is_moe),
+                        decode_microbatch=args.decode_microbatch,
                         seed=1234 + rank // tp_size)
     if device == "cpu":
         ecfg.num_kv_blocks = (args.sessions *

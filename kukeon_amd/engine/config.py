@@ -88,6 +88,8 @@ class EngineConfig:
     use_graphs: bool = True
     graph_buckets: tuple = (8, 16, 32, 64, 96, 128, 192, 256)
     decode_splits: int = 1          # split-KV factor for small-batch decode
+    decode_microbatch: int = 16     # decode steps per host sync (self-
+                                    # advancing graph replay train)
     seed: int = 1234
     tp_size: int = 1
 

@@ -106,6 +106,11 @@ class LLMEngine:
         self.max_splits = 16
         self.d_tmp_out = {}
         self.d_tmp_ml = {}
+        # self-advancing decode: token ring + device step counter
+        self.max_microbatch = 256
+        self.d_ring = torch.zeros(self.max_microbatch * self.Bmax,
+                                  dtype=torch.int32, device=d)
+        self.d_step = torch.zeros(1, dtype=torch.int32, device=d)
         pin = self.is_cuda
         self.h_ids = torch.zeros(self.Bmax, dtype=torch.int32, pin_memory=pin)
         self.h_pos = torch.zeros(self.Bmax, dtype=torch.int32, pin_memory=pin)
@@ -351,23 +356,32 @@ class LLMEngine:
         self.waiting.insert(0, req)
 
     def _run_decode(self) -> List[StepOutput]:
-        # alloc phase first: on KV exhaustion preempt from the back
-        # (recompute policy) and retry
-        for row in range(self.Bmax):
+        """A micro-batch of k decode steps with ONE host sync: the graph is
+        self-advancing (sampled tokens feed back on device, cursors advance
+        in-kernel, tokens land in a ring buffer)."""
+        active0 = [r for r in self._rows if r is not None]
+        k = min(min(r.sampling.max_new_tokens - len(r.output_tokens)
+                    for r in active0),
+                max(1, self.ecfg.decode_microbatch), self.max_microbatch)
+        # ---- alloc k tokens of KV per request (preempt on exhaustion) ----
+        row = 0
+        while row < self.Bmax:
             r = self._rows[row]
             if r is None:
+                row += 1
                 continue
-            try:
-                slot = self._alloc_for(r, 1)[0]
-            except MemoryError:
-                actives = [x for x in self._rows if x is not None]
-                victim = actives[-1] if actives[-1] is not r else r
-                self._preempt(victim)
-                if victim is r:
+            need = r.kv.blocks_needed(k)
+            if need:
+                try:
+                    r.kv.blocks.extend(self.kv.allocator.alloc(need))
+                except MemoryError:
+                    actives = [x for x in self._rows if x is not None]
+                    victim = actives[-1] if actives[-1] is not r else r
+                    self._preempt(victim)
+                    if victim is r:
+                        row += 1
                     continue
-                slot = self._alloc_for(r, 1)[0]
-            r._slot = slot
-            r.kv.history.append(r.output_tokens[-1])
+            row += 1
         if self.num_running == 0:
             return []
         maxrow = max(i for i, r in enumerate(self._rows) if r is not None)
@@ -376,27 +390,26 @@ class LLMEngine:
         for row in range(nrows):
             r = self._rows[row]
             if r is None:
-                self.n_slots[row] = -1
                 self.n_seq_lens[row] = 0
                 continue
-            active.append(r)
+            active.append((row, r))
+            start = r.kv.num_tokens
             self.n_ids[row] = r.output_tokens[-1]
-            self.n_pos[row] = r.kv.num_tokens - 1
-            self.n_slots[row] = r._slot
-            self.n_seq_lens[row] = r.kv.num_tokens
+            self.n_pos[row] = start
+            self.n_seq_lens[row] = start + 1
             n = len(r.kv.blocks)
             if n > r.bt_written:
                 self.n_bt[row, r.bt_written: n] = r.kv.blocks[r.bt_written:]
                 r.bt_written = n
         use_graph = self.is_cuda and self.ecfg.use_graphs
         bucket = self._bucket(nrows) if use_graph else nrows
-        nb = bucket
-        for row in range(nrows, nb):
-            self.n_slots[row] = -1
+        for row in range(nrows, bucket):
             self.n_seq_lens[row] = 0
+        if use_graph and bucket not in self.graphs:
+            self._capture_safely(bucket)
+        nb = bucket
         self.d_ids[:nb].copy_(self.h_ids[:nb], non_blocking=True)
         self.d_pos[:nb].copy_(self.h_pos[:nb], non_blocking=True)
-        self.d_slots[:nb].copy_(self.h_slots[:nb], non_blocking=True)
         self.d_seq_lens[:nb].copy_(self.h_seq_lens[:nb], non_blocking=True)
         self.d_bt[:nb].copy_(self.h_bt[:nb], non_blocking=True)
         if self._params_dirty:
@@ -405,15 +418,26 @@ class LLMEngine:
                               non_blocking=True)
             self.d_topp.copy_(self.h_params[:, 2], non_blocking=True)
             self._params_dirty = False
-
+        self.d_step.zero_()
         if use_graph:
-            if bucket not in self.graphs:
-                self._capture(bucket)
-            self.graphs[bucket].replay()
+            g = self.graphs[bucket]
+            for _ in range(k):
+                g.replay()
         else:
-            self._decode_forward(bucket)
-        toks = self.d_tokens[:nrows].cpu()
-        return [self._append_token(r, int(toks[r.row])) for r in active]
+            for _ in range(k):
+                self._decode_forward(bucket)
+        slab = self.d_ring[: k * bucket].view(k, bucket)[:, :nrows].cpu()
+        outs = []
+        for row, r in active:
+            toks = [int(t) for t in slab[:, row]]
+            prev = r.output_tokens[-1]
+            r.kv.history.extend([prev] + toks[:-1])
+            r.kv.num_tokens += k
+            fin = False
+            for t in toks:
+                fin = self._append_token(r, t).finished
+            outs.append(StepOutput(r.req_id, toks, fin))
+        return outs
 
     def _decode_forward(self, B: int) -> None:
         splits = self._decode_splits(B)
@@ -428,6 +452,22 @@ class LLMEngine:
         ops.sample(self.d_tokens[:B], logits, self.d_temps[:B],
                    self.d_topk[:B], self.d_topp[:B], self.d_seed,
                    self.d_ws[:B])
+        ops.decode_advance(self.d_ids[:B], self.d_pos[:B],
+                           self.d_seq_lens[:B], self.d_tokens[:B],
+                           self.d_ring, self.d_step)
+
+    def _capture_safely(self, bucket: int) -> None:
+        """Capture a bucket graph mid-serving: the warmup forward mutates
+        the self-advancing cursors, so snapshot and restore them."""
+        saved = (self.d_ids.clone(), self.d_pos.clone(),
+                 self.d_seq_lens.clone(), self.d_step.clone())
+        self._capture(bucket)
+        self.d_ids.copy_(saved[0])
+        self.d_pos.copy_(saved[1])
+        self.d_seq_lens.copy_(saved[2])
+        self.d_step.copy_(saved[3])
+        if self.is_cuda:
+            torch.cuda.synchronize()
 
     def _capture(self, bucket: int) -> None:
         log.info("capturing decode graph for bucket %d", bucket)

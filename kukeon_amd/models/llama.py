@@ -62,7 +62,9 @@ class LlamaAttention(nn.Module):
                 kc: torch.Tensor, vc: torch.Tensor, meta: AttnMeta):
         qkv = ops.linear(x, self.qkv_w)
         ops.rope_kv_append(qkv, kc, vc, cos_sin, meta.positions,
-                           meta.slot_mapping, self.hq, self.hk, self.d)
+                           meta.slot_mapping, self.hq, self.hk, self.d,
+                           block_table=(meta.block_table
+                                        if meta.mode == "decode" else None))
         out = torch.empty(x.shape[0], self.hq * self.d, dtype=x.dtype,
                           device=x.device)
         if meta.mode == "decode":

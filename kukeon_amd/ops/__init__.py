@@ -49,10 +49,17 @@ def silu_mul(out, gate_up) -> None:
 
 
 def rope_kv_append(qkv, k_cache, v_cache, cos_sin, positions, slot_mapping,
-                   num_q_heads: int, num_kv_heads: int, head_dim: int) -> None:
+                   num_q_heads: int, num_kv_heads: int, head_dim: int,
+                   block_table=None) -> None:
+    if block_table is None:
+        block_table = torch.empty(0, dtype=torch.int32, device=qkv.device)
     _impl(qkv).rope_kv_append(qkv, k_cache, v_cache, cos_sin, positions,
                               slot_mapping, num_q_heads, num_kv_heads,
-                              head_dim)
+                              head_dim, block_table)
+
+
+def decode_advance(ids, pos, seq_lens, tokens, ring, counter) -> None:
+    _impl(ids).decode_advance(ids, pos, seq_lens, tokens, ring, counter)
 
 
 def paged_attention(out, q, k_cache, v_cache, block_table, seq_lens,

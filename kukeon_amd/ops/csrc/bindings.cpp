@@ -11,7 +11,10 @@ void rope_kv_append(torch::Tensor qkv, torch::Tensor k_cache,
                     torch::Tensor v_cache, torch::Tensor cos_sin,
                     torch::Tensor positions, torch::Tensor slot_mapping,
                     int64_t num_q_heads, int64_t num_kv_heads,
-                    int64_t head_dim);
+                    int64_t head_dim, torch::Tensor block_table);
+void decode_advance(torch::Tensor ids, torch::Tensor pos,
+                    torch::Tensor seq_lens, torch::Tensor tokens,
+                    torch::Tensor ring, torch::Tensor counter);
 void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                      torch::Tensor v_cache, torch::Tensor block_table,
                      torch::Tensor seq_lens, int64_t q_offset,
@@ -54,6 +57,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &kukeon::skinny_gemm,
         "weight-streaming decode GEMM (M<=64)");
   m.def("sample", &kukeon::sample, "top-k/top-p/temperature sampling");
+  m.def("decode_advance", &kukeon::decode_advance,
+        "on-device decode cursor advance (self-advancing graph)");
   m.def("moe_gather_tokens", &kukeon::moe_gather_tokens, "MoE permute");
   m.def("moe_scatter_tokens", &kukeon::moe_scatter_tokens,
         "MoE unpermute + weighted combine");

@@ -22,6 +22,8 @@ __global__ void rope_kv_kernel(
     const float* __restrict__ cos_sin,       // [max_pos, D]
     const int* __restrict__ positions,       // [T]
     const int* __restrict__ slot_mapping,    // [T] flat slot = block*BS + off
+    const int* __restrict__ block_table,     // [T, max_blocks] decode mode:
+    int max_blocks,                          //   slot derived on device
     int Hq, int Hk, int D, int BS) {
   const long t = blockIdx.x;
   // grid.y splits the per-token chunk loops so small decode batches still
@@ -32,7 +34,15 @@ __global__ void rope_kv_kernel(
   const long row_stride = (long)(Hq + 2 * Hk) * D;
   unsigned short* row = qkv + t * row_stride;
   const int pos = positions[t];
-  const int slot = slot_mapping[t];
+  int slot;
+  if (block_table != nullptr) {
+    // decode: the device owns the sequence cursor (self-advancing graph)
+    slot = (pos >= 0)
+               ? block_table[t * (long)max_blocks + pos / BS] * BS + pos % BS
+               : -1;
+  } else {
+    slot = slot_mapping[t];
+  }
   const float* cs = cos_sin + (long)pos * D;
 
   // ---- rotate q (in place) ----
@@ -99,7 +109,7 @@ void rope_kv_append(torch::Tensor qkv, torch::Tensor k_cache,
                     torch::Tensor v_cache, torch::Tensor cos_sin,
                     torch::Tensor positions, torch::Tensor slot_mapping,
                     int64_t num_q_heads, int64_t num_kv_heads,
-                    int64_t head_dim) {
+                    int64_t head_dim, torch::Tensor block_table) {
   TORCH_CHECK(qkv.is_contiguous() && qkv.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(cos_sin.scalar_type() == torch::kFloat32);
   TORCH_CHECK(positions.scalar_type() == torch::kInt32);
@@ -112,13 +122,17 @@ void rope_kv_append(torch::Tensor qkv, torch::Tensor k_cache,
   const int BS = (int)k_cache.size(2);
   TORCH_CHECK(k_cache.size(1) == num_kv_heads && k_cache.size(3) == D);
   auto stream = c10::hip::getCurrentHIPStream().stream();
+  const bool table_mode = block_table.dim() == 2;
+  const int* bt = table_mode ? block_table.data_ptr<int>() : nullptr;
+  const int mb = table_mode ? (int)block_table.size(1) : 0;
   const unsigned ysplit = T <= 128 ? 4 : 1;
   rope_kv_kernel<<<dim3((unsigned)T, ysplit), 256, 0, stream>>>(
       reinterpret_cast<unsigned short*>(qkv.data_ptr()),
       reinterpret_cast<unsigned short*>(k_cache.data_ptr()),
       reinterpret_cast<unsigned short*>(v_cache.data_ptr()),
       cos_sin.data_ptr<float>(), positions.data_ptr<int>(),
-      slot_mapping.data_ptr<int>(), (int)num_q_heads, (int)num_kv_heads, D, BS);
+      slot_mapping.data_ptr<int>(), bt, mb, (int)num_q_heads,
+      (int)num_kv_heads, D, BS);
   HIP_CHECK_KERNEL();
 }
 

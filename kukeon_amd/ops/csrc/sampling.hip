@@ -325,6 +325,40 @@ __global__ void k7_finalize(int* __restrict__ tokens,
   if (b == 0) seed[0] += 1;
 }
 
+// ---- self-advancing decode support: after sampling, feed the token back
+// as the next step's input, advance the device-side cursors, and append to
+// the ring buffer — so a micro-batch of decode steps replays with no host
+// round-trip (one sync per micro-batch).
+__global__ void decode_advance_kernel(int* __restrict__ ids,
+                                      int* __restrict__ pos,
+                                      int* __restrict__ seq_lens,
+                                      const int* __restrict__ tokens,
+                                      int* __restrict__ ring,
+                                      int* __restrict__ counter, int B) {
+  const int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  const int step = counter[0];
+  if (seq_lens[b] > 0) {  // active rows only (padded rows stay parked)
+    ids[b] = tokens[b];
+    pos[b] += 1;
+    seq_lens[b] += 1;
+    ring[step * B + b] = tokens[b];
+  }
+  if (b == 0) counter[0] = step + 1;
+}
+
+void decode_advance(torch::Tensor ids, torch::Tensor pos,
+                    torch::Tensor seq_lens, torch::Tensor tokens,
+                    torch::Tensor ring, torch::Tensor counter) {
+  const int B = ids.size(0);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  decode_advance_kernel<<<(B + 255) / 256, 256, 0, stream>>>(
+      ids.data_ptr<int>(), pos.data_ptr<int>(), seq_lens.data_ptr<int>(),
+      tokens.data_ptr<int>(), ring.data_ptr<int>(), counter.data_ptr<int>(),
+      B);
+  HIP_CHECK_KERNEL();
+}
+
 void sample(torch::Tensor tokens, torch::Tensor logits, torch::Tensor temps,
             torch::Tensor top_k, torch::Tensor top_p, torch::Tensor seed,
             torch::Tensor workspace) {

@@ -47,7 +47,8 @@ def _rotate(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tens
 def rope_kv_append(qkv: torch.Tensor, k_cache: torch.Tensor,
                    v_cache: torch.Tensor, cos_sin: torch.Tensor,
                    positions: torch.Tensor, slot_mapping: torch.Tensor,
-                   num_q_heads: int, num_kv_heads: int, head_dim: int) -> None:
+                   num_q_heads: int, num_kv_heads: int, head_dim: int,
+                   block_table=None) -> None:
     T = qkv.shape[0]
     D = head_dim
     half = D // 2
@@ -62,10 +63,17 @@ def rope_kv_append(qkv: torch.Tensor, k_cache: torch.Tensor,
     qkv[:, : num_q_heads * D] = qr.reshape(T, -1)
     qkv[:, num_q_heads * D: (num_q_heads + num_kv_heads) * D] = kr.reshape(T, -1)
     BS = k_cache.shape[2]
+    table_mode = block_table is not None and block_table.dim() == 2
     for t in range(T):
-        slot = int(slot_mapping[t])
-        if slot < 0:
-            continue
+        if table_mode:
+            pos_t = int(positions[t])
+            if pos_t < 0:
+                continue
+            slot = int(block_table[t, pos_t // BS]) * BS + pos_t % BS
+        else:
+            slot = int(slot_mapping[t])
+            if slot < 0:
+                continue
         blk, off = slot // BS, slot % BS
         k_cache[blk, :, off] = kr[t]
         v_cache[blk, :, off] = v[t]
@@ -177,6 +185,18 @@ def sample(tokens: torch.Tensor, logits: torch.Tensor, temps: torch.Tensor,
         probs = probs / probs.sum()
         tokens[b] = int(torch.multinomial(probs, 1))
     seed += 1
+
+
+def decode_advance(ids, pos, seq_lens, tokens, ring, counter) -> None:
+    step = int(counter[0])
+    B = ids.shape[0]
+    for b in range(B):
+        if int(seq_lens[b]) > 0:
+            ids[b] = tokens[b]
+            pos[b] += 1
+            seq_lens[b] += 1
+            ring[step * B + b] = tokens[b]
+    counter[0] = step + 1
 
 
 def moe_gather_tokens(out: torch.Tensor, input: torch.Tensor,
