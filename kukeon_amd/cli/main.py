@@ -647,6 +647,65 @@ def daemon_status(ctx):
 
 
 @cli.command()
+@click.option("--yes", is_flag=True, help="skip confirmation")
+@pass_ctx
+def uninstall(ctx, yes):
+    """Full teardown: kill cells, remove the run path state and cgroups."""
+    if not yes:
+        click.confirm(f"remove all kukeon state under {ctx.run_path}?",
+                      abort=True)
+    ctl = ctx._controller()
+    for realm in list(ctl.store.list_children(ctl.store.data_root)):
+        try:
+            ctl.delete_realm(realm, cascade=True)
+        except Exception as e:
+            click.echo(f"warn: realm {realm}: {e}", err=True)
+    import shutil as sh
+    for sub in ("data", "sessions", "images", "persist", "s", "bin"):
+        sh.rmtree(Path(ctx.run_path) / sub, ignore_errors=True)
+    for f in ("gpus.json", "kukeond.pid", ".kukeon-instance.json",
+              "kukeond.sock", "modelhub.sock"):
+        with __import__("contextlib").suppress(OSError):
+            (Path(ctx.run_path) / f).unlink()
+    ctl.cgroups.delete("")
+    click.echo(f"kukeon state removed from {ctx.run_path}")
+
+
+@cli.command()
+@click.argument("shell", type=click.Choice(["bash", "zsh", "fish"]),
+                default="bash")
+def autocomplete(shell):
+    """Print shell completion setup for kuke."""
+    var = {"bash": "bash_source", "zsh": "zsh_source",
+           "fish": "fish_source"}[shell]
+    click.echo(f'eval "$(_KUKE_COMPLETE={var} kuke)"')
+
+
+@cli.command()
+@click.option("-f", "--file", "files", multiple=True, required=True,
+              type=click.Path(exists=True, allow_dash=True))
+@pass_ctx
+def create(ctx, files):
+    """Create resources from manifests (create-only: fails on existing)."""
+    for fp in files:
+        text = sys.stdin.read() if fp == "-" else Path(fp).read_text()
+        from kukeon_amd.controller import parser as docparser
+        for doc in docparser.parse_documents(text):
+            method = {"Realm": "CreateRealm", "Space": "CreateSpace",
+                      "Stack": "CreateStack", "Cell": "CreateCell",
+                      "Session": "CreateSession"}.get(doc.kind)
+            try:
+                if method:
+                    ctx.client.call(method, doc=doc.to_dict())
+                else:
+                    ctx.client.ApplyDocuments(
+                        yaml=yaml.safe_dump(doc.to_dict()))
+                click.echo(f"{doc.kind.lower()}/{doc.metadata.name} created")
+            except errors.KukeonError as e:
+                _die(e)
+
+
+@cli.command()
 @pass_ctx
 def refresh(ctx):
     """Re-derive all statuses from live runtime state."""

@@ -53,6 +53,8 @@ class ModelhubServer:
         self._stop = threading.Event()
         self._srv = None
         self._threads = []
+        self.metrics = {"tokens_generated": 0, "turns_completed": 0,
+                        "engine_steps": 0, "busy_seconds": 0.0}
 
     # ---- engine loop --------------------------------------------------
     def _engine_loop(self):
@@ -71,14 +73,20 @@ class ModelhubServer:
                 pass
             if not self.engine.has_work():
                 continue
+            import time as _time
+            t0 = _time.perf_counter()
             outs = self.engine.step()
+            self.metrics["busy_seconds"] += _time.perf_counter() - t0
+            self.metrics["engine_steps"] += 1
             with self._lock:
                 for o in outs:
+                    self.metrics["tokens_generated"] += len(o.new_tokens)
                     p = self._pending.get(o.req_id)
                     if p is None:
                         continue
                     p.tokens.extend(o.new_tokens)
                     if o.finished:
+                        self.metrics["turns_completed"] += 1
                         del self._pending[o.req_id]
                         p.reply.put({"tokens": p.tokens})
 
@@ -94,6 +102,7 @@ class ModelhubServer:
                 "waiting": len(self.engine.waiting),
                 "kv_blocks_total": kv.num_blocks,
                 "kv_blocks_free": kv.allocator.num_free,
+                **self.metrics,
             }
         if method == "release":
             name = params["session"]
