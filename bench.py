@@ -36,6 +36,7 @@ def main() -> None:
     ap.add_argument("--max-model-len", type=int, default=4096)
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--decode-microbatch", type=int, default=32)
+    ap.add_argument("--kv-dtype", choices=["bf16", "fp8"], default="bf16")
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
 
@@ -65,6 +66,7 @@ def main() -> None:
                         use_graphs=(not args.no_graphs and device != "cpu"
                                     and not cfg.is_moe),
                         decode_microbatch=args.decode_microbatch,
+                        kv_dtype=args.kv_dtype,
                         seed=1234 + rank // tp_size)
     if device == "cpu":
         ecfg.num_kv_blocks = (args.sessions *

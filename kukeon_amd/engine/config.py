@@ -90,6 +90,8 @@ class EngineConfig:
     decode_splits: int = 1          # split-KV factor for small-batch decode
     decode_microbatch: int = 16     # decode steps per host sync (self-
                                     # advancing graph replay train)
+    kv_dtype: str = "bf16"          # "bf16" | "fp8" (OCP e4m3 cache: half
+                                    # the KV bandwidth, 2x the capacity)
     seed: int = 1234
     tp_size: int = 1
 

@@ -64,6 +64,9 @@ class LLMEngine:
         self.hk = cfg.num_kv_heads // max(1, parallel.tp_size())
         self.hq = cfg.num_q_heads // max(1, parallel.tp_size())
 
+        kv_torch_dtype = (torch.uint8 if ecfg.kv_dtype == "fp8"
+                          else torch.bfloat16)
+        kv_bytes = 1 if ecfg.kv_dtype == "fp8" else 2
         nblocks = ecfg.num_kv_blocks
         if nblocks <= 0:
             if self.is_cuda:
@@ -71,12 +74,13 @@ class LLMEngine:
                 budget = int(free * ecfg.kv_mem_fraction) - (2 << 30)
                 nblocks = PagedKVCache.blocks_from_bytes(
                     max(budget, 1 << 28), cfg.num_layers, self.hk,
-                    ecfg.block_size, cfg.head_dim)
+                    ecfg.block_size, cfg.head_dim, dtype_bytes=kv_bytes)
                 nblocks = min(nblocks, 4_000_000)
             else:
                 nblocks = 512
         self.kv = PagedKVCache(cfg.num_layers, nblocks, self.hk,
-                               ecfg.block_size, cfg.head_dim, self.device)
+                               ecfg.block_size, cfg.head_dim, self.device,
+                               dtype=kv_torch_dtype)
         log.info("KV cache: %d blocks (%.1f GiB for K+V)", nblocks,
                  self.kv.k.nbytes * 2 / (1 << 30))
 

@@ -56,6 +56,39 @@ DEV_INLINE uint4 pack_bf16x8(const float* f) {
   return r;
 }
 
+// ---------- fp8 (OCP e4m3fn) helpers ----------
+typedef __attribute__((__vector_size__(2 * sizeof(float)))) float f32x2_cvt_t;
+
+DEV_INLINE unsigned int cvtpk_bf16(float lo, float hi) {
+  __hip_bfloat162 h = __float22bfloat162_rn(float2{lo, hi});
+  unsigned int r;
+  __builtin_memcpy(&r, &h, 4);
+  return r;
+}
+
+// 4 fp8 bytes (one dword) -> 4 bf16 (uint2)
+DEV_INLINE uint2 fp8x4_to_bf16x4(unsigned int w) {
+  f32x2_cvt_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(w, false);
+  f32x2_cvt_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(w, true);
+  return uint2{cvtpk_bf16(lo[0], lo[1]), cvtpk_bf16(hi[0], hi[1])};
+}
+
+// 8 fp8 bytes (uint2) -> 8 bf16 (uint4)
+DEV_INLINE uint4 fp8x8_to_bf16x8(uint2 w) {
+  const uint2 a = fp8x4_to_bf16x4(w.x);
+  const uint2 b = fp8x4_to_bf16x4(w.y);
+  return uint4{a.x, a.y, b.x, b.y};
+}
+
+// 8 f32 -> 8 fp8 bytes (uint2)
+DEV_INLINE uint2 pack_fp8x8(const float* f) {
+  unsigned int w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[0], f[1], 0u, false);
+  w0 = __builtin_amdgcn_cvt_pk_fp8_f32(f[2], f[3], w0, true);
+  unsigned int w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[4], f[5], 0u, false);
+  w1 = __builtin_amdgcn_cvt_pk_fp8_f32(f[6], f[7], w1, true);
+  return uint2{w0, w1};
+}
+
 // ---------- wave reductions (wave64) ----------
 DEV_INLINE float wave_sum(float v) {
 #pragma unroll

@@ -26,14 +26,14 @@ namespace kukeon {
 
 typedef __attribute__((__vector_size__(2 * sizeof(short)))) short bf16x2_t;
 
-template <int D, int BS, int GPW>
+template <int D, int BS, int GPW, bool FP8>
 __global__ __launch_bounds__(256) void paged_attn_kernel(
     unsigned short* __restrict__ out,     // [B, Hq*D] bf16 (splits==1)
     float* __restrict__ tmp_out,          // [B, Hq, S, D] f32 (splits>1)
     float* __restrict__ tmp_ml,           // [B, Hq, S, 2]
     const unsigned short* __restrict__ q, // [B, q_stride] (fused qkv row)
-    const unsigned short* __restrict__ k_cache, // [NB, Hk, BS, D]
-    const unsigned short* __restrict__ v_cache,
+    const void* __restrict__ k_cache,     // [NB, Hk, BS, D] bf16|fp8
+    const void* __restrict__ v_cache,
     const int* __restrict__ block_table,  // [B, max_blocks]
     const int* __restrict__ seq_lens,     // [B]
     long q_stride, int Hq, int Hk, int max_blocks, int num_splits,
@@ -85,33 +85,56 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
     m[g] = -INFINITY; l[g] = 0.f; o0[g] = 0.f; o1[g] = 0.f;
   }
 
-  // stage geometry: each thread owns one 16B chunk of K and V
+  // stage geometry: each thread owns one 16B (bf16) / 8B (fp8) chunk
   const int stg_t = threadIdx.x / (D / 8);      // token 0..15
   const int stg_d0 = (threadIdx.x % (D / 8)) * 8;
   const int stg_dsw = (((stg_d0 / 32) ^ (stg_t & 3)) * 32) + (stg_d0 % 32);
+  const unsigned short* kc16 =
+      reinterpret_cast<const unsigned short*>(k_cache);
+  const unsigned short* vc16 =
+      reinterpret_cast<const unsigned short*>(v_cache);
+  const unsigned char* kc8 = reinterpret_cast<const unsigned char*>(k_cache);
+  const unsigned char* vc8 = reinterpret_cast<const unsigned char*>(v_cache);
   uint4 kreg, vreg;
+  uint2 kreg8, vreg8;
   if (blk_begin < blk_end) {
     {  // prologue: fetch the first block into registers
       const int pb0 = block_table[(long)b * max_blocks + blk_begin];
-      kreg = *reinterpret_cast<const uint4*>(
-          k_cache + (((long)pb0 * Hk + hk) * BS + stg_t) * D + stg_d0);
-      vreg = *reinterpret_cast<const uint4*>(
-          v_cache + (((long)pb0 * Hk + hk) * BS + stg_t) * D + stg_d0);
+      const long off = (((long)pb0 * Hk + hk) * BS + stg_t) * D + stg_d0;
+      if (FP8) {
+        kreg8 = *reinterpret_cast<const uint2*>(kc8 + off);
+        vreg8 = *reinterpret_cast<const uint2*>(vc8 + off);
+      } else {
+        kreg = *reinterpret_cast<const uint4*>(kc16 + off);
+        vreg = *reinterpret_cast<const uint4*>(vc16 + off);
+      }
     }
     for (int bi = blk_begin; bi < blk_end; ++bi) {
-      // ---- write the prefetched block, then issue the next block's
-      // loads so their HBM latency hides under this block's compute
-      // (guide T14 async-stage split) ----
+      // ---- write the prefetched block (fp8 converts to the bf16 LDS
+      // image here, so the compute path is dtype-independent), then issue
+      // the next block's loads so their HBM latency hides under this
+      // block's compute (guide T14 async-stage split) ----
       __syncthreads();
-      *reinterpret_cast<uint4*>(&kbuf[stg_t * D + stg_dsw]) = kreg;
-      *reinterpret_cast<uint4*>(&vbuf[stg_t * D + stg_d0]) = vreg;
+      if (FP8) {
+        *reinterpret_cast<uint4*>(&kbuf[stg_t * D + stg_dsw]) =
+            fp8x8_to_bf16x8(kreg8);
+        *reinterpret_cast<uint4*>(&vbuf[stg_t * D + stg_d0]) =
+            fp8x8_to_bf16x8(vreg8);
+      } else {
+        *reinterpret_cast<uint4*>(&kbuf[stg_t * D + stg_dsw]) = kreg;
+        *reinterpret_cast<uint4*>(&vbuf[stg_t * D + stg_d0]) = vreg;
+      }
       __syncthreads();
       if (bi + 1 < blk_end) {
         const int pbn = block_table[(long)b * max_blocks + bi + 1];
-        kreg = *reinterpret_cast<const uint4*>(
-            k_cache + (((long)pbn * Hk + hk) * BS + stg_t) * D + stg_d0);
-        vreg = *reinterpret_cast<const uint4*>(
-            v_cache + (((long)pbn * Hk + hk) * BS + stg_t) * D + stg_d0);
+        const long off = (((long)pbn * Hk + hk) * BS + stg_t) * D + stg_d0;
+        if (FP8) {
+          kreg8 = *reinterpret_cast<const uint2*>(kc8 + off);
+          vreg8 = *reinterpret_cast<const uint2*>(vc8 + off);
+        } else {
+          kreg = *reinterpret_cast<const uint4*>(kc16 + off);
+          vreg = *reinterpret_cast<const uint4*>(vc16 + off);
+        }
       }
       if (!active) continue;
       // ---- scores ----
@@ -258,16 +281,16 @@ void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
   float* tml = num_splits > 1 ? tmp_ml.data_ptr<float>() : nullptr;
   auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
 
-#define PA_LAUNCH(GPW)                                                        \
-  paged_attn_kernel<128, 16, GPW><<<grid, 256, 0, stream>>>(                  \
-      op, tov, tml, qp, reinterpret_cast<unsigned short*>(k_cache.data_ptr()),\
-      reinterpret_cast<unsigned short*>(v_cache.data_ptr()),                  \
+  const bool fp8 = k_cache.scalar_type() == torch::kUInt8;
+#define PA_LAUNCH(GPW, FP8_)                                                  \
+  paged_attn_kernel<128, 16, GPW, FP8_><<<grid, 256, 0, stream>>>(            \
+      op, tov, tml, qp, k_cache.data_ptr(), v_cache.data_ptr(),               \
       block_table.data_ptr<int>(), seq_lens.data_ptr<int>(), q_stride, Hq,    \
       Hk, max_blocks, (int)num_splits, (float)scale)
   if (G <= 4) {
-    PA_LAUNCH(1);
+    if (fp8) { PA_LAUNCH(1, true); } else { PA_LAUNCH(1, false); }
   } else if (G == 8) {
-    PA_LAUNCH(2);
+    if (fp8) { PA_LAUNCH(2, true); } else { PA_LAUNCH(2, false); }
   } else {
     TORCH_CHECK(false, "unsupported GQA group size ", G);
   }

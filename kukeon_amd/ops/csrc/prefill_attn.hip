@@ -122,12 +122,12 @@ DEV_INLINE unsigned int cvt_pk_bf16(float lo, float hi) {
 
 // ---------- the kernel ----------
 // D=128, QBLK=32 rows per workgroup, KV tile = 32 tokens (2 cache blocks).
-template <int BS>
+template <int BS, bool FP8>
 __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
     unsigned short* __restrict__ out,       // [T, Hq*D]
     const unsigned short* __restrict__ q,   // [T, q_stride] fused qkv rows
-    const unsigned short* __restrict__ k_cache,  // [NB, Hk, BS, D]
-    const unsigned short* __restrict__ v_cache,
+    const void* __restrict__ k_cache,       // [NB, Hk, BS, D] bf16|fp8
+    const void* __restrict__ v_cache,
     const int* __restrict__ block_table,    // [nseq, max_blocks]
     const int* __restrict__ seq_lens,       // [nseq] total ctx after append
     const int* __restrict__ q_starts,       // [nseq, 2] = (abs_start, row0)
@@ -193,12 +193,20 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
       const int kvpos = kvbase + tok;
       const int bidx = min(kvpos / BS, nblocks - 1);
       const int pblk = block_table[(long)s * max_blocks + bidx];
-      const unsigned short* kg =
-          k_cache + (((long)pblk * Hk + hk) * BS + (kvpos % BS)) * D;
+      const long koff = (((long)pblk * Hk + hk) * BS + (kvpos % BS)) * D;
       const int swz = (tok & 15) << 4;
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        const uint4 v = *reinterpret_cast<const uint4*>(kg + d0 + c * 8);
+        uint4 v;
+        if (FP8) {
+          v = fp8x8_to_bf16x8(*reinterpret_cast<const uint2*>(
+              reinterpret_cast<const unsigned char*>(k_cache) + koff + d0 +
+              c * 8));
+        } else {
+          v = *reinterpret_cast<const uint4*>(
+              reinterpret_cast<const unsigned short*>(k_cache) + koff + d0 +
+              c * 8);
+        }
         const int byte = tok * 256 + (((d0 + c * 8) * 2) ^ swz);
         *reinterpret_cast<uint4*>(
             reinterpret_cast<char*>(kbuf) + byte) = v;
@@ -210,14 +218,26 @@ __global__ __launch_bounds__(256, 1) void prefill_attn_kernel(
       const int vpos1 = vpos0 + 1;
       const int vb0 = min(vpos0 / BS, nblocks - 1);
       const int vb1 = min(vpos1 / BS, nblocks - 1);
-      const unsigned short* vg0 =
-          v_cache + (((long)block_table[(long)s * max_blocks + vb0] * Hk + hk)
-                     * BS + (vpos0 % BS)) * D;
-      const unsigned short* vg1 =
-          v_cache + (((long)block_table[(long)s * max_blocks + vb1] * Hk + hk)
-                     * BS + (vpos1 % BS)) * D;
-      uint4 r0 = *reinterpret_cast<const uint4*>(vg0 + vd0);
-      uint4 r1 = *reinterpret_cast<const uint4*>(vg1 + vd0);
+      const long voff0 =
+          (((long)block_table[(long)s * max_blocks + vb0] * Hk + hk) * BS +
+           (vpos0 % BS)) * D;
+      const long voff1 =
+          (((long)block_table[(long)s * max_blocks + vb1] * Hk + hk) * BS +
+           (vpos1 % BS)) * D;
+      uint4 r0, r1;
+      if (FP8) {
+        const unsigned char* v8 =
+            reinterpret_cast<const unsigned char*>(v_cache);
+        r0 = fp8x8_to_bf16x8(
+            *reinterpret_cast<const uint2*>(v8 + voff0 + vd0));
+        r1 = fp8x8_to_bf16x8(
+            *reinterpret_cast<const uint2*>(v8 + voff1 + vd0));
+      } else {
+        const unsigned short* v16 =
+            reinterpret_cast<const unsigned short*>(v_cache);
+        r0 = *reinterpret_cast<const uint4*>(v16 + voff0 + vd0);
+        r1 = *reinterpret_cast<const uint4*>(v16 + voff1 + vd0);
+      }
       // zero V rows past ctx: a 0*NaN in the PV mfma would poison O
       if (vpos0 >= ctx) r0 = uint4{0, 0, 0, 0};
       if (vpos1 >= ctx) r1 = uint4{0, 0, 0, 0};
@@ -350,15 +370,18 @@ void prefill_attention(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(qb_seq.scalar_type() == torch::kInt32);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   dim3 grid(nqb, Hk, (G + 3) / 4);
-  prefill_attn_kernel<16><<<grid, 256, 0, stream>>>(
-      reinterpret_cast<unsigned short*>(out.data_ptr()),
-      reinterpret_cast<const unsigned short*>(q.data_ptr()),
-      reinterpret_cast<const unsigned short*>(k_cache.data_ptr()),
-      reinterpret_cast<const unsigned short*>(v_cache.data_ptr()),
-      block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
-      q_starts.data_ptr<int>(), qb_seq.data_ptr<int>(),
-      qb_start.data_ptr<int>(), q.stride(0), q_offset, Hq, Hk,
-      (int)block_table.size(1), (int)out.stride(0), (float)scale);
+  const bool fp8 = k_cache.scalar_type() == torch::kUInt8;
+#define PF_LAUNCH(FP8_)                                                      \
+  prefill_attn_kernel<16, FP8_><<<grid, 256, 0, stream>>>(                   \
+      reinterpret_cast<unsigned short*>(out.data_ptr()),                     \
+      reinterpret_cast<const unsigned short*>(q.data_ptr()),                 \
+      k_cache.data_ptr(), v_cache.data_ptr(), block_table.data_ptr<int>(),   \
+      seq_lens.data_ptr<int>(), q_starts.data_ptr<int>(),                    \
+      qb_seq.data_ptr<int>(), qb_start.data_ptr<int>(), q.stride(0),         \
+      q_offset, Hq, Hk, (int)block_table.size(1), (int)out.stride(0),        \
+      (float)scale)
+  if (fp8) { PF_LAUNCH(true); } else { PF_LAUNCH(false); }
+#undef PF_LAUNCH
   HIP_CHECK_KERNEL();
 }
 

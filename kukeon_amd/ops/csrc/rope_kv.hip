@@ -15,10 +15,11 @@
 
 namespace kukeon {
 
+template <bool FP8>
 __global__ void rope_kv_kernel(
     unsigned short* __restrict__ qkv,        // [T, (Hq+2Hk)*D]
-    unsigned short* __restrict__ k_cache,    // [NB, Hk, BS, D]
-    unsigned short* __restrict__ v_cache,    // [NB, Hk, BS, D]
+    void* __restrict__ k_cache,              // [NB, Hk, BS, D] bf16|fp8
+    void* __restrict__ v_cache,
     const float* __restrict__ cos_sin,       // [max_pos, D]
     const int* __restrict__ positions,       // [T]
     const int* __restrict__ slot_mapping,    // [T] flat slot = block*BS + off
@@ -85,9 +86,16 @@ __global__ void rope_kv_kernel(
     *reinterpret_cast<uint4*>(base + d0) = px;
     *reinterpret_cast<uint4*>(base + half + d0) = py;
     if (slot >= 0) {
-      unsigned short* kc = k_cache + cache_tok_base + (long)h * BS * D;
-      *reinterpret_cast<uint4*>(kc + d0) = px;
-      *reinterpret_cast<uint4*>(kc + half + d0) = py;
+      const long off = cache_tok_base + (long)h * BS * D;
+      if (FP8) {
+        unsigned char* kc = reinterpret_cast<unsigned char*>(k_cache) + off;
+        *reinterpret_cast<uint2*>(kc + d0) = pack_fp8x8(xo);
+        *reinterpret_cast<uint2*>(kc + half + d0) = pack_fp8x8(yo);
+      } else {
+        unsigned short* kc = reinterpret_cast<unsigned short*>(k_cache) + off;
+        *reinterpret_cast<uint4*>(kc + d0) = px;
+        *reinterpret_cast<uint4*>(kc + half + d0) = py;
+      }
     }
   }
 
@@ -98,9 +106,19 @@ __global__ void rope_kv_kernel(
       const int h = c / (D / 8);
       const int d0 = (c % (D / 8)) * 8;
       const unsigned short* base = row + (long)(Hq + Hk + h) * D;
-      unsigned short* vc = v_cache + cache_tok_base + (long)h * BS * D;
-      *reinterpret_cast<uint4*>(vc + d0) =
-          *reinterpret_cast<const uint4*>(base + d0);
+      const long off = cache_tok_base + (long)h * BS * D;
+      if (FP8) {
+        bf16x8 v = load_bf16x8(base + d0);
+        float f[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) f[j] = v.f(j);
+        unsigned char* vc = reinterpret_cast<unsigned char*>(v_cache) + off;
+        *reinterpret_cast<uint2*>(vc + d0) = pack_fp8x8(f);
+      } else {
+        unsigned short* vc = reinterpret_cast<unsigned short*>(v_cache) + off;
+        *reinterpret_cast<uint4*>(vc + d0) =
+            *reinterpret_cast<const uint4*>(base + d0);
+      }
     }
   }
 }
@@ -125,14 +143,11 @@ void rope_kv_append(torch::Tensor qkv, torch::Tensor k_cache,
   const bool table_mode = block_table.dim() == 2;
   const int* bt = table_mode ? block_table.data_ptr<int>() : nullptr;
   const int mb = table_mode ? (int)block_table.size(1) : 0;
+  const bool fp8 = k_cache.scalar_type() == torch::kUInt8;
   const unsigned ysplit = T <= 128 ? 4 : 1;
-  rope_kv_kernel<<<dim3((unsigned)T, ysplit), 256, 0, stream>>>(
-      reinterpret_cast<unsigned short*>(qkv.data_ptr()),
-      reinterpret_cast<unsigned short*>(k_cache.data_ptr()),
-      reinterpret_cast<unsigned short*>(v_cache.data_ptr()),
-      cos_sin.data_ptr<float>(), positions.data_ptr<int>(),
-      slot_mapping.data_ptr<int>(), bt, mb, (int)num_q_heads,
-      (int)num_kv_heads, D, BS);
+#define RK_LAUNCH(FP8_)                                                         rope_kv_kernel<FP8_><<<dim3((unsigned)T, ysplit), 256, 0, stream>>>(              reinterpret_cast<unsigned short*>(qkv.data_ptr()), k_cache.data_ptr(),        v_cache.data_ptr(), cos_sin.data_ptr<float>(),                                positions.data_ptr<int>(), slot_mapping.data_ptr<int>(), bt, mb,              (int)num_q_heads, (int)num_kv_heads, D, BS)
+  if (fp8) { RK_LAUNCH(true); } else { RK_LAUNCH(false); }
+#undef RK_LAUNCH
   HIP_CHECK_KERNEL();
 }
 

@@ -63,6 +63,10 @@ def rope_kv_append(qkv: torch.Tensor, k_cache: torch.Tensor,
     qkv[:, : num_q_heads * D] = qr.reshape(T, -1)
     qkv[:, num_q_heads * D: (num_q_heads + num_kv_heads) * D] = kr.reshape(T, -1)
     BS = k_cache.shape[2]
+    fp8 = k_cache.dtype == torch.uint8
+    if fp8:
+        kr = kr.float().to(torch.float8_e4m3fn).view(torch.uint8)
+        v = v.float().to(torch.float8_e4m3fn).view(torch.uint8)
     table_mode = block_table is not None and block_table.dim() == 2
     for t in range(T):
         if table_mode:
@@ -81,11 +85,13 @@ def rope_kv_append(qkv: torch.Tensor, k_cache: torch.Tensor,
 
 def _gather_kv(cache: torch.Tensor, block_table: torch.Tensor, ctx: int,
                b: int) -> torch.Tensor:
-    """-> [ctx, Hk, D] from paged cache [NB, Hk, BS, D]."""
+    """-> [ctx, Hk, D] from paged cache [NB, Hk, BS, D] (bf16 or fp8)."""
     BS = cache.shape[2]
     nb = (ctx + BS - 1) // BS
     blocks = block_table[b, :nb].long()
     flat = cache[blocks]                      # [nb, Hk, BS, D]
+    if flat.dtype == torch.uint8:
+        flat = flat.view(torch.float8_e4m3fn).float()
     flat = flat.permute(0, 2, 1, 3).reshape(nb * BS, cache.shape[1], -1)
     return flat[:ctx]
 

@@ -188,3 +188,40 @@ def test_preemption_recompute_matches_naive():
     assert preempted, "expected a preemption under KV exhaustion"
     assert outs[0] == naive_greedy(engine.model, cfg, p1, 6)
     assert outs[1] == naive_greedy(engine.model, cfg, p2, 6)
+
+
+def test_fp8_kv_cache_generates_consistently():
+    """fp8 KV path: engine greedy output matches a naive recompute that
+    quantizes its cache identically."""
+    torch.manual_seed(0)
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=512, max_sessions=8, num_kv_blocks=256,
+                        use_graphs=False, kv_dtype="fp8")
+    model = LlamaModel(cfg, device="cpu")
+    engine = LLMEngine(model, cfg, ecfg, device="cpu")
+    assert engine.kv.k.dtype == torch.uint8
+    prompt = [7, 3, 99, 140, 11, 42, 17, 23]
+    kv = SequenceKV(ecfg.block_size)
+    engine.add_request(kv, prompt,
+                       SamplingParams(temperature=0.0, max_new_tokens=5))
+    got = []
+    while engine.has_work():
+        for o in engine.step():
+            got.extend(o.new_tokens)
+    # naive with an fp8 cache of its own
+    tokens = list(prompt)
+    for _ in range(5):
+        kvc = PagedKVCache(cfg.num_layers, 64, cfg.num_kv_heads, 16,
+                           cfg.head_dim, "cpu", dtype=torch.uint8)
+        T = len(tokens)
+        bt = torch.arange((T + 15) // 16, dtype=torch.int32).unsqueeze(0)
+        meta = AttnMeta(mode="prefill",
+                        positions=torch.arange(T, dtype=torch.int32),
+                        slot_mapping=torch.arange(T, dtype=torch.int32),
+                        block_table=bt,
+                        seq_lens=torch.tensor([T], dtype=torch.int32),
+                        q_starts=torch.tensor([[0, 0]], dtype=torch.int32))
+        hidden = model.forward(torch.tensor(tokens, dtype=torch.int32),
+                               kvc.k, kvc.v, meta)
+        tokens.append(int(model.compute_logits(hidden[-1:]).float().argmax()))
+    assert got == tokens[len(prompt):]

@@ -71,9 +71,12 @@ def test_silu_mul():
                                atol=2e-2)
 
 
-def _make_cache(nb, hk, bs, d):
+def _make_cache(nb, hk, bs, d, fp8=False):
     k = torch.randn(nb, hk, bs, d, dtype=torch.bfloat16, device=DEV)
     v = torch.randn(nb, hk, bs, d, dtype=torch.bfloat16, device=DEV)
+    if fp8:
+        k = k.float().to(torch.float8_e4m3fn).view(torch.uint8)
+        v = v.float().to(torch.float8_e4m3fn).view(torch.uint8)
     return k, v
 
 
@@ -102,14 +105,17 @@ def test_rope_kv_append():
     torch.testing.assert_close(vc.cpu().float(), vc_ref.float(), rtol=0, atol=0)
 
 
-@pytest.mark.parametrize("G,splits", [(4, 1), (4, 4), (8, 1), (8, 3)])
-def test_paged_attention(G, splits):
+@pytest.mark.parametrize("G,splits,fp8", [(4, 1, False), (4, 4, False),
+                                          (8, 1, False), (8, 3, False),
+                                          (4, 1, True), (4, 3, True)])
+def test_paged_attention(G, splits, fp8):
     torch.manual_seed(5)
     B, Hk, D, BS = 5, 2, 128, 16
     Hq = G * Hk
     ctxs = [1, 16, 17, 100, 250]
     NB = sum((c + BS - 1) // BS for c in ctxs) + 4
-    kc, vc = _make_cache(NB, Hk, BS, D)
+    kc, vc = _make_cache(NB, Hk, BS, D, fp8=fp8)
+    tol = 7e-2 if fp8 else 2e-2
     maxb = max((c + BS - 1) // BS for c in ctxs)
     bt = torch.zeros(B, maxb, dtype=torch.int32, device=DEV)
     nxt = 0
@@ -128,8 +134,8 @@ def test_paged_attention(G, splits):
     ref = torch.empty(B, Hq * D, dtype=torch.bfloat16)
     reference.paged_attention(ref, q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
                               seq_lens.cpu(), 0, 1, scale)
-    torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=2e-2,
-                               atol=2e-2)
+    torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=tol,
+                               atol=tol)
 
 
 def test_paged_attention_q_offset():
@@ -230,14 +236,15 @@ def test_moe_gather_scatter():
                                atol=2e-2)
 
 
-def _prefill_case(ctxs_starts, Hq, Hk, seed=11):
+def _prefill_case(ctxs_starts, Hq, Hk, seed=11, fp8=False):
     """ctxs_starts: list of (total_ctx, q_start). Builds caches+q and runs
     both impls."""
     torch.manual_seed(seed)
     D, BS = 128, 16
     nseq = len(ctxs_starts)
     NB = sum((c + BS - 1) // BS for c, _ in ctxs_starts) + 2
-    kc, vc = _make_cache(NB, Hk, BS, D)
+    kc, vc = _make_cache(NB, Hk, BS, D, fp8=fp8)
+    tol = 7e-2 if fp8 else 2.5e-2
     maxb = max((c + BS - 1) // BS for c, _ in ctxs_starts)
     bt = torch.zeros(nseq, maxb, dtype=torch.int32, device=DEV)
     nxt = 0
@@ -270,8 +277,8 @@ def _prefill_case(ctxs_starts, Hq, Hk, seed=11):
     reference.prefill_attention(ref, q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
                                 seq_lens.cpu(), q_starts.cpu(), None, None,
                                 0, scale)
-    torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=2.5e-2,
-                               atol=2.5e-2)
+    torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=tol,
+                               atol=tol)
 
 
 def test_prefill_single_seq_aligned():
@@ -289,6 +296,33 @@ def test_prefill_continuation():
 
 def test_prefill_gqa8():
     _prefill_case([(77, 0), (130, 40)], Hq=8, Hk=1)
+
+
+def test_prefill_fp8_kv():
+    _prefill_case([(100, 0), (90, 64)], Hq=8, Hk=2, fp8=True)
+
+
+def test_rope_kv_append_fp8():
+    torch.manual_seed(14)
+    T, Hq, Hk, D, BS, NB = 6, 4, 2, 128, 16, 4
+    qkv = torch.randn(T, (Hq + 2 * Hk) * D, dtype=torch.bfloat16, device=DEV)
+    kc, vc = _make_cache(NB, Hk, BS, D, fp8=True)
+    inv = 1.0 / (500000.0 ** (torch.arange(0, D, 2).float() / D))
+    fr = torch.outer(torch.arange(64).float(), inv)
+    cos_sin = torch.cat([fr.cos(), fr.sin()], dim=1).to(DEV)
+    pos = torch.randint(0, 64, (T,), dtype=torch.int32, device=DEV)
+    slots = torch.randperm(NB * BS)[:T].to(torch.int32).to(DEV)
+    kc_ref, vc_ref = kc.cpu().clone(), vc.cpu().clone()
+    qkv_ref = qkv.cpu().clone()
+    ops.rope_kv_append(qkv, kc, vc, cos_sin, pos, slots, Hq, Hk, D)
+    reference.rope_kv_append(qkv_ref, kc_ref, vc_ref, cos_sin.cpu(),
+                             pos.cpu(), slots.cpu(), Hq, Hk, D)
+    k_gpu = kc.cpu().view(torch.float8_e4m3fn).float()
+    k_ref = kc_ref.view(torch.float8_e4m3fn).float()
+    torch.testing.assert_close(k_gpu, k_ref, rtol=8e-2, atol=8e-2)
+    v_gpu = vc.cpu().view(torch.float8_e4m3fn).float()
+    v_ref = vc_ref.view(torch.float8_e4m3fn).float()
+    torch.testing.assert_close(v_gpu, v_ref, rtol=8e-2, atol=8e-2)
 
 
 def test_prefill_single_token_turns():
