@@ -59,12 +59,10 @@ def main() -> None:
 
     cfg = MODEL_PRESETS[args.model]()
     tp_size = world if args.model == "llama-3-70b" else 1
-    # MoE decode runs eager: per-expert GEMM sizes are data-dependent.
-    # TP ranks share a sampling seed (identical draws, no collective needed).
+    # TP ranks share a sampling seed (identical draws, no collective needed)
     ecfg = EngineConfig(max_model_len=args.max_model_len,
                         max_sessions=max(args.sessions, 8),
-                        use_graphs=(not args.no_graphs and device != "cpu"
-                                    and not cfg.is_moe),
+                        use_graphs=(not args.no_graphs and device != "cpu"),
                         decode_microbatch=args.decode_microbatch,
                         kv_dtype=args.kv_dtype,
                         seed=1234 + rank // tp_size)

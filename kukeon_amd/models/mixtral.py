@@ -41,6 +41,38 @@ class MixtralMoE(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T = x.shape[0]
+        if x.is_cuda and T <= 128:
+            return self._forward_dense(x)
+        return self._forward_sparse(x)
+
+    def _forward_dense(self, x: torch.Tensor) -> torch.Tensor:
+        """Decode path: run EVERY expert on the whole (small) batch as one
+        batched GEMM and combine with the dense routing-weight matrix.
+        Decode is weight-bandwidth-bound — all 8 experts' weights stream
+        through HBM regardless — so the 8x redundant flops are free, and
+        this path has no router host-sync, no permute, and is
+        hipGraph-capturable (the sparse path's per-expert loop is neither).
+        """
+        T = x.shape[0]
+        logits = F.linear(x, self.router_w).float()            # [T, E]
+        probs = torch.softmax(logits, dim=-1)
+        topv, topi = probs.topk(self.K, dim=-1)                # [T, K]
+        topv = topv / topv.sum(dim=-1, keepdim=True)
+        wdense = torch.zeros(T, self.E, dtype=torch.float32,
+                             device=x.device)
+        wdense.scatter_(1, topi, topv)                         # [T, E]
+        x_rep = x.unsqueeze(0).expand(self.E, T, x.shape[1])
+        gu = torch.bmm(x_rep, self.gate_up_w.transpose(1, 2))  # [E,T,2I]
+        act = torch.empty(self.E * T, self.inter, dtype=x.dtype,
+                          device=x.device)
+        ops.silu_mul(act, gu.reshape(self.E * T, 2 * self.inter))
+        y = torch.bmm(act.view(self.E, T, self.inter),
+                      self.down_w.transpose(1, 2))             # [E,T,H]
+        out = (y.float() * wdense.t().unsqueeze(-1)).sum(dim=0)
+        return parallel.tp_all_reduce(out.to(x.dtype))
+
+    def _forward_sparse(self, x: torch.Tensor) -> torch.Tensor:
+        T = x.shape[0]
         logits = F.linear(x, self.router_w).float()           # [T, E]
         probs = torch.softmax(logits, dim=-1)
         topv, topi = probs.topk(self.K, dim=-1)               # [T, K]

@@ -51,3 +51,15 @@ def test_mixtral_engine_generates():
             toks.extend(o.new_tokens)
     assert len(toks) == 4
     assert all(0 <= t < cfg.vocab_size for t in toks)
+
+
+def test_dense_and_sparse_moe_paths_agree():
+    torch.manual_seed(1)
+    cfg = tiny_mixtral()
+    moe = __import__("kukeon_amd.models.mixtral",
+                     fromlist=["MixtralMoE"]).MixtralMoE(cfg, "cpu")
+    x = torch.randn(11, cfg.hidden_size, dtype=torch.bfloat16)
+    sparse = moe._forward_sparse(x.clone())
+    dense = moe._forward_dense(x.clone())
+    torch.testing.assert_close(dense.float(), sparse.float(), rtol=3e-2,
+                               atol=3e-2)
