@@ -28,7 +28,6 @@
 namespace kukeon {
 
 constexpr int WS_ROW = 528;
-constexpr int S_MAX = 0;  // (sentinel comment: max mapped u32 starts at 0)
 
 DEV_INLINE unsigned int map16(unsigned short us) {
   return (us & 0x8000u) ? (unsigned int)(~us & 0xffffu)
