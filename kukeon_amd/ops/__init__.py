@@ -96,10 +96,14 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
             and w.shape[0] % 64 == 0 and w.shape[1] % 32 == 0
             and x.dtype == torch.bfloat16):
         N = w.shape[0]
+        K = w.shape[1]
         out = torch.empty(rows, N, dtype=x.dtype, device=x.device)
         key = (x.device.index or 0)
         ws = _SKINNY_WS.get(key)
-        need = 64 * N
+        ntiles = N // 64
+        nslices = -(-K // 512)
+        splitk = 1 if ntiles >= 512 else min(nslices, -(-512 // ntiles))
+        need = max(1, splitk) * 64 * N
         if ws is None or ws.numel() < need:
             # grown only outside graph capture (engine warmup runs eager)
             ws = torch.empty(need, dtype=torch.float32, device=x.device)

@@ -67,20 +67,30 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   const int n0 = blockIdx.x * 64 + wid * 16;     // this wave's 16 N rows
   const int row16 = lane & 15;                   // A row / B col
   const int kgrp = lane >> 4;                    // 0..3 -> k = 8*kgrp + j
-  const long ks = (long)blockIdx.y * KSLICE;
-  const int klen = (int)min((long)KSLICE, K - ks);
-  const int nbatch = (klen + U * 32 - 1) / (U * 32);
-
-  const unsigned short* wrow = w + (long)(n0 + row16) * K + ks + 8 * kgrp;
-
-  // ---- batch 0 of W goes in flight before anything else ----
-  u32x4_t wa[U];
+  f32x4_t acc[MT];
 #pragma unroll
-  for (int u = 0; u < U; ++u)
-    wa[u] = nt_load16(wrow + u * 32);
+  for (int m = 0; m < MT; ++m) acc[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+
+  // a block walks slices {blockIdx.y, +gridDim.y, ...} and accumulates
+  // locally, so grid.y is an occupancy choice, not forced to K/512
+  for (long ks = (long)blockIdx.y * KSLICE; ks < K;
+       ks += (long)gridDim.y * KSLICE) {
+  const int klen = (int)min((long)KSLICE, K - ks);
+  const unsigned short* wrow = w + (long)(n0 + row16) * K + ks + 8 * kgrp;
+  const int nfull = klen / (U * 32);
+
+  // this slice's first W batch goes in flight BEFORE the staging barrier
+  // (no LDS dependence), so its HBM latency hides under the x staging
+  u32x4_t wa0[U];
+  if (nfull >= 1) {
+#pragma unroll
+    for (int u = 0; u < U; ++u)
+      wa0[u] = nt_load16(wrow + u * 32);
+  }
 
   // ---- stage x[0..63][ks..ks+klen) into LDS (loads batched, then
   // writes, so the 8 L2 round trips overlap) ----
+  __syncthreads();  // previous slice's readers are done
   {
     const int xr = threadIdx.x >> 2;             // row 0..63
     const int c0 = (threadIdx.x & 3) * 8;
@@ -101,22 +111,37 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   }
   __syncthreads();
 
-  f32x4_t acc[MT];
+  // Batch 0 consumes the pre-staged wa0; later batches keep loads and
+  // consume in the SAME iteration with no explicit double-buffer: hipcc
+  // then emits counted vmcnt(N) per u so each load's latency hides under
+  // the consumption of earlier ones (the explicit next-batch prefetch
+  // variant got hoisted to the loop bottom and drained with one vmcnt(0)
+  // at the top — nothing overlapped; 45us vs 13us on the
+  // ingredient-identical pattern probe). No runtime condition inside the
+  // unrolled body either (guide §5 ".s-level traps" (c)).
+  if (nfull >= 1) {
 #pragma unroll
-  for (int m = 0; m < MT; ++m) acc[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
-
-  u32x4_t wb[U];
-  for (int b = 0; b < nbatch; ++b) {
-    // issue the next batch while this one feeds the MFMAs
-    if (b + 1 < nbatch) {
+    for (int u = 0; u < U; ++u) {
+      const int kc = u * 32 + 8 * kgrp;
+      const bf16x8_t afrag = frag_of(wa0[u]);
 #pragma unroll
-      for (int u = 0; u < U; ++u)
-        wb[u] = nt_load16(wrow + (b + 1) * U * 32 + u * 32);
+      for (int m = 0; m < MT; ++m) {
+        const int xr = m * 16 + row16;
+        const uint4 xb = *reinterpret_cast<const uint4*>(
+            reinterpret_cast<const char*>(xbuf) + xswz(xr, kc * 2));
+        acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, frag_of(xb), acc[m], 0, 0, 0);
+      }
     }
+  }
+  for (int b = 1; b < nfull; ++b) {
+    u32x4_t wa[U];
+#pragma unroll
+    for (int u = 0; u < U; ++u)
+      wa[u] = nt_load16(wrow + b * U * 32 + u * 32);
 #pragma unroll
     for (int u = 0; u < U; ++u) {
       const int kc = b * U * 32 + u * 32 + 8 * kgrp;
-      if (kc >= klen) break;
       const bf16x8_t afrag = frag_of(wa[u]);
 #pragma unroll
       for (int m = 0; m < MT; ++m) {
@@ -127,13 +152,28 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
             afrag, frag_of(xb), acc[m], 0, 0, 0);
       }
     }
-#pragma unroll
-    for (int u = 0; u < U; ++u) wa[u] = wb[u];
   }
+  // tail chunks (klen not a multiple of U*32), one chunk at a time
+  for (int kc0 = nfull * U * 32; kc0 < klen; kc0 += 32) {
+    const bf16x8_t afrag = frag_of(nt_load16(wrow + kc0));
+    const int kc = kc0 + 8 * kgrp;
+#pragma unroll
+    for (int m = 0; m < MT; ++m) {
+      const int xr = m * 16 + row16;
+      const uint4 xb = *reinterpret_cast<const uint4*>(
+          reinterpret_cast<const char*>(xbuf) + xswz(xr, kc * 2));
+      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          afrag, frag_of(xb), acc[m], 0, 0, 0);
+    }
+  }
+  }  // slice loop
 
   // C layout (16x16): lane -> col = lane&15 (the M index here),
-  // row = 4*(lane>>4) + r (the N index)
+  // row = 4*(lane>>4) + r (the N index). Split-K partials go to plain
+  // per-slice slabs (atomicAdd here measured ~35us of L2 RMW serialization
+  // on the 3M-element qkv epilogue; slabs + a reduce pass are ~4us).
   const int ncol = n0 + 4 * kgrp;
+  float* slab = SPLIT ? ws + (long)blockIdx.y * M * N : nullptr;
 #pragma unroll
   for (int m = 0; m < MT; ++m) {
     const int mrow = m * 16 + row16;
@@ -142,7 +182,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     for (int r = 0; r < 4; ++r) {
       const long off = (long)mrow * N + ncol + r;
       if (SPLIT) {
-        atomicAdd(ws + off, acc[m][r]);
+        slab[off] = acc[m][r];
       } else {
         out[off] = f2us(acc[m][r]);
       }
@@ -150,16 +190,26 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   }
 }
 
-__global__ void skinny_cast_kernel(unsigned short* __restrict__ out,
-                                   const float* __restrict__ ws, long n) {
+// reduce the split-K slabs and cast to bf16
+__global__ void skinny_reduce_kernel(unsigned short* __restrict__ out,
+                                     const float* __restrict__ ws, long n,
+                                     int splitk) {
   const long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   if (i + 7 < n) {
-    float4 a = *reinterpret_cast<const float4*>(ws + i);
-    float4 b = *reinterpret_cast<const float4*>(ws + i + 4);
-    float v[8] = {a.x, a.y, a.z, a.w, b.x, b.y, b.z, b.w};
+    float v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int s = 0; s < splitk; ++s) {
+      const float4 a = *reinterpret_cast<const float4*>(ws + s * n + i);
+      const float4 b = *reinterpret_cast<const float4*>(ws + s * n + i + 4);
+      v[0] += a.x; v[1] += a.y; v[2] += a.z; v[3] += a.w;
+      v[4] += b.x; v[5] += b.y; v[6] += b.z; v[7] += b.w;
+    }
     *reinterpret_cast<uint4*>(out + i) = pack_bf16x8(v);
   } else {
-    for (long j = i; j < n; ++j) out[j] = f2us(ws[j]);
+    for (long j = i; j < n; ++j) {
+      float acc = 0.f;
+      for (int s = 0; s < splitk; ++s) acc += ws[s * n + j];
+      out[j] = f2us(acc);
+    }
   }
 }
 
@@ -174,7 +224,9 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int ntiles = N / 64;
-  const int splitk = (int)((K + KSLICE - 1) / KSLICE);
+  const int nslices = (int)((K + KSLICE - 1) / KSLICE);
+  int splitk = 1;
+  if (ntiles < 512) splitk = min(nslices, (512 + ntiles - 1) / ntiles);
   const int MT = (M + 15) / 16;
   dim3 grid(ntiles, splitk);
   auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
@@ -188,13 +240,12 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
         op, nullptr, xp, wp, M, N, K);                                       \
   } else {                                                                   \
     float* wsp = ws.data_ptr<float>();                                       \
-    TORCH_CHECK(ws.numel() >= total, "skinny_gemm workspace too small");     \
-    skinny_zero_kernel<<<dim3((unsigned)((total / 4 + 255) / 256)), 256, 0,  \
-                         stream>>>(wsp, total);                              \
+    TORCH_CHECK(ws.numel() >= total * splitk,                                \
+                "skinny_gemm workspace too small");                          \
     skinny_gemm_kernel<MT_, true><<<grid, 256, 0, stream>>>(                 \
         nullptr, wsp, xp, wp, M, N, K);                                      \
-    skinny_cast_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256, 0,  \
-                         stream>>>(op, wsp, total);                          \
+    skinny_reduce_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256,   \
+                           0, stream>>>(op, wsp, total, splitk);             \
   }
   switch (MT) {
     case 1: SK_LAUNCH(1); break;

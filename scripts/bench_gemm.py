@@ -17,7 +17,7 @@ for (M, N, K, tag) in [(64, 6144, 4096, "qkv"), (64, 4096, 4096, "o"),
     x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") * 0.3
     w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.05
     out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
-    ws = torch.empty(64 * N, dtype=torch.float32, device="cuda")
+    ws = torch.empty(16 * 64 * N, dtype=torch.float32, device="cuda")
     us_sk = t(lambda: _C.skinny_gemm(out, x, w, ws))
     us_bl = t(lambda: F.linear(x, w))
     floor = N * K * 2 / 6.3e12 * 1e6

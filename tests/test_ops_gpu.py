@@ -351,7 +351,7 @@ def test_skinny_gemm(M, N, K):
     x = (torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.5)
     w = (torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05)
     out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
-    ws = torch.empty(64 * N, dtype=torch.float32, device=DEV)
+    ws = torch.empty(16 * 64 * N, dtype=torch.float32, device=DEV)
     _C.skinny_gemm(out, x, w, ws)
     ref = (x.float() @ w.float().T)
     torch.testing.assert_close(out.float().cpu(), ref.cpu(), rtol=3e-2,
