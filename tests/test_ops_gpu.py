@@ -356,3 +356,33 @@ def test_skinny_gemm(M, N, K):
     ref = (x.float() @ w.float().T)
     torch.testing.assert_close(out.float().cpu(), ref.cpu(), rtol=3e-2,
                                atol=3e-2)
+
+
+def test_decode_graphs_match_eager():
+    """Self-advancing graph decode must produce the same tokens as the
+    eager path (same seeds, greedy)."""
+    from kukeon_amd.engine.config import EngineConfig, SamplingParams, tiny_llama
+    from kukeon_amd.engine.engine import LLMEngine
+    from kukeon_amd.engine.kv_cache import SequenceKV
+    from kukeon_amd.models.llama import LlamaModel
+
+    cfg = tiny_llama()
+    prompt = [7, 3, 99, 140, 11, 42, 17, 23, 5, 81]
+    outs = {}
+    for use_graphs in (False, True):
+        torch.manual_seed(0)
+        ecfg = EngineConfig(max_model_len=256, max_sessions=4,
+                            num_kv_blocks=128, use_graphs=use_graphs,
+                            decode_microbatch=4)
+        model = LlamaModel(cfg, device=DEV)
+        engine = LLMEngine(model, cfg, ecfg, device=DEV)
+        kv = SequenceKV(ecfg.block_size)
+        engine.add_request(kv, prompt,
+                           SamplingParams(temperature=0.0, max_new_tokens=9))
+        toks = []
+        while engine.has_work():
+            for o in engine.step():
+                toks.extend(o.new_tokens)
+        outs[use_graphs] = toks
+    assert outs[False] == outs[True], outs
+    assert len(outs[True]) == 9
