@@ -323,7 +323,7 @@ class LLMEngine:
     def _bucket(self, b: int) -> int:
         for s in self.ecfg.graph_buckets:
             if b <= s:
-                return s
+                return min(s, self.Bmax)
         return self.Bmax
 
     def _decode_splits(self, bucket: int) -> int:
@@ -495,5 +495,6 @@ class LLMEngine:
         self.d_ids.zero_()
         self.d_pos.zero_()
         for b in sorted(self.ecfg.graph_buckets, reverse=True):
+            b = min(b, self.Bmax)
             if b not in self.graphs:
                 self._capture(b)
