@@ -407,10 +407,12 @@ class LLMEngine:
                 r.bt_written = n
         use_graph = self.is_cuda and self.ecfg.use_graphs
         bucket = self._bucket(nrows) if use_graph else nrows
+        for row in range(nrows):
+            if self._rows[row] is None:
+                self.n_pos[row] = -1   # rope kernel skips the KV write
         for row in range(nrows, bucket):
             self.n_seq_lens[row] = 0
-        if use_graph and bucket not in self.graphs:
-            self._capture_safely(bucket)
+            self.n_pos[row] = -1
         nb = bucket
         self.d_ids[:nb].copy_(self.h_ids[:nb], non_blocking=True)
         self.d_pos[:nb].copy_(self.h_pos[:nb], non_blocking=True)
@@ -423,6 +425,12 @@ class LLMEngine:
             self.d_topp.copy_(self.h_params[:, 2], non_blocking=True)
             self._params_dirty = False
         self.d_step.zero_()
+        if use_graph and bucket not in self.graphs:
+            # capture AFTER the staging copies: the warmup forward then runs
+            # on the real current state (its KV writes land exactly where
+            # the first replay re-writes the same values), never on stale
+            # block tables pointing at freed blocks.
+            self._capture_safely(bucket)
         if use_graph:
             g = self.graphs[bucket]
             for _ in range(k):
@@ -493,7 +501,7 @@ class LLMEngine:
         self.d_slots.fill_(-1)
         self.d_seq_lens.zero_()
         self.d_ids.zero_()
-        self.d_pos.zero_()
+        self.d_pos.fill_(-1)   # rope kernel skips KV writes for pos < 0
         for b in sorted(self.ecfg.graph_buckets, reverse=True):
             b = min(b, self.Bmax)
             if b not in self.graphs:

@@ -35,12 +35,11 @@ __global__ void rope_kv_kernel(
   const long row_stride = (long)(Hq + 2 * Hk) * D;
   unsigned short* row = qkv + t * row_stride;
   const int pos = positions[t];
+  if (pos < 0) return;  // inactive decode row: no rotate, no KV write
   int slot;
   if (block_table != nullptr) {
     // decode: the device owns the sequence cursor (self-advancing graph)
-    slot = (pos >= 0)
-               ? block_table[t * (long)max_blocks + pos / BS] * BS + pos % BS
-               : -1;
+    slot = block_table[t * (long)max_blocks + pos / BS] * BS + pos % BS;
   } else {
     slot = slot_mapping[t];
   }

@@ -56,12 +56,17 @@ def rope_kv_append(qkv: torch.Tensor, k_cache: torch.Tensor,
     k = qkv[:, num_q_heads * D: (num_q_heads + num_kv_heads) * D].view(
         T, num_kv_heads, D).float()
     v = qkv[:, (num_q_heads + num_kv_heads) * D:].view(T, num_kv_heads, D)
-    cs = cos_sin[positions.long()]
+    cs = cos_sin[positions.long().clamp_min(0)]
     cos, sin = cs[:, :half], cs[:, half:]
     qr = _rotate(q, cos, sin).to(qkv.dtype)
     kr = _rotate(k, cos, sin).to(qkv.dtype)
-    qkv[:, : num_q_heads * D] = qr.reshape(T, -1)
-    qkv[:, num_q_heads * D: (num_q_heads + num_kv_heads) * D] = kr.reshape(T, -1)
+    # inactive decode rows (pos < 0) are skipped entirely, like the kernel
+    act = (positions >= 0).view(T, 1)
+    qkv[:, : num_q_heads * D] = torch.where(
+        act, qr.reshape(T, -1), qkv[:, : num_q_heads * D])
+    qkv[:, num_q_heads * D: (num_q_heads + num_kv_heads) * D] = torch.where(
+        act, kr.reshape(T, -1),
+        qkv[:, num_q_heads * D: (num_q_heads + num_kv_heads) * D])
     BS = k_cache.shape[2]
     fp8 = k_cache.dtype == torch.uint8
     if fp8:
