@@ -73,8 +73,13 @@ class PagedKVCache:
         self.num_blocks = num_blocks
         self.block_size = block_size
         shape = (num_layers, num_blocks, num_kv_heads, block_size, head_dim)
-        self.k = torch.empty(shape, dtype=dtype, device=device)
-        self.v = torch.empty(shape, dtype=dtype, device=device)
+        # zeros, NOT empty: attention masks out-of-range tail-block tokens
+        # by score (-inf -> weight 0), but their V still enters the
+        # accumulation as 0*value — uninitialized bytes can decode to NaN
+        # bf16 and 0*NaN poisons the whole row. Zero-init guarantees every
+        # slot is 0.0 or a previously written finite value.
+        self.k = torch.zeros(shape, dtype=dtype, device=device)
+        self.v = torch.zeros(shape, dtype=dtype, device=device)
         self.allocator = BlockAllocator(num_blocks)
 
     @staticmethod
