@@ -82,19 +82,27 @@ def sample(tokens, logits, temps, top_k, top_p, seed, workspace) -> None:
 
 
 _SKINNY_WS = {}
-# The hand-written skinny GEMM currently trails hipBLASLt (see
-# profiles/; best variant ~2.5x off the roofline vs hipBLASLt's 36-100%)
-# so it is opt-in until a future round closes the gap.
+# Per-shape dispatch, from the measured sweep (profiles/r01_progress.md,
+# scripts/sweep_splitk.py on MI355X): the glds-staged skinny kernel beats
+# hipBLASLt on square o-projection shapes (N==K: 16.6us vs 19.3 at M=64,
+# 11.9 vs 19.2 at M=16 on 4096x4096) and trails it on qkv/gate_up/down/
+# lm_head, where hipBLASLt is 55-100% of the HBM floor. KUKEON_SKINNY_GEMM=1
+# forces the skinny kernel for every eligible shape (benchmarking).
 _USE_SKINNY = __import__("os").environ.get("KUKEON_SKINNY_GEMM", "0") == "1"
+
+
+def _skinny_wins(rows: int, N: int, K: int) -> bool:
+    return N == K == 4096
 
 
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """F.linear with the weight-streaming skinny-GEMM kernel on the decode
-    path (CUDA, rows <= 64, aligned shapes, opt-in); hipBLASLt otherwise."""
+    shapes where it measured faster than hipBLASLt; hipBLASLt otherwise."""
     rows = x.shape[0]
-    if (_USE_SKINNY and x.is_cuda and x.dim() == 2 and rows <= 64
+    if (x.is_cuda and x.dim() == 2 and rows <= 64
             and w.shape[0] % 64 == 0 and w.shape[1] % 32 == 0
-            and x.dtype == torch.bfloat16):
+            and x.dtype == torch.bfloat16
+            and (_USE_SKINNY or _skinny_wins(rows, w.shape[0], w.shape[1]))):
         N = w.shape[0]
         K = w.shape[1]
         out = torch.empty(rows, N, dtype=x.dtype, device=x.device)
@@ -102,7 +110,7 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
         ws = _SKINNY_WS.get(key)
         ntiles = N // 64
         nslices = -(-K // 512)
-        splitk = 1 if ntiles >= 512 else min(nslices, -(-512 // ntiles))
+        splitk = 1 if ntiles >= 512 else min(nslices, -(-256 // ntiles))
         need = max(1, splitk) * 64 * N
         if ws is None or ws.numel() < need:
             # grown only outside graph capture (engine warmup runs eager)

@@ -29,6 +29,7 @@ void mfma_probe(torch::Tensor out, torch::Tensor a, torch::Tensor b);
 void mfma_probe16(torch::Tensor out, torch::Tensor a, torch::Tensor b);
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws);
+void glds_probe(torch::Tensor out, torch::Tensor src);
 void sample(torch::Tensor tokens, torch::Tensor logits, torch::Tensor temps,
             torch::Tensor top_k, torch::Tensor top_p, torch::Tensor seed,
             torch::Tensor workspace);
@@ -56,6 +57,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "16x16x32 bf16 MFMA fragment-layout probe");
   m.def("skinny_gemm", &kukeon::skinny_gemm,
         "weight-streaming decode GEMM (M<=64)");
+  m.def("glds_probe", &kukeon::glds_probe,
+        "asm global_load_lds round-trip validator");
   m.def("sample", &kukeon::sample, "top-k/top-p/temperature sampling");
   m.def("decode_advance", &kukeon::decode_advance,
         "on-device decode cursor advance (self-advancing graph)");

@@ -53,6 +53,87 @@ DEV_INLINE int xswz(int row, int byte_in_row) {
   return row * (KSLICE * 2) + (byte_in_row ^ ((row & 15) << 4));
 }
 
+// ---- async global->LDS staging (gfx950 LDS-DMA) ----
+// One instruction stages 64 lanes x 16 B = 1 KiB to a WAVE-UNIFORM LDS
+// base + lane*16 (lane-linear dest — guide §5.4 rule 21); the swizzle
+// therefore goes on the per-lane SOURCE address. Raw asm, not the
+// builtin: hipcc tracks the builtin in its s_waitcnt bookkeeping and
+// demotes every later counted vmcnt(N) to vmcnt(0) while one is in
+// flight, which would serialize the W-stream pipeline this staging is
+// meant to hide under. m0 (the LDS-DMA destination base) is saved and
+// restored in the same statement (it is compiler-reserved).
+DEV_INLINE void glds16(const unsigned short* gsrc, unsigned lds_dst) {
+  unsigned keep;
+  asm volatile(
+      "s_mov_b32 %0, m0\n\t"
+      "s_mov_b32 m0, %2\n\t"
+      "s_nop 0\n\t"
+      "global_load_lds_dwordx4 %1, off\n\t"
+      "s_mov_b32 m0, %0"
+      : "=&s"(keep)
+      : "v"(gsrc), "s"(lds_dst)
+      : "memory");
+}
+
+DEV_INLINE unsigned lds_addr_of(const void* p) {
+  // LDS aperture is 2^32-aligned: the low 32 bits of a generic pointer
+  // into LDS are the LDS byte address m0/LDS-DMA expects (validated by
+  // glds_probe_kernel below against a read-back).
+  return (unsigned)(unsigned long long)p;
+}
+
+// Stage rows [0,64) x KSLICE cols of x into the xswz LDS image with 16
+// asm glds per wave (wave w owns rows 16w..16w+15). Completion rides the
+// VM counter; callers drain with s_waitcnt vmcnt + barrier. For tail
+// slices (klen < KSLICE) the per-lane source byte offset is clamped
+// in-bounds — the over-staged slots hold junk the consumer never reads
+// (it bounds every read by klen).
+DEV_INLINE void glds_stage_x(unsigned short* xbuf,
+                             const unsigned short* __restrict__ x, long K,
+                             long ks, int klen, int M, int wid, int lane) {
+  const int cap = klen * 2 - 16;  // klen%32==0 -> 16B-aligned
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    const int row = wid * 16 + i;
+    const int byte_in_row = min((lane * 16) ^ ((row & 15) << 4), cap);
+    const unsigned short* g =
+        x + (long)min(M - 1, row) * K + ks + byte_in_row / 2;
+    glds16(g, __builtin_amdgcn_readfirstlane(
+                  lds_addr_of(xbuf + row * KSLICE)));
+  }
+}
+
+// Round-trip validator for the asm path: stage via glds_stage_x, read
+// back through the same xswz addresses, store linear. out must equal src.
+__global__ __launch_bounds__(256) void glds_probe_kernel(
+    unsigned short* __restrict__ out, const unsigned short* __restrict__ src) {
+  __shared__ __align__(16) unsigned short xbuf[64 * KSLICE];
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  glds_stage_x(xbuf, src, KSLICE, 0, KSLICE, 64, wid, lane);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  const int xr = threadIdx.x >> 2;
+  const int c0 = (threadIdx.x & 3) * 8;
+#pragma unroll
+  for (int i = 0; i < KSLICE / 32; ++i) {
+    const int c = c0 + i * 32;
+    const uint4 v = *reinterpret_cast<const uint4*>(
+        reinterpret_cast<const char*>(xbuf) + xswz(xr, c * 2));
+    *reinterpret_cast<uint4*>(out + (long)xr * KSLICE + c) = v;
+  }
+}
+
+void glds_probe(torch::Tensor out, torch::Tensor src) {
+  TORCH_CHECK(src.size(0) == 64 && src.size(1) == KSLICE &&
+              src.is_contiguous() && out.is_contiguous());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  glds_probe_kernel<<<dim3(1), 256, 0, stream>>>(
+      reinterpret_cast<unsigned short*>(out.data_ptr()),
+      reinterpret_cast<const unsigned short*>(src.data_ptr()));
+  HIP_CHECK_KERNEL();
+}
+
 // MT = number of 16-row M subtiles (1 => M<=16, 4 => M<=64)
 template <int MT, bool SPLIT>
 __global__ __launch_bounds__(256) void skinny_gemm_kernel(
@@ -61,7 +142,13 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     const unsigned short* __restrict__ x,  // [M, K]
     const unsigned short* __restrict__ w,  // [N, K]
     int M, int N, long K) {
-  __shared__ __align__(16) unsigned short xbuf[64 * KSLICE];
+  // Two x buffers (128 KiB LDS, 1 block/CU): slice s+1's x is staged by
+  // async LDS-DMA (glds_stage_x) issued at the END of slice s, so the
+  // 64 KiB L2 read flies under slice s's tail + slice s+1's W prologue
+  // instead of serializing all four waves behind a load+ds_write pass
+  // between two barriers (that serial stage was ~half the kernel's time
+  // on multi-slice shapes: gate_up measured 73us vs its 37us HBM floor).
+  __shared__ __align__(16) unsigned short xbuf[2][64 * KSLICE];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int n0 = blockIdx.x * 64 + wid * 16;     // this wave's 16 N rows
@@ -73,41 +160,31 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 
   // a block walks slices {blockIdx.y, +gridDim.y, ...} and accumulates
   // locally, so grid.y is an occupancy choice, not forced to K/512
-  for (long ks = (long)blockIdx.y * KSLICE; ks < K;
-       ks += (long)gridDim.y * KSLICE) {
+  const long kstride = (long)gridDim.y * KSLICE;
+  const long ks0 = (long)blockIdx.y * KSLICE;
+  if (ks0 < K) {
+    glds_stage_x(xbuf[0], x, K, ks0, (int)min((long)KSLICE, K - ks0), M,
+                 wid, lane);
+  }
+  int cur = 0;
+  for (long ks = ks0; ks < K; ks += kstride, cur ^= 1) {
   const int klen = (int)min((long)KSLICE, K - ks);
   const unsigned short* wrow = w + (long)(n0 + row16) * K + ks + 8 * kgrp;
   const int nfull = klen / (U * 32);
+  const unsigned short* xb = xbuf[cur];
 
-  // this slice's first W batch goes in flight BEFORE the staging barrier
-  // (no LDS dependence), so its HBM latency hides under the x staging
+  // this slice's first W batch goes in flight BEFORE the drain+barrier
+  // (no LDS dependence), so its HBM latency hides under them
   u32x4_t wa0[U];
   if (nfull >= 1) {
 #pragma unroll
     for (int u = 0; u < U; ++u)
       wa0[u] = nt_load16(wrow + u * 32);
-  }
-
-  // ---- stage x[0..63][ks..ks+klen) into LDS (loads batched, then
-  // writes, so the 8 L2 round trips overlap) ----
-  __syncthreads();  // previous slice's readers are done
-  {
-    const int xr = threadIdx.x >> 2;             // row 0..63
-    const int c0 = (threadIdx.x & 3) * 8;
-    const unsigned short* xs = x + (long)min(M - 1, xr) * K + ks;
-    uint4 v[KSLICE / 32];
-#pragma unroll
-    for (int i = 0; i < KSLICE / 32; ++i) {
-      const int c = c0 + i * 32;
-      v[i] = (c < klen) ? *reinterpret_cast<const uint4*>(xs + c)
-                        : uint4{0, 0, 0, 0};
-    }
-#pragma unroll
-    for (int i = 0; i < KSLICE / 32; ++i) {
-      const int c = c0 + i * 32;
-      *reinterpret_cast<uint4*>(reinterpret_cast<char*>(xbuf) +
-                                xswz(xr, c * 2)) = v[i];
-    }
+    // retire this buffer's 16 glds (older than the 8 wa0 just issued;
+    // hipcc cannot count the asm DMAs, so the wait is explicit)
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   }
   __syncthreads();
 
@@ -127,10 +204,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 #pragma unroll
       for (int m = 0; m < MT; ++m) {
         const int xr = m * 16 + row16;
-        const uint4 xb = *reinterpret_cast<const uint4*>(
-            reinterpret_cast<const char*>(xbuf) + xswz(xr, kc * 2));
+        const uint4 xv = *reinterpret_cast<const uint4*>(
+            reinterpret_cast<const char*>(xb) + xswz(xr, kc * 2));
         acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag, frag_of(xb), acc[m], 0, 0, 0);
+            afrag, frag_of(xv), acc[m], 0, 0, 0);
       }
     }
   }
@@ -146,10 +223,10 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 #pragma unroll
       for (int m = 0; m < MT; ++m) {
         const int xr = m * 16 + row16;
-        const uint4 xb = *reinterpret_cast<const uint4*>(
-            reinterpret_cast<const char*>(xbuf) + xswz(xr, kc * 2));
+        const uint4 xv = *reinterpret_cast<const uint4*>(
+            reinterpret_cast<const char*>(xb) + xswz(xr, kc * 2));
         acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag, frag_of(xb), acc[m], 0, 0, 0);
+            afrag, frag_of(xv), acc[m], 0, 0, 0);
       }
     }
   }
@@ -160,10 +237,23 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 #pragma unroll
     for (int m = 0; m < MT; ++m) {
       const int xr = m * 16 + row16;
-      const uint4 xb = *reinterpret_cast<const uint4*>(
-          reinterpret_cast<const char*>(xbuf) + xswz(xr, kc * 2));
+      const uint4 xv = *reinterpret_cast<const uint4*>(
+          reinterpret_cast<const char*>(xb) + xswz(xr, kc * 2));
       acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          afrag, frag_of(xb), acc[m], 0, 0, 0);
+          afrag, frag_of(xv), acc[m], 0, 0, 0);
+    }
+  }
+  // async-stage the NEXT slice's x into the other buffer: issued after
+  // every W wait of this slice, so nothing forces these DMAs to retire
+  // before the next iteration's explicit vmcnt(8) — they fly under the
+  // loop turnaround and the next slice's wa0 prologue. Safe without a
+  // barrier: every wave finished READING that buffer before it crossed
+  // this slice's top barrier.
+  {
+    const long ksn = ks + kstride;
+    if (ksn < K) {
+      glds_stage_x(xbuf[cur ^ 1], x, K, ksn,
+                   (int)min((long)KSLICE, K - ksn), M, wid, lane);
     }
   }
   }  // slice loop
@@ -225,8 +315,15 @@ void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int ntiles = N / 64;
   const int nslices = (int)((K + KSLICE - 1) / KSLICE);
+  // decomposition targets ~256 blocks (1x the CU count): per-block
+  // latency is ~constant in N in this regime, so more split-K past chip
+  // fill only adds reduce traffic (measured in scripts/sweep_splitk.py)
   int splitk = 1;
-  if (ntiles < 512) splitk = min(nslices, (512 + ntiles - 1) / ntiles);
+  if (ntiles < 512) splitk = min(nslices, (256 + ntiles - 1) / ntiles);
+  if (const char* ov = getenv("KUKEON_SKINNY_SPLITK")) {
+    const int v = atoi(ov);
+    if (v > 0) splitk = min(nslices, v);
+  }
   const int MT = (M + 15) / 16;
   dim3 grid(ntiles, splitk);
   auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());

@@ -342,6 +342,17 @@ def test_mfma_probe16_layout():
     torch.testing.assert_close(out.cpu(), ref.cpu(), rtol=2e-2, atol=2e-2)
 
 
+def test_glds_probe_roundtrip():
+    """asm global_load_lds staging: swizzled-source/lane-linear-dest LDS
+    image must read back exactly through the xswz addresses."""
+    from kukeon_amd import _C
+    torch.manual_seed(7)
+    src = torch.randn(64, 512, dtype=torch.bfloat16, device=DEV)
+    out = torch.zeros_like(src)
+    _C.glds_probe(out, src)
+    assert torch.equal(out, src)
+
+
 @pytest.mark.parametrize("M,N,K", [(1, 4096, 4096), (8, 6144, 4096),
                                    (33, 4096, 14336), (64, 28672, 4096),
                                    (64, 128256, 4096), (64, 4096, 4096)])
