@@ -35,7 +35,7 @@ def main() -> None:
     ap.add_argument("--ctx-cap", type=int, default=3584)
     ap.add_argument("--max-model-len", type=int, default=4096)
     ap.add_argument("--no-graphs", action="store_true")
-    ap.add_argument("--decode-microbatch", type=int, default=32)
+    ap.add_argument("--decode-microbatch", type=int, default=64)
     ap.add_argument("--kv-dtype", choices=["bf16", "fp8"], default="bf16")
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
