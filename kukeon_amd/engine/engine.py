@@ -327,7 +327,11 @@ class LLMEngine:
         return self.Bmax
 
     def _decode_splits(self, bucket: int) -> int:
-        want = (2 * 256 + bucket * self.hk - 1) // (bucket * self.hk)
+        # target ~8x the 256 CUs in workgroups: the micro-bench
+        # (scripts/bench_paged_attn.py) shows 2048-4096 WGs runs 5-9%
+        # faster than the 512-WG minimum (imbalance absorption across
+        # the 8 XCDs), flat beyond
+        want = (8 * 256 + bucket * self.hk - 1) // (bucket * self.hk)
         return max(1, min(self.max_splits, want))
 
     def _tmp_for(self, bucket: int, splits: int):

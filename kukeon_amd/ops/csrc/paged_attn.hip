@@ -141,24 +141,29 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
       const int tok_global = bi * BS + tok;
       const bool valid = tok_global < ctx;
       float s[GPW];
-#pragma unroll
-      for (int g = 0; g < GPW; ++g) s[g] = 0.f;
       {
-        // packed bf16 dot: one v_dot2_f32_bf16 per 2 elems (no unpack VALU)
+        // packed bf16 dot: one v_dot2_f32_bf16 per 2 elems (no unpack
+        // VALU). FOUR independent partial accumulators per head: a single
+        // running accumulator made this a 16-deep RAW chain of dependent
+        // v_dot2 (~5 cyc each), and at the ~2 resident waves/SIMD this
+        // kernel runs at, that latency is exposed as issue stalls
+        // (SQ_WAIT_INST_ANY was 49% of wave cycles).
         const unsigned short* kp = &kbuf[tok * D + ((sub ^ (tok & 3)) * 32)];
 #pragma unroll
-        for (int c = 0; c < 4; ++c) {
-          const uint4 kv = *reinterpret_cast<const uint4*>(kp + c * 8);
-          const unsigned int kw[4] = {kv.x, kv.y, kv.z, kv.w};
+        for (int g = 0; g < GPW; ++g) {
+          float sj[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-          for (int j = 0; j < 4; ++j) {
+          for (int c = 0; c < 4; ++c) {
+            const uint4 kv = *reinterpret_cast<const uint4*>(kp + c * 8);
+            const unsigned int kw[4] = {kv.x, kv.y, kv.z, kv.w};
 #pragma unroll
-            for (int g = 0; g < GPW; ++g)
-              s[g] = __builtin_amdgcn_fdot2_f32_bf16(
+            for (int j = 0; j < 4; ++j)
+              sj[j] = __builtin_amdgcn_fdot2_f32_bf16(
                   __builtin_bit_cast(bf16x2_t, kw[j]),
                   __builtin_bit_cast(bf16x2_t, qpk[g][c * 4 + j]),
-                  s[g], false);
+                  sj[j], false);
           }
+          s[g] = (sj[0] + sj[1]) + (sj[2] + sj[3]);
         }
       }
 #pragma unroll
@@ -186,17 +191,26 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
         // so each token group contributes exactly once
         l[g] = l[g] * alpha + psum;
         m[g] = mn;
-        o0[g] *= alpha;
-        o1[g] *= alpha;
-        // ---- V accumulate: lane owns dims (2*lane, 2*lane+1) ----
+        // ---- V accumulate: lane owns dims (2*lane, 2*lane+1). Block-
+        // local partials in two independent chains per dim (a single
+        // running o0/o1 was a 16-deep dependent FMA chain — same issue-
+        // stall story as the score dot above). ----
+        float va0 = 0.f, va1 = 0.f, vb0 = 0.f, vb1 = 0.f;
 #pragma unroll
-        for (int t2 = 0; t2 < BS; ++t2) {
-          const float pt = __shfl(p, t2 * 4, WAVE);
-          const unsigned int vv =
+        for (int t2 = 0; t2 < BS; t2 += 2) {
+          const float pa = __shfl(p, t2 * 4, WAVE);
+          const float pb = __shfl(p, (t2 + 1) * 4, WAVE);
+          const unsigned int v1 =
               *reinterpret_cast<const unsigned int*>(&vbuf[t2 * D + lane * 2]);
-          o0[g] = fmaf(pt, us2f((unsigned short)(vv & 0xffffu)), o0[g]);
-          o1[g] = fmaf(pt, us2f((unsigned short)(vv >> 16)), o1[g]);
+          const unsigned int v2 = *reinterpret_cast<const unsigned int*>(
+              &vbuf[(t2 + 1) * D + lane * 2]);
+          va0 = fmaf(pa, us2f((unsigned short)(v1 & 0xffffu)), va0);
+          va1 = fmaf(pa, us2f((unsigned short)(v1 >> 16)), va1);
+          vb0 = fmaf(pb, us2f((unsigned short)(v2 & 0xffffu)), vb0);
+          vb1 = fmaf(pb, us2f((unsigned short)(v2 >> 16)), vb1);
         }
+        o0[g] = fmaf(o0[g], alpha, va0 + vb0);
+        o1[g] = fmaf(o1[g], alpha, va1 + vb1);
       }
     }
   }
