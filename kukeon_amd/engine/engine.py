@@ -17,6 +17,7 @@ inference plane at all):
 from __future__ import annotations
 
 import logging
+import weakref
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 
@@ -86,6 +87,10 @@ class LLMEngine:
 
         self.waiting: List[Request] = []
         self._next_id = 0
+        # every SequenceKV ever submitted (weak): idle-session KV is
+        # evictable when admission starves on blocks (history recompute)
+        self._known_kvs: "weakref.WeakValueDictionary[int, SequenceKV]" = \
+            weakref.WeakValueDictionary()
 
         # ---- sticky-row decode state ----
         self.Bmax = ecfg.max_sessions
@@ -144,10 +149,16 @@ class LLMEngine:
                     sampling: SamplingParams) -> int:
         rid = self._next_id
         self._next_id += 1
+        self._known_kvs[id(kv)] = kv
         prompt = list(prompt_tokens)
         if kv.pending_token is not None:
             prompt.insert(0, kv.pending_token)
             kv.pending_token = None
+        if kv.num_tokens == 0 and kv.history:
+            # session was evicted while idle: rebuild the whole context
+            # by prefilling its history ahead of the new tokens
+            prompt = list(kv.history) + prompt
+            kv.history = []
         if kv.num_tokens + len(prompt) + sampling.max_new_tokens > \
                 self.ecfg.max_model_len:
             raise ValueError("request exceeds max_model_len")
@@ -168,19 +179,54 @@ class LLMEngine:
     # ------------------------------------------------------------------
     def step(self) -> List[StepOutput]:
         if self.waiting:
-            batch = self._admit_prefill()
+            batch, blocked = self._admit_prefill()
+            if not batch and blocked and self._evict_idle_kv():
+                batch, blocked = self._admit_prefill()
             if batch:
                 return self._run_prefill(batch)
+            if blocked and self.num_running == 0:
+                # nothing running to make progress, nothing left to evict:
+                # the head request alone cannot ever fit
+                req = self.waiting[0]
+                raise MemoryError(
+                    f"request {req.req_id} cannot be admitted: needs more "
+                    "KV blocks than the pool holds even with every idle "
+                    "session evicted")
         if self.num_running > 0:
             return self._run_decode()
         return []
 
+    def _evict_idle_kv(self) -> bool:
+        """Evict the KV of an idle session (resident context, no active
+        request) to unblock admission; its history stays so the next
+        request on that session transparently re-prefills the context.
+        Sessions idle in HBM are the norm for agent serving — without
+        this, resident-but-idle context starves new admissions forever."""
+        active = {id(r.kv) for r in self.waiting}
+        active.update(id(r.kv) for r in self._rows if r is not None)
+        victim = None
+        for kv in self._known_kvs.values():
+            if id(kv) in active or not kv.blocks:
+                continue
+            if victim is None or len(kv.blocks) > len(victim.blocks):
+                victim = kv
+        if victim is None:
+            return False
+        log.info("evicting idle session KV (%d blocks) to unblock admission",
+                 len(victim.blocks))
+        self.kv.allocator.free(victim.blocks)
+        victim.blocks = []
+        victim.num_tokens = 0
+        return True
+
     # ------------------------------------------------------------------
-    def _admit_prefill(self) -> List[Tuple[Request, int]]:
-        """-> [(request, chunk_len)]; a chunk smaller than the remaining
-        prompt keeps the request at the front of the queue (chunked
-        prefill for prompts beyond the per-step token budget)."""
+    def _admit_prefill(self):
+        """-> ([(request, chunk_len)], blocked_on_blocks); a chunk smaller
+        than the remaining prompt keeps the request at the front of the
+        queue (chunked prefill for prompts beyond the per-step token
+        budget)."""
         batch: List[Tuple[Request, int]] = []
+        blocked = False
         tokens = 0
         free = self.kv.allocator.num_free
         avail_rows = len(self._free_rows)
@@ -196,9 +242,7 @@ class LLMEngine:
                                else 0)
             need = req.kv.blocks_needed(reserve)
             if need > free:
-                if not batch:
-                    # can't make progress on the head request right now
-                    break
+                blocked = True
                 break
             free -= need
             tokens += chunk
@@ -208,7 +252,7 @@ class LLMEngine:
             else:
                 batch.append((req, chunk))
                 break  # a partial chunk consumes the whole budget
-        return batch
+        return batch, blocked
 
     def _alloc_for(self, req: Request, new_tokens: int) -> List[int]:
         need = req.kv.blocks_needed(new_tokens)

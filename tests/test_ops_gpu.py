@@ -435,3 +435,54 @@ def test_mixtral_gpu_engine_and_moe_paths():
             toks.extend(o.new_tokens)
     assert len(toks) == 6
     assert all(0 <= t < cfg.vocab_size for t in toks)
+
+
+def test_engine_gpu_stress_churn():
+    """Session churn under KV pressure on the HIP path: random-length
+    requests with mixed sampling params, a KV pool small enough to force
+    preemption-by-recompute, and graphs on. Every request must finish
+    with exactly its requested token count."""
+    import random
+
+    from kukeon_amd.engine.config import (EngineConfig, SamplingParams,
+                                          tiny_llama)
+    from kukeon_amd.engine.engine import LLMEngine
+    from kukeon_amd.engine.kv_cache import SequenceKV
+    from kukeon_amd.models.llama import LlamaModel
+
+    rng = random.Random(0)
+    torch.manual_seed(0)
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=192, max_sessions=8, num_kv_blocks=48,
+                        use_graphs=True, decode_microbatch=8,
+                        graph_buckets=(2, 4, 8))
+    model = LlamaModel(cfg, device=DEV)
+    engine = LLMEngine(model, cfg, ecfg, device=DEV)
+
+    want = {}
+    got = {}
+    sessions = {}
+    for i in range(24):
+        kv = SequenceKV(ecfg.block_size)
+        sessions[i] = kv
+        n = rng.randint(2, 12)
+        rid = engine.add_request(
+            kv, [rng.randrange(cfg.vocab_size) for _ in range(
+                rng.randint(3, 40))],
+            SamplingParams(temperature=rng.choice([0.0, 0.7, 1.0]),
+                           top_k=rng.choice([0, 5, 50]),
+                           top_p=rng.choice([1.0, 0.9]),
+                           max_new_tokens=n))
+        want[rid] = n
+        got[rid] = 0
+    steps = 0
+    while engine.has_work():
+        for o in engine.step():
+            got[o.req_id] += len(o.new_tokens)
+        steps += 1
+        assert steps < 2000
+    assert got == want
+    for kv in sessions.values():
+        assert all(0 <= t < cfg.vocab_size for t in kv.history)
+        engine.free_sequence(kv)
+    assert engine.kv.allocator.num_free == ecfg.num_kv_blocks
