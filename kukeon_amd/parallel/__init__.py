@@ -23,14 +23,20 @@ def init_distributed(backend: str | None = None) -> tuple[int, int]:
         return dist.get_rank(), dist.get_world_size()
     if "RANK" not in os.environ:
         return 0, 1
+    ndev = torch.cuda.device_count() if torch.cuda.is_available() else 0
+    local_world = int(os.environ.get(
+        "LOCAL_WORLD_SIZE", os.environ.get("WORLD_SIZE", "1")))
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # RCCL needs one DISTINCT GPU per rank; an oversubscribed smoke
+        # run (more local ranks than GPUs) falls back to gloo so the
+        # multi-rank harness still executes on a small box
+        backend = "nccl" if 0 < local_world <= ndev else "gloo"
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
     dist.init_process_group(backend=backend)
     rank = dist.get_rank()
-    if torch.cuda.is_available():
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+    if ndev > 0:
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)) % ndev)
     return rank, dist.get_world_size()
 
 
