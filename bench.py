@@ -52,7 +52,8 @@ def main() -> None:
     if args.device:
         device = args.device
     elif torch.cuda.is_available():
-        device = f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}"
+        dev_i = int(os.environ.get("LOCAL_RANK", 0)) % torch.cuda.device_count()
+        device = f"cuda:{dev_i}"
         torch.cuda.set_device(device)
     else:
         device = "cpu"
