@@ -381,6 +381,29 @@ def restart(ctx, name):
 
 @cli.command()
 @click.argument("name")
+@pass_ctx
+def top(ctx, name):
+    """Live per-container CPU/RSS for a cell (TaskMetrics parity)."""
+    try:
+        m = ctx.client.CellMetrics(realm=ctx.realm, space=ctx.space,
+                                   stack=ctx.stack, name=name)
+    except errors.KukeonError as e:
+        _die(e)
+    click.echo(f"{'CONTAINER':<20} {'PID':>7} {'CPU(s)':>9} "
+               f"{'RSS(MB)':>9} {'THREADS':>8}")
+    for cname, c in m["containers"].items():
+        if not c.get("running"):
+            click.echo(f"{cname:<20} {'-':>7} {'-':>9} {'-':>9} {'-':>8}")
+            continue
+        click.echo(f"{cname:<20} {c['pid']:>7} {c['cpuSeconds']:>9.1f} "
+                   f"{c['rssBytes'] / 1e6:>9.1f} {c['threads']:>8}")
+    t = m["total"]
+    click.echo(f"{'TOTAL':<20} {'':>7} {t['cpuSeconds']:>9.1f} "
+               f"{t['rssBytes'] / 1e6:>9.1f} {t['threads']:>8}")
+
+
+@cli.command()
+@click.argument("name")
 @click.option("--container", default="")
 @click.option("-f", "--follow", is_flag=True)
 @click.option("-n", "--lines", default=100)
@@ -606,9 +629,9 @@ def daemon():
 @pass_ctx
 def daemon_serve(ctx, reconcile_interval, configuration, foreground):
     import logging
-    logging.basicConfig(
-        level=logging.INFO,
-        format="%(asctime)s %(levelname)s %(name)s %(message)s")
+
+    from kukeon_amd.utils import logging as klog
+    klog.setup(logging.INFO)
     server_cfg = None
     if configuration:
         raw = yaml.safe_load(Path(configuration).read_text())

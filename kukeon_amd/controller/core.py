@@ -265,6 +265,28 @@ class Controller:
             raise errors.CellNotFound(f"{realm}/{space}/{stack}/{name}")
         return api.CellDoc.from_dict(data)
 
+    def cell_metrics(self, realm, space, stack, name) -> dict:
+        """Live per-container resource metrics for a cell (reference
+        parity: ctr TaskMetrics surfaced per task; here sampled from /proc
+        for the cell's host processes)."""
+        from kukeon_amd.runtime import proc as procutil
+        doc = self.get_cell(realm, space, stack, name)
+        containers = {}
+        total = {"cpuSeconds": 0.0, "rssBytes": 0, "threads": 0}
+        for i, cs in enumerate(doc.status.containers):
+            cname = (doc.spec.containers[i].id
+                     if i < len(doc.spec.containers) else f"c{i}")
+            m = procutil.metrics(cs.pid) if cs.pid > 0 else None
+            if m is None:
+                containers[cname] = {"running": False}
+                continue
+            containers[cname] = {"running": True, **m}
+            total["cpuSeconds"] = round(total["cpuSeconds"] +
+                                        m["cpuSeconds"], 3)
+            total["rssBytes"] += m["rssBytes"]
+            total["threads"] += m["threads"]
+        return {"cell": name, "containers": containers, "total": total}
+
     def list_cells(self, realm, space, stack) -> List[api.CellDoc]:
         return [self.get_cell(realm, space, stack, n) for n in
                 self.store.list_children(

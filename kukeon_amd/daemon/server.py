@@ -144,6 +144,10 @@ class Service:
         return [d.to_dict() for d in
                 self.ctl.list_cells(p["realm"], p["space"], p["stack"])]
 
+    def CellMetrics(self, p):
+        return self.ctl.cell_metrics(p["realm"], p["space"], p["stack"],
+                                     p["name"])
+
     def StartCell(self, p):
         return self.ctl.start_cell(p["realm"], p["space"], p["stack"],
                                    p["name"]).to_dict()

@@ -29,6 +29,28 @@ def proc_starttime(pid: int) -> Optional[int]:
         return None
 
 
+def metrics(pid: int) -> Optional[dict]:
+    """Per-process resource metrics (reference parity: ctr TaskMetrics,
+    ctr/client.go:123 — there from containerd's cgroup sampling; here from
+    /proc since cells are host processes). None if the pid is gone."""
+    rest = _stat_fields(pid)
+    if rest is None or rest[0] == "Z":
+        return None
+    try:
+        tck = os.sysconf("SC_CLK_TCK")
+        page = os.sysconf("SC_PAGE_SIZE")
+        # stat fields after comm/state: index 11/12 are utime/stime
+        # (fields 14/15 1-based), 17 is num_threads (field 20)
+        cpu_s = (int(rest[11]) + int(rest[12])) / tck
+        threads = int(rest[17])
+        with open(f"/proc/{pid}/statm") as f:
+            rss_pages = int(f.read().split()[1])
+        return {"pid": pid, "cpuSeconds": round(cpu_s, 3),
+                "rssBytes": rss_pages * page, "threads": threads}
+    except (OSError, ValueError, IndexError):
+        return None
+
+
 def alive(pid: int, starttime: Optional[int] = None) -> bool:
     if pid <= 0:
         return False

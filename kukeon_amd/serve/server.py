@@ -245,7 +245,8 @@ def main(argv: Optional[List[str]] = None) -> None:
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--kv-blocks", type=int, default=0)
     args = ap.parse_args(argv)
-    logging.basicConfig(level=logging.INFO)
+    from kukeon_amd.utils import logging as klog
+    klog.setup(logging.INFO)
     import torch
     device = args.device or ("cuda:0" if torch.cuda.is_available() else "cpu")
     cfg = MODEL_PRESETS[args.model]()

@@ -223,3 +223,20 @@ def test_error_mapping_over_rpc(harness):
                        name="nope")
     with pytest.raises(errors.RealmNotFound):
         client.GetRealm(name="ghost")
+
+
+def test_cell_metrics_live(harness):
+    """`kuke top` backing verb: live /proc metrics for a running cell."""
+    ctl, srv, client = harness
+    client.ApplyDocuments(yaml=CELL_YAML)
+    client.StartCell(realm="default", space="default", stack="default",
+                     name="busy")
+    m = client.CellMetrics(realm="default", space="default",
+                           stack="default", name="busy")
+    assert m["cell"] == "busy"
+    running = [c for c in m["containers"].values() if c.get("running")]
+    assert running, m
+    assert all(c["rssBytes"] > 0 and c["threads"] >= 1 for c in running)
+    assert m["total"]["rssBytes"] > 0
+    client.KillCell(realm="default", space="default", stack="default",
+                    name="busy")
