@@ -19,7 +19,7 @@ import yaml
 
 from kukeon_amd.api import errors
 from kukeon_amd.api import v1beta1 as api
-from kukeon_amd.api.client import DEFAULT_SOCKET, LocalClient, UnixClient
+from kukeon_amd.api.client import LocalClient, UnixClient
 
 DEFAULT_RUN_PATH = os.environ.get("KUKEON_RUN_PATH", "/run/kukeon")
 

@@ -8,7 +8,7 @@ from __future__ import annotations
 
 import json
 import threading
-from typing import Dict, Optional, Tuple
+from typing import Dict, Optional
 
 from kukeon_amd.api import errors
 from kukeon_amd.state.store import Store

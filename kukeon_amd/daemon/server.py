@@ -25,7 +25,7 @@ from typing import Any, Dict, Optional
 
 from kukeon_amd.api import errors
 from kukeon_amd.api import v1beta1 as api
-from kukeon_amd.controller.core import Controller, ResourceResult
+from kukeon_amd.controller.core import Controller
 
 log = logging.getLogger("kukeon.daemon")
 

@@ -6,7 +6,7 @@ bf16 (no network; the benchmark contract is synthetic data / random weights).
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 
 @dataclass

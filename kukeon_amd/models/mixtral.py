@@ -18,8 +18,6 @@ all-reduce is per-link bound.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

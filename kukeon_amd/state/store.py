@@ -24,7 +24,6 @@ reference's correctness envelope (metadata/lock.go WriteMetadataCAS).
 from __future__ import annotations
 
 import contextlib
-import errno
 import fcntl
 import hashlib
 import json

@@ -7,7 +7,6 @@ from __future__ import annotations
 
 import contextlib
 import os
-import shutil
 import subprocess
 from pathlib import Path
 from string import Template
