@@ -369,9 +369,10 @@ def test_skinny_gemm(M, N, K):
                                atol=3e-2)
 
 
-def test_decode_graphs_match_eager():
+@pytest.mark.parametrize("kv_dtype", ["bf16", "fp8"])
+def test_decode_graphs_match_eager(kv_dtype):
     """Self-advancing graph decode must produce the same tokens as the
-    eager path (same seeds, greedy)."""
+    eager path (same seeds, greedy), for both KV cache dtypes."""
     from kukeon_amd.engine.config import EngineConfig, SamplingParams, tiny_llama
     from kukeon_amd.engine.engine import LLMEngine
     from kukeon_amd.engine.kv_cache import SequenceKV
@@ -388,7 +389,8 @@ def test_decode_graphs_match_eager():
         # argmax ties on a random-init tiny model).
         ecfg = EngineConfig(max_model_len=256, max_sessions=4,
                             num_kv_blocks=128, use_graphs=use_graphs,
-                            decode_microbatch=4, graph_buckets=(1, 2, 4))
+                            decode_microbatch=4, graph_buckets=(1, 2, 4),
+                            kv_dtype=kv_dtype)
         model = LlamaModel(cfg, device=DEV)
         engine = LLMEngine(model, cfg, ecfg, device=DEV)
         kv = SequenceKV(ecfg.block_size)
@@ -486,3 +488,44 @@ def test_engine_gpu_stress_churn():
         assert all(0 <= t < cfg.vocab_size for t in kv.history)
         engine.free_sequence(kv)
     assert engine.kv.allocator.num_free == ecfg.num_kv_blocks
+
+
+def test_engine_gpu_preemption_recompute():
+    """Mid-decode KV exhaustion on the HIP path: a growing multi-turn
+    session forces preempt-by-recompute; every turn must still deliver
+    its full token count and end with a consistent context length."""
+    from kukeon_amd.engine.config import (EngineConfig, SamplingParams,
+                                          tiny_llama)
+    from kukeon_amd.engine.engine import LLMEngine
+    from kukeon_amd.engine.kv_cache import SequenceKV
+    from kukeon_amd.models.llama import LlamaModel
+
+    torch.manual_seed(3)
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=512, max_sessions=2, num_kv_blocks=14,
+                        use_graphs=True, decode_microbatch=8,
+                        graph_buckets=(2, 4))
+    model = LlamaModel(cfg, device=DEV)
+    engine = LLMEngine(model, cfg, ecfg, device=DEV)
+    kvs = [SequenceKV(ecfg.block_size) for _ in range(2)]
+    expect = [0, 0]
+    for turn in range(4):
+        rids = {}
+        for i, kv in enumerate(kvs):
+            rid = engine.add_request(
+                kv, [11 + turn, 7, 5], SamplingParams(temperature=0.0,
+                                                      max_new_tokens=24))
+            rids[rid] = i
+        got = {rid: 0 for rid in rids}
+        steps = 0
+        while engine.has_work():
+            for o in engine.step():
+                got[o.req_id] += len(o.new_tokens)
+            steps += 1
+            assert steps < 3000
+        assert all(v == 24 for v in got.values()), got
+        for i in range(2):
+            expect[i] += 3 + 24
+    for i, kv in enumerate(kvs):
+        # context length: all tokens of all turns, last one pending
+        assert len(kv.history) + kv.num_tokens >= expect[i] - 1
