@@ -376,7 +376,9 @@ class LLMEngine:
         # faster than the 512-WG minimum (imbalance absorption across
         # the 8 XCDs), flat beyond
         want = (8 * 256 + bucket * self.hk - 1) // (bucket * self.hk)
-        return max(1, min(self.max_splits, want))
+        # multiple of 4: the MFMA decode-attention kernel gives each of a
+        # workgroup's 4 autonomous waves its own split slot
+        return max(4, min(self.max_splits, ((want + 3) // 4) * 4))
 
     def _tmp_for(self, bucket: int, splits: int):
         key = (bucket, splits)

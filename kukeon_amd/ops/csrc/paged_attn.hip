@@ -25,6 +25,21 @@
 namespace kukeon {
 
 typedef __attribute__((__vector_size__(2 * sizeof(short)))) short bf16x2_t;
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8_v;
+typedef __attribute__((__vector_size__(16 * sizeof(float)))) float f32x16_t;
+
+struct u4s_ { unsigned int x[4]; };
+DEV_INLINE bf16x8_v as_frag32(unsigned int w0, unsigned int w1,
+                              unsigned int w2, unsigned int w3) {
+  u4s_ v{{w0, w1, w2, w3}};
+  return __builtin_bit_cast(bf16x8_v, v);
+}
+DEV_INLINE unsigned int cvt_pk_bf16d(float lo, float hi) {
+  __hip_bfloat162 h = __float22bfloat162_rn(float2{lo, hi});
+  unsigned int r;
+  __builtin_memcpy(&r, &h, 4);
+  return r;
+}
 
 template <int D, int BS, int GPW, bool FP8>
 __global__ __launch_bounds__(256) void paged_attn_kernel(
@@ -269,6 +284,294 @@ __global__ void paged_attn_reduce_kernel(
   out[(long)b * Hq * D + (long)h * D + d] = f2us(L > 0.f ? acc / L : 0.f);
 }
 
+// ===================================================================
+// MFMA decode attention (wave-autonomous split-KV).
+//
+// The v_dot2 kernel above is issue-stall bound (PMC: SQ_WAIT_INST_ANY 49%
+// of wave cycles — ~28 VALU instructions per KV token in dependent
+// chains). This variant moves QK^T and PV onto the matrix cores using the
+// prefill kernel's fragment machinery (swapped S^T = K.Q^T so softmax is
+// lane-local; P via cvt_pk + permlane32_swap; PV as O^T = V^T.P^T with a
+// transposed V image), at ~5 wave-instructions per token.
+//
+// Decode-specific structure: the 4 waves of a workgroup are fully
+// AUTONOMOUS — each owns one of 4 consecutive tmp slots (the engine's
+// split count is a multiple of 4) and walks its own KV-block range,
+// staging 32-token tiles into its private LDS quarter. No barriers at
+// all: while one wave waits on its stage, the other three compute, so
+// stage latency hides at the CU level. All S slots merge in one reduce
+// pass (every slot is written each call — empty ranges write m=-inf).
+//
+// The q-head dimension (G <= 32) rides the MFMA N axis; padding columns
+// are garbage but PER-LANE, so nothing crosses into real heads and only
+// cols < G are written back.
+template <bool FP8>
+__global__ __launch_bounds__(256, 2) void paged_attn_mfma_kernel(
+    float* __restrict__ tmp_out,          // [B, Hq, S, D] f32
+    float* __restrict__ tmp_ml,           // [B, Hq, S, 2]
+    const unsigned short* __restrict__ q, // [B, q_stride]
+    const void* __restrict__ k_cache,     // [NB, Hk, 16, D]
+    const void* __restrict__ v_cache,
+    const int* __restrict__ block_table,  // [B, max_blocks]
+    const int* __restrict__ seq_lens,     // [B]
+    long q_stride, int Hq, int Hk, int max_blocks, int S, float scale) {
+  constexpr int D = 128;
+  constexpr int BS = 16;
+  constexpr int KVT = 32;
+  const int b = blockIdx.x;
+  const int hk = blockIdx.y;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int hi = lane >> 5;
+  const int qcol = lane & 31;
+  const int G = Hq / Hk;
+  const int slot = blockIdx.z * 4 + wid;
+  const int ctx = seq_lens[b];
+  if (ctx <= 0 || slot >= S) return;
+
+  // per-wave private LDS quarters — no cross-wave sharing, no barriers
+  __shared__ __align__(16) unsigned short kbuf[4][KVT * D];
+  __shared__ __align__(16) unsigned short vtbuf[4][D * KVT];
+  unsigned short* kb = kbuf[wid];
+  unsigned short* vt = vtbuf[wid];
+
+  const int nblocks = (ctx + BS - 1) / BS;
+  const int per_slot = (nblocks + S - 1) / S;
+  const int blk_begin = slot * per_slot;
+  const int blk_end = min(nblocks, blk_begin + per_slot);
+
+  // Q row pointer: fragments are re-read per tile from L2 (the row is
+  // hot — every workgroup of this sequence reads it) instead of pinned
+  // in 32 registers for the kernel's lifetime; this keeps the wave under
+  // the 256-VGPR boundary for 2 waves/SIMD.
+  const int qh = min(qcol, G - 1);
+  const unsigned short* qp =
+      q + (long)b * q_stride + (long)(hk * G + qh) * D;
+
+  float m = -INFINITY, l_acc = 0.f;
+  f32x16_t oacc[4] = {{}, {}, {}, {}};
+
+  const unsigned short* kc16 =
+      reinterpret_cast<const unsigned short*>(k_cache);
+  const unsigned short* vc16 =
+      reinterpret_cast<const unsigned short*>(v_cache);
+  const unsigned char* kc8 = reinterpret_cast<const unsigned char*>(k_cache);
+  const unsigned char* vc8 = reinterpret_cast<const unsigned char*>(v_cache);
+  const int kv_hi = min(ctx, blk_end * BS);
+
+  // per-lane staging geometry (fixed across tiles)
+  const int stg_tok = lane >> 4;            // +4 per pass
+  const int stg_d0 = (lane & 15) * 8;
+
+  // T14 within the wave: tile bi's K is PREFETCHED into registers during
+  // tile bi-1's compute; V loads issue right after the K LDS write and
+  // fly under QK^T + softmax. Without this the wave serializes a full
+  // HBM fetch per tile (measured 2.8x slower than the v_dot2 kernel).
+  uint4 kpre[8];
+  auto kfetch = [&](int bi_f, uint4* dst) {
+#pragma unroll
+    for (int pass = 0; pass < 8; ++pass) {
+      const int tok = pass * 4 + stg_tok;
+      const int kvpos = bi_f * BS + tok;
+      const int bidx = min(kvpos / BS, nblocks - 1);
+      const int pblk = block_table[(long)b * max_blocks + bidx];
+      const long koff =
+          (((long)pblk * Hk + hk) * BS + (kvpos % BS)) * D + stg_d0;
+      if (FP8) {
+        dst[pass] =
+            fp8x8_to_bf16x8(*reinterpret_cast<const uint2*>(kc8 + koff));
+      } else {
+        dst[pass] = *reinterpret_cast<const uint4*>(kc16 + koff);
+      }
+    }
+  };
+  if (blk_begin < blk_end) kfetch(blk_begin, kpre);
+
+  for (int bi = blk_begin; bi < blk_end; bi += 2) {
+    const int kvbase = bi * BS;
+    // ---- write the prefetched K tile (swizzled image) ----
+#pragma unroll
+    for (int pass = 0; pass < 8; ++pass) {
+      const int tok = pass * 4 + stg_tok;
+      const int byte = tok * 256 + ((stg_d0 * 2) ^ ((tok & 15) << 4));
+      *reinterpret_cast<uint4*>(reinterpret_cast<char*>(kb) + byte) =
+          kpre[pass];
+    }
+    // ---- issue next tile's K prefetch + this tile's V loads ----
+    if (bi + 2 < blk_end) kfetch(bi + 2, kpre);
+    uint4 vr[4][2];
+#pragma unroll
+    for (int pass = 0; pass < 4; ++pass) {
+      const int kp = pass * 4 + stg_tok;              // kv pair 0..15
+      const int vpos0 = kvbase + 2 * kp;
+      const int vpos1 = vpos0 + 1;
+      const int vb0 = min(vpos0 / BS, nblocks - 1);
+      const int vb1 = min(vpos1 / BS, nblocks - 1);
+      const long voff0 =
+          (((long)block_table[(long)b * max_blocks + vb0] * Hk + hk) * BS +
+           (vpos0 % BS)) * D + stg_d0;
+      const long voff1 =
+          (((long)block_table[(long)b * max_blocks + vb1] * Hk + hk) * BS +
+           (vpos1 % BS)) * D + stg_d0;
+      if (FP8) {
+        vr[pass][0] =
+            fp8x8_to_bf16x8(*reinterpret_cast<const uint2*>(vc8 + voff0));
+        vr[pass][1] =
+            fp8x8_to_bf16x8(*reinterpret_cast<const uint2*>(vc8 + voff1));
+      } else {
+        vr[pass][0] = *reinterpret_cast<const uint4*>(vc16 + voff0);
+        vr[pass][1] = *reinterpret_cast<const uint4*>(vc16 + voff1);
+      }
+    }
+
+    // ---- S^T = K . Q^T ----
+    f32x16_t sacc = {};
+#pragma unroll
+    for (int st = 0; st < 8; ++st) {
+      const int row = qcol;  // A row = kv token in tile
+      const int byte =
+          row * 256 + (((st * 16 + hi * 8) * 2) ^ ((row & 15) << 4));
+      const uint4 kf = *reinterpret_cast<const uint4*>(
+          reinterpret_cast<const char*>(kb) + byte);
+      const uint4 qv = *reinterpret_cast<const uint4*>(qp + st * 16 + hi * 8);
+      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          as_frag32(kf.x, kf.y, kf.z, kf.w),
+          as_frag32(qv.x, qv.y, qv.z, qv.w),
+          sacc, 0, 0, 0);
+    }
+
+    // ---- online softmax over this lane's 16 kv rows (lane-local) ----
+    float sv[16];
+    float tmax = -INFINITY;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kvrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const bool valid = (kvbase + kvrow) < kv_hi;
+      sv[r] = valid ? sacc[r] * scale : -INFINITY;
+      tmax = fmaxf(tmax, sv[r]);
+    }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
+    const float mn = fmaxf(m, tmax);
+    const float alpha = (m == -INFINITY) ? 0.f : __expf(m - mn);
+    m = mn;
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      sv[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - mn);
+      psum += sv[r];
+    }
+    psum += __shfl_xor(psum, 32, WAVE);
+    l_acc = l_acc * alpha + psum;
+#pragma unroll
+    for (int db = 0; db < 4; ++db)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) oacc[db][r] *= alpha;
+
+    // ---- write the V^T image (loads issued before QK have landed) ----
+#pragma unroll
+    for (int pass = 0; pass < 4; ++pass) {
+      const int kp = pass * 4 + stg_tok;
+      const int vpos0 = kvbase + 2 * kp;
+      const int vpos1 = vpos0 + 1;
+      uint4 r0 = vr[pass][0];
+      uint4 r1 = vr[pass][1];
+      if (vpos0 >= kv_hi) r0 = uint4{0, 0, 0, 0};
+      if (vpos1 >= kv_hi) r1 = uint4{0, 0, 0, 0};
+      const unsigned int* a0 = reinterpret_cast<const unsigned int*>(&r0);
+      const unsigned int* a1 = reinterpret_cast<const unsigned int*>(&r1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const unsigned short e0 = (a0[j >> 1] >> ((j & 1) * 16)) & 0xffffu;
+        const unsigned short e1 = (a1[j >> 1] >> ((j & 1) * 16)) & 0xffffu;
+        const unsigned int packed = (unsigned)e0 | ((unsigned)e1 << 16);
+        const int d = stg_d0 + j;
+        const int byte = d * 64 + ((4 * kp) ^ ((d & 3) << 4));
+        *reinterpret_cast<unsigned int*>(
+            reinterpret_cast<char*>(vt) + byte) = packed;
+      }
+    }
+
+    // ---- P -> bf16 B-fragments via cvt_pk + permlane32_swap ----
+    unsigned int cp[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      cp[j] = cvt_pk_bf16d(sv[2 * j], sv[2 * j + 1]);
+    unsigned int pb[2][4];
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      const int b0 = 4 * c;
+      auto s1 = __builtin_amdgcn_permlane32_swap(cp[b0 + 0], cp[b0 + 2],
+                                                 false, false);
+      auto s2 = __builtin_amdgcn_permlane32_swap(cp[b0 + 1], cp[b0 + 3],
+                                                 false, false);
+      pb[c][0] = s1[0]; pb[c][1] = s2[0];
+      pb[c][2] = s1[1]; pb[c][3] = s2[1];
+    }
+
+    // ---- O^T += V^T . P^T ----
+#pragma unroll
+    for (int db = 0; db < 4; ++db) {
+      const int d = db * 32 + qcol;
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        const int byte = d * 64 + (((c * 32) + hi * 16) ^ ((d & 3) << 4));
+        const uint4 vf = *reinterpret_cast<const uint4*>(
+            reinterpret_cast<const char*>(vt) + byte);
+        oacc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            as_frag32(vf.x, vf.y, vf.z, vf.w),
+            as_frag32(pb[c][0], pb[c][1], pb[c][2], pb[c][3]),
+            oacc[db], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: unnormalized O + (m, l) into this wave's slot ----
+  if (qcol >= G) return;
+  const int head = hk * G + qcol;
+  float* top = tmp_out + (((long)b * Hq + head) * S + slot) * D;
+#pragma unroll
+  for (int db = 0; db < 4; ++db) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int d = db * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      top[d] = oacc[db][r];
+    }
+  }
+  if (hi == 0) {
+    float* ml = tmp_ml + (((long)b * Hq + head) * S + slot) * 2;
+    ml[0] = m;
+    ml[1] = l_acc;
+  }
+}
+
+// Reduce for the MFMA path: every one of the S slots is written on every
+// call (empty ranges write m = -inf), so all are read unconditionally —
+// no per-split block math, no zero-ctx division hazard.
+template <int D>
+__global__ void paged_attn_reduce_all_kernel(
+    unsigned short* __restrict__ out, const float* __restrict__ tmp_out,
+    const float* __restrict__ tmp_ml, const int* __restrict__ seq_lens,
+    int Hq, int S) {
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;
+  const int d = threadIdx.x;
+  if (seq_lens[b] <= 0) return;
+  const float* ml = tmp_ml + (((long)b * Hq + h) * S) * 2;
+  float M = -INFINITY;
+  for (int s = 0; s < S; ++s) M = fmaxf(M, ml[s * 2]);
+  float L = 0.f;
+  for (int s = 0; s < S; ++s)
+    L += (ml[s * 2] == -INFINITY ? 0.f : __expf(ml[s * 2] - M)) *
+         ml[s * 2 + 1];
+  const float* top = tmp_out + (((long)b * Hq + h) * S) * D;
+  float acc = 0.f;
+  for (int s = 0; s < S; ++s) {
+    const float w = ml[s * 2] == -INFINITY ? 0.f : __expf(ml[s * 2] - M);
+    acc += w * top[(long)s * D + d];
+  }
+  out[(long)b * Hq * D + (long)h * D + d] = f2us(L > 0.f ? acc / L : 0.f);
+}
+
 void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                      torch::Tensor v_cache, torch::Tensor block_table,
                      torch::Tensor seq_lens, int64_t q_offset,
@@ -296,6 +599,34 @@ void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
   auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
 
   const bool fp8 = k_cache.scalar_type() == torch::kUInt8;
+  // Opt-in (KUKEON_ATTN_MFMA=1): correct and ~5x fewer wave-instructions
+  // per token, but measured 25-30% behind the v_dot2 kernel end to end —
+  // its 64 KiB LDS caps residency at 2 workgroups/CU while the v_dot2
+  // kernel's 8 KiB image keeps ~8x more waves in flight to hide its
+  // issue stalls. Round-2 lever: 16-token tiles on 16x16x32 MFMAs
+  // (quarter the LDS) or a hand-scheduled pipeline.
+  const char* mfma_env = getenv("KUKEON_ATTN_MFMA");
+  const bool use_mfma = (mfma_env && mfma_env[0] == '1') &&
+                        num_splits % 4 == 0 && G <= 32 &&
+                        tmp_out.numel() >= (long)B * Hq * num_splits * 128;
+  if (use_mfma) {
+    dim3 mgrid(B, Hk, (unsigned)(num_splits / 4));
+    float* tovm = tmp_out.data_ptr<float>();
+    float* tmlm = tmp_ml.data_ptr<float>();
+#define PAM_LAUNCH(FP8_)                                                     \
+    paged_attn_mfma_kernel<FP8_><<<mgrid, 256, 0, stream>>>(                 \
+        tovm, tmlm, qp, k_cache.data_ptr(), v_cache.data_ptr(),              \
+        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(), q_stride,     \
+        Hq, Hk, max_blocks, (int)num_splits, (float)scale)
+    if (fp8) { PAM_LAUNCH(true); } else { PAM_LAUNCH(false); }
+#undef PAM_LAUNCH
+    HIP_CHECK_KERNEL();
+    dim3 rgrid(B, Hq);
+    paged_attn_reduce_all_kernel<128><<<rgrid, 128, 0, stream>>>(
+        op, tovm, tmlm, seq_lens.data_ptr<int>(), Hq, (int)num_splits);
+    HIP_CHECK_KERNEL();
+    return;
+  }
 #define PA_LAUNCH(GPW, FP8_)                                                  \
   paged_attn_kernel<128, 16, GPW, FP8_><<<grid, 256, 0, stream>>>(            \
       op, tov, tml, qp, k_cache.data_ptr(), v_cache.data_ptr(),               \

@@ -42,13 +42,21 @@ for ctx_lo, ctx_hi in [(512, 3584), (1500, 1500), (3584, 3584)]:
     scale = D ** -0.5
     kv_bytes = sum(ctxs) * Hk * D * 2 * 2
     print(f"ctx {ctx_lo}-{ctx_hi} (KV {kv_bytes/1e6:.0f} MB/call):")
+    import os as _os
     for splits in (1, 2, 4, 8, 16):
         tmp_out = torch.zeros(B, Hq, splits, D, dtype=torch.float32,
                               device=DEV)
         tmp_ml = torch.zeros(B, Hq, splits, 2, dtype=torch.float32,
                              device=DEV)
-        us = t(lambda: ops.paged_attention(out, q, kc, vc, bt, seq_lens, 0,
-                                           splits, scale, tmp_out, tmp_ml))
-        bw = kv_bytes / (us * 1e-6) / 1e12
-        print(f"  splits={splits:2d} ({B*Hk*splits:5d} WGs): {us:7.1f}us  "
-              f"{bw:5.2f} TB/s ({bw/6.3*100:4.1f}% roofline)", flush=True)
+        line = f"  splits={splits:2d}:"
+        for mode in ("0", "1"):
+            _os.environ["KUKEON_ATTN_MFMA"] = mode
+            us = t(lambda: ops.paged_attention(out, q, kc, vc, bt, seq_lens,
+                                               0, splits, scale, tmp_out,
+                                               tmp_ml))
+            bw = kv_bytes / (us * 1e-6) / 1e12
+            tag = "mfma" if mode == "1" else "dot2"
+            line += (f"  {tag} {us:7.1f}us {bw:5.2f}TB/s"
+                     f" ({bw/6.3*100:4.1f}%)")
+        _os.environ.pop("KUKEON_ATTN_MFMA", None)
+        print(line, flush=True)
