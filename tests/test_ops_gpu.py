@@ -529,3 +529,37 @@ def test_engine_gpu_preemption_recompute():
     for i, kv in enumerate(kvs):
         # context length: all tokens of all turns, last one pending
         assert len(kv.history) + kv.num_tokens >= expect[i] - 1
+
+
+@pytest.mark.parametrize("splits,fp8", [(4, False), (8, False), (4, True)])
+def test_paged_attention_mfma_path(splits, fp8, monkeypatch):
+    """The opt-in MFMA decode-attention kernel (KUKEON_ATTN_MFMA=1) must
+    match the fp32 reference like the default v_dot2 kernel does."""
+    monkeypatch.setenv("KUKEON_ATTN_MFMA", "1")
+    torch.manual_seed(5)
+    B, Hk, D, BS = 5, 2, 128, 16
+    Hq = 4 * Hk
+    ctxs = [1, 16, 17, 100, 250]
+    NB = sum((c + BS - 1) // BS for c in ctxs) + 4
+    kc, vc = _make_cache(NB, Hk, BS, D, fp8=fp8)
+    tol = 7e-2 if fp8 else 2e-2
+    maxb = max((c + BS - 1) // BS for c in ctxs)
+    bt = torch.zeros(B, maxb, dtype=torch.int32, device=DEV)
+    nxt = 0
+    for b, c in enumerate(ctxs):
+        n = (c + BS - 1) // BS
+        bt[b, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    seq_lens = torch.tensor(ctxs, dtype=torch.int32, device=DEV)
+    q = torch.randn(B, Hq * D, dtype=torch.bfloat16, device=DEV)
+    out = torch.empty(B, Hq * D, dtype=torch.bfloat16, device=DEV)
+    scale = D ** -0.5
+    tmp_out = torch.zeros(B, Hq, splits, D, dtype=torch.float32, device=DEV)
+    tmp_ml = torch.zeros(B, Hq, splits, 2, dtype=torch.float32, device=DEV)
+    ops.paged_attention(out, q, kc, vc, bt, seq_lens, 0, splits, scale,
+                        tmp_out, tmp_ml)
+    ref = torch.empty(B, Hq * D, dtype=torch.bfloat16)
+    reference.paged_attention(ref, q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                              seq_lens.cpu(), 0, 1, scale)
+    torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=tol,
+                               atol=tol)
