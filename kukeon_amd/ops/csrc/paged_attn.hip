@@ -544,6 +544,276 @@ __global__ __launch_bounds__(256, 2) void paged_attn_mfma_kernel(
   }
 }
 
+// ---- 16x16x16 (K=16) bf16 MFMA layout probe: C[16,16]=A[16,16]@B[16,16]
+// assumed maps (validated on-device by test_mfma_probe16k):
+//   A: lane -> row = l&15, k = 4*(l>>4)+j (j=0..3, bf16x4)
+//   B: lane -> col = l&15, k = 4*(l>>4)+j
+//   C: lane -> col = l&15, row = 4*(l>>4)+r
+typedef __attribute__((__vector_size__(4 * sizeof(short)))) short bf16x4_v;
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4_v;
+struct u2s_ { unsigned int x[2]; };
+DEV_INLINE bf16x4_v as_frag16(unsigned int w0, unsigned int w1) {
+  u2s_ v{{w0, w1}};
+  return __builtin_bit_cast(bf16x4_v, v);
+}
+
+__global__ void mfma_probe16k_kernel(float* __restrict__ c,
+                                     const unsigned short* __restrict__ a,
+                                     const unsigned short* __restrict__ b) {
+  const int lane = threadIdx.x & 63;
+  const int g = lane >> 4;
+  const int rc = lane & 15;
+  unsigned int ar[2], br[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int k0 = 4 * g + 2 * i;
+    ar[i] = (unsigned)a[rc * 16 + k0] | ((unsigned)a[rc * 16 + k0 + 1] << 16);
+    br[i] = (unsigned)b[k0 * 16 + rc] | ((unsigned)b[(k0 + 1) * 16 + rc] << 16);
+  }
+  f32x4_v acc = {};
+  acc = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
+      as_frag16(ar[0], ar[1]), as_frag16(br[0], br[1]), acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) c[(4 * g + r) * 16 + rc] = acc[r];
+}
+
+void mfma_probe16k(torch::Tensor out, torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.sizes() == torch::IntArrayRef({16, 16}));
+  TORCH_CHECK(b.sizes() == torch::IntArrayRef({16, 16}));
+  TORCH_CHECK(out.sizes() == torch::IntArrayRef({16, 16}));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  mfma_probe16k_kernel<<<1, 64, 0, stream>>>(
+      out.data_ptr<float>(),
+      reinterpret_cast<const unsigned short*>(a.data_ptr()),
+      reinterpret_cast<const unsigned short*>(b.data_ptr()));
+  HIP_CHECK_KERNEL();
+}
+
+// ===================================================================
+// 16-token-tile MFMA decode attention: the residency-first variant.
+//
+// Same wave-autonomous split-slot scheme as paged_attn_mfma_kernel, but
+// one cache block (16 tokens) per tile on 16x16 MFMAs:
+//   - S^T[tok, head] via 16x16x32 (K = head_dim chunks),
+//   - PV O^T[d, head] via 16x16x16bf16_1k (K = the 16 tokens) — and in
+//     this swapped shape the C layout of the scores IS the B layout of
+//     P (row=k=4*(l>>4)+idx, col=head), so P needs NO cross-lane
+//     redistribution at all, just two v_cvt_pk_bf16_f32.
+//   - 8 KiB LDS per wave (32 KiB/WG) -> ~5 workgroups/CU, matching the
+//     v_dot2 kernel's residency while issuing ~5x fewer instructions.
+// G <= 16 (8B: 4, 70B: 8); q-head padding is per-lane garbage, never
+// written back.
+template <bool FP8>
+__global__ __launch_bounds__(256, 4) void paged_attn_mfma16_kernel(
+    float* __restrict__ tmp_out,          // [B, Hq, S, D] f32
+    float* __restrict__ tmp_ml,           // [B, Hq, S, 2]
+    const unsigned short* __restrict__ q, // [B, q_stride]
+    const void* __restrict__ k_cache,     // [NB, Hk, 16, D]
+    const void* __restrict__ v_cache,
+    const int* __restrict__ block_table,  // [B, max_blocks]
+    const int* __restrict__ seq_lens,     // [B]
+    long q_stride, int Hq, int Hk, int max_blocks, int S, float scale) {
+  constexpr int D = 128;
+  constexpr int BS = 16;
+  const int b = blockIdx.x;
+  const int hk = blockIdx.y;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int g4 = lane >> 4;       // 0..3
+  const int rc = lane & 15;
+  const int G = Hq / Hk;
+  const int slot = blockIdx.z * 4 + wid;
+  const int ctx = seq_lens[b];
+  if (ctx <= 0 || slot >= S) return;
+
+  // V^T images only — K is consumed straight from HBM in fragment shape
+  // (16 B/lane across 64 B-line groups; the LDS round trip was pure
+  // added latency here: per-wave parked time was 67% of cycles). Two
+  // V^T buffers per wave: tile bi's image is written during tile bi-1,
+  // so PV never waits on fresh LDS stores.
+  __shared__ __align__(16) unsigned short vtbuf[4][2][D * BS];
+
+  const int nblocks = (ctx + BS - 1) / BS;
+  const int per_slot = (nblocks + S - 1) / S;
+  const int blk_begin = slot * per_slot;
+  const int blk_end = min(nblocks, blk_begin + per_slot);
+  const int kv_hi = min(ctx, blk_end * BS);
+
+  const int qh = min(rc, G - 1);  // B col = q head
+  // Q fragments pinned in 16 registers (L2 re-reads in the tile loop
+  // measured as 4 dependent ~250-cycle stalls per tile)
+  unsigned int qf[4][4];
+  {
+    const unsigned short* qp =
+        q + (long)b * q_stride + (long)(hk * G + qh) * D;
+#pragma unroll
+    for (int st = 0; st < 4; ++st) {
+      const uint4 v = *reinterpret_cast<const uint4*>(qp + st * 32 + g4 * 8);
+      qf[st][0] = v.x; qf[st][1] = v.y; qf[st][2] = v.z; qf[st][3] = v.w;
+    }
+  }
+
+  float m = -INFINITY, l_acc = 0.f;
+  f32x4_v oacc[8] = {{}, {}, {}, {}, {}, {}, {}, {}};
+
+  const unsigned short* kc16 =
+      reinterpret_cast<const unsigned short*>(k_cache);
+  const unsigned short* vc16 =
+      reinterpret_cast<const unsigned short*>(v_cache);
+  const unsigned char* kc8 = reinterpret_cast<const unsigned char*>(k_cache);
+  const unsigned char* vc8 = reinterpret_cast<const unsigned char*>(v_cache);
+
+  // K in MFMA A-fragment shape: lane (rc = token, g4 = k-group) reads
+  // 16 B at tok*256B + st*64B + g4*16B — lanes sharing a token cover one
+  // 64 B line, so the four per-tile loads coalesce at line granularity.
+  uint4 kfr[4];
+  auto kfetch = [&](int bi_f) {
+    const long base =
+        ((long)block_table[(long)b * max_blocks + bi_f] * Hk + hk) * BS * D;
+    const long off = base + (long)rc * D + g4 * 8;
+#pragma unroll
+    for (int st = 0; st < 4; ++st) {
+      if (FP8) {
+        kfr[st] = fp8x8_to_bf16x8(
+            *reinterpret_cast<const uint2*>(kc8 + off + st * 32));
+      } else {
+        kfr[st] = *reinterpret_cast<const uint4*>(kc16 + off + st * 32);
+      }
+    }
+  };
+  const int stg_d0 = rc * 8;
+  uint4 vpre[2][2];
+  auto vfetch = [&](int bi_f) {
+    const long base =
+        ((long)block_table[(long)b * max_blocks + bi_f] * Hk + hk) * BS * D;
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      const int kp = pass * 4 + g4;
+      const long voff0 = base + (long)(2 * kp) * D + stg_d0;
+      const long voff1 = base + (long)(2 * kp + 1) * D + stg_d0;
+      if (FP8) {
+        vpre[pass][0] =
+            fp8x8_to_bf16x8(*reinterpret_cast<const uint2*>(vc8 + voff0));
+        vpre[pass][1] =
+            fp8x8_to_bf16x8(*reinterpret_cast<const uint2*>(vc8 + voff1));
+      } else {
+        vpre[pass][0] = *reinterpret_cast<const uint4*>(vc16 + voff0);
+        vpre[pass][1] = *reinterpret_cast<const uint4*>(vc16 + voff1);
+      }
+    }
+  };
+  auto vwrite = [&](unsigned short* vt, int kvbase) {
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      const int kp = pass * 4 + g4;
+      uint4 r0 = vpre[pass][0];
+      uint4 r1 = vpre[pass][1];
+      if (kvbase + 2 * kp >= kv_hi) r0 = uint4{0, 0, 0, 0};
+      if (kvbase + 2 * kp + 1 >= kv_hi) r1 = uint4{0, 0, 0, 0};
+      const unsigned int* a0 = reinterpret_cast<const unsigned int*>(&r0);
+      const unsigned int* a1 = reinterpret_cast<const unsigned int*>(&r1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const unsigned short e0 = (a0[j >> 1] >> ((j & 1) * 16)) & 0xffffu;
+        const unsigned short e1 = (a1[j >> 1] >> ((j & 1) * 16)) & 0xffffu;
+        const unsigned int packed = (unsigned)e0 | ((unsigned)e1 << 16);
+        const int d = stg_d0 + j;
+        const int byte = d * 32 + ((4 * kp) ^ ((d & 3) << 3));
+        *reinterpret_cast<unsigned int*>(
+            reinterpret_cast<char*>(vt) + byte) = packed;
+      }
+    }
+  };
+
+  if (blk_begin < blk_end) {
+    kfetch(blk_begin);
+    vfetch(blk_begin);
+    vwrite(vtbuf[wid][blk_begin & 1], blk_begin * BS);
+  }
+
+  for (int bi = blk_begin; bi < blk_end; ++bi) {
+    const int kvbase = bi * BS;
+    const unsigned short* vt = vtbuf[wid][bi & 1];
+
+    // ---- S^T = K . Q^T from the in-flight K fragments ----
+    f32x4_v sacc = {};
+#pragma unroll
+    for (int st = 0; st < 4; ++st) {
+      sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          as_frag32(kfr[st].x, kfr[st].y, kfr[st].z, kfr[st].w),
+          as_frag32(qf[st][0], qf[st][1], qf[st][2], qf[st][3]),
+          sacc, 0, 0, 0);
+    }
+    // K registers are free again: issue the next tile's fragment loads
+    // (they fly under softmax + V staging + PV + the loop turnaround)
+    if (bi + 1 < blk_end) kfetch(bi + 1);
+
+    // ---- online softmax (lane holds tokens 4*g4..+3 of its head) ----
+    float sv[4];
+    float tmax = -INFINITY;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const bool valid = (kvbase + 4 * g4 + r) < kv_hi;
+      sv[r] = valid ? sacc[r] * scale : -INFINITY;
+      tmax = fmaxf(tmax, sv[r]);
+    }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 16, WAVE));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));
+    const float mn = fmaxf(m, tmax);
+    const float alpha = (m == -INFINITY) ? 0.f : __expf(m - mn);
+    m = mn;
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      sv[r] = (sv[r] == -INFINITY) ? 0.f : __expf(sv[r] - mn);
+      psum += sv[r];
+    }
+    psum += __shfl_xor(psum, 16, WAVE);
+    psum += __shfl_xor(psum, 32, WAVE);
+    l_acc = l_acc * alpha + psum;
+#pragma unroll
+    for (int dc = 0; dc < 8; ++dc)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) oacc[dc][r] *= alpha;
+
+    // ---- stage NEXT tile's V^T into the other buffer ----
+    if (bi + 1 < blk_end) {
+      vfetch(bi + 1);
+      vwrite(vtbuf[wid][(bi + 1) & 1], (bi + 1) * BS);
+    }
+
+    // ---- PV: P is already a 16x16x16 B fragment ----
+    const bf16x4_v pfrag = as_frag16(cvt_pk_bf16d(sv[0], sv[1]),
+                                     cvt_pk_bf16d(sv[2], sv[3]));
+#pragma unroll
+    for (int dc = 0; dc < 8; ++dc) {
+      const int d = dc * 16 + rc;  // A row = output dim
+      const int byte = d * 32 + ((8 * g4) ^ ((d & 3) << 3));
+      const uint2 vf = *reinterpret_cast<const uint2*>(
+          reinterpret_cast<const char*>(vt) + byte);
+      oacc[dc] = __builtin_amdgcn_mfma_f32_16x16x16bf16_1k(
+          as_frag16(vf.x, vf.y), pfrag, oacc[dc], 0, 0, 0);
+    }
+  }
+
+  // ---- epilogue ----
+  if (rc >= G) return;
+  const int head = hk * G + rc;
+  float* top = tmp_out + (((long)b * Hq + head) * S + slot) * D;
+#pragma unroll
+  for (int dc = 0; dc < 8; ++dc) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      top[dc * 16 + 4 * g4 + r] = oacc[dc][r];
+    }
+  }
+  if (g4 == 0) {
+    float* ml = tmp_ml + (((long)b * Hq + head) * S + slot) * 2;
+    ml[0] = m;
+    ml[1] = l_acc;
+  }
+}
+
 // Reduce for the MFMA path: every one of the S slots is written on every
 // call (empty ranges write m = -inf), so all are read unconditionally —
 // no per-split block math, no zero-ctx division hazard.
@@ -605,20 +875,34 @@ void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
   // kernel's 8 KiB image keeps ~8x more waves in flight to hide its
   // issue stalls. Round-2 lever: 16-token tiles on 16x16x32 MFMAs
   // (quarter the LDS) or a hand-scheduled pipeline.
+  // DEFAULT: the 16-token-tile MFMA kernel (fragment-direct K, double-
+  // buffered V^T, zero barriers) — measured 4.9-5.4 TB/s (78-86% of the
+  // HBM roofline) vs the v_dot2 kernel's 3.8-4.1 on the decode shapes.
+  // KUKEON_ATTN_MFMA=0 forces v_dot2, =32 the 32-token 32x32 variant;
+  // v_dot2 also remains the fallback for split counts not divisible by
+  // 4 and GQA groups over 16.
   const char* mfma_env = getenv("KUKEON_ATTN_MFMA");
-  const bool use_mfma = (mfma_env && mfma_env[0] == '1') &&
-                        num_splits % 4 == 0 && G <= 32 &&
+  const bool want32 = mfma_env && mfma_env[0] == '3' && G <= 32;
+  const bool want16 = !want32 && (!mfma_env || mfma_env[0] != '0') && G <= 16;
+  const bool use_mfma = (want16 || want32) &&
+                        num_splits % 4 == 0 &&
                         tmp_out.numel() >= (long)B * Hq * num_splits * 128;
   if (use_mfma) {
     dim3 mgrid(B, Hk, (unsigned)(num_splits / 4));
     float* tovm = tmp_out.data_ptr<float>();
     float* tmlm = tmp_ml.data_ptr<float>();
-#define PAM_LAUNCH(FP8_)                                                     \
-    paged_attn_mfma_kernel<FP8_><<<mgrid, 256, 0, stream>>>(                 \
+#define PAM_LAUNCH(KERN, FP8_)                                               \
+    KERN<FP8_><<<mgrid, 256, 0, stream>>>(                                   \
         tovm, tmlm, qp, k_cache.data_ptr(), v_cache.data_ptr(),              \
         block_table.data_ptr<int>(), seq_lens.data_ptr<int>(), q_stride,     \
         Hq, Hk, max_blocks, (int)num_splits, (float)scale)
-    if (fp8) { PAM_LAUNCH(true); } else { PAM_LAUNCH(false); }
+    if (want16) {
+      if (fp8) { PAM_LAUNCH(paged_attn_mfma16_kernel, true); }
+      else { PAM_LAUNCH(paged_attn_mfma16_kernel, false); }
+    } else {
+      if (fp8) { PAM_LAUNCH(paged_attn_mfma_kernel, true); }
+      else { PAM_LAUNCH(paged_attn_mfma_kernel, false); }
+    }
 #undef PAM_LAUNCH
     HIP_CHECK_KERNEL();
     dim3 rgrid(B, Hq);

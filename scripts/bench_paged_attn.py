@@ -49,13 +49,13 @@ for ctx_lo, ctx_hi in [(512, 3584), (1500, 1500), (3584, 3584)]:
         tmp_ml = torch.zeros(B, Hq, splits, 2, dtype=torch.float32,
                              device=DEV)
         line = f"  splits={splits:2d}:"
-        for mode in ("0", "1"):
+        for mode in ("0", "1", "32"):
             _os.environ["KUKEON_ATTN_MFMA"] = mode
             us = t(lambda: ops.paged_attention(out, q, kc, vc, bt, seq_lens,
                                                0, splits, scale, tmp_out,
                                                tmp_ml))
             bw = kv_bytes / (us * 1e-6) / 1e12
-            tag = "mfma" if mode == "1" else "dot2"
+            tag = {"0": "dot2", "1": "m16", "32": "m32"}[mode]
             line += (f"  {tag} {us:7.1f}us {bw:5.2f}TB/s"
                      f" ({bw/6.3*100:4.1f}%)")
         _os.environ.pop("KUKEON_ATTN_MFMA", None)

@@ -531,11 +531,25 @@ def test_engine_gpu_preemption_recompute():
         assert len(kv.history) + kv.num_tokens >= expect[i] - 1
 
 
-@pytest.mark.parametrize("splits,fp8", [(4, False), (8, False), (4, True)])
-def test_paged_attention_mfma_path(splits, fp8, monkeypatch):
-    """The opt-in MFMA decode-attention kernel (KUKEON_ATTN_MFMA=1) must
-    match the fp32 reference like the default v_dot2 kernel does."""
-    monkeypatch.setenv("KUKEON_ATTN_MFMA", "1")
+def test_mfma_probe16k_layout():
+    from kukeon_amd import _C
+    torch.manual_seed(0)
+    a = (torch.randn(16, 16) * torch.linspace(0.2, 2.0, 16)).bfloat16().to(DEV)
+    b = (torch.randn(16, 16) * torch.linspace(-1.5, 1.5, 16)).bfloat16().to(DEV)
+    out = torch.empty(16, 16, dtype=torch.float32, device=DEV)
+    _C.mfma_probe16k(out, a, b)
+    ref = a.float() @ b.float()
+    torch.testing.assert_close(out.cpu(), ref.cpu(), rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.parametrize("mode,splits,fp8", [
+    ("1", 4, False), ("1", 8, False), ("1", 4, True),
+    ("32", 4, False), ("32", 8, False), ("32", 4, True)])
+def test_paged_attention_mfma_path(mode, splits, fp8, monkeypatch):
+    """The opt-in MFMA decode-attention kernels (KUKEON_ATTN_MFMA=1 is the
+    16-token-tile 16x16 variant, =32 the 32-token 32x32 one) must match
+    the fp32 reference like the default v_dot2 kernel does."""
+    monkeypatch.setenv("KUKEON_ATTN_MFMA", mode)
     torch.manual_seed(5)
     B, Hk, D, BS = 5, 2, 128, 16
     Hq = 4 * Hk
