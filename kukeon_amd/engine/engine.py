@@ -375,9 +375,11 @@ class LLMEngine:
         # (scripts/bench_paged_attn.py) shows 2048-4096 WGs runs 5-9%
         # faster than the 512-WG minimum (imbalance absorption across
         # the 8 XCDs), flat beyond
-        want = (8 * 256 + bucket * self.hk - 1) // (bucket * self.hk)
-        # multiple of 4: the MFMA decode-attention kernel gives each of a
-        # workgroup's 4 autonomous waves its own split slot
+        # target ~4096 split-units: the MFMA decode kernel runs B*Hk*S
+        # autonomous waves (4 per workgroup), and 4096 exactly fills the
+        # chip at its 4-waves/SIMD occupancy (multiple of 4 required —
+        # each workgroup's waves own consecutive split slots)
+        want = (16 * 256 + bucket * self.hk - 1) // (bucket * self.hk)
         return max(4, min(self.max_splits, ((want + 3) // 4) * 4))
 
     def _tmp_for(self, bucket: int, splits: int):
