@@ -78,6 +78,28 @@ class LlamaAttention(nn.Module):
         o = ops.linear(out, self.o_w)
         return parallel.tp_all_reduce(o)
 
+    def forward_pre_o(self, x: torch.Tensor, cos_sin: torch.Tensor,
+                      kc: torch.Tensor, vc: torch.Tensor, meta: AttnMeta):
+        """Attention WITHOUT the output projection: the layer fuses the
+        o-proj with the post-attention add+RMSNorm (one epilogue kernel
+        instead of reduce + norm)."""
+        qkv = ops.linear(x, self.qkv_w)
+        ops.rope_kv_append(qkv, kc, vc, cos_sin, meta.positions,
+                           meta.slot_mapping, self.hq, self.hk, self.d,
+                           block_table=(meta.block_table
+                                        if meta.mode == "decode" else None))
+        out = torch.empty(x.shape[0], self.hq * self.d, dtype=x.dtype,
+                          device=x.device)
+        if meta.mode == "decode":
+            ops.paged_attention(out, qkv, kc, vc, meta.block_table,
+                                meta.seq_lens, 0, meta.num_splits, self.scale,
+                                meta.tmp_out, meta.tmp_ml)
+        else:
+            ops.prefill_attention(out, qkv, kc, vc, meta.block_table,
+                                  meta.seq_lens, meta.q_starts, meta.qb_seq,
+                                  meta.qb_start, 0, self.scale)
+        return out
+
 
 class LlamaMLP(nn.Module):
     def __init__(self, cfg: ModelConfig, device):
@@ -117,8 +139,9 @@ class LlamaLayer(nn.Module):
         else:
             ops.fused_add_rmsnorm(x, residual, self.input_norm, self.eps)
             h = x
-        h = self.attn.forward(h, cos_sin, kc, vc, meta)
-        ops.fused_add_rmsnorm(h, residual, self.post_norm, self.eps)
+        attn_out = self.attn.forward_pre_o(h, cos_sin, kc, vc, meta)
+        h = ops.linear_add_rmsnorm(attn_out, self.attn.o_w, residual,
+                                   self.post_norm, self.eps)
         h = self.mlp.forward(h)
         return h, residual
 

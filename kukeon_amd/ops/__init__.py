@@ -121,6 +121,41 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.linear(x, w)
 
 
+def linear_add_rmsnorm(x: torch.Tensor, w: torch.Tensor,
+                       residual: torch.Tensor, norm_weight: torch.Tensor,
+                       eps: float) -> torch.Tensor:
+    """out-projection + TP all-reduce + residual add + RMSNorm as one
+    fused path. On the skinny decode shapes (TP=1) the split-K reduce,
+    residual add and norm run in a single kernel — the separate epilogue
+    pair cost two launches at the ~4.5us in-graph dispatch floor for
+    <1us of work. Mutates `residual` (+= x@w.T) and returns the normed
+    activations, exactly like linear() followed by fused_add_rmsnorm().
+    """
+    from kukeon_amd import parallel
+    rows = x.shape[0]
+    N, K = w.shape
+    if (x.is_cuda and parallel.tp_size() == 1 and x.dim() == 2
+            and rows <= 64 and N % 2048 == 0 and N <= 8192 and K % 32 == 0
+            and x.dtype == torch.bfloat16
+            and (_USE_SKINNY or _skinny_wins(rows, N, K))):
+        key = (x.device.index or 0)
+        ws = _SKINNY_WS.get(key)
+        ntiles = N // 64
+        nslices = -(-K // 512)
+        splitk = min(nslices, -(-256 // ntiles))
+        need = max(1, splitk) * 64 * N
+        if ws is None or ws.numel() < need:
+            ws = torch.empty(need, dtype=torch.float32, device=x.device)
+            _SKINNY_WS[key] = ws
+        normed = torch.empty(rows, N, dtype=x.dtype, device=x.device)
+        _native().skinny_gemm_fused_norm(normed, x, w, ws, residual,
+                                         norm_weight, eps)
+        return normed
+    h = parallel.tp_all_reduce(linear(x, w))
+    fused_add_rmsnorm(h, residual, norm_weight, eps)
+    return h
+
+
 def moe_gather_tokens(out, input, row_map) -> None:
     _impl(input).moe_gather_tokens(out, input, row_map)
 

@@ -31,6 +31,10 @@ void mfma_probe16k(torch::Tensor out, torch::Tensor a, torch::Tensor b);
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws);
 void glds_probe(torch::Tensor out, torch::Tensor src);
+void skinny_gemm_fused_norm(torch::Tensor normed, torch::Tensor x,
+                            torch::Tensor w, torch::Tensor ws,
+                            torch::Tensor residual, torch::Tensor nw,
+                            double eps);
 void sample(torch::Tensor tokens, torch::Tensor logits, torch::Tensor temps,
             torch::Tensor top_k, torch::Tensor top_p, torch::Tensor seed,
             torch::Tensor workspace);
@@ -61,7 +65,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &kukeon::skinny_gemm,
         "weight-streaming decode GEMM (M<=64)");
   m.def("glds_probe", &kukeon::glds_probe,
-        "asm global_load_lds round-trip validator");
+        "asm global_lds round-trip validator");
+  m.def("skinny_gemm_fused_norm", &kukeon::skinny_gemm_fused_norm,
+        "skinny GEMM + split-K reduce + residual add + RMSNorm");
   m.def("sample", &kukeon::sample, "top-k/top-p/temperature sampling");
   m.def("decode_advance", &kukeon::decode_advance,
         "on-device decode cursor advance (self-advancing graph)");

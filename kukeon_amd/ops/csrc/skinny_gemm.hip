@@ -303,6 +303,103 @@ __global__ void skinny_reduce_kernel(unsigned short* __restrict__ out,
   }
 }
 
+// Split-K reduce fused with residual-add + RMSNorm: the decode o-proj's
+// epilogue pair (skinny_reduce 5.0us + fused_add_rmsnorm 5.4us) was two
+// launches at the ~4.5us in-graph dispatch floor each for <1us of real
+// work. One block per row: sum the slabs, update the residual in place,
+// one block-reduce for the mean square, write the normed activations.
+__global__ __launch_bounds__(256) void skinny_reduce_add_rmsnorm_kernel(
+    unsigned short* __restrict__ normed,    // [M, N] bf16
+    unsigned short* __restrict__ residual,  // [M, N] bf16, updated
+    const float* __restrict__ ws,           // [splitk, M, N] f32
+    const unsigned short* __restrict__ nw,  // [N]
+    int N, long total, int splitk, float eps) {
+  const int t = blockIdx.x;
+  __shared__ float red[16];
+  float vals[32];  // up to N = 8192 at 256 threads x 8 elems
+  const int nchunk = N / (256 * 8);
+  float sq = 0.f;
+  for (int c = 0; c < nchunk; ++c) {
+    const int n0 = (c * 256 + threadIdx.x) * 8;
+    const long off = (long)t * N + n0;
+    const bf16x8 r = load_bf16x8(residual + off);
+    float acc[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] = r.f(j);
+    for (int s = 0; s < splitk; ++s) {
+      const float4 a =
+          *reinterpret_cast<const float4*>(ws + (long)s * total + off);
+      const float4 b =
+          *reinterpret_cast<const float4*>(ws + (long)s * total + off + 4);
+      acc[0] += a.x; acc[1] += a.y; acc[2] += a.z; acc[3] += a.w;
+      acc[4] += b.x; acc[5] += b.y; acc[6] += b.z; acc[7] += b.w;
+    }
+    *reinterpret_cast<uint4*>(residual + off) = pack_bf16x8(acc);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      // norm of the bf16-rounded residual, matching the two-kernel path
+      const float v = us2f(f2us(acc[j]));
+      vals[c * 8 + j] = v;
+      sq += v * v;
+    }
+  }
+  sq = block_reduce(sq, red, SumOp{}, 0.f);
+  const float rs = __frsqrt_rn(sq / N + eps);
+  for (int c = 0; c < nchunk; ++c) {
+    const int n0 = (c * 256 + threadIdx.x) * 8;
+    const bf16x8 wv = load_bf16x8(nw + n0);
+    float o[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = vals[c * 8 + j] * rs * wv.f(j);
+    *reinterpret_cast<uint4*>(normed + (long)t * N + n0) = pack_bf16x8(o);
+  }
+}
+
+void skinny_gemm_fused_norm(torch::Tensor normed, torch::Tensor x,
+                            torch::Tensor w, torch::Tensor ws,
+                            torch::Tensor residual, torch::Tensor nw,
+                            double eps) {
+  const int M = x.size(0);
+  const long K = x.size(1);
+  const int N = w.size(0);
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 2048 == 0 && N <= 8192 &&
+              K % 32 == 0);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() &&
+              residual.is_contiguous() && normed.is_contiguous());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int ntiles = N / 64;
+  const int nslices = (int)((K + KSLICE - 1) / KSLICE);
+  int splitk = min(nslices, (256 + ntiles - 1) / ntiles);
+  if (const char* ov = getenv("KUKEON_SKINNY_SPLITK")) {
+    const int v = atoi(ov);
+    if (v > 0) splitk = min(nslices, v);
+  }
+  const int MT = (M + 15) / 16;
+  dim3 grid(ntiles, splitk);
+  auto* xp = reinterpret_cast<const unsigned short*>(x.data_ptr());
+  auto* wp = reinterpret_cast<const unsigned short*>(w.data_ptr());
+  const long total = (long)M * N;
+  float* wsp = ws.data_ptr<float>();
+  TORCH_CHECK(ws.numel() >= total * splitk, "workspace too small");
+#define SKF_LAUNCH(MT_)                                                      \
+  skinny_gemm_kernel<MT_, true><<<grid, 256, 0, stream>>>(                   \
+      nullptr, wsp, xp, wp, M, N, K)
+  switch (MT) {
+    case 1: SKF_LAUNCH(1); break;
+    case 2: SKF_LAUNCH(2); break;
+    case 3: SKF_LAUNCH(3); break;
+    default: SKF_LAUNCH(4); break;
+  }
+#undef SKF_LAUNCH
+  HIP_CHECK_KERNEL();
+  skinny_reduce_add_rmsnorm_kernel<<<dim3((unsigned)M), 256, 0, stream>>>(
+      reinterpret_cast<unsigned short*>(normed.data_ptr()),
+      reinterpret_cast<unsigned short*>(residual.data_ptr()), wsp,
+      reinterpret_cast<const unsigned short*>(nw.data_ptr()), N, total,
+      splitk, (float)eps);
+  HIP_CHECK_KERNEL();
+}
+
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws) {
   const int M = x.size(0);

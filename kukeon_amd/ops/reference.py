@@ -198,6 +198,12 @@ def sample(tokens: torch.Tensor, logits: torch.Tensor, temps: torch.Tensor,
     seed += 1
 
 
+def linear_add_rmsnorm(x, w, residual, norm_weight, eps: float):
+    h = torch.nn.functional.linear(x, w)
+    fused_add_rmsnorm(h, residual, norm_weight, eps)
+    return h
+
+
 def decode_advance(ids, pos, seq_lens, tokens, ring, counter) -> None:
     step = int(counter[0])
     B = ids.shape[0]

@@ -577,3 +577,23 @@ def test_paged_attention_mfma_path(mode, splits, fp8, monkeypatch):
                               seq_lens.cpu(), 0, 1, scale)
     torch.testing.assert_close(out.cpu().float(), ref.float(), rtol=tol,
                                atol=tol)
+
+
+@pytest.mark.parametrize("M", [1, 17, 64])
+def test_linear_add_rmsnorm_fused(M):
+    """The fused o-proj epilogue (split-K reduce + residual add + RMSNorm
+    in one kernel) must match linear() + fused_add_rmsnorm()."""
+    torch.manual_seed(11)
+    N = K = 4096
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.3
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.02
+    resid = torch.randn(M, N, dtype=torch.bfloat16, device=DEV)
+    nw = torch.randn(N, dtype=torch.bfloat16, device=DEV)
+    resid_ref = resid.cpu().clone()
+    h_ref = (x.cpu().float() @ w.cpu().float().T).to(torch.bfloat16)
+    reference.fused_add_rmsnorm(h_ref, resid_ref, nw.cpu(), 1e-5)
+    out = ops.linear_add_rmsnorm(x, w, resid, nw, 1e-5)
+    torch.testing.assert_close(resid.cpu().float(), resid_ref.float(),
+                               rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(out.cpu().float(), h_ref.float(), rtol=2e-2,
+                               atol=3e-2)
