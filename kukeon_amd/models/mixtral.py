@@ -133,8 +133,9 @@ class MixtralLayer(nn.Module):
         else:
             ops.fused_add_rmsnorm(x, residual, self.input_norm, self.eps)
             h = x
-        h = self.attn.forward(h, cos_sin, kc, vc, meta)
-        ops.fused_add_rmsnorm(h, residual, self.post_norm, self.eps)
+        attn_out = self.attn.forward_pre_o(h, cos_sin, kc, vc, meta)
+        h = ops.linear_add_rmsnorm(attn_out, self.attn.o_w, residual,
+                                   self.post_norm, self.eps)
         h = self.moe.forward(h)
         return h, residual
 
