@@ -109,7 +109,7 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
         key = (x.device.index or 0)
         ws = _SKINNY_WS.get(key)
         ntiles = N // 64
-        nslices = -(-K // 512)
+        nslices = -(-K // 256)
         splitk = 1 if ntiles >= 512 else min(nslices, -(-256 // ntiles))
         need = max(1, splitk) * 64 * N
         if ws is None or ws.numel() < need:
@@ -141,7 +141,7 @@ def linear_add_rmsnorm(x: torch.Tensor, w: torch.Tensor,
         key = (x.device.index or 0)
         ws = _SKINNY_WS.get(key)
         ntiles = N // 64
-        nslices = -(-K // 512)
+        nslices = -(-K // 256)
         splitk = min(nslices, -(-256 // ntiles))
         need = max(1, splitk) * 64 * N
         if ws is None or ws.numel() < need:

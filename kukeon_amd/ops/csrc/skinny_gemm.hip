@@ -37,8 +37,13 @@ DEV_INLINE u32x4_t nt_load16(const void* p) {
   return __builtin_nontemporal_load(reinterpret_cast<const u32x4_t*>(p));
 }
 
-constexpr int KSLICE = 512;
-constexpr int U = 8;  // W k-chunks (of 32) per batch; 2 batches per slice
+constexpr int KSLICE = 256;
+constexpr int U = 8;  // W k-chunks (of 32) per batch; 1 batch per slice
+// KSLICE 512 -> 256: with 512-wide slices the two x buffers took 128 KiB
+// of LDS = 1 workgroup/CU = ONE wave per SIMD, and PMC showed the waves
+// parked on memory 74% of their cycles with nothing to hide behind.
+// 256-wide slices halve the LDS (2 workgroups/CU, 2 waves/SIMD) at the
+// cost of twice the slice turnarounds.
 
 __global__ void skinny_zero_kernel(float* __restrict__ ws, long n) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -92,14 +97,16 @@ DEV_INLINE void glds_stage_x(unsigned short* xbuf,
                              const unsigned short* __restrict__ x, long K,
                              long ks, int klen, int M, int wid, int lane) {
   const int cap = klen * 2 - 16;  // klen%32==0 -> 16B-aligned
+  // one glds stages 1 KiB = TWO 512-B rows: lanes 0-31 cover the even
+  // row, 32-63 the odd row (dest stays lane-linear)
 #pragma unroll
-  for (int i = 0; i < 16; ++i) {
-    const int row = wid * 16 + i;
-    const int byte_in_row = min((lane * 16) ^ ((row & 15) << 4), cap);
+  for (int i = 0; i < 8; ++i) {
+    const int row = wid * 16 + 2 * i + (lane >> 5);
+    const int byte_in_row = min(((lane & 31) * 16) ^ ((row & 15) << 4), cap);
     const unsigned short* g =
         x + (long)min(M - 1, row) * K + ks + byte_in_row / 2;
     glds16(g, __builtin_amdgcn_readfirstlane(
-                  lds_addr_of(xbuf + row * KSLICE)));
+                  lds_addr_of(xbuf + (row & ~1) * KSLICE)));
   }
 }
 

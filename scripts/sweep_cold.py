@@ -41,8 +41,8 @@ for (M, N, K, tag) in SHAPES:
     ws_list = [torch.randn(N, K, dtype=torch.bfloat16, device="cuda") * 0.05
                for _ in range(nw)]
     out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
-    nslices = (K + 511) // 512
-    wrk = torch.empty(nslices * M * N, dtype=torch.float32, device="cuda")
+    nslices = (K + 255) // 256
+    wrk = torch.empty(min(16, nslices) * M * N, dtype=torch.float32, device="cuda")
     floor = wbytes / 6.3e12 * 1e6
     us_bl = t_rot(lambda i: F.linear(x, ws_list[i]), nw)
     line = (f"{tag:>8} ({nw} copies): blas {us_bl:6.1f}  floor {floor:6.1f}")
