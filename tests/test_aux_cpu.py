@@ -1,0 +1,121 @@
+"""Aux-subsystem tests mirroring the reference's injectable-runner style
+(iptables paths asserted by command line, frozen clocks,
+fake samplers — no root, no iptables, no amdgpu needed)."""
+import logging
+
+import pytest
+
+from kukeon_amd.api import errors
+from kukeon_amd.api import v1beta1 as api
+from kukeon_amd.netpolicy import (IptablesEnforcer, MASTER_CHAIN, Policy,
+                                  Rule, build_policy, space_chain)
+from kukeon_amd.runtime import proc
+from kukeon_amd.runtime.devices import GPUAllocator, visible_devices_env
+from kukeon_amd.runtime.diskpressure import Guard, Sample
+
+
+class RecordingRunner:
+    """Every command 'exists' checks fail (-C/-nL return 1) so the
+    enforcer takes its create path; all other commands succeed."""
+
+    def __init__(self):
+        self.cmds = []
+
+    def __call__(self, args):
+        self.cmds.append(" ".join(args))
+        return 1 if args[1] in ("-C", "-nL") else 0
+
+
+def test_egress_enforcer_command_lines():
+    r = RecordingRunner()
+    enf = IptablesEnforcer(runner=r)
+    pol = Policy(default_deny=True,
+                 rules=[Rule(cidr="140.82.0.0/16", ports=[443]),
+                        Rule(cidr="10.0.0.5/32", ports=[])])
+    enf.apply("default", "dev", "10.88.3.0/24", pol)
+    chain = space_chain("default", "dev")
+    joined = "\n".join(r.cmds)
+    assert f"iptables -N {MASTER_CHAIN}" in joined
+    assert "iptables -I FORWARD -j KUKEON-EGRESS" in joined
+    assert f"iptables -N {chain}" in joined
+    assert f"iptables -A {MASTER_CHAIN} -s 10.88.3.0/24 -j {chain}" in joined
+    assert (f"iptables -A {chain} -d 140.82.0.0/16 -p tcp --dport 443 "
+            "-j ACCEPT") in joined
+    assert f"iptables -A {chain} -d 10.0.0.5/32 -j ACCEPT" in joined
+    # established return traffic precedes the default drop
+    est = joined.index("ESTABLISHED,RELATED -j ACCEPT")
+    assert f"iptables -A {chain} -j DROP" in joined
+    assert est < joined.index(f"iptables -A {chain} -j DROP")
+    r2 = RecordingRunner()
+    IptablesEnforcer(runner=r2).remove("default", "dev")
+    assert f"iptables -X {chain}" in "\n".join(r2.cmds)
+
+
+def test_build_policy_resolves_hosts():
+    def fake_resolver(host, port):
+        assert host == "api.example.com"
+        return [(2, 1, 6, "", ("1.2.3.4", 0)), (2, 1, 6, "", ("1.2.3.4", 0)),
+                (10, 1, 6, "", ("::1", 0, 0, 0))]
+
+    eg = api.EgressPolicy(default="deny", allow=[
+        api.EgressAllowRule(host="api.example.com", ports=[443]),
+        api.EgressAllowRule(cidr="9.9.9.0/24")])
+    p = build_policy(eg, resolver=fake_resolver)
+    assert p.default_deny
+    # v6 and duplicate addresses dropped; host becomes /32
+    assert [r.cidr for r in p.rules] == ["1.2.3.4/32", "9.9.9.0/24"]
+
+
+def test_gpu_allocator_persistence_and_exhaustion(tmp_path):
+    path = str(tmp_path / "gpus.json")
+    a = GPUAllocator(path, devices=[0, 1, 2, 3])
+    assert a.allocate("sess-a", 2) == [0, 1]
+    assert a.allocate("sess-a", 2) == [0, 1]  # idempotent per owner
+    assert a.allocate("sess-b", 1) == [2]
+    with pytest.raises(errors.GPUUnavailable):
+        a.allocate("sess-c", 2)
+    # persisted: a fresh allocator sees the same assignments
+    b = GPUAllocator(path, devices=[0, 1, 2, 3])
+    assert b.free == [3]
+    b.release("sess-a")
+    assert b.free == [0, 1, 3]
+    assert GPUAllocator(path, devices=[0, 1, 2, 3]).free == [0, 1, 3]
+    env = dict(e.split("=", 1) for e in visible_devices_env([2]))
+    assert env["ROCR_VISIBLE_DEVICES"] == "2"
+
+
+def test_disk_pressure_block_warn_and_ratelimit(caplog):
+    t = [0.0]
+    pct = [50.0]
+
+    def sampler(path):
+        return Sample(total_bytes=100, used_bytes=int(pct[0]))
+
+    g = Guard("/x", warn_percent=85.0, block_percent=95.0, sampler=sampler,
+              now_fn=lambda: t[0])
+    g.check()  # calm
+    pct[0] = 96.0
+    with pytest.raises(errors.DiskPressure):
+        g.check()
+    g.check(ignore=True)  # ignoreDiskPressure override
+    pct[0] = 90.0
+    with caplog.at_level(logging.WARNING, logger="kukeon.diskpressure"):
+        t[0] = 1000.0
+        g.check()
+        t[0] = 1060.0
+        g.check()  # inside the 5-minute re-emit window: suppressed
+        warns = [r for r in caplog.records if "warn threshold" in r.message]
+        assert len(warns) == 1
+        t[0] = 1000.0 + 301.0
+        g.check()
+        warns = [r for r in caplog.records if "warn threshold" in r.message]
+        assert len(warns) == 2
+
+
+def test_proc_metrics_self():
+    import os
+    m = proc.metrics(os.getpid())
+    assert m is not None
+    assert m["rssBytes"] > 1 << 20 and m["threads"] >= 1
+    assert m["cpuSeconds"] >= 0.0
+    assert proc.metrics(2 ** 22 + 12345) is None  # unlikely pid
