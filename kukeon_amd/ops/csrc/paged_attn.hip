@@ -604,7 +604,9 @@ void mfma_probe16k(torch::Tensor out, torch::Tensor a, torch::Tensor b) {
 // G <= 16 (8B: 4, 70B: 8); q-head padding is per-lane garbage, never
 // written back.
 template <bool FP8>
-__global__ __launch_bounds__(256, 4) void paged_attn_mfma16_kernel(
+// fp8 needs headroom for the cvt_pk conversions (at 4 waves/SIMD the
+// 128-reg bound put a 2-VGPR spill in the hot loop)
+__global__ __launch_bounds__(256, FP8 ? 3 : 4) void paged_attn_mfma16_kernel(
     float* __restrict__ tmp_out,          // [B, Hq, S, D] f32
     float* __restrict__ tmp_ml,           // [B, Hq, S, 2]
     const unsigned short* __restrict__ q, // [B, q_stride]
