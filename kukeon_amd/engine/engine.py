@@ -112,7 +112,10 @@ class LLMEngine:
         self.d_seed = torch.full((1,), ecfg.seed, dtype=torch.int64, device=d)
         self.d_ws = torch.zeros(self.Bmax, 528, dtype=torch.float32,
                                 device=d)  # sampling workspace rows
-        self.max_splits = 16
+        # 32 split slots: small decode batches (16-session / 70B configs)
+        # need B*Hk*S ~ 4096 wave-units to fill the MFMA attention kernel
+        # at its 4-waves/SIMD occupancy; 16 capped them at half the chip
+        self.max_splits = 32
         self.d_tmp_out = {}
         self.d_tmp_ml = {}
         # self-advancing decode: token ring + device step counter
