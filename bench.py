@@ -37,6 +37,10 @@ def main() -> None:
     ap.add_argument("--no-graphs", action="store_true")
     ap.add_argument("--decode-microbatch", type=int, default=64)
     ap.add_argument("--kv-dtype", choices=["bf16", "fp8"], default="bf16")
+    ap.add_argument("--moe-ep", action="store_true",
+                    help="MoE expert parallelism: experts partitioned "
+                         "across ranks, tokens routed over all-to-all "
+                         "(RCCL over xGMI); sessions stay rank-local")
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
 
@@ -74,6 +78,10 @@ def main() -> None:
     if args.model == "llama-3-70b":
         parallel.init_tensor_parallel(world)
         ecfg.tp_size = world
+    use_ep = args.moe_ep and world > 1
+    if use_ep:
+        parallel.init_expert_parallel(world)
+        ecfg.use_graphs = False  # all-to-all splits are data-dependent
     if cfg.is_moe:
         from kukeon_amd.models.mixtral import MixtralModel
         model = MixtralModel(cfg, device=device)
@@ -144,7 +152,8 @@ def main() -> None:
                 "model": args.model,
                 "global_batch": sessions_per_engine * dp,
                 "seq_len": args.ctx_cap,
-                "parallelism": (f"tp{tp}" if tp > 1 else f"dp{dp}"),
+                "parallelism": (f"ep{world}" if use_ep else
+                                f"tp{tp}" if tp > 1 else f"dp{dp}"),
                 "sessions_per_gpu_group": sessions_per_engine,
                 "prompt_len": args.prompt_len,
                 "followup_len": args.followup_len,

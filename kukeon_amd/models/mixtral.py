@@ -31,25 +31,101 @@ class MixtralMoE(nn.Module):
     def __init__(self, cfg: ModelConfig, device):
         super().__init__()
         tp = parallel.tp_size()
+        self.ep = parallel.ep_size()
         assert cfg.intermediate_size % tp == 0
         self.E = cfg.num_experts
         self.K = cfg.top_k_experts
         self.inter = cfg.intermediate_size // tp
         self.router_w = _init_weight((self.E, cfg.hidden_size), device)
-        self.gate_up_w = nn.Parameter(
-            torch.empty(self.E, 2 * self.inter, cfg.hidden_size,
-                        dtype=torch.bfloat16, device=device).normal_(0, 0.02),
-            requires_grad=False)
-        self.down_w = nn.Parameter(
-            torch.empty(self.E, cfg.hidden_size, self.inter,
-                        dtype=torch.bfloat16, device=device).normal_(0, 0.02),
-            requires_grad=False)
+        gate_up = torch.empty(self.E, 2 * self.inter, cfg.hidden_size,
+                              dtype=torch.bfloat16,
+                              device=device).normal_(0, 0.02)
+        down = torch.empty(self.E, cfg.hidden_size, self.inter,
+                           dtype=torch.bfloat16,
+                           device=device).normal_(0, 0.02)
+        if self.ep > 1:
+            # expert-parallel: this rank OWNS E/ep whole experts (full
+            # intermediate dim); tokens travel to their experts over
+            # all-to-all. Full-size init then slice keeps every rank's
+            # RNG stream identical to the single-rank module, so EP is
+            # numerically the same model.
+            assert tp == 1 and self.E % self.ep == 0
+            self.e_local = self.E // self.ep
+            e0 = parallel.ep_rank() * self.e_local
+            gate_up = gate_up[e0: e0 + self.e_local].contiguous()
+            down = down[e0: e0 + self.e_local].contiguous()
+        else:
+            self.e_local = self.E
+        self.gate_up_w = nn.Parameter(gate_up, requires_grad=False)
+        self.down_w = nn.Parameter(down, requires_grad=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T = x.shape[0]
+        if self.ep > 1:
+            return self._forward_ep(x)
         if x.is_cuda and T <= 128:
             return self._forward_dense(x)
         return self._forward_sparse(x)
+
+    def _expert_mlp(self, xs: torch.Tensor, e: int) -> torch.Tensor:
+        gu = F.linear(xs, self.gate_up_w[e])
+        act = torch.empty(xs.shape[0], self.inter, dtype=xs.dtype,
+                          device=xs.device)
+        ops.silu_mul(act, gu)
+        return F.linear(act, self.down_w[e])
+
+    def _forward_ep(self, x: torch.Tensor) -> torch.Tensor:
+        """Expert-parallel MoE (BASELINE config 5: expert all-to-all over
+        xGMI): route each (token, expert) pair to the rank owning that
+        expert, compute there, route back, combine with the gate weights
+        locally. Two all-to-alls per layer; splits are data-dependent so
+        this path is eager (not hipGraph-captured)."""
+        import torch.distributed as dist
+        T, H = x.shape
+        ep = self.ep
+        logits = F.linear(x, self.router_w).float()
+        probs = torch.softmax(logits, dim=-1)
+        topv, topi = probs.topk(self.K, dim=-1)
+        topv = topv / topv.sum(dim=-1, keepdim=True)
+        flat_expert = topi.reshape(-1)                      # [T*K]
+        dest = flat_expert // self.e_local                  # owner rank
+        order = torch.argsort(dest, stable=True)
+        send_rows = x[order // self.K].contiguous()
+        send_eid = (flat_expert[order] % self.e_local).contiguous()
+        in_splits = torch.bincount(dest, minlength=ep)
+        out_splits = torch.empty_like(in_splits)
+        dist.all_to_all_single(out_splits, in_splits.contiguous())
+        isl = in_splits.tolist()
+        osl = out_splits.tolist()
+        nrecv = int(sum(osl))
+        recv_rows = torch.empty(nrecv, H, dtype=x.dtype, device=x.device)
+        recv_eid = torch.empty(nrecv, dtype=send_eid.dtype, device=x.device)
+        dist.all_to_all_single(recv_rows, send_rows, osl, isl)
+        dist.all_to_all_single(recv_eid, send_eid, osl, isl)
+        # group the received tokens by local expert, run each expert once
+        order2 = torch.argsort(recv_eid, stable=True)
+        grouped = recv_rows[order2]
+        counts = torch.bincount(recv_eid, minlength=self.e_local).tolist()
+        processed = torch.empty_like(grouped)
+        start = 0
+        for e in range(self.e_local):
+            n = counts[e]
+            if n == 0:
+                continue
+            processed[start: start + n] = self._expert_mlp(
+                grouped[start: start + n], e)
+            start += n
+        # un-group to the received order, send results home
+        unsorted = torch.empty_like(processed)
+        unsorted[order2] = processed
+        back = torch.empty(T * self.K, H, dtype=x.dtype, device=x.device)
+        dist.all_to_all_single(back, unsorted, isl, osl)
+        # back[i] answers send slot i (= expanded slot order[i])
+        expanded = torch.empty_like(back)
+        expanded[order] = back
+        out = (expanded.view(T, self.K, H).float() *
+               topv.unsqueeze(-1)).sum(dim=1)
+        return out.to(x.dtype)
 
     def _forward_dense(self, x: torch.Tensor) -> torch.Tensor:
         """Decode path: run EVERY expert on the whole (small) batch as one

@@ -80,3 +80,32 @@ def tp_all_reduce(t: torch.Tensor) -> torch.Tensor:
 def barrier() -> None:
     if dist.is_initialized():
         dist.barrier()
+
+
+# ---- expert parallelism (MoE): experts partitioned by id across the
+# world, tokens routed to their experts' owners with all-to-all over
+# xGMI (RCCL) — the alternative to TP-sharding every expert's
+# intermediate dim. EP spans the whole world; TP must be 1.
+_EP_SIZE = 1
+_EP_RANK = 0
+
+
+def init_expert_parallel(ep_size: int) -> None:
+    global _EP_SIZE, _EP_RANK
+    assert _TP_SIZE == 1, "EP spans the world; combine with TP is not wired"
+    _EP_SIZE = ep_size
+    _EP_RANK = dist.get_rank() if (ep_size > 1 and dist.is_initialized()) \
+        else 0
+
+
+def ep_size() -> int:
+    return _EP_SIZE
+
+
+def ep_rank() -> int:
+    return _EP_RANK
+
+
+def ep_all_to_all(out: torch.Tensor, inp: torch.Tensor,
+                  out_splits, in_splits) -> None:
+    dist.all_to_all_single(out, inp, out_splits, in_splits)

@@ -76,3 +76,46 @@ def _tp_engine_worker(rank, world, port, result_dir):
 def test_tp_world2(worker, tmp_path):
     port = 29600 + (os.getpid() + (0 if worker is _tp_linear_worker else 7)) % 500
     mp.spawn(worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+
+
+def _ep_moe_worker(rank, world, port, result_dir):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from kukeon_amd import parallel
+    from kukeon_amd.engine.config import tiny_mixtral
+    from kukeon_amd.models.mixtral import MixtralMoE
+
+    cfg = tiny_mixtral()
+
+    # single-rank reference module (seed S, all experts local)
+    torch.manual_seed(31)
+    ref_moe = MixtralMoE(cfg, "cpu")
+
+    # expert-parallel module from the SAME rng stream -> same model
+    parallel.init_expert_parallel(world)
+    torch.manual_seed(31)
+    ep_moe = MixtralMoE(cfg, "cpu")
+    assert ep_moe.e_local == cfg.num_experts // world
+    torch.testing.assert_close(
+        ep_moe.gate_up_w,
+        ref_moe.gate_up_w[rank * ep_moe.e_local:(rank + 1) * ep_moe.e_local])
+
+    # each rank routes ITS OWN tokens; results must match the single-rank
+    # module on those tokens exactly (same experts, same math)
+    torch.manual_seed(100 + rank)
+    x = torch.randn(7 + rank, cfg.hidden_size, dtype=torch.bfloat16)
+    out_ep = ep_moe._forward_ep(x.clone())
+    out_ref = ref_moe._forward_sparse(x.clone())
+    torch.testing.assert_close(out_ep.float(), out_ref.float(), rtol=3e-2,
+                               atol=3e-2)
+    parallel.init_expert_parallel(1)
+    dist.destroy_process_group()
+
+
+def test_expert_parallel_moe_all_to_all(tmp_path):
+    """EP MoE (BASELINE config 5 wording: expert all-to-all) matches the
+    single-rank sparse path on every rank's tokens — gloo world 2."""
+    port = 29600 + (os.getpid() + 13) % 500
+    mp.spawn(_ep_moe_worker, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
