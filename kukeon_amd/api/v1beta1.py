@@ -1,8 +1,10 @@
 """v1beta1 API model: external YAML/JSON document types for every kind.
 
 Mirrors the reference's manifest contract (field names, kinds, state strings
-— /root/reference/pkg/api/model/v1beta1/{cell,container,space}.go and
-consts.go are the parity spec; studied, not copied) with two first-class
+— pkg/api/model/v1beta1: kinds consts.go:24-84, CellSpec cell.go:45-118,
+ContainerSpec container.go:34-181, restart policy container.go:126-141,
+state enums cell.go:242-271, SpaceSpec egress space.go:38-79; studied, not
+copied) with two first-class
 additions from the agent-native proposal: the Session and Interactive kinds
 (docs/site/proposals/agent-native-orchestration.md:179-211), plus the GPU
 fields the MI355X runtime needs (ContainerSpec.gpus → amdgpu device pinning

@@ -1,7 +1,12 @@
 """The controller: verb orchestration, desired-state apply, reconcile.
 
 The capability map of the reference's internal/controller + runner (SURVEY.md
-§2.3/2.4) re-built on the process-cell runtime: provisioning (cgroups,
+§2.3/2.4; key parity anchors: Bootstrap internal/controller/controller.go:168-247,
+apply ordering apply.go:97-101 + per-resource results apply.go:40-48,
+idempotent start runner/start.go:587-620 with markCellFailed rollback
+start.go:414, restart backoff floor/retry cap v1beta1/container.go:126-141,
+reconcile tick runner/refresh.go:655-1010, OutOfSync re-derivation
+reconcile_outofsync.go:65-120) re-built on the process-cell runtime: provisioning (cgroups,
 per-space subnets, egress policy), cell lifecycle with idempotent start and
 markCellFailed rollback, restart policies with backoff floor + retry cap,
 AutoDelete, Session lifetime enforcement (wallClock/idleTimeout + onEnd

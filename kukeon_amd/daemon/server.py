@@ -4,7 +4,9 @@ Wire protocol: newline-delimited JSON frames
     {"id": N, "method": "CreateCell", "params": {...}}
  -> {"id": N, "result": ...} | {"id": N, "error": {"code", "message"}}
 
-Background loops (reference: internal/daemon/server.go): a cell-reconcile
+Background loops (reference: internal/daemon/server.go:224-342 — eager
+first reconcile on startup, then periodic passes; RPC service surface
+mirrors rpcservice.go:38-475; instance pinning instance.go:59-141): a cell-reconcile
 pass every `reconcile_interval` seconds (eager first pass on startup so
 state converges after a host restart), a session-lifetime pass, and a
 space-network re-assert pass; each pass is exception-guarded so one bad

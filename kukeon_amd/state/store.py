@@ -1,7 +1,10 @@
 """On-disk metadata store: the durable state tree.
 
 Layout (byte-layout contract from the reference — SURVEY.md §5 checkpoint
-row; studied from /root/reference/internal/util/fs + internal/metadata):
+row; studied from /root/reference/internal/util/fs/metadata.go:39-527 and
+internal/metadata/lock.go:41-193 — sidecar flock, CAS writes, K8s-style
+Generation/ObservedGeneration from v1beta1/cell.go:39-42,216-219; socket
+symlink dir for the SUN_PATH 107-byte limit from consts.go:105-124):
 
     <run_path>/data/<realm>/metadata.json
     <run_path>/data/<realm>/<space>/metadata.json

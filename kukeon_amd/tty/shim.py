@@ -1,7 +1,10 @@
 """Container shim: the kuketty/kukepause analog for process cells.
 
 One shim per container (the runtime spawns `python -m kukeon_amd.tty.shim
---dir <container_dir>` in its own session/process group). It:
+--dir <container_dir>` in its own session/process group; reference parity:
+cmd/kuketty/main.go:57-117 PTY server + metadata contract, stages
+stages.go:62-88, kukepause PID-1 semantics cmd/kukepause/main.go:17-60 via
+--pause mode). It:
 
 * reads the spawn spec (`spawn.json`) the runner rendered — argv, env, cwd,
   attachable flag, capture path — and records `runtime.json` (shim pid +
