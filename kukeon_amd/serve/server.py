@@ -11,7 +11,10 @@ interleave across clients — the engine continuously batches them):
       "tokens": [...], "max_new_tokens": 128, "temperature": 0.7,
       "top_k": 50, "top_p": 0.9}}
   -> {"id": N, "result": {"tokens": [...], "context_len": M}}
-  other methods: ping, stats, release (free a session's KV).
+  other methods: ping, stats, release (free a session's KV). Passing
+  "stream": true on generate delivers incremental
+  {"id": N, "delta": [tok, ...]} frames (one per decode micro-batch)
+  ahead of the final result frame — time-to-first-token for agents.
 """
 from __future__ import annotations
 
@@ -39,6 +42,7 @@ class _Pending:
     req_id: int
     reply: "queue.Queue"
     tokens: List[int]
+    stream: bool = False
 
 
 class ModelhubServer:
@@ -85,13 +89,22 @@ class ModelhubServer:
                     if p is None:
                         continue
                     p.tokens.extend(o.new_tokens)
+                    if p.stream and o.new_tokens and not o.finished:
+                        p.reply.put({"delta": list(o.new_tokens)})
                     if o.finished:
                         self.metrics["turns_completed"] += 1
                         del self._pending[o.req_id]
+                        if p.stream:
+                            p.reply.put({"delta": list(o.new_tokens),
+                                         "_final_stub": True})
                         p.reply.put({"tokens": p.tokens})
 
     # ---- request handling --------------------------------------------
-    def handle(self, method: str, params: dict) -> dict:
+    def handle(self, method: str, params: dict, emit=None) -> dict:
+        """Dispatch one request. `emit`, when provided, receives
+        incremental frames for streaming methods (written by the
+        connection handler as `{"id": N, "delta": [tokens...]}` lines
+        ahead of the final result frame)."""
         if method == "ping":
             return {"ok": True, "pid": os.getpid()}
         if method == "stats":
@@ -116,12 +129,13 @@ class ModelhubServer:
             self._submit.put(_do_release)
             return done.get(timeout=60)
         if method == "generate":
-            return self._generate(params)
+            return self._generate(params, emit)
         raise ValueError(f"unknown method {method}")
 
-    def _generate(self, params: dict) -> dict:
+    def _generate(self, params: dict, emit=None) -> dict:
         name = params["session"]
         tokens = list(params["tokens"])
+        stream = bool(params.get("stream", False)) and emit is not None
         sp = SamplingParams(
             temperature=float(params.get("temperature", 0.7)),
             top_k=int(params.get("top_k", 50)),
@@ -140,11 +154,17 @@ class ModelhubServer:
                 reply.put({"error": str(e)})
                 return
             with self._lock:
-                self._pending[rid] = _Pending(rid, reply, [])
+                self._pending[rid] = _Pending(rid, reply, [], stream=stream)
         self._submit.put(_do_submit)
-        out = reply.get(timeout=600)
-        if "error" in out:
-            raise ValueError(out["error"])
+        while True:
+            out = reply.get(timeout=600)
+            if "error" in out:
+                raise ValueError(out["error"])
+            if "delta" in out:
+                if not out.get("_final_stub"):
+                    emit({"delta": out["delta"]})
+                continue
+            break
         kv = self.sessions.get(name)
         out["context_len"] = kv.num_tokens if kv else 0
         return out
@@ -165,9 +185,16 @@ class ModelhubServer:
                         continue
                     try:
                         req = json.loads(line)
+                        rid = req.get("id")
+
+                        def emit(frame, _rid=rid):
+                            self.wfile.write((json.dumps(
+                                {"id": _rid, **frame}) + "\n").encode())
+                            self.wfile.flush()
                         result = hub.handle(req.get("method", ""),
-                                            req.get("params") or {})
-                        resp = {"id": req.get("id"), "result": result}
+                                            req.get("params") or {},
+                                            emit=emit)
+                        resp = {"id": rid, "result": result}
                     except Exception as e:  # noqa: BLE001
                         resp = {"id": req.get("id") if isinstance(req, dict)
                                 else None, "error": str(e)}
@@ -230,6 +257,28 @@ class ModelhubClient:
 
     def generate(self, session: str, tokens: List[int], **kw):
         return self.call("generate", session=session, tokens=tokens, **kw)
+
+    def generate_stream(self, session: str, tokens: List[int], **kw):
+        """Yield token deltas as the engine produces them; the final
+        yield is the full result dict (with "tokens" and "context_len")."""
+        with self._lock:
+            self._id += 1
+            self.sock.sendall((json.dumps(
+                {"id": self._id, "method": "generate",
+                 "params": {"session": session, "tokens": tokens,
+                            "stream": True, **kw}}) + "\n").encode())
+            while True:
+                line = self._rf.readline()
+                if not line:
+                    raise ConnectionError("modelhub closed the connection")
+                resp = json.loads(line)
+                if "error" in resp:
+                    raise RuntimeError(resp["error"])
+                if "delta" in resp:
+                    yield {"delta": resp["delta"]}
+                    continue
+                yield resp["result"]
+                return
 
     def close(self):
         self.sock.close()

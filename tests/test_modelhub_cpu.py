@@ -85,3 +85,41 @@ def test_modelhub_as_system_cell(tmp_path):
     assert len(r["tokens"]) == 3
     client.close()
     ctl.kill_cell("kuke-system", "kukeon", "kukeon", "modelhub")
+
+
+def test_streaming_generate(tmp_path):
+    """stream=True delivers incremental token deltas ahead of the final
+    result, and the concatenated deltas equal the final token list."""
+    import torch
+    from kukeon_amd.engine.config import EngineConfig, tiny_llama
+    from kukeon_amd.models.llama import LlamaModel
+    from kukeon_amd.serve.server import ModelhubClient, ModelhubServer
+
+    torch.manual_seed(0)
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=256, max_sessions=4, num_kv_blocks=128,
+                        use_graphs=False, decode_microbatch=2)
+    sock = f"/tmp/mhs-{uuid.uuid4().hex[:8]}.sock"
+    hub = ModelhubServer(LlamaModel(cfg, device="cpu"), cfg, ecfg, sock,
+                         device="cpu")
+    hub.start()
+    try:
+        c = ModelhubClient(sock, timeout=120)
+        deltas = []
+        final = None
+        for frame in c.generate_stream("s1", [5, 6, 7], max_new_tokens=9,
+                                       temperature=0.0):
+            if "delta" in frame:
+                deltas.extend(frame["delta"])
+                assert final is None
+            else:
+                final = frame
+        assert final is not None and len(final["tokens"]) == 9
+        assert deltas == final["tokens"][: len(deltas)]
+        assert len(deltas) >= 2  # micro-batch 2 => incremental frames
+        # non-streaming call still works on the same connection
+        r = c.generate("s1", [9], max_new_tokens=3, temperature=0.0)
+        assert len(r["tokens"]) == 3
+        c.close()
+    finally:
+        hub.stop()
