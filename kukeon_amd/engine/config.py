@@ -88,7 +88,7 @@ class EngineConfig:
     use_graphs: bool = True
     graph_buckets: tuple = (8, 16, 32, 64, 96, 128, 192, 256)
     decode_splits: int = 1          # split-KV factor for small-batch decode
-    decode_microbatch: int = 16     # decode steps per host sync (self-
+    decode_microbatch: int = 32     # decode steps per host sync (self-
                                     # advancing graph replay train)
     kv_dtype: str = "bf16"          # "bf16" | "fp8" (OCP e4m3 cache: half
                                     # the KV bandwidth, 2x the capacity)
