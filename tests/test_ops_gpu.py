@@ -597,3 +597,55 @@ def test_linear_add_rmsnorm_fused(M):
                                rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(out.cpu().float(), h_ref.float(), rtol=2e-2,
                                atol=3e-2)
+
+
+def test_modelhub_server_gpu_streaming():
+    """The serving stack end to end on the GPU: engine-loop thread with
+    hipGraphs, unix-socket protocol, concurrent sessions, streaming."""
+    import threading
+    import uuid
+
+    from kukeon_amd.engine.config import EngineConfig, tiny_llama
+    from kukeon_amd.models.llama import LlamaModel
+    from kukeon_amd.serve.server import ModelhubClient, ModelhubServer
+
+    torch.manual_seed(0)
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=256, max_sessions=8, num_kv_blocks=256,
+                        use_graphs=True, decode_microbatch=4,
+                        graph_buckets=(2, 4, 8))
+    sock = f"/tmp/mhg-{uuid.uuid4().hex[:8]}.sock"
+    hub = ModelhubServer(LlamaModel(cfg, device=DEV), cfg, ecfg, sock,
+                         device=DEV)
+    hub.start()
+    try:
+        results = {}
+
+        def worker(name):
+            c = ModelhubClient(sock, timeout=120)
+            deltas, final = [], None
+            for frame in c.generate_stream(name, [3, 1, 4, 1, 5],
+                                           max_new_tokens=8,
+                                           temperature=0.0):
+                if "delta" in frame:
+                    deltas.extend(frame["delta"])
+                else:
+                    final = frame
+            r2 = c.generate(name, [9, 2], max_new_tokens=4, temperature=0.7)
+            results[name] = (deltas, final, r2)
+            c.close()
+
+        ts = [threading.Thread(target=worker, args=(f"s{i}",))
+              for i in range(4)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join(timeout=120)
+        assert len(results) == 4
+        for deltas, final, r2 in results.values():
+            assert final is not None and len(final["tokens"]) == 8
+            assert deltas == final["tokens"][: len(deltas)]
+            assert len(r2["tokens"]) == 4
+            assert r2["context_len"] > 8  # persistent multi-turn KV
+    finally:
+        hub.stop()
