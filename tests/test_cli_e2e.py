@@ -1,0 +1,96 @@
+"""Black-box CLI e2e: drive the real `bin/kuke` binary as a subprocess
+against a per-test run path (the reference's e2e style — SURVEY.md §4
+tier 3: CLI output, on-disk metadata and process state verified together).
+Uses --local (in-process verbs) so no daemon needs to be running."""
+import json
+import os
+import subprocess
+import time
+from pathlib import Path
+
+import yaml
+
+REPO = Path(__file__).resolve().parent.parent
+KUKE = str(REPO / "bin" / "kuke")
+
+CELL_YAML = """
+apiVersion: v1beta1
+kind: Cell
+metadata: {name: cliy}
+spec:
+  realmId: default
+  spaceId: default
+  stackId: default
+  containers:
+    - id: main
+      image: busybox
+      command: sleep
+      args: ["30"]
+"""
+
+
+def kuke(run, *args, check=True, env_extra=None):
+    env = dict(os.environ)
+    env["KUKE_CONFIG"] = str(Path(run) / "nonexistent-kuke.yaml")
+    if env_extra:
+        env.update(env_extra)
+    r = subprocess.run([KUKE, "--run-path", run, "--local", *args],
+                       capture_output=True, text=True, timeout=60, env=env)
+    if check:
+        assert r.returncode == 0, (args, r.stdout, r.stderr)
+    return r
+
+
+def test_cli_full_lifecycle(tmp_path):
+    run = str(tmp_path / "run")
+    kuke(run, "init")
+    assert (Path(run) / "data" / "default" / "metadata.json").exists()
+
+    spec = tmp_path / "cell.yaml"
+    spec.write_text(CELL_YAML)
+    out = kuke(run, "apply", "-f", str(spec)).stdout
+    assert "created" in out
+
+    out = kuke(run, "get", "cells").stdout
+    assert "cliy" in out
+
+    kuke(run, "start", "cliy")
+    doc = yaml.safe_load(kuke(run, "get", "cell", "cliy", "-o",
+                              "yaml").stdout)
+    assert doc["status"]["state"] == "Ready"
+    pid = doc["status"]["containers"][0]["pid"]
+    assert pid > 0
+
+    top = kuke(run, "top", "cliy").stdout
+    assert "TOTAL" in top and "main" in top
+
+    out = kuke(run, "status", check=False)
+    assert "state tree" in out.stdout
+
+    kuke(run, "stop", "cliy")
+    deadline = time.monotonic() + 5
+    while time.monotonic() < deadline:
+        try:
+            os.kill(pid, 0)
+            time.sleep(0.05)
+        except ProcessLookupError:
+            break
+    doc = yaml.safe_load(kuke(run, "get", "cell", "cliy", "-o",
+                              "yaml").stdout)
+    assert doc["status"]["state"] == "Stopped"
+
+    kuke(run, "delete", "cell", "cliy")
+    r = kuke(run, "get", "cell", "cliy", check=False)
+    assert r.returncode != 0
+
+
+def test_cli_apply_validation_error(tmp_path):
+    run = str(tmp_path / "run")
+    kuke(run, "init")
+    bad = tmp_path / "bad.yaml"
+    bad.write_text("apiVersion: v1beta1\nkind: Cell\n"
+                   "metadata: {name: BAD NAME}\n"
+                   "spec: {realmId: default, spaceId: default, "
+                   "stackId: default}\n")
+    r = kuke(run, "apply", "-f", str(bad), check=False)
+    assert r.returncode != 0 or "failed" in r.stdout
