@@ -94,3 +94,41 @@ def test_cli_apply_validation_error(tmp_path):
                    "stackId: default}\n")
     r = kuke(run, "apply", "-f", str(bad), check=False)
     assert r.returncode != 0 or "failed" in r.stdout
+
+
+def test_cli_against_live_daemon(tmp_path):
+    """kuke over the unix socket to a live kukeond (not --local)."""
+    import uuid
+
+    from kukeon_amd.controller.core import Controller
+    from kukeon_amd.daemon.server import Server
+
+    run = str(tmp_path / "run")
+    sock = f"/tmp/kukecli-{uuid.uuid4().hex[:8]}.sock"
+    ctl = Controller(run, gpu_devices=[])
+    ctl.bootstrap()
+    srv = Server(ctl, sock, reconcile_interval=0)
+    srv.start()
+    try:
+        env = dict(os.environ)
+        env["KUKE_CONFIG"] = str(tmp_path / "none.yaml")
+
+        def kd(*args, check=True):
+            r = subprocess.run([KUKE, "--run-path", run, "--socket", sock,
+                                *args], capture_output=True, text=True,
+                               timeout=60, env=env)
+            if check:
+                assert r.returncode == 0, (args, r.stdout, r.stderr)
+            return r
+
+        spec = tmp_path / "cell.yaml"
+        spec.write_text(CELL_YAML)
+        assert "created" in kd("apply", "-f", str(spec)).stdout
+        kd("start", "cliy")
+        assert "Ready" in kd("get", "cell", "cliy", "-o", "yaml").stdout
+        st = kd("status")
+        assert "daemon rpc" in st.stdout and "[  ok]" in st.stdout
+        kd("kill", "cliy")
+        kd("delete", "cell", "cliy")
+    finally:
+        srv.stop()
