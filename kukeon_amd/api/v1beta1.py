@@ -140,6 +140,10 @@ def _from_dict(cls, data):
         return None
     if not dataclasses.is_dataclass(cls):
         return data
+    if not isinstance(data, dict):
+        raise ValueError(
+            f"{cls.__name__}: expected a mapping, got "
+            f"{type(data).__name__} ({data!r})")
     kwargs = {}
     for f in dataclasses.fields(cls):
         key = _json_key(f)

@@ -318,3 +318,13 @@ def test_session_modelhub_env_injection(ctl):
         ctl.get_cell("default", "default", "default", "c2"),
         cell.spec.containers[0], [])
     assert not any(e.startswith("KUKEON_MODELHUB") for e in env2)
+
+
+def test_from_dict_scalar_for_mapping_is_validation_error():
+    """A scalar where the schema wants a struct (e.g. `tty: true`) must
+    raise a clean error, not an internal TypeError."""
+    with pytest.raises(ValueError, match="expected a mapping"):
+        api.CellDoc.from_dict(
+            {"apiVersion": "v1beta1", "kind": "Cell",
+             "metadata": {"name": "x"},
+             "spec": {"containers": [{"id": "m", "tty": True}]}})
