@@ -328,3 +328,27 @@ def test_from_dict_scalar_for_mapping_is_validation_error():
             {"apiVersion": "v1beta1", "kind": "Cell",
              "metadata": {"name": "x"},
              "spec": {"containers": [{"id": "m", "tty": True}]}})
+
+
+def test_all_examples_parse():
+    """Every manifest in examples/ must stay valid against the API model
+    (kuketeam.yaml is a kuketeams.io doc, parsed by the teams parser)."""
+    from pathlib import Path
+
+    import yaml as _yaml
+
+    from kukeon_amd.controller.parser import parse_documents
+    from kukeon_amd.teams import parse_team_doc
+
+    exdir = Path(__file__).resolve().parent.parent / "examples"
+    seen = 0
+    for f in sorted(exdir.glob("*.yaml")):
+        raw = _yaml.safe_load(f.read_text())
+        if raw.get("apiVersion", "").startswith("kuketeams.io"):
+            team = parse_team_doc(raw)
+            assert team.roles
+        else:
+            docs = parse_documents(f.read_text())
+            assert docs and docs[0].metadata.name
+        seen += 1
+    assert seen >= 6
