@@ -18,7 +18,6 @@ import contextlib
 import json
 import logging
 import os
-import socket
 import socketserver
 import threading
 import time

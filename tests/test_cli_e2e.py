@@ -2,7 +2,6 @@
 against a per-test run path (the reference's e2e style — SURVEY.md §4
 tier 3: CLI output, on-disk metadata and process state verified together).
 Uses --local (in-process verbs) so no daemon needs to be running."""
-import json
 import os
 import subprocess
 import time

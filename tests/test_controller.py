@@ -6,7 +6,7 @@ from kukeon_amd.api import errors
 from kukeon_amd.api import v1beta1 as api
 from kukeon_amd.controller.core import Controller
 from kukeon_amd.controller import naming
-from kukeon_amd.runtime.process import ROOT_CONTAINER, FakeRuntime
+from kukeon_amd.runtime.process import FakeRuntime
 
 
 class Clock:

@@ -4,7 +4,6 @@ on a CPU-only host) as a test. Reference harness pattern: per-test daemon on
 a short /tmp socket with --reconcile-interval 0 so tests don't race the
 loop."""
 import json
-import os
 import socket
 import time
 import uuid

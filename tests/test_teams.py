@@ -1,9 +1,7 @@
 """Team distribution pipeline tests: local agents source -> render -> apply
 with per-team prune, seeds, layered secrets, image registration."""
-from pathlib import Path
 
 import pytest
-import yaml
 
 from kukeon_amd.api import errors
 from kukeon_amd.api import v1beta1 as api
