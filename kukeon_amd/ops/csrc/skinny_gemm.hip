@@ -1,20 +1,22 @@
 // Skinny decode GEMM for gfx950: out[M,N] = x[M,K] @ W[N,K]^T, M <= 64.
 //
 // Decode is weight-bandwidth-bound (the whole W streams through HBM once
-// per step); hipBLASLt's skinny tiles measured 36-54% of the roofline on
-// the o/down/qkv shapes (profiles/). Design learned from the pattern probe
-// (scripts/pattern_probe.hip: the 16-row-strided nt read reaches 5.6 TB/s,
-// so the access pattern is not the limit — duty cycle is):
-//   * x (tiny, L2-resident) is staged into LDS ONCE per kernel for the
-//     whole 512-deep k-slice (XOR-swizzled, conflict-free ds_read_b128
+// per step). Design (ledger: profiles/r01_progress.md):
+//   * x (tiny, L2-resident) is staged into per-slice LDS images by async
+//     LDS-DMA (glds), double-buffered so slice s+1's stage flies under
+//     slice s's W stream (XOR-swizzled, conflict-free ds_read_b128
 //     B-fragments that consume no vmcnt slots),
-//   * W streams with nt 16B/lane loads in 8-deep batches, software-
-//     pipelined so batch b+1 is in flight while batch b feeds the MFMAs,
-//     and batch 0 is issued BEFORE the staging barrier (it has no LDS
-//     dependence) so even the prologue overlaps,
-//   * split-K (slice = 512) gives every shape >= 512 workgroups; partials
-//     combine through f32 atomicAdd + a cast pass.
-// Grid: (N/64, K/512); block = 4 waves, wave w owns N rows [64b+16w, +16).
+//   * W streams with nt 16B/lane loads in 8-deep probe-shaped batches
+//     (loads + consume in one iteration => hipcc emits counted vmcnt(N)),
+//     batch 0 issued BEFORE the drain+barrier so the prologue overlaps,
+//   * split-K over 256-deep slices, sized to ~256 blocks (chip fill);
+//     partials combine through per-slice f32 slabs + a reduce pass
+//     (atomicAdd measured ~35us of L2 RMW serialization), or fused
+//     straight into the residual-add+RMSNorm epilogue on the decode
+//     o-projection path.
+// Grid: (N/64, splitk); block = 4 waves, wave w owns N rows [64b+16w, +16)
+// of every slice the block walks. Dispatched per shape where it beats
+// hipBLASLt cold-LLC (ops/__init__.py _skinny_wins).
 #include "common.h"
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
