@@ -12,7 +12,10 @@ inference plane at all):
 * requests hold *sticky rows* in the static decode buffers, so the per-step
   host work is O(active rows) scalar updates + one slab H2D copy — no
   per-step tensor rebuilds,
-* split-KV decode attention keeps >=512 workgroups in flight at small batch.
+* split-KV decode attention (matrix-core kernel, 4 autonomous wave-slots
+  per workgroup) is split to ~4096 wave-units so the chip stays full even
+  at small batch; idle sessions' KV evicts under pressure (history
+  recompute) and running rows preempt-by-recompute on exhaustion.
 """
 from __future__ import annotations
 

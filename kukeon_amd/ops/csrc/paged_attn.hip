@@ -1,6 +1,13 @@
 // Paged decode attention for gfx950 (MI355X).
 //
-// One workgroup (4 waves, 256 threads) per (sequence, kv-head, kv-split).
+// TWO kernels share this file: paged_attn_mfma16_kernel (the default —
+// matrix-core scores/PV, wave-autonomous split slots, 86% of the HBM
+// roofline; see its header further down) and the v_dot2 kernel described
+// below (the fallback for GQA groups > 16 or split counts not divisible
+// by 4, and the reference point the MFMA design was measured against).
+//
+// v_dot2 kernel: one workgroup (4 waves, 256 threads) per
+// (sequence, kv-head, kv-split).
 // The whole GQA group (G q-heads sharing one kv-head) is computed by the
 // workgroup so each K/V block is read from HBM exactly once and shared via
 // LDS — KV bandwidth is the decode cost, never multiply it by G.
