@@ -52,9 +52,14 @@ class AgentSession:
 
     def start_turn(self) -> int:
         """Submit one turn; returns the request id."""
-        fresh = self.kv.num_tokens == 0
-        n = self.first_prompt if fresh else self.followup_prompt
-        if self.kv.num_tokens + n + self.decode_len > self.ctx_cap:
+        # effective context: cached tokens, or — if the engine evicted this
+        # idle session's KV under pressure — the history add_request would
+        # transparently re-prefill (plus the pending sampled token)
+        ctx = self.kv.num_tokens or (
+            len(self.kv.history) + (1 if self.kv.pending_token is not None
+                                    else 0))
+        n = self.first_prompt if ctx == 0 else self.followup_prompt
+        if ctx + n + self.decode_len > self.ctx_cap:
             # context compaction: drop history, re-seed with a fresh prompt
             self.engine.free_sequence(self.kv)
             n = self.first_prompt

@@ -269,3 +269,26 @@ def test_idle_session_kv_eviction():
     big = SequenceKV(ecfg.block_size)
     with pytest.raises((MemoryError, ValueError)):
         run(big, list(range(100)) * 2, 50)
+
+
+def test_session_driver_survives_idle_eviction():
+    """An idle-evicted session's next turn must compact, not overflow
+    max_model_len with the transparent history re-prefill."""
+    from kukeon_amd.serve.sessions import TurnDriver
+
+    torch.manual_seed(0)
+    cfg = tiny_llama()
+    # pool sized so two sessions' contexts cannot both stay resident
+    ecfg = EngineConfig(max_model_len=96, max_sessions=2, num_kv_blocks=7,
+                        use_graphs=False)
+    model = LlamaModel(cfg, device="cpu")
+    engine = LLMEngine(model, cfg, ecfg, device="cpu")
+    driver = TurnDriver(engine, 2, cfg.vocab_size, first_prompt=20,
+                        followup_prompt=10, decode_len=8, ctx_cap=80,
+                        sampling=SamplingParams(temperature=0.0,
+                                                max_new_tokens=8))
+    total = 0
+    for _ in range(6):  # enough rounds to force eviction + compaction
+        total += driver.run_round()
+    assert total == 12
+    assert len(driver.turn_latencies) == 12
