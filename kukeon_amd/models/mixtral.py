@@ -79,7 +79,14 @@ class MixtralMoE(nn.Module):
         xGMI): route each (token, expert) pair to the rank owning that
         expert, compute there, route back, combine with the gate weights
         locally. Two all-to-alls per layer; splits are data-dependent so
-        this path is eager (not hipGraph-captured)."""
+        this path is eager (not hipGraph-captured).
+
+        NOTE: the all-to-alls make every MoE forward a COLLECTIVE — all
+        EP ranks must step in lockstep (the bench's TurnDriver rounds
+        satisfy this: equal session counts and fixed prompt/decode
+        lengths give identical step sequences). Free-running engines per
+        rank would need a synchronized scheduler; that is a round-2 item
+        if EP serving (rather than TP) is chosen for production MoE."""
         import torch.distributed as dist
         T, H = x.shape
         ep = self.ep
