@@ -82,6 +82,10 @@ def sample(tokens, logits, temps, top_k, top_p, seed, workspace) -> None:
 
 
 _SKINNY_WS = {}
+# replaced (outgrown) workspaces are pinned here forever: a hipGraph
+# captured earlier may still hold the old tensor's device pointer, and
+# letting it free would put a use-after-free inside every later replay
+_SKINNY_WS_RETIRED = []
 # Per-shape dispatch, from the measured sweep (profiles/r01_progress.md,
 # scripts/sweep_splitk.py on MI355X): the glds-staged skinny kernel beats
 # hipBLASLt on square o-projection shapes (N==K: 16.6us vs 19.3 at M=64,
@@ -114,6 +118,8 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
         need = max(1, splitk) * 64 * N
         if ws is None or ws.numel() < need:
             # grown only outside graph capture (engine warmup runs eager)
+            if ws is not None:
+                _SKINNY_WS_RETIRED.append(ws)
             ws = torch.empty(need, dtype=torch.float32, device=x.device)
             _SKINNY_WS[key] = ws
         _native().skinny_gemm(out, x, w, ws)
@@ -145,6 +151,8 @@ def linear_add_rmsnorm(x: torch.Tensor, w: torch.Tensor,
         splitk = min(nslices, -(-256 // ntiles))
         need = max(1, splitk) * 64 * N
         if ws is None or ws.numel() < need:
+            if ws is not None:
+                _SKINNY_WS_RETIRED.append(ws)
             ws = torch.empty(need, dtype=torch.float32, device=x.device)
             _SKINNY_WS[key] = ws
         normed = torch.empty(rows, N, dtype=x.dtype, device=x.device)
