@@ -905,7 +905,8 @@ class Controller:
         crashed_terminal = 0
         restarting = 0
         statuses = []
-        for c in doc.spec.containers:
+        prior = {i: cs for i, cs in enumerate(doc.status.containers)}
+        for i, c in enumerate(doc.spec.containers):
             cdir = cell_dir / (c.id or "main")
             probe = self.runtime.probe(cdir)
             key = str(cdir)
@@ -915,6 +916,12 @@ class Controller:
                                      started_at=probe.started_at,
                                      finished_at=probe.finished_at,
                                      restart_count=count)
+            if i in prior:
+                # carry start-time facts the probe cannot re-derive:
+                # GPU pinning (drives the restart env) and repo statuses
+                st.gpu_ids = list(prior[i].gpu_ids)
+                st.repos = prior[i].repos
+                st.last_restart_at = prior[i].last_restart_at
             if probe.running:
                 running += 1
                 if count and probe.started_at and \
@@ -935,6 +942,7 @@ class Controller:
                         self._restart_state[key] = (now, count + 1)
                         st.restart_count = count + 1
                         st.state = api.STATE_READY
+                        st.last_restart_at = now_iso(now)
                         restarting += 1
                 elif should:
                     restarting += 1  # waiting out the backoff floor

@@ -197,6 +197,7 @@ class FakeRuntime(Runtime):
 
     def __init__(self):
         self.started: List[str] = []
+        self.started_envs: List[List[str]] = []
         self.stopped: List[str] = []
         self.killed: List[str] = []
         self.states: Dict[str, ContainerProbe] = {}
@@ -216,6 +217,7 @@ class FakeRuntime(Runtime):
             raise self.fail_on[key]
         self._pid += 1
         self.started.append(key)
+        self.started_envs.append(list(env))
         self.states[key] = ContainerProbe(exists=True, running=True,
                                           pid=self._pid, started_at=_now())
         cdir.mkdir(parents=True, exist_ok=True)

@@ -352,3 +352,40 @@ def test_all_examples_parse():
             assert docs and docs[0].metadata.name
         seen += 1
     assert seen >= 6
+
+
+def test_reconcile_preserves_gpu_pinning(tmp_path):
+    """Status rebuild during reconcile must carry gpu_ids forward — a
+    restart after a crash re-derives its ROCR_VISIBLE_DEVICES env from
+    them."""
+    rt = FakeRuntime()
+    clock = Clock()
+    ctl = Controller(str(tmp_path / "run"), runtime=rt,
+                     gpu_devices=[0, 1, 2, 3], now_fn=clock)
+    ctl.bootstrap()
+    doc = api.CellDoc(
+        metadata=api.Metadata(name="gpuc"),
+        spec=api.CellSpec(realm_id="default", space_id="default",
+                          stack_id="default",
+                          containers=[api.ContainerSpec(
+                              id="main", image="x", command="work",
+                              gpus=2,
+                              restart_policy=api.RESTART_ALWAYS)]))
+    ctl.create_cell(doc)
+    started = ctl.start_cell("default", "default", "default", "gpuc")
+    pinned = started.status.containers[0].gpu_ids
+    assert len(pinned) == 2
+    # healthy reconcile pass: pinning survives the status rebuild
+    d2 = ctl.reconcile_cell("default", "default", "default", "gpuc")
+    assert d2.status.containers[0].gpu_ids == pinned
+    # crash; backoff elapses; the restart env re-pins the same GPUs
+    rt.mark_exited(ctl.store.cell_dir("default", "default", "default",
+                                      "gpuc") / "main", 3)
+    clock.t += 31.0
+    d3 = ctl.reconcile_cell("default", "default", "default", "gpuc")
+    assert d3.status.containers[0].gpu_ids == pinned
+    envs = [e for e in rt.started_envs
+            if any(x.startswith("ROCR_VISIBLE_DEVICES=") for x in e)]
+    assert envs, "restart must carry the GPU env"
+    last = dict(x.split("=", 1) for x in envs[-1])
+    assert last["ROCR_VISIBLE_DEVICES"] == ",".join(str(g) for g in pinned)
