@@ -932,10 +932,7 @@ class Controller:
                 rc = probe.exit_code
                 should = self._should_restart(c, rc, count)
                 if should and now - last >= self._backoff(c):
-                    env_gpus = [s.gpu_ids for s in doc.status.containers
-                                if s.gpu_ids]
-                    env = self._container_env(
-                        doc, c, env_gpus[0] if env_gpus else [])
+                    env = self._container_env(doc, c, list(st.gpu_ids))
                     with contextlib.suppress(Exception):
                         self.runtime.start_container(
                             cdir, c, env, self._cell_cgroup(doc))
