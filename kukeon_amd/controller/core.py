@@ -414,6 +414,9 @@ class Controller:
                     self.runtime.kill(cdir)
             with contextlib.suppress(Exception):
                 self.runtime.kill(cell_dir / ROOT_CONTAINER)
+            with contextlib.suppress(Exception):
+                # a failed start must not hold GPU reservations
+                self.gpus.release(self._gpu_owner(doc))
             doc.status.state = api.STATE_FAILED
             doc.status.message = str(e)
             with contextlib.suppress(Exception):

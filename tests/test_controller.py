@@ -389,3 +389,24 @@ def test_reconcile_preserves_gpu_pinning(tmp_path):
     assert envs, "restart must carry the GPU env"
     last = dict(x.split("=", 1) for x in envs[-1])
     assert last["ROCR_VISIBLE_DEVICES"] == ",".join(str(g) for g in pinned)
+
+
+def test_failed_start_releases_gpu_reservation(tmp_path):
+    rt = FakeRuntime()
+    ctl = Controller(str(tmp_path / "run"), runtime=rt,
+                     gpu_devices=[0, 1], now_fn=Clock())
+    ctl.bootstrap()
+    doc = api.CellDoc(
+        metadata=api.Metadata(name="gfail"),
+        spec=api.CellSpec(realm_id="default", space_id="default",
+                          stack_id="default",
+                          containers=[api.ContainerSpec(
+                              id="main", image="x", command="work",
+                              gpus=2)]))
+    ctl.create_cell(doc)
+    cdir = ctl.store.cell_dir("default", "default", "default",
+                              "gfail") / "main"
+    rt.fail_on[str(cdir)] = RuntimeError("boom")
+    started = ctl.start_cell("default", "default", "default", "gfail")
+    assert started.status.state == api.STATE_FAILED
+    assert ctl.gpus.free == [0, 1], "failed start must not hold GPUs"
