@@ -407,6 +407,8 @@ def test_failed_start_releases_gpu_reservation(tmp_path):
     cdir = ctl.store.cell_dir("default", "default", "default",
                               "gfail") / "main"
     rt.fail_on[str(cdir)] = RuntimeError("boom")
-    started = ctl.start_cell("default", "default", "default", "gfail")
-    assert started.status.state == api.STATE_FAILED
+    with pytest.raises(RuntimeError):
+        ctl.start_cell("default", "default", "default", "gfail")
+    doc2 = ctl.get_cell("default", "default", "default", "gfail")
+    assert doc2.status.state == api.STATE_FAILED
     assert ctl.gpus.free == [0, 1], "failed start must not hold GPUs"
