@@ -122,12 +122,27 @@ class ModelhubServer:
             done: "queue.Queue" = queue.Queue()
 
             def _do_release():
-                kv = self.sessions.pop(name, None)
-                if kv is not None:
-                    self.engine.free_sequence(kv)
+                kv = self.sessions.get(name)
+                if kv is None:
+                    done.put({})
+                    return
+                busy = any(r is not None and r.kv is kv
+                           for r in self.engine._rows) or \
+                    any(r.kv is kv for r in self.engine.waiting)
+                if busy:
+                    # freeing blocks under an in-flight request would let
+                    # the decode graph write into reallocated memory
+                    done.put({"error": f"session {name} has an active "
+                                       "request; release after it finishes"})
+                    return
+                self.sessions.pop(name, None)
+                self.engine.free_sequence(kv)
                 done.put({})
             self._submit.put(_do_release)
-            return done.get(timeout=60)
+            out = done.get(timeout=60)
+            if "error" in out:
+                raise ValueError(out["error"])
+            return out
         if method == "generate":
             return self._generate(params, emit)
         raise ValueError(f"unknown method {method}")

@@ -123,3 +123,52 @@ def test_streaming_generate(tmp_path):
         c.close()
     finally:
         hub.stop()
+
+
+def test_release_refuses_while_request_in_flight(tmp_path):
+    """Freeing a session's KV under an in-flight request would hand its
+    blocks to another sequence mid-decode; release must refuse."""
+    import threading
+    import time as _time
+
+    import torch
+
+    from kukeon_amd.engine.config import EngineConfig, tiny_llama
+    from kukeon_amd.models.llama import LlamaModel
+    from kukeon_amd.serve.server import ModelhubClient, ModelhubServer
+
+    torch.manual_seed(0)
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=256, max_sessions=4, num_kv_blocks=128,
+                        use_graphs=False, decode_microbatch=1)
+    sock = f"/tmp/mhr-{uuid.uuid4().hex[:8]}.sock"
+    hub = ModelhubServer(LlamaModel(cfg, device="cpu"), cfg, ecfg, sock,
+                         device="cpu")
+    hub.start()
+    try:
+        c1 = ModelhubClient(sock, timeout=120)
+        c2 = ModelhubClient(sock, timeout=120)
+        t = threading.Thread(
+            target=lambda: c1.generate("busy", [1, 2, 3],
+                                       max_new_tokens=40, temperature=0.0))
+        t.start()
+        # wait until the request is actually running on the engine
+        deadline = _time.monotonic() + 30
+        while _time.monotonic() < deadline:
+            if hub.engine.num_running > 0 or hub.engine.waiting:
+                break
+            _time.sleep(0.005)
+        refused = False
+        try:
+            c2.call("release", session="busy")
+        except RuntimeError as e:
+            refused = "active request" in str(e)
+        t.join(timeout=60)
+        # either we raced past completion, or it refused while in flight
+        assert refused or hub.engine.num_running == 0
+        # after completion release succeeds
+        c2.call("release", session="busy")
+        assert hub.engine.kv.allocator.num_free == ecfg.num_kv_blocks
+        c1.close(); c2.close()
+    finally:
+        hub.stop()
