@@ -517,8 +517,17 @@ class Controller:
         """Force residual-state removal even when metadata is damaged."""
         with contextlib.suppress(Exception):
             self.kill_cell(realm, space, stack, name)
+        # kill by DISCOVERY too: kill_cell needs the cell document, which
+        # is exactly what may be corrupt — walk the on-disk container dirs
+        # and kill from their runtime records so no process is orphaned
+        cell_dir = self.store.cell_dir(realm, space, stack, name)
+        if cell_dir.is_dir():
+            for cdir in cell_dir.iterdir():
+                if cdir.is_dir():
+                    with contextlib.suppress(Exception):
+                        self.runtime.kill(cdir)
         self.gpus.release(f"cell:{realm}/{space}/{stack}/{name}")
-        self.store.delete_tree(self.store.cell_dir(realm, space, stack, name))
+        self.store.delete_tree(cell_dir)
         self.cgroups.delete(f"{realm}/{space}/{stack}/{name}")
 
     def recreate_cell(self, doc: api.CellDoc) -> api.CellDoc:

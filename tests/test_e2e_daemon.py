@@ -239,3 +239,31 @@ def test_cell_metrics_live(harness):
     assert m["total"]["rssBytes"] > 0
     client.KillCell(realm="default", space="default", stack="default",
                     name="busy")
+
+
+def test_purge_recovers_from_corrupted_metadata(harness):
+    """purge must clear residual state even when metadata.json is damaged
+    (the reference's purge contract: force residual-state removal)."""
+    ctl, srv, client = harness
+    client.ApplyDocuments(yaml=CELL_YAML)
+    cell = client.StartCell(realm="default", space="default",
+                            stack="default", name="busy")
+    pid = cell["status"]["containers"][0]["pid"]
+    assert proc.alive(pid)
+    # corrupt the cell document on disk
+    mpath = ctl.store.cell_dir("default", "default", "default",
+                               "busy") / "metadata.json"
+    mpath.write_text("{ not json !!!")
+    # normal get now fails...
+    with pytest.raises(Exception):
+        client.GetCell(realm="default", space="default", stack="default",
+                       name="busy")
+    # ...but purge clears the process and the residual tree
+    client.PurgeCell(realm="default", space="default", stack="default",
+                     name="busy")
+    deadline = time.monotonic() + 5
+    while proc.alive(pid) and time.monotonic() < deadline:
+        time.sleep(0.05)
+    assert not proc.alive(pid)
+    assert not ctl.store.cell_dir("default", "default", "default",
+                                  "busy").exists()
