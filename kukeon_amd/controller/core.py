@@ -922,6 +922,16 @@ class Controller:
             cdir = cell_dir / (c.id or "main")
             probe = self.runtime.probe(cdir)
             key = str(cdir)
+            if key not in self._restart_state and i in prior and \
+                    prior[i].restart_count:
+                # daemon restarted: re-seed crash-loop bookkeeping from the
+                # persisted status so retry caps survive restarts
+                seed_last = 0.0
+                if prior[i].last_restart_at:
+                    with contextlib.suppress(Exception):
+                        seed_last = parse_iso(prior[i].last_restart_at)
+                self._restart_state[key] = (seed_last,
+                                            prior[i].restart_count)
             last, count = self._restart_state.get(key, (0.0, 0))
             st = api.ContainerStatus(state=probe.state, pid=probe.pid,
                                      exit_code=probe.exit_code,

@@ -412,3 +412,41 @@ def test_failed_start_releases_gpu_reservation(tmp_path):
     doc2 = ctl.get_cell("default", "default", "default", "gfail")
     assert doc2.status.state == api.STATE_FAILED
     assert ctl.gpus.free == [0, 1], "failed start must not hold GPUs"
+
+
+def test_restart_retry_cap_survives_daemon_restart(tmp_path):
+    """on-failure retry caps must not reset when the daemon restarts:
+    the reconcile bookkeeping re-seeds from the persisted status."""
+    rt = FakeRuntime()
+    clock = Clock()
+    ctl = Controller(str(tmp_path / "run"), runtime=rt, gpu_devices=[],
+                     now_fn=clock)
+    ctl.bootstrap()
+    doc = api.CellDoc(
+        metadata=api.Metadata(name="loopy"),
+        spec=api.CellSpec(realm_id="default", space_id="default",
+                          stack_id="default",
+                          containers=[api.ContainerSpec(
+                              id="main", image="x", command="crash",
+                              restart_policy=api.RESTART_ON_FAILURE,
+                              restart_max_retries=2,
+                              restart_backoff_seconds=10)]))
+    ctl.create_cell(doc)
+    ctl.start_cell("default", "default", "default", "loopy")
+    cdir = ctl.store.cell_dir("default", "default", "default",
+                              "loopy") / "main"
+    # two crash/restart cycles exhaust the cap
+    for expect in (1, 2):
+        rt.mark_exited(cdir, 1)
+        clock.t += 11.0
+        d = ctl.reconcile_cell("default", "default", "default", "loopy")
+        assert d.status.containers[0].restart_count == expect
+    # daemon restart: fresh controller, empty in-memory bookkeeping
+    ctl2 = Controller(str(tmp_path / "run"), runtime=rt, gpu_devices=[],
+                      now_fn=clock)
+    rt.mark_exited(cdir, 1)
+    clock.t += 11.0
+    d = ctl2.reconcile_cell("default", "default", "default", "loopy")
+    # cap (2) already consumed: must NOT restart again
+    assert d.status.containers[0].restart_count == 2
+    assert d.status.state == api.STATE_ERROR
