@@ -770,12 +770,21 @@ class Controller:
                             shutil.copytree(src, dst, dirs_exist_ok=True)
                         else:
                             shutil.copy2(src, dst)
-        # destroy the stack's cells (ephemeral by default)
-        with contextlib.suppress(errors.StackNotFound):
-            for cell in self.store.list_children(
-                    self.store.stack_dir(realm, space, stack)):
-                with contextlib.suppress(Exception):
-                    self.delete_cell(realm, space, stack, cell, force=True)
+        # destroy the stack's cells (a session OWNS its stack — but never
+        # sweep the shared default/system stacks, where unrelated cells
+        # live; sessions placed there keep their cells and only release
+        # GPUs + state)
+        if stack not in (naming.DEFAULT_STACK, naming.SYSTEM_STACK):
+            with contextlib.suppress(errors.StackNotFound):
+                for cell in self.store.list_children(
+                        self.store.stack_dir(realm, space, stack)):
+                    with contextlib.suppress(Exception):
+                        self.delete_cell(realm, space, stack, cell,
+                                         force=True)
+        else:
+            log.warning("session %s closed on shared stack %s: cells are "
+                        "kept (use a dedicated stack for ephemeral "
+                        "session workspaces)", name, stack)
         self.gpus.release(f"session:{name}")
         doc.status.state = state
         doc.status.ended_at = now_iso(self.now())

@@ -239,29 +239,32 @@ def test_blueprint_run_and_outofsync(ctl):
 
 
 def test_session_wallclock_deadline(ctl):
+    # sessions own a DEDICATED stack; its cells are swept on teardown
     ses = api.SessionDoc(
         metadata=api.Metadata(name="s1"),
-        spec=api.SessionSpec(stack_id="default", gpus=1,
+        spec=api.SessionSpec(stack_id="s1", gpus=1,
                              lifetime=api.SessionLifetime(wall_clock="30m")))
     ctl.create_session(ses)
-    got = ctl.get_session("default", "default", "default", "s1")
+    got = ctl.get_session("default", "default", "s1", "s1")
     assert got.status.state == api.STATE_RUNNING
     assert got.status.gpu_ids == [0]
-    # a cell lives in the stack; the session teardown must remove it
-    ctl.create_cell(make_cell())
-    ctl.start_cell("default", "default", "default", "c1")
+    # a cell lives in the session's stack; teardown must remove it
+    cell = make_cell()
+    cell.spec.stack_id = "s1"
+    ctl.create_cell(cell)
+    ctl.start_cell("default", "default", "s1", "c1")
     ctl.now.t += 29 * 60
     ctl.reconcile_sessions()
-    assert ctl.get_session("default", "default", "default",
+    assert ctl.get_session("default", "default", "s1",
                            "s1").status.state == api.STATE_RUNNING
     ctl.now.t += 2 * 60
     ctl.reconcile_sessions()
-    got = ctl.get_session("default", "default", "default", "s1")
+    got = ctl.get_session("default", "default", "s1", "s1")
     assert got.status.state == api.STATE_TERMINATED
     assert got.status.ended_at
     assert ctl.gpus.free == [0, 1, 2, 3]
     with pytest.raises(errors.CellNotFound):
-        ctl.get_cell("default", "default", "default", "c1")
+        ctl.get_cell("default", "default", "s1", "c1")
 
 
 def test_secret_env_injection(ctl):
@@ -450,3 +453,40 @@ def test_restart_retry_cap_survives_daemon_restart(tmp_path):
     # cap (2) already consumed: must NOT restart again
     assert d.status.containers[0].restart_count == 2
     assert d.status.state == api.STATE_ERROR
+
+
+def test_session_close_never_sweeps_shared_stack(ctl):
+    """Closing a session placed on the shared default stack must not
+    destroy unrelated cells there; a dedicated stack IS swept."""
+    # unrelated cell on the default stack
+    cell = api.CellDoc(
+        metadata=api.Metadata(name="innocent"),
+        spec=api.CellSpec(realm_id="default", space_id="default",
+                          stack_id="default",
+                          containers=[api.ContainerSpec(
+                              id="main", image="x", command="sleep")]))
+    ctl.create_cell(cell)
+    ses = api.SessionDoc(
+        metadata=api.Metadata(name="s-shared"),
+        spec=api.SessionSpec(realm_id="default", space_id="default",
+                             stack_id="default", owner="t", task="t"))
+    ctl.create_session(ses)
+    ctl.close_session("default", "default", "default", "s-shared")
+    assert ctl.get_cell("default", "default", "default",
+                        "innocent").metadata.name == "innocent"
+    # dedicated stack: cells are swept with the session
+    ses2 = api.SessionDoc(
+        metadata=api.Metadata(name="s-own"),
+        spec=api.SessionSpec(realm_id="default", space_id="default",
+                             stack_id="s-own", owner="t", task="t"))
+    ctl.create_session(ses2)
+    c2 = api.CellDoc(
+        metadata=api.Metadata(name="workspace"),
+        spec=api.CellSpec(realm_id="default", space_id="default",
+                          stack_id="s-own",
+                          containers=[api.ContainerSpec(
+                              id="main", image="x", command="sleep")]))
+    ctl.create_cell(c2)
+    ctl.close_session("default", "default", "s-own", "s-own")
+    with pytest.raises(errors.CellNotFound):
+        ctl.get_cell("default", "default", "s-own", "workspace")
