@@ -502,12 +502,17 @@ def session():
 @click.option("--gpus", default=0)
 @click.option("--wall-clock", default="")
 @click.option("--idle-timeout", default="")
+@click.option("--session-stack", default="",
+              help="stack the session owns (default: a dedicated stack "
+                   "named after the session, swept on close)")
 @pass_ctx
-def session_create(ctx, name, task, owner, gpus, wall_clock, idle_timeout):
+def session_create(ctx, name, task, owner, gpus, wall_clock, idle_timeout,
+                   session_stack):
     doc = api.SessionDoc(
         metadata=api.Metadata(name=name),
         spec=api.SessionSpec(
-            realm_id=ctx.realm, space_id=ctx.space, stack_id=ctx.stack,
+            realm_id=ctx.realm, space_id=ctx.space,
+            stack_id=session_stack or name,
             owner=owner, task=task, gpus=gpus,
             lifetime=api.SessionLifetime(wall_clock=wall_clock,
                                          idle_timeout=idle_timeout)
@@ -524,10 +529,12 @@ def session_create(ctx, name, task, owner, gpus, wall_clock, idle_timeout):
 @session.command("close")
 @click.argument("name")
 @pass_ctx
-def session_close(ctx, name):
+@click.option("--session-stack", default="")
+def session_close(ctx, name, session_stack):
     try:
         res = ctx.client.CloseSession(realm=ctx.realm, space=ctx.space,
-                                      stack=ctx.stack, name=name)
+                                      stack=session_stack or name,
+                                      name=name)
     except errors.KukeonError as e:
         _die(e)
     click.echo(f"session {name}: {res['status']['state']}")

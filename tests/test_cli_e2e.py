@@ -131,3 +131,15 @@ def test_cli_against_live_daemon(tmp_path):
         kd("delete", "cell", "cliy")
     finally:
         srv.stop()
+
+
+def test_cli_session_lifecycle(tmp_path):
+    run = str(tmp_path / "run")
+    kuke(run, "init")
+    out = kuke(run, "session", "create", "agent-7", "--gpus", "0",
+               "--wall-clock", "30m", "--task", "demo").stdout
+    assert "Running" in out
+    # dedicated stack named after the session was created
+    assert (Path(run) / "data" / "default" / "default" / "agent-7").exists()
+    out = kuke(run, "session", "close", "agent-7").stdout
+    assert "Completed" in out
