@@ -152,6 +152,15 @@ def run(ctx, file_, blueprint, config_, params, env_, name, auto_delete,
                 cell = ctx.client.GetCell(
                     realm=doc.spec.realm_id, space=doc.spec.space_id,
                     stack=doc.spec.stack_id, name=doc.metadata.name)
+                # divergence warning (reference run.go create-or-attach):
+                # the file's spec differs from the live cell
+                from kukeon_amd.controller import diff as diffmod
+                d = diffmod.diff_cell(doc, api.CellDoc.from_dict(cell))
+                if d.change_type is not diffmod.ChangeType.NONE:
+                    click.echo(
+                        "warning: cell exists with a diverging spec "
+                        f"({', '.join(d.paths)}) — reusing the live cell; "
+                        "`kuke apply -f` to update it", err=True)
                 click.echo(f"cell {doc.metadata.name} exists — starting")
             except errors.NotFound:
                 ctx.client.CreateCell(doc=doc.to_dict(),

@@ -143,3 +143,19 @@ def test_cli_session_lifecycle(tmp_path):
     assert (Path(run) / "data" / "default" / "default" / "agent-7").exists()
     out = kuke(run, "session", "close", "agent-7").stdout
     assert "Completed" in out
+
+
+def test_cli_run_divergence_warning(tmp_path):
+    run = str(tmp_path / "run")
+    kuke(run, "init")
+    spec = tmp_path / "cell.yaml"
+    spec.write_text(CELL_YAML)
+    kuke(run, "run", "-f", str(spec), "--no-attach")
+    # same file again: no divergence warning
+    r = kuke(run, "run", "-f", str(spec), "--no-attach")
+    assert "diverging" not in r.stderr
+    # changed args: reuse with a warning
+    spec.write_text(CELL_YAML.replace('["30"]', '["60"]'))
+    r = kuke(run, "run", "-f", str(spec), "--no-attach")
+    assert "diverging spec" in r.stderr and "args" in r.stderr
+    kuke(run, "kill", "busy")
