@@ -158,4 +158,4 @@ def test_cli_run_divergence_warning(tmp_path):
     spec.write_text(CELL_YAML.replace('["30"]', '["60"]'))
     r = kuke(run, "run", "-f", str(spec), "--no-attach")
     assert "diverging spec" in r.stderr and "args" in r.stderr
-    kuke(run, "kill", "busy")
+    kuke(run, "kill", "cliy")
