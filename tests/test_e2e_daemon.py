@@ -4,6 +4,7 @@ on a CPU-only host) as a test. Reference harness pattern: per-test daemon on
 a short /tmp socket with --reconcile-interval 0 so tests don't race the
 loop."""
 import json
+import os
 import socket
 import time
 import uuid
@@ -267,3 +268,47 @@ def test_purge_recovers_from_corrupted_metadata(harness):
     assert not proc.alive(pid)
     assert not ctl.store.cell_dir("default", "default", "default",
                                   "busy").exists()
+
+
+def test_background_reconcile_loop_restarts(tmp_path):
+    """The daemon's own periodic loop (no manual ReconcileCells calls)
+    must pick up a crash and restart the container."""
+    import uuid as _uuid
+
+    sock = f"/tmp/kuke-bg-{_uuid.uuid4().hex[:8]}.sock"
+    ctl = Controller(str(tmp_path / "run"), gpu_devices=[])
+    ctl.bootstrap()
+    srv = Server(ctl, sock, reconcile_interval=0.1)
+    srv.start()
+    client = UnixClient(sock, timeout=15.0)
+    try:
+        doc = api.CellDoc(
+            metadata=api.Metadata(name="bg"),
+            spec=api.CellSpec(
+                realm_id="default", space_id="default", stack_id="default",
+                containers=[api.ContainerSpec(
+                    id="main", image="busybox", command="/bin/sh",
+                    args=["-c", "sleep 60"], restart_policy="always",
+                    restart_backoff_seconds=0)]))
+        client.CreateCell(doc=doc.to_dict())
+        cell = client.StartCell(realm="default", space="default",
+                                stack="default", name="bg")
+        pid = cell["status"]["containers"][0]["pid"]
+        os.kill(pid, 9)
+        deadline = time.monotonic() + 15
+        restarted = False
+        while time.monotonic() < deadline:
+            time.sleep(0.2)
+            cur = client.GetCell(realm="default", space="default",
+                                 stack="default", name="bg")
+            cs = cur["status"]["containers"][0]
+            if cs.get("restartCount", 0) >= 1 and cs["pid"] != pid and \
+                    proc.alive(cs["pid"]):
+                restarted = True
+                break
+        assert restarted, "background loop never restarted the container"
+        client.KillCell(realm="default", space="default", stack="default",
+                        name="bg")
+    finally:
+        client.close()
+        srv.stop()
