@@ -119,3 +119,37 @@ def test_proc_metrics_self():
     assert m["rssBytes"] > 1 << 20 and m["threads"] >= 1
     assert m["cpuSeconds"] >= 0.0
     assert proc.metrics(2 ** 22 + 12345) is None  # unlikely pid
+
+
+def test_cgroup_v2_tree_against_injected_root(tmp_path):
+    """CgroupManager drives a v2 tree: chain creation, subtree-controller
+    delegation up the ancestors, memory.max / cpu.weight knobs, cleanup —
+    asserted against an injected fs root (no root privileges)."""
+    from kukeon_amd.runtime.cgroup import CgroupManager
+
+    root = tmp_path / "cg"
+    root.mkdir()
+    (root / "cgroup.controllers").write_text("cpu memory io pids hugetlb")
+    m = CgroupManager(root=str(root))
+    assert m.mode == "v2"
+    created = m.create("default/dev/stack/cell-a")
+    assert created and (root / "kukeon" / "default" / "dev" / "stack" /
+                        "cell-a").is_dir()
+    # delegation wrote the resource subset (not hugetlb) on the ancestors
+    sub = (root / "kukeon" / "default" / "dev" /
+           "cgroup.subtree_control").read_text()
+    assert "+cpu" in sub and "+memory" in sub and "hugetlb" not in sub
+    m.set_memory_limit("default/dev/stack/cell-a", 1 << 30)
+    assert (root / "kukeon" / "default" / "dev" / "stack" / "cell-a" /
+            "memory.max").read_text() == str(1 << 30)
+    m.set_cpu_shares("default/dev/stack/cell-a", 1024)
+    w = int((root / "kukeon" / "default" / "dev" / "stack" / "cell-a" /
+             "cpu.weight").read_text())
+    assert 1 <= w <= 10000
+    # on real cgroupfs the knob files are virtual and vanish with rmdir;
+    # on the injected plain fs they must be cleared first
+    leaf = root / "kukeon" / "default" / "dev" / "stack" / "cell-a"
+    for f in leaf.iterdir():
+        f.unlink()
+    m.delete("default/dev/stack/cell-a")
+    assert not leaf.exists()
