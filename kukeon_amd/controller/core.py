@@ -949,10 +949,15 @@ class Controller:
                                      restart_count=count)
             if i in prior:
                 # carry start-time facts the probe cannot re-derive:
-                # GPU pinning (drives the restart env) and repo statuses
+                # GPU pinning (drives the restart env) and restart stamps
                 st.gpu_ids = list(prior[i].gpu_ids)
                 st.repos = prior[i].repos
                 st.last_restart_at = prior[i].last_restart_at
+            setup = self.store.read(cdir / "setup.json")
+            if setup:
+                # the shim's repo-setup record is authoritative
+                st.repos = [api.RepoStatus.from_dict(r)
+                            for r in setup.get("repos", [])]
             if probe.running:
                 running += 1
                 if count and probe.started_at and \

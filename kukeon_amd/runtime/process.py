@@ -124,6 +124,10 @@ class ProcessRuntime(Runtime):
             "cwd": spec.working_dir or str(cdir),
             "attachable": bool(spec.attachable),
             "init_script": (spec.tty.init_script if spec.tty else ""),
+            "repos": [{"url": r.url, "path": r.path, "ref": r.ref}
+                      for r in (spec.repos or [])],
+            "git": ({"name": spec.git.name, "email": spec.git.email}
+                    if spec.git else {}),
         }
         # clear stale exit state from a previous run
         with contextlib.suppress(FileNotFoundError):

@@ -15,6 +15,10 @@ stages.go:62-88, kukepause PID-1 semantics cmd/kukepause/main.go:17-60 via
   at <dir>/tty/socket (multi-client byte pump, capture file for `kuke log`,
   activity timestamps for Session idleTimeout), runs the optional tty init
   script first,
+* runs container setup before the workload (the kuketty runOn:create
+  role): clones/fetches declared git repos with the container-local git
+  identity and records per-repo state to `setup.json`, which the
+  reconcile loop surfaces into `ContainerStatus.repos`,
 * `--pause` mode is the cell root: a signal-wait PID holding the cell's
   process group open (kukepause analog; exits 0 on SIGTERM/SIGINT).
 """
@@ -83,6 +87,54 @@ class Shim:
     def record_status(self, rc: int) -> None:
         write_json(self.dir / "status.json",
                    {"exitCode": rc, "finishedAt": now_iso()})
+
+    def run_setup(self) -> None:
+        """Clone/fetch declared repos; write per-repo states to
+        setup.json. Failures are recorded, not fatal — the workload still
+        starts and the agent can inspect status.repos."""
+        repos = self.spec.get("repos") or []
+        if not repos:
+            return
+        import subprocess
+        git_id = self.spec.get("git") or {}
+        env = self.child_env()
+        if git_id.get("name"):
+            env["GIT_AUTHOR_NAME"] = env["GIT_COMMITTER_NAME"] = \
+                git_id["name"]
+        if git_id.get("email"):
+            env["GIT_AUTHOR_EMAIL"] = env["GIT_COMMITTER_EMAIL"] = \
+                git_id["email"]
+        states = []
+        for r in repos:
+            url = r.get("url", "")
+            dest = Path(r.get("path") or Path(url).name or "repo")
+            if not dest.is_absolute():
+                dest = self.dir / dest
+            st = {"url": url, "state": "", "error": ""}
+            try:
+                if (dest / ".git").is_dir():
+                    subprocess.run(["git", "-C", str(dest), "fetch",
+                                    "--all", "--tags"], env=env, timeout=120,
+                                   capture_output=True, check=True)
+                    st["state"] = "fetched"
+                else:
+                    cmd = ["git", "clone", url, str(dest)]
+                    subprocess.run(cmd, env=env, timeout=300,
+                                   capture_output=True, check=True)
+                    st["state"] = "cloned"
+                if r.get("ref"):
+                    subprocess.run(["git", "-C", str(dest), "checkout",
+                                    r["ref"]], env=env, timeout=60,
+                                   capture_output=True, check=True)
+            except subprocess.CalledProcessError as e:
+                st["state"] = "failed"
+                st["error"] = (e.stderr or b"").decode(
+                    "utf-8", "replace")[-500:]
+            except Exception as e:  # timeout, missing git, ...
+                st["state"] = "failed"
+                st["error"] = str(e)[-500:]
+            states.append(st)
+        write_json(self.dir / "setup.json", {"repos": states})
 
     def child_env(self) -> dict:
         env = dict(os.environ)
@@ -276,6 +328,7 @@ def main() -> int:
                 pass
 
     signal.signal(signal.SIGTERM, on_term)
+    shim.run_setup()
     if shim.spec.get("attachable"):
         return shim.run_attachable()
     return shim.run_plain()

@@ -312,3 +312,44 @@ def test_background_reconcile_loop_restarts(tmp_path):
     finally:
         client.close()
         srv.stop()
+
+
+def test_container_repo_setup(harness, tmp_path):
+    """The shim clones declared repos before the workload (kuketty
+    runOn:create parity) and the reconcile loop surfaces per-repo state
+    into ContainerStatus.repos."""
+    import subprocess
+
+    ctl, srv, client = harness
+    src = tmp_path / "srcrepo"
+    src.mkdir()
+    subprocess.run(["git", "init", "-q", str(src)], check=True)
+    (src / "hello.txt").write_text("hi")
+    subprocess.run(["git", "-C", str(src), "add", "."], check=True)
+    subprocess.run(["git", "-C", str(src), "-c", "user.name=t",
+                    "-c", "user.email=t@x", "commit", "-qm", "init"],
+                   check=True)
+    doc = api.CellDoc(
+        metadata=api.Metadata(name="repocell"),
+        spec=api.CellSpec(
+            realm_id="default", space_id="default", stack_id="default",
+            containers=[api.ContainerSpec(
+                id="main", image="busybox", command="sleep", args=["30"],
+                repos=[api.ContainerRepo(url=str(src), path="work")],
+                git=api.ContainerGit(name="agent", email="a@x"))]))
+    client.CreateCell(doc=doc.to_dict())
+    client.StartCell(realm="default", space="default", stack="default",
+                     name="repocell")
+    cdir = ctl.store.cell_dir("default", "default", "default",
+                              "repocell") / "main"
+    deadline = time.monotonic() + 20
+    while time.monotonic() < deadline and not (cdir / "setup.json").exists():
+        time.sleep(0.1)
+    assert (cdir / "work" / ".git").is_dir()
+    assert (cdir / "work" / "hello.txt").read_text() == "hi"
+    cell = client.ReconcileCells() and client.GetCell(
+        realm="default", space="default", stack="default", name="repocell")
+    repos = cell["status"]["containers"][0]["repos"]
+    assert repos and repos[0]["state"] == "cloned", repos
+    client.KillCell(realm="default", space="default", stack="default",
+                    name="repocell")
