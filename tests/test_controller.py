@@ -490,3 +490,66 @@ def test_session_close_never_sweeps_shared_stack(ctl):
     ctl.close_session("default", "default", "s-own", "s-own")
     with pytest.raises(errors.CellNotFound):
         ctl.get_cell("default", "default", "s-own", "workspace")
+
+
+def test_controller_fuzz_invariants(tmp_path):
+    """Seeded random verb storm against the controller (FakeRuntime):
+    whatever the sequence, GPU reservations stay conserved, the state
+    tree stays parseable, and terminal cleanup releases everything."""
+    import random
+
+    rng = random.Random(1234)
+    rt = FakeRuntime()
+    ctl = Controller(str(tmp_path / "run"), runtime=rt,
+                     gpu_devices=[0, 1, 2, 3], now_fn=Clock())
+    ctl.bootstrap()
+    names = [f"f{i}" for i in range(6)]
+
+    def mk(name):
+        return api.CellDoc(
+            metadata=api.Metadata(name=name),
+            spec=api.CellSpec(realm_id="default", space_id="default",
+                              stack_id="default",
+                              containers=[api.ContainerSpec(
+                                  id="main", image="x", command="sleep",
+                                  gpus=rng.choice([0, 0, 1, 2]))]))
+
+    ops_done = 0
+    for _ in range(300):
+        name = rng.choice(names)
+        op = rng.choice(["create", "start", "stop", "kill", "delete",
+                         "reconcile", "purge"])
+        try:
+            if op == "create":
+                ctl.create_cell(mk(name))
+            elif op == "start":
+                ctl.start_cell("default", "default", "default", name)
+            elif op == "stop":
+                ctl.stop_cell("default", "default", "default", name)
+            elif op == "kill":
+                ctl.kill_cell("default", "default", "default", name)
+            elif op == "delete":
+                ctl.delete_cell("default", "default", "default", name,
+                                force=True)
+            elif op == "purge":
+                ctl.purge_cell("default", "default", "default", name)
+            else:
+                ctl.reconcile_cells()
+            ops_done += 1
+        except errors.KukeonError:
+            pass  # invalid transitions are allowed to fail cleanly
+        # invariant: every parseable doc; GPU books balance
+        used = {g for v in ctl.gpus.assignments.values() for g in v}
+        assert len(used) + len(ctl.gpus.free) == 4
+        for cell in ctl.store.list_children(
+                ctl.store.stack_dir("default", "default", "default")):
+            ctl.get_cell("default", "default", "default", cell)
+    assert ops_done > 150
+    # terminal cleanup returns every GPU
+    for name in names:
+        try:
+            ctl.delete_cell("default", "default", "default", name,
+                            force=True)
+        except errors.KukeonError:
+            pass
+    assert sorted(ctl.gpus.free) == [0, 1, 2, 3]
