@@ -113,6 +113,11 @@ def paged_attention(out: torch.Tensor, q: torch.Tensor, k_cache: torch.Tensor,
     for b in range(B):
         ctx = int(seq_lens[b])
         if ctx <= 0:
+            # inactive decode row: write zeros so downstream logits stay
+            # finite (the row's sample is discarded by the advance guard,
+            # but NaNs from uninitialized memory would crash the CPU
+            # sampler's multinomial)
+            out.view(B, -1)[b] = 0
             continue
         k = _gather_kv(k_cache, block_table, ctx, b).float()  # [ctx,Hk,D]
         v = _gather_kv(v_cache, block_table, ctx, b).float()

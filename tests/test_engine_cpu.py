@@ -292,3 +292,52 @@ def test_session_driver_survives_idle_eviction():
         total += driver.run_round()
     assert total == 12
     assert len(driver.turn_latencies) == 12
+
+
+@pytest.mark.parametrize("seed", [3, 99])
+def test_engine_fuzz_kv_conservation(seed):
+    """Random request storms on a tiny KV pool: block accounting must
+    balance at every step (free + every sequence's held blocks == pool),
+    and every request must finish with its full token count."""
+    import random
+
+    rng = random.Random(seed)
+    torch.manual_seed(0)
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=160, max_sessions=3, num_kv_blocks=20,
+                        use_graphs=False)
+    model = LlamaModel(cfg, device="cpu")
+    engine = LLMEngine(model, cfg, ecfg, device="cpu")
+    kvs = [SequenceKV(ecfg.block_size) for _ in range(5)]
+    want, got = {}, {}
+    live = set()
+    steps = 0
+    for round_ in range(8):
+        for kv in rng.sample(kvs, rng.randint(1, 3)):
+            if id(kv) in live:
+                continue
+            n = rng.randint(2, 10)
+            ctx = kv.num_tokens or len(kv.history)
+            if ctx + 12 + n > ecfg.max_model_len:
+                engine.free_sequence(kv)
+            rid = engine.add_request(
+                kv, [rng.randrange(cfg.vocab_size)
+                     for _ in range(rng.randint(3, 12))],
+                SamplingParams(temperature=rng.choice([0.0, 0.8]),
+                               max_new_tokens=n))
+            want[rid], got[rid] = n, 0
+            live.add(id(kv))
+        while engine.has_work():
+            for o in engine.step():
+                got[o.req_id] += len(o.new_tokens)
+                if o.finished:
+                    pass
+            steps += 1
+            assert steps < 5000
+            held = sum(len(kv.blocks) for kv in kvs)
+            assert held + engine.kv.allocator.num_free == ecfg.num_kv_blocks
+        live.clear()
+    assert got == want
+    for kv in kvs:
+        engine.free_sequence(kv)
+    assert engine.kv.allocator.num_free == ecfg.num_kv_blocks
