@@ -172,3 +172,65 @@ def test_release_refuses_while_request_in_flight(tmp_path):
         c1.close(); c2.close()
     finally:
         hub.stop()
+
+
+def test_server_concurrent_storm(tmp_path):
+    """Thread storm against the server: interleaved generate/release/
+    stats from many clients on a tiny KV pool (forces eviction) must
+    neither deadlock nor corrupt accounting."""
+    import random
+    import threading
+
+    import torch
+
+    from kukeon_amd.engine.config import EngineConfig, tiny_llama
+    from kukeon_amd.models.llama import LlamaModel
+    from kukeon_amd.serve.server import ModelhubClient, ModelhubServer
+
+    torch.manual_seed(0)
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=128, max_sessions=4, num_kv_blocks=24,
+                        use_graphs=False, decode_microbatch=2)
+    sock = f"/tmp/mhst-{uuid.uuid4().hex[:8]}.sock"
+    hub = ModelhubServer(LlamaModel(cfg, device="cpu"), cfg, ecfg, sock,
+                         device="cpu")
+    hub.start()
+    errs = []
+
+    def worker(wid):
+        rng = random.Random(wid)
+        try:
+            c = ModelhubClient(sock, timeout=120)
+            for i in range(6):
+                op = rng.choice(["gen", "gen", "gen", "rel", "stats"])
+                sess = f"s{rng.randint(0, 5)}"
+                if op == "gen":
+                    r = c.generate(sess,
+                                   [rng.randrange(cfg.vocab_size)
+                                    for _ in range(rng.randint(2, 8))],
+                                   max_new_tokens=rng.randint(2, 6),
+                                   temperature=0.0)
+                    assert r["tokens"]
+                elif op == "rel":
+                    try:
+                        c.call("release", session=sess)
+                    except RuntimeError as e:
+                        assert ("active request" in str(e) or
+                                "exceeds" in str(e)), e
+                else:
+                    st = c.call("stats")
+                    assert st["kv_blocks_total"] == 24
+            c.close()
+        except Exception as e:  # noqa: BLE001
+            errs.append((wid, repr(e)))
+
+    ts = [threading.Thread(target=worker, args=(i,)) for i in range(6)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=180)
+    assert not any(t.is_alive() for t in ts), "storm deadlocked"
+    allowed = ("exceeds max_model_len", "cannot be admitted")
+    real = [e for e in errs if not any(a in e[1] for a in allowed)]
+    assert not real, real
+    hub.stop()
