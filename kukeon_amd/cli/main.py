@@ -480,23 +480,50 @@ def status(ctx):
 @click.argument("what", default="all")
 @pass_ctx
 def doctor(ctx, what):
-    """Preflight checks (cgroups, gpus, tooling)."""
+    """Preflight checks with required/optional classification (reference
+    cgroupcheck probe-and-classify); exits non-zero when a required
+    capability is missing."""
     from kukeon_amd.runtime.cgroup import CgroupManager
     from kukeon_amd.runtime.devices import discover_gpus
     import shutil as sh
+    ok = True
+
+    def row(name, good, detail, required=False):
+        nonlocal ok
+        mark = "ok" if good else ("FAIL" if required else "warn")
+        if required and not good:
+            ok = False
+        click.echo(f"[{mark:>4}] {name:<16} {detail}")
+
     cg = CgroupManager()
-    click.echo(f"cgroups: mode={cg.mode} "
-               f"controllers={','.join(cg.available_controllers()) or '-'}")
+    ctrls = set(cg.available_controllers())
+    row("cgroups", cg.mode != "none",
+        f"mode={cg.mode} controllers={','.join(sorted(ctrls)) or '-'}")
+    for c in ("cpu", "memory", "pids"):
+        row(f"cgroup:{c}", c in ctrls or cg.mode == "none",
+            "available" if c in ctrls else "missing (limits degrade)")
     gpus = discover_gpus()
-    click.echo(f"amdgpu: {len(gpus)} device(s) {gpus}")
-    click.echo(f"kfd: {'present' if os.path.exists('/dev/kfd') else 'absent'}")
-    click.echo(f"iptables: {'present' if sh.which('iptables') else 'absent'}")
+    row("amdgpu", True, f"{len(gpus)} device(s) {gpus}")
+    row("kfd", os.path.exists("/dev/kfd"),
+        "present" if os.path.exists("/dev/kfd") else
+        "absent (no GPU compute on this host)")
+    row("iptables", bool(sh.which("iptables")),
+        "present" if sh.which("iptables") else
+        "absent (egress policy degrades to noop)")
+    row("git", bool(sh.which("git")),
+        "present" if sh.which("git") else
+        "absent (container repo setup unavailable)")
+    run = Path(ctx.run_path)
+    writable = os.access(run if run.exists() else run.parent, os.W_OK)
+    row("run path", writable, str(run), required=True)
     try:
         import kukeon_amd.ops as ops
-        click.echo(f"hip extension: "
-                   f"{'loaded' if ops.native_available() else 'MISSING'}")
-    except Exception as e:
-        click.echo(f"hip extension: error ({e})")
+        row("hip extension", ops.native_available(),
+            "loaded" if ops.native_available() else
+            "not built (python -m kukeon_amd.ops.build)")
+    except Exception as e:  # noqa: BLE001
+        row("hip extension", False, f"error ({e})")
+    sys.exit(0 if ok else 1)
 
 
 @cli.group()

@@ -159,3 +159,12 @@ def test_cli_run_divergence_warning(tmp_path):
     r = kuke(run, "run", "-f", str(spec), "--no-attach")
     assert "diverging spec" in r.stderr and "args" in r.stderr
     kuke(run, "kill", "cliy")
+
+
+def test_cli_doctor(tmp_path):
+    run = str(tmp_path / "run")
+    r = kuke(run, "doctor")
+    out = r.stdout
+    for field in ("cgroups", "cgroup:memory", "amdgpu", "kfd", "iptables",
+                  "git", "run path", "hip extension"):
+        assert field in out, out
