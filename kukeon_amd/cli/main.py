@@ -254,8 +254,17 @@ def get(ctx, kind, name, output):
             docs = [c.GetCell(realm=r, space=s, stack=st, name=name)] \
                 if name else c.ListCells(realm=r, space=s, stack=st)
         elif k == "Session":
-            docs = [c.GetSession(realm=r, space=s, stack=st, name=name)] \
-                if name else c.ListSessions()
+            if name:
+                try:
+                    docs = [c.GetSession(realm=r, space=s, stack=st,
+                                         name=name)]
+                except errors.NotFound:
+                    # dedicated-stack convention: the session's stack is
+                    # named after it
+                    docs = [c.GetSession(realm=r, space=s, stack=name,
+                                         name=name)]
+            else:
+                docs = c.ListSessions()
         elif k == "Secret":
             docs = ([c.GetSecret(realm=r, space=s, name=name)] if name else
                     [{"metadata": {"name": n}, "kind": "Secret",

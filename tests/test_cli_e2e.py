@@ -141,6 +141,8 @@ def test_cli_session_lifecycle(tmp_path):
     assert "Running" in out
     # dedicated stack named after the session was created
     assert (Path(run) / "data" / "default" / "default" / "agent-7").exists()
+    out = kuke(run, "get", "session", "agent-7", "-o", "yaml").stdout
+    assert "Running" in out
     out = kuke(run, "session", "close", "agent-7").stdout
     assert "Completed" in out
 
