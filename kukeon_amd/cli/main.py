@@ -313,7 +313,11 @@ def _scope_verb(ctx, verb, kind, name, cascade=False, force=False):
                 c.DeleteCell(realm=r, space=s, stack=st, name=name,
                              force=force)
             elif k == "Session":
-                c.DeleteSession(realm=r, space=s, stack=st, name=name)
+                try:
+                    c.DeleteSession(realm=r, space=s, stack=st, name=name)
+                except errors.NotFound:
+                    # dedicated-stack convention
+                    c.DeleteSession(realm=r, space=s, stack=name, name=name)
             elif k == "Secret":
                 c.DeleteSecret(realm=r, space=s, name=name)
             elif k == "CellBlueprint":
