@@ -153,3 +153,50 @@ def test_cgroup_v2_tree_against_injected_root(tmp_path):
         f.unlink()
     m.delete("default/dev/stack/cell-a")
     assert not leaf.exists()
+
+
+def test_store_cas_multiprocess_no_lost_updates(tmp_path):
+    """Hard part #1: the flock+CAS envelope under REAL
+    multi-process contention — 4 processes each add their increments to
+    a counter field via optimistic CAS with retry; no update may be
+    lost and generations must count every successful write."""
+    import subprocess
+    import sys as _sys
+
+    doc = tmp_path / "metadata.json"
+    worker = tmp_path / "worker.py"
+    worker.write_text(f'''
+import sys
+sys.path.insert(0, {str(repr(str(__import__('pathlib').Path(__file__).resolve().parent.parent)))})
+from pathlib import Path
+from kukeon_amd.state.store import Store
+from kukeon_amd.api import errors
+store = Store(sys.argv[1])
+path = Path(sys.argv[2])
+wid, n = int(sys.argv[3]), int(sys.argv[4])
+done = 0
+while done < n:
+    cur = store.read(path) or {{"metadata": {{"generation": 0}},
+                               "count": 0, "by": {{}}}}
+    gen = cur["metadata"].get("generation", 0)
+    cur["count"] = cur.get("count", 0) + 1
+    by = cur.setdefault("by", {{}})
+    by[str(wid)] = by.get(str(wid), 0) + 1
+    try:
+        store.write_cas(path, cur, expected_generation=gen)
+        done += 1
+    except errors.StaleResource:
+        pass  # lost the race: retry with a fresh read
+''')
+    store_root = str(tmp_path / "run")
+    from kukeon_amd.state.store import Store
+    Store(store_root)  # create root
+    procs = [subprocess.Popen([_sys.executable, str(worker), store_root,
+                               str(doc), str(i), "40"])
+             for i in range(4)]
+    for p in procs:
+        assert p.wait(timeout=120) == 0
+    final = Store(store_root).read(doc)
+    assert final["count"] == 160, final
+    assert final["metadata"]["generation"] == 160
+    assert sorted(final["by"].items()) == [(str(i), 40) for i in range(4)]
