@@ -200,3 +200,27 @@ while done < n:
     assert final["count"] == 160, final
     assert final["metadata"]["generation"] == 160
     assert sorted(final["by"].items()) == [(str(i), 40) for i in range(4)]
+
+
+def test_subnet_allocator_persistence_release_exhaustion(tmp_path):
+    from kukeon_amd.state.store import Store
+    from kukeon_amd.controller.subnet import SubnetAllocator
+
+    store = Store(str(tmp_path / "run"))
+    a = SubnetAllocator(store)
+    s1 = a.allocate("default", "alpha")
+    s2 = a.allocate("default", "beta")
+    assert s1 != s2 and s1.endswith(".0/24")
+    assert a.allocate("default", "alpha") == s1   # idempotent
+    # persisted: a fresh allocator over the same tree sees it
+    b = SubnetAllocator(Store(str(tmp_path / "run")))
+    assert b.lookup("default", "alpha") == s1
+    b.release("default", "alpha")
+    assert b.lookup("default", "alpha") is None
+    s3 = b.allocate("default", "gamma")
+    assert s3 == s1  # the freed /24 is reusable
+    # exhaustion: burn the rest of the 256-slot pool (beta + gamma held)
+    for i in range(254):
+        b.allocate("default", f"sp{i}")
+    with pytest.raises(errors.KukeonError, match="exhausted"):
+        b.allocate("default", "overflow")
