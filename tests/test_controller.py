@@ -554,3 +554,22 @@ def test_controller_fuzz_invariants(tmp_path, seed):
         except errors.KukeonError:
             pass
     assert sorted(ctl.gpus.free) == [0, 1, 2, 3]
+
+
+def test_server_configuration_wires_knobs(tmp_path):
+    """ServerConfiguration (kukeond --configuration) feeds GPU devices and
+    disk-pressure thresholds into the controller."""
+    cfg = api.ServerConfigurationSpec(
+        gpu_devices=[4, 5], disk_pressure_warn_percent=50,
+        disk_pressure_block_percent=60)
+    ctl = Controller(str(tmp_path / "run"), runtime=FakeRuntime(),
+                     server_config=cfg, now_fn=Clock())
+    ctl.bootstrap()
+    assert ctl.gpus.devices == [4, 5]
+    assert ctl.disk_guard.warn_percent == 50
+    assert ctl.disk_guard.block_percent == 60
+    # a doc round-trips through YAML with the daemon's loader shape
+    doc = api.ServerConfigurationDoc(
+        metadata=api.Metadata(name="srv"), spec=cfg)
+    again = api.ServerConfigurationDoc.from_dict(doc.to_dict())
+    assert again.spec.gpu_devices == [4, 5]
