@@ -144,7 +144,10 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            # weights/compute are always bf16; a reduced-precision KV
+            # cache is reported so the line never overstates the dtype
+            "dtype": ("bf16" if args.kv_dtype == "bf16"
+                      else f"bf16+kv_{args.kv_dtype}"),
             "data": "synthetic",
             "p50_turn_latency_ms": round(p50, 1) if p50 else None,
             "config": {

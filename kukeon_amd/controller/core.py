@@ -759,17 +759,33 @@ class Controller:
         if doc.spec.on_end:
             for p in doc.spec.on_end.persist:
                 src = Path(p.volume)
-                if src.is_absolute():
-                    # cell-relative volumes: look under the cell dirs
-                    src = Path(str(src))
-                if src.exists():
-                    dst = persist_dir / src.name
-                    dst.parent.mkdir(parents=True, exist_ok=True)
-                    with contextlib.suppress(Exception):
-                        if src.is_dir():
-                            shutil.copytree(src, dst, dirs_exist_ok=True)
-                        else:
-                            shutil.copy2(src, dst)
+                if not src.is_absolute():
+                    # non-absolute entries name a Volume doc in the
+                    # session's scope: resolve to its backing path
+                    try:
+                        vol = self.get_volume(realm, space, p.volume)
+                        src = Path(vol.status.path)
+                    except errors.NotFound:
+                        log.warning(
+                            "session %s persist entry %r: no such volume "
+                            "in %s/%s; skipped", name, p.volume, realm,
+                            space)
+                        continue
+                if not src.exists():
+                    log.warning("session %s persist entry %r: path %s "
+                                "does not exist; skipped", name, p.volume,
+                                src)
+                    continue
+                dst = persist_dir / src.name
+                dst.parent.mkdir(parents=True, exist_ok=True)
+                try:
+                    if src.is_dir():
+                        shutil.copytree(src, dst, dirs_exist_ok=True)
+                    else:
+                        shutil.copy2(src, dst)
+                except OSError as exc:
+                    log.warning("session %s persist of %s failed: %s",
+                                name, src, exc)
         # destroy the stack's cells (a session OWNS its stack — but never
         # sweep the shared default/system stacks, where unrelated cells
         # live; sessions placed there keep their cells and only release

@@ -186,7 +186,11 @@ class LLMEngine:
     def step(self) -> List[StepOutput]:
         if self.waiting:
             batch, blocked = self._admit_prefill()
-            if not batch and blocked and self._evict_idle_kv():
+            # several small idle sessions may need to go before the head
+            # request fits: keep evicting until admission unblocks or no
+            # victims remain (single-evict crashed on spread-out
+            # idle context even though capacity existed)
+            while not batch and blocked and self._evict_idle_kv():
                 batch, blocked = self._admit_prefill()
             if batch:
                 return self._run_prefill(batch)

@@ -72,13 +72,19 @@ class GPUAllocator:
         return [d for d in self.devices if d not in used]
 
     def allocate(self, owner: str, count: int) -> List[int]:
-        if owner in self.assignments:
-            return self.assignments[owner]
+        have = self.assignments.get(owner, [])
+        if have and len(have) >= count:
+            return have
+        # top up an existing (shorter) assignment after a spec change
+        # requesting more GPUs — never return a stale short list, which
+        # would make start_cell silently pin fewer GPUs than spec
         free = self.free
-        if count > len(free):
+        need = count - len(have)
+        if need > len(free):
             raise errors.GPUUnavailable(
-                f"want {count} GPUs, {len(free)} free of {len(self.devices)}")
-        got = free[:count]
+                f"want {count} GPUs ({need} more), {len(free)} free of "
+                f"{len(self.devices)}")
+        got = have + free[:need]
         self.assignments[owner] = got
         self._save()
         return got

@@ -79,7 +79,28 @@ class ModelhubServer:
                 continue
             import time as _time
             t0 = _time.perf_counter()
-            outs = self.engine.step()
+            try:
+                outs = self.engine.step()
+            except Exception as exc:  # noqa: BLE001
+                # A raising step must never silently kill this thread: the
+                # process would stay alive with every pending (and future)
+                # request hung to its reply timeout and RestartPolicy never
+                # firing. Fail all pending replies, then either keep
+                # serving (admission-level MemoryError leaves the engine
+                # consistent) or exit so restart_policy revives the cell.
+                log.exception("engine.step failed: %s", exc)
+                with self._lock:
+                    for rid, p in list(self._pending.items()):
+                        p.reply.put({"error": f"engine step failed: {exc}"})
+                        del self._pending[rid]
+                if isinstance(exc, MemoryError):
+                    # drop the unadmittable head request(s); engine state
+                    # is consistent, keep serving the rest
+                    self.engine.waiting.clear()
+                    continue
+                log.critical("engine thread fatal; exiting for restart "
+                             "policy")
+                os._exit(70)
             self.metrics["busy_seconds"] += _time.perf_counter() - t0
             self.metrics["engine_steps"] += 1
             with self._lock:

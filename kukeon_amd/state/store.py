@@ -167,12 +167,16 @@ class Store:
             self._atomic_write(path, data, mode)
 
     def delete(self, path: Path) -> bool:
+        # The sidecar .lock is left behind as a tombstone on purpose:
+        # unlinking it while held would let a process already blocked in
+        # flock() on the old inode acquire a stale lock concurrently with
+        # a new holder that recreated the file — two writers in the CAS
+        # critical section. delete_tree() reclaims them with
+        # the directory.
         with self.lock(path):
             existed = path.exists()
             with contextlib.suppress(FileNotFoundError):
                 path.unlink()
-            with contextlib.suppress(FileNotFoundError):
-                Path(str(path) + ".lock").unlink()
             return existed
 
     def delete_tree(self, d: Path) -> None:

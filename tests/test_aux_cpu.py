@@ -224,3 +224,41 @@ def test_subnet_allocator_persistence_release_exhaustion(tmp_path):
         b.allocate("default", f"sp{i}")
     with pytest.raises(errors.KukeonError, match="exhausted"):
         b.allocate("default", "overflow")
+
+
+def test_gpu_allocator_top_up_after_spec_growth(tmp_path):
+    """An owner whose spec grows to more GPUs gets topped up from free,
+    never a stale shorter list."""
+    path = str(tmp_path / "gpus.json")
+    a = GPUAllocator(path, devices=[0, 1, 2, 3])
+    assert a.allocate("sess-a", 1) == [0]
+    # spec change: now wants 3 — keeps its existing GPU, gains two
+    assert a.allocate("sess-a", 3) == [0, 1, 2]
+    # persisted
+    assert GPUAllocator(path, devices=[0, 1, 2, 3]).allocate(
+        "sess-a", 3) == [0, 1, 2]
+    # impossible top-up raises, assignment unchanged
+    a.allocate("sess-b", 1)
+    with pytest.raises(errors.GPUUnavailable):
+        a.allocate("sess-a", 5)
+    assert a.allocate("sess-a", 3) == [0, 1, 2]
+
+
+def test_store_delete_leaves_lock_tombstone(tmp_path):
+    """delete() must not unlink the sidecar lock file: a process blocked
+    in flock on the old inode would otherwise hold a stale lock
+    concurrently with a new holder."""
+    from kukeon_amd.state.store import Store
+
+    store = Store(str(tmp_path / "run"))
+    doc = tmp_path / "run" / "thing.json"
+    store.create_exclusive(doc, {"spec": {}})
+    lockp = tmp_path / "run" / "thing.json.lock"
+    assert lockp.exists()
+    assert store.delete(doc)
+    assert not doc.exists()
+    assert lockp.exists()  # tombstone stays
+    # doc can be recreated and the same lock inode still guards it
+    ino_before = lockp.stat().st_ino
+    store.create_exclusive(doc, {"spec": {}})
+    assert lockp.stat().st_ino == ino_before

@@ -341,3 +341,33 @@ def test_engine_fuzz_kv_conservation(seed):
     for kv in kvs:
         engine.free_sequence(kv)
     assert engine.kv.allocator.num_free == ecfg.num_kv_blocks
+
+
+def test_multi_victim_eviction_unblocks_admission():
+    """Several small idle sessions whose combined KV would satisfy the
+    head request: the engine must keep evicting until admission unblocks
+    rather than raise MemoryError after one victim."""
+    torch.manual_seed(0)
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=512, max_sessions=8, num_kv_blocks=8,
+                        use_graphs=False)
+    model = LlamaModel(cfg, device="cpu")
+    engine = LLMEngine(model, cfg, ecfg, device="cpu")
+    idle = [SequenceKV(ecfg.block_size) for _ in range(3)]
+    for kv in idle:  # 24 tokens -> 2 blocks each; 6 of 8 blocks held idle
+        engine.add_request(kv, list(range(20)),
+                           SamplingParams(temperature=0.0, max_new_tokens=4))
+        while engine.has_work():
+            engine.step()
+    assert engine.kv.allocator.num_free == 2
+    # head request needs 5 blocks (72 prompt + 8 decode): two victims
+    big = SequenceKV(ecfg.block_size)
+    engine.add_request(big, list(range(72)),
+                       SamplingParams(temperature=0.0, max_new_tokens=8))
+    got = []
+    while engine.has_work():
+        for o in engine.step():
+            got.extend(o.new_tokens)
+    assert len(got) == 8
+    # at least two idle sessions were evicted
+    assert sum(1 for kv in idle if not kv.blocks) >= 2

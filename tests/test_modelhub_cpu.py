@@ -234,3 +234,29 @@ def test_server_concurrent_storm(tmp_path):
     real = [e for e in errs if not any(a in e[1] for a in allowed)]
     assert not real, real
     hub.stop()
+
+
+def test_engine_step_error_fails_pending_and_keeps_serving(hub):
+    """A raising engine.step must not silently wedge the daemon: pending
+    requests get an error frame and the loop keeps serving afterwards
+    (ADVICE r01). MemoryError (admission-level) -> continue."""
+    h, sock = hub
+    orig_step = h.engine.step
+    fired = {"n": 0}
+
+    def boom():
+        if fired["n"] == 0:
+            fired["n"] += 1
+            raise MemoryError("synthetic: cannot admit")
+        return orig_step()
+
+    h.engine.step = boom
+    c = ModelhubClient(sock, timeout=30)
+    import pytest as _pytest
+    with _pytest.raises(Exception) as ei:
+        c.generate("doomed", [1, 2, 3], max_new_tokens=2, temperature=0.0)
+    assert "engine step failed" in str(ei.value)
+    # the loop survived: a fresh request completes normally
+    r = c.generate("alive", [4, 5, 6], max_new_tokens=2, temperature=0.0)
+    assert len(r["tokens"]) == 2
+    c.close()
