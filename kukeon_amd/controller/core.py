@@ -29,6 +29,8 @@ from kukeon_amd.api import v1beta1 as api
 from kukeon_amd.controller import blueprint as bpmod
 from kukeon_amd.controller import diff as diffmod
 from kukeon_amd.controller import naming, parser
+from kukeon_amd.controller.locks import ScopeLocks, locked_cell
+from kukeon_amd.controller.spechash import SPEC_HASH_LABEL, spec_hash
 from kukeon_amd.controller.subnet import SubnetAllocator
 from kukeon_amd.netpolicy import Enforcer, NoopEnforcer, build_policy
 from kukeon_amd.runtime import diskpressure
@@ -84,6 +86,10 @@ class Controller:
             block_percent=self.server_config.disk_pressure_block_percent)
         self.now = now_fn
         self._restart_state: Dict[str, Tuple[float, int]] = {}
+        # per-cell scope locks: the daemon is multi-threaded (RPC threads
+        # + reconcile loop); every cell mutation holds its cell's lock —
+        # reference runner/runner.go:333-340
+        self.cell_locks = ScopeLocks()
 
     # ==================================================================
     # bootstrap
@@ -298,9 +304,16 @@ class Controller:
                     self.store.stack_dir(realm, space, stack))]
 
     def _persist_cell(self, doc: api.CellDoc) -> None:
+        """Status persist, generation-guarded: if the on-disk doc's
+        generation moved since `doc` was read (a concurrent apply/update
+        bumped the spec), raise StaleResource instead of silently
+        overwriting the newer spec with this stale copy (reference
+        runner/refresh.go:37-121 generation-guarded persist)."""
         path = self._cell_path(doc.spec.realm_id, doc.spec.space_id,
                                doc.spec.stack_id, doc.metadata.name)
-        self.store.write_cas(path, doc.to_dict(), bump=False)
+        expected = doc.metadata.generation or None
+        self.store.write_cas(path, doc.to_dict(), bump=False,
+                             expected_generation=expected)
 
     def _container_env(self, doc: api.CellDoc, c: api.ContainerSpec,
                        gpu_ids: List[int]) -> List[str]:
@@ -360,12 +373,16 @@ class Controller:
                 return api.SessionDoc.from_dict(data)
         return None
 
+    @locked_cell
     def start_cell(self, realm, space, stack, name) -> api.CellDoc:
         doc = self.get_cell(realm, space, stack, name)
         cell_dir = self.store.cell_dir(realm, space, stack, name)
         cg = self._cell_cgroup(doc)
-        # idempotency guard: all containers running -> no-op
-        if doc.status.state == api.STATE_READY and self._all_running(doc):
+        # idempotency guard: all containers running AND no spawn-spec
+        # drift -> no-op (reference start.go:587-620 + the spec-hash
+        # reuse contract of start.go:867+)
+        if doc.status.state == api.STATE_READY and self._all_running(doc) \
+                and not self._spec_drifted(doc):
             return doc
         self.cgroups.create(cg)
         started = []
@@ -386,10 +403,22 @@ class Controller:
                 cursor += c.gpus
                 env = self._container_env(doc, c, mine)
                 env += self._mount_volumes(doc, c, cdir)
+                h = spec_hash(c)
+                prev = self.store.read(cdir / METADATA_FILE) or {}
+                prev_hash = prev.get("metadata", {}).get(
+                    "labels", {}).get(SPEC_HASH_LABEL, "")
                 cdoc = api.ContainerDoc(
-                    metadata=api.Metadata(name=c.id or "main"), spec=c)
+                    metadata=api.Metadata(name=c.id or "main",
+                                          labels={SPEC_HASH_LABEL: h}),
+                    spec=c)
                 self.store.write(cdir / METADATA_FILE, cdoc.to_dict())
                 probe = self.runtime.probe(cdir)
+                if probe.running and prev_hash and prev_hash != h:
+                    # live container whose spawn spec drifted: recreate
+                    # (reference start.go:867+ spec-hash compare — reuse
+                    # only when the hash matches)
+                    self.runtime.kill(cdir)
+                    probe = self.runtime.probe(cdir)
                 if not probe.running:
                     self.runtime.start_container(cdir, c, env, cg)
                 started.append(cdir)
@@ -460,6 +489,21 @@ class Controller:
         s = doc.spec
         return f"cell:{s.realm_id}/{s.space_id}/{s.stack_id}/{doc.metadata.name}"
 
+    def _spec_drifted(self, doc: api.CellDoc) -> bool:
+        """True when any live container's stored spawn-spec hash differs
+        from the current spec's — start must then recreate it. Containers
+        without a stored hash (pre-hash records) count as un-drifted."""
+        cell_dir = self.store.cell_dir(doc.spec.realm_id, doc.spec.space_id,
+                                       doc.spec.stack_id, doc.metadata.name)
+        for c in doc.spec.containers:
+            prev = self.store.read(
+                cell_dir / (c.id or "main") / METADATA_FILE) or {}
+            stored = prev.get("metadata", {}).get(
+                "labels", {}).get(SPEC_HASH_LABEL, "")
+            if stored and stored != spec_hash(c):
+                return True
+        return False
+
     def _all_running(self, doc: api.CellDoc) -> bool:
         cell_dir = self.store.cell_dir(doc.spec.realm_id, doc.spec.space_id,
                                        doc.spec.stack_id, doc.metadata.name)
@@ -468,6 +512,7 @@ class Controller:
                 return False
         return True
 
+    @locked_cell
     def stop_cell(self, realm, space, stack, name,
                   grace_seconds: float = 10.0) -> api.CellDoc:
         doc = self.get_cell(realm, space, stack, name)
@@ -479,6 +524,7 @@ class Controller:
         self._persist_cell(doc)
         return doc
 
+    @locked_cell
     def kill_cell(self, realm, space, stack, name) -> api.CellDoc:
         doc = self.get_cell(realm, space, stack, name)
         cell_dir = self.store.cell_dir(realm, space, stack, name)
@@ -494,6 +540,7 @@ class Controller:
         self.stop_cell(realm, space, stack, name)
         return self.start_cell(realm, space, stack, name)
 
+    @locked_cell
     def delete_cell(self, realm, space, stack, name,
                     force: bool = False) -> None:
         try:
@@ -513,6 +560,7 @@ class Controller:
         self.store.delete_tree(self.store.cell_dir(realm, space, stack, name))
         self.cgroups.delete(self._cell_cgroup(doc))
 
+    @locked_cell
     def purge_cell(self, realm, space, stack, name) -> None:
         """Force residual-state removal even when metadata is damaged."""
         with contextlib.suppress(Exception):
@@ -530,6 +578,7 @@ class Controller:
         self.store.delete_tree(cell_dir)
         self.cgroups.delete(f"{realm}/{space}/{stack}/{name}")
 
+    @locked_cell
     def recreate_cell(self, doc: api.CellDoc) -> api.CellDoc:
         was_running = False
         try:
@@ -892,24 +941,29 @@ class Controller:
                 self.create_session(doc)
                 return ResourceResult(kind, name, "created")
         if kind == api.KIND_CELL:
-            try:
-                cur = self.get_cell(doc.spec.realm_id, doc.spec.space_id,
-                                    doc.spec.stack_id, name)
-            except errors.CellNotFound:
-                self.create_cell(doc)
-                return ResourceResult(kind, name, "created")
-            d = diffmod.diff_cell(doc, cur)
-            ct = d.change_type
-            if ct == diffmod.ChangeType.NONE:
-                return ResourceResult(kind, name, "unchanged")
-            if ct == diffmod.ChangeType.BREAKING:
-                self.recreate_cell(doc)
-                return ResourceResult(kind, name, "recreated")
-            cur.spec = doc.spec
-            self.store.write_cas(
-                self._cell_path(doc.spec.realm_id, doc.spec.space_id,
-                                doc.spec.stack_id, name), cur.to_dict())
-            return ResourceResult(kind, name, "updated")
+            with self.cell_locks.hold((doc.spec.realm_id, doc.spec.space_id,
+                                       doc.spec.stack_id, name)):
+                try:
+                    cur = self.get_cell(doc.spec.realm_id, doc.spec.space_id,
+                                        doc.spec.stack_id, name)
+                except errors.CellNotFound:
+                    self.create_cell(doc)
+                    return ResourceResult(kind, name, "created")
+                d = diffmod.diff_cell(doc, cur)
+                ct = d.change_type
+                if ct == diffmod.ChangeType.NONE:
+                    return ResourceResult(kind, name, "unchanged")
+                if ct == diffmod.ChangeType.BREAKING:
+                    self.recreate_cell(doc)
+                    return ResourceResult(kind, name, "recreated")
+                cur.spec = doc.spec
+                # spec update: bump generation, guarded against a racing
+                # writer between our read and this write
+                self.store.write_cas(
+                    self._cell_path(doc.spec.realm_id, doc.spec.space_id,
+                                    doc.spec.stack_id, name), cur.to_dict(),
+                    expected_generation=cur.metadata.generation or None)
+                return ResourceResult(kind, name, "updated")
         raise errors.ValidationError(f"kind {kind} is not applyable")
 
     # ==================================================================
@@ -930,6 +984,7 @@ class Controller:
                             visited += 1
         return visited
 
+    @locked_cell
     def reconcile_cell(self, realm, space, stack, name) -> api.CellDoc:
         doc = self.get_cell(realm, space, stack, name)
         if doc.status.state in (api.STATE_PENDING, api.STATE_STOPPED,
@@ -1015,7 +1070,16 @@ class Controller:
             doc.status.state = api.STATE_DEGRADED
         self._reconcile_outofsync(doc)
         doc.status.observed_generation = doc.metadata.generation
-        self._persist_cell(doc)
+        try:
+            self._persist_cell(doc)
+        except errors.StaleResource:
+            # a spec update landed between our read and this persist:
+            # skip the tick; the next pass re-derives against the new spec
+            # (reference refresh.go:51 ErrStaleResource skip)
+            log.info("reconcile of %s/%s/%s/%s skipped: resource moved "
+                     "underneath (stale generation)", realm, space, stack,
+                     name)
+            return doc
         if doc.spec.auto_delete and doc.status.state in \
                 api.TERMINAL_CELL_STATES:
             self.delete_cell(realm, space, stack, name, force=True)
