@@ -289,6 +289,186 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   }
 }
 
+// ===========================================================================
+// skinny GEMM v2 — G-tile walk with hand-counted W stream (round 2).
+//
+// v1's duty-cycle hole (ledger): the per-slice x re-stage was issued at the
+// END of slice s and drained at the TOP of slice s+1, so every slice paid
+// ~1.3-1.5k cycles of LDS-DMA landing latency in the open; and the W stream
+// was capped at 8 loads in flight per wave with a full drain per slice.
+// v2 restructures around a constant-count VMEM pipeline (all W loads are
+// inline-asm, invisible to hipcc, so its own counted waits can never
+// over-wait on the staging DMAs — guide §5 ".s-level traps" (b)):
+//   * each block owns G=2 N-tiles of 64 rows and walks K slices, so one
+//     16-glds x stage serves 2x the W stream, and the stage for slice s+1
+//     is issued at the TOP of slice s — its landing deadline is the end of
+//     the slice, giving it the full 2-tile consume (~2k cycles) as cover;
+//   * W chunks carry hand-counted s_waitcnt vmcnt(K) per consume with the
+//     glds depth folded into the constants, so nothing ever drains early;
+//   * tile0's loads for slice s+1 are issued mid-slice-s (between the two
+//     consumes), keeping 8+ W loads in flight across the barrier — the
+//     stream never goes dry at a slice boundary.
+// Requires K%256==0, N%128==0 (all decode shapes qualify).
+// ===========================================================================
+
+__global__ void skinny_reduce_kernel(unsigned short* __restrict__ out,
+                                     const float* __restrict__ ws, long n,
+                                     int splitk);
+
+DEV_INLINE void issue_w8(u32x4_t (&reg)[8], const unsigned short* p) {
+  // 8 nt 16B loads at 64 B stride (one 256-k slice of one W row per lane
+  // group); hipcc does not count these — waits are hand-placed.
+#define SK2_LD(u, off)                                                    \
+  asm volatile("global_load_dwordx4 %0, %1, off offset:" #off " nt"      \
+               : "=&v"(reg[u]) : "v"(p) : "memory")
+  SK2_LD(0, 0);   SK2_LD(1, 64);  SK2_LD(2, 128); SK2_LD(3, 192);
+  SK2_LD(4, 256); SK2_LD(5, 320); SK2_LD(6, 384); SK2_LD(7, 448);
+#undef SK2_LD
+}
+
+// consume one tile's 8 W chunks: BASE = VMEM ops younger than chunk 7 at
+// its wait (so chunk u waits vmcnt(BASE + 7 - u)).
+template <int MT, int BASE>
+DEV_INLINE void consume8(u32x4_t (&wreg)[8], const unsigned short* xb,
+                         f32x4_t (&acc)[MT], int row16, int kgrp) {
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    asm volatile("s_waitcnt vmcnt(%c1)"
+                 : "+v"(wreg[u]) : "i"(BASE + 7 - u));
+    const bf16x8_t af = frag_of(wreg[u]);
+    const int kc = u * 32 + 8 * kgrp;
+#pragma unroll
+    for (int m = 0; m < MT; ++m) {
+      const int xr = m * 16 + row16;
+      const uint4 xv = *reinterpret_cast<const uint4*>(
+          reinterpret_cast<const char*>(xb) + xswz(xr, kc * 2));
+      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af, frag_of(xv), acc[m], 0, 0, 0);
+    }
+  }
+}
+
+template <int MT, bool SPLIT>
+__global__ __launch_bounds__(256) void skinny2_kernel(
+    unsigned short* __restrict__ out,      // [M, N] bf16 (SPLIT=false)
+    float* __restrict__ ws,                // [splitk, M, N] f32 (SPLIT)
+    const unsigned short* __restrict__ x,  // [M, K]
+    const unsigned short* __restrict__ w,  // [N, K]
+    int M, int N, long K) {
+  __shared__ __align__(16) unsigned short xbuf[2][64 * KSLICE];
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int row16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int n0t0 = (blockIdx.x * 2 + 0) * 64 + wid * 16;
+  const int n0t1 = (blockIdx.x * 2 + 1) * 64 + wid * 16;
+  f32x4_t acc0[MT], acc1[MT];
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    acc0[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+    acc1[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+  }
+  const long kadv = (long)gridDim.y * KSLICE;   // strided slice walk
+  const long ks0 = (long)blockIdx.y * KSLICE;
+  if (ks0 >= K) return;
+  u32x4_t w0[8], w1[8];
+  const unsigned short* p0 = w + (long)(n0t0 + row16) * K + ks0 + 8 * kgrp;
+  const unsigned short* p1 = w + (long)(n0t1 + row16) * K + ks0 + 8 * kgrp;
+
+  // prologue: x slice 0 stage + tile0 W in flight => loop-top vmcnt(8)
+  // waits exactly the 16 glds
+  glds_stage_x(xbuf[0], x, K, ks0, KSLICE, M, wid, lane);
+  issue_w8(w0, p0);
+  int cur = 0;
+  for (long ks = ks0; ks < K; ks += kadv, cur ^= 1) {
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    __syncthreads();
+    const unsigned short* xb = xbuf[cur];
+    const long ksn = ks + kadv;
+    // stage slice s+1 NOW: its deadline is the end of this slice's
+    // consume (the vmcnt(8) at tile1's final chunk), ~2k cycles away.
+    // Always 16 DMAs (clamped to slice 0 past the end) so every wait
+    // constant below holds on every iteration.
+    glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
+                 wid, lane);
+    issue_w8(w1, p1);                       // outstanding: 8+16+8
+    consume8<MT, 24>(w0, xb, acc0, row16, kgrp);
+    const unsigned short* p0n = (ksn < K) ? p0 + kadv : p0;
+    issue_w8(w0, p0n);                      // tile0 of the NEXT slice
+    consume8<MT, 8>(w1, xb, acc1, row16, kgrp);
+    p0 = p0n;
+    if (ksn < K) p1 += kadv;
+  }
+
+  const int ncol0 = n0t0 + 4 * kgrp;
+  const int ncol1 = n0t1 + 4 * kgrp;
+  float* slab = SPLIT ? ws + (long)blockIdx.y * M * N : nullptr;
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    const int mrow = m * 16 + row16;
+    if (mrow >= M) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long off0 = (long)mrow * N + ncol0 + r;
+      const long off1 = (long)mrow * N + ncol1 + r;
+      if (SPLIT) {
+        slab[off0] = acc0[m][r];
+        slab[off1] = acc1[m][r];
+      } else {
+        out[off0] = f2us(acc0[m][r]);
+        out[off1] = f2us(acc1[m][r]);
+      }
+    }
+  }
+}
+
+void skinny_gemm2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                  torch::Tensor ws) {
+  const int M = x.size(0);
+  const long K = x.size(1);
+  const int N = w.size(0);
+  TORCH_CHECK(M >= 1 && M <= 64, "skinny_gemm2: M in [1,64]");
+  TORCH_CHECK(N % 128 == 0 && K % KSLICE == 0,
+              "skinny_gemm2: N%128, K%256");
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int ngroups = N / 128;
+  const int nslices = (int)(K / KSLICE);
+  int splitk = 1;
+  if (ngroups < 256) splitk = min(nslices, (256 + ngroups - 1) / ngroups);
+  if (const char* ov = getenv("KUKEON_SK2_SPLITK")) {
+    const int v = atoi(ov);
+    if (v > 0) splitk = min(nslices, v);
+  }
+  const int MT = (M + 15) / 16;
+  dim3 grid(ngroups, splitk);
+  auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
+  auto* xp = reinterpret_cast<const unsigned short*>(x.data_ptr());
+  auto* wp = reinterpret_cast<const unsigned short*>(w.data_ptr());
+  const long total = (long)M * N;
+#define SK2_LAUNCH(MT_)                                                      \
+  if (splitk == 1) {                                                         \
+    skinny2_kernel<MT_, false><<<grid, 256, 0, stream>>>(                    \
+        op, nullptr, xp, wp, M, N, K);                                       \
+  } else {                                                                   \
+    float* wsp = ws.data_ptr<float>();                                       \
+    TORCH_CHECK(ws.numel() >= total * splitk, "sk2 workspace too small");    \
+    skinny2_kernel<MT_, true><<<grid, 256, 0, stream>>>(                     \
+        nullptr, wsp, xp, wp, M, N, K);                                      \
+    skinny_reduce_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256,   \
+                           0, stream>>>(op, wsp, total, splitk);             \
+  }
+  switch (MT) {
+    case 1: SK2_LAUNCH(1); break;
+    case 2: SK2_LAUNCH(2); break;
+    case 3: SK2_LAUNCH(3); break;
+    default: SK2_LAUNCH(4); break;
+  }
+#undef SK2_LAUNCH
+  HIP_CHECK_KERNEL();
+}
+
 // reduce the split-K slabs and cast to bf16
 __global__ void skinny_reduce_kernel(unsigned short* __restrict__ out,
                                      const float* __restrict__ ws, long n,

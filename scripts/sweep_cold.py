@@ -54,4 +54,12 @@ for (M, N, K, tag) in SHAPES:
         us = t_rot(lambda i: _C.skinny_gemm(out, x, ws_list[i], wrk), nw)
         line += f"  sk{sk or 'A'}={us:6.1f}"
     os.environ.pop("KUKEON_SKINNY_SPLITK", None)
+    for sk in (0, 1, 2, 4, 5, 8, 16):
+        if sk:
+            os.environ["KUKEON_SK2_SPLITK"] = str(sk)
+        else:
+            os.environ.pop("KUKEON_SK2_SPLITK", None)
+        us = t_rot(lambda i: _C.skinny_gemm2(out, x, ws_list[i], wrk), nw)
+        line += f"  v2k{sk or 'A'}={us:6.1f}"
+    os.environ.pop("KUKEON_SK2_SPLITK", None)
     print(line, flush=True)

@@ -369,6 +369,28 @@ def test_skinny_gemm(M, N, K):
                                atol=3e-2)
 
 
+@pytest.mark.parametrize("M,N,K,splitk", [
+    (1, 4096, 4096, 0), (8, 6144, 4096, 0), (33, 4096, 14336, 0),
+    (64, 28672, 4096, 0), (64, 128256, 4096, 0), (64, 4096, 4096, 0),
+    (64, 4096, 14336, 8), (64, 6144, 4096, 5), (64, 28672, 4096, 2),
+    (17, 6144, 4096, 3)])
+def test_skinny_gemm2(M, N, K, splitk, monkeypatch):
+    """v2 G-walk hand-counted kernel vs fp32 reference, incl. strided
+    split-K walks with uneven slice counts per block."""
+    from kukeon_amd import _C
+    if splitk:
+        monkeypatch.setenv("KUKEON_SK2_SPLITK", str(splitk))
+    torch.manual_seed(13)
+    x = (torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.5)
+    w = (torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05)
+    out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+    ws = torch.empty(16 * 64 * N, dtype=torch.float32, device=DEV)
+    _C.skinny_gemm2(out, x, w, ws)
+    ref = (x.float() @ w.float().T)
+    torch.testing.assert_close(out.float().cpu(), ref.cpu(), rtol=3e-2,
+                               atol=3e-2)
+
+
 @pytest.mark.parametrize("kv_dtype", ["bf16", "fp8"])
 def test_decode_graphs_match_eager(kv_dtype):
     """Self-advancing graph decode must produce the same tokens as the
