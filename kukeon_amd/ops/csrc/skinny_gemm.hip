@@ -399,6 +399,10 @@ __global__ __launch_bounds__(256) void skinny2_kernel(
     p0 = p0n;
     if (ksn < K) p1 += kadv;
   }
+  // drain the dangling tile0-next loads (hipcc cannot see them, so it
+  // emits no vmcnt(0) before s_endpgm; a load landing after the wave
+  // slot is re-issued to the NEXT kernel's waves corrupts their VGPRs)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 
   const int ncol0 = n0t0 + 4 * kgrp;
   const int ncol1 = n0t1 + 4 * kgrp;
