@@ -89,8 +89,8 @@ DEV_INLINE unsigned lds_addr_of(const void* p) {
   return (unsigned)(unsigned long long)p;
 }
 
-// Stage rows [0,64) x KSLICE cols of x into the xswz LDS image with 16
-// asm glds per wave (wave w owns rows 16w..16w+15). Completion rides the
+// Stage rows [0,64) x KSLICE cols of x into the xswz LDS image with 8
+// asm glds per wave (each 1 KiB instruction covers two 512-B rows) (wave w owns rows 16w..16w+15). Completion rides the
 // VM counter; callers drain with s_waitcnt vmcnt + barrier. For tail
 // slices (klen < KSLICE) the per-lane source byte offset is clamped
 // in-bounds — the over-staged slots hold junk the consumer never reads
@@ -376,7 +376,7 @@ __global__ __launch_bounds__(256) void skinny2_kernel(
   const unsigned short* p1 = w + (long)(n0t1 + row16) * K + ks0 + 8 * kgrp;
 
   // prologue: x slice 0 stage + tile0 W in flight => loop-top vmcnt(8)
-  // waits exactly the 16 glds
+  // waits exactly the 8 glds (one per row pair)
   glds_stage_x(xbuf[0], x, K, ks0, KSLICE, M, wid, lane);
   issue_w8(w0, p0);
   int cur = 0;
@@ -387,12 +387,12 @@ __global__ __launch_bounds__(256) void skinny2_kernel(
     const long ksn = ks + kadv;
     // stage slice s+1 NOW: its deadline is the end of this slice's
     // consume (the vmcnt(8) at tile1's final chunk), ~2k cycles away.
-    // Always 16 DMAs (clamped to slice 0 past the end) so every wait
+    // Always 8 DMAs (clamped to slice 0 past the end) so every wait
     // constant below holds on every iteration.
     glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
                  wid, lane);
-    issue_w8(w1, p1);                       // outstanding: 8+16+8
-    consume8<MT, 24>(w0, xb, acc0, row16, kgrp);
+    issue_w8(w1, p1);      // outstanding: w0(8) + glds(8) + w1(8)
+    consume8<MT, 16>(w0, xb, acc0, row16, kgrp);
     const unsigned short* p0n = (ksn < K) ? p0 + kadv : p0;
     issue_w8(w0, p0n);                      // tile0 of the NEXT slice
     consume8<MT, 8>(w1, xb, acc1, row16, kgrp);
