@@ -348,7 +348,7 @@ DEV_INLINE void consume8(u32x4_t (&wreg)[8], const unsigned short* xb,
   }
 }
 
-template <int MT, bool SPLIT>
+template <int MT, bool SPLIT, bool SERIAL = false>
 __global__ __launch_bounds__(256) void skinny2_kernel(
     unsigned short* __restrict__ out,      // [M, N] bf16 (SPLIT=false)
     float* __restrict__ ws,                // [splitk, M, N] f32 (SPLIT)
@@ -375,27 +375,35 @@ __global__ __launch_bounds__(256) void skinny2_kernel(
   const unsigned short* p0 = w + (long)(n0t0 + row16) * K + ks0 + 8 * kgrp;
   const unsigned short* p1 = w + (long)(n0t1 + row16) * K + ks0 + 8 * kgrp;
 
-  // prologue: x slice 0 stage + tile0 W in flight => loop-top vmcnt(8)
-  // waits exactly the 8 glds (one per row pair)
-  glds_stage_x(xbuf[0], x, K, ks0, KSLICE, M, wid, lane);
+  // prologue: tile0 W issued FIRST so its destination registers are
+  // live when the glds source addresses are computed — the allocator
+  // must then keep them disjoint (a load landing into a register the
+  // DMA engine still holds as a source address was the round-2 wild-
+  // write fault). One full drain before the loop keeps the loop-top
+  // wait constant.
   issue_w8(w0, p0);
+  glds_stage_x(xbuf[0], x, K, ks0, KSLICE, M, wid, lane);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   int cur = 0;
   for (long ks = ks0; ks < K; ks += kadv, cur ^= 1) {
+    // glds(s+1) is older than w0(s+1): <=8 outstanding => glds landed
     asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
     __syncthreads();
     const unsigned short* xb = xbuf[cur];
     const long ksn = ks + kadv;
-    // stage slice s+1 NOW: its deadline is the end of this slice's
-    // consume (the vmcnt(8) at tile1's final chunk), ~2k cycles away.
-    // Always 8 DMAs (clamped to slice 0 past the end) so every wait
-    // constant below holds on every iteration.
+    issue_w8(w1, p1);
+    // stage slice s+1 with BOTH W register sets live (disjoint regs);
+    // its landing deadline is the next loop-top vmcnt(8), a whole
+    // 2-tile consume away. Always 8 DMAs (clamped to slice 0 past the
+    // end) so every wait constant holds on every iteration.
     glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
                  wid, lane);
-    issue_w8(w1, p1);      // outstanding: w0(8) + glds(8) + w1(8)
+    if (SERIAL) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     consume8<MT, 16>(w0, xb, acc0, row16, kgrp);
     const unsigned short* p0n = (ksn < K) ? p0 + kadv : p0;
     issue_w8(w0, p0n);                      // tile0 of the NEXT slice
-    consume8<MT, 8>(w1, xb, acc1, row16, kgrp);
+    if (SERIAL) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    consume8<MT, 16>(w1, xb, acc1, row16, kgrp);
     p0 = p0n;
     if (ksn < K) p1 += kadv;
   }
@@ -462,6 +470,13 @@ void skinny_gemm2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
         nullptr, wsp, xp, wp, M, N, K);                                      \
     skinny_reduce_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256,   \
                            0, stream>>>(op, wsp, total, splitk);             \
+  }
+  const bool serial = getenv("KUKEON_SK2_SERIAL") != nullptr;
+  if (serial && MT == 4 && splitk == 1) {
+    skinny2_kernel<4, false, true><<<grid, 256, 0, stream>>>(
+        op, nullptr, xp, wp, M, N, K);
+    HIP_CHECK_KERNEL();
+    return;
   }
   switch (MT) {
     case 1: SK2_LAUNCH(1); break;
