@@ -348,7 +348,7 @@ DEV_INLINE void consume8(u32x4_t (&wreg)[8], const unsigned short* xb,
   }
 }
 
-template <int MT, bool SPLIT, bool SERIAL = false>
+template <int MT, bool SPLIT, int SMASK = 0>
 __global__ __launch_bounds__(256) void skinny2_kernel(
     unsigned short* __restrict__ out,      // [M, N] bf16 (SPLIT=false)
     float* __restrict__ ws,                // [splitk, M, N] f32 (SPLIT)
@@ -398,12 +398,13 @@ __global__ __launch_bounds__(256) void skinny2_kernel(
     // end) so every wait constant holds on every iteration.
     glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
                  wid, lane);
-    if (SERIAL) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (SMASK & 1) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     consume8<MT, 16>(w0, xb, acc0, row16, kgrp);
     const unsigned short* p0n = (ksn < K) ? p0 + kadv : p0;
     issue_w8(w0, p0n);                      // tile0 of the NEXT slice
-    if (SERIAL) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (SMASK & 4) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     consume8<MT, 16>(w1, xb, acc1, row16, kgrp);
+    if (SMASK & 2) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     p0 = p0n;
     if (ksn < K) p1 += kadv;
   }
@@ -471,10 +472,21 @@ void skinny_gemm2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
     skinny_reduce_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256,   \
                            0, stream>>>(op, wsp, total, splitk);             \
   }
-  const bool serial = getenv("KUKEON_SK2_SERIAL") != nullptr;
-  if (serial && MT == 4 && splitk == 1) {
-    skinny2_kernel<4, false, true><<<grid, 256, 0, stream>>>(
-        op, nullptr, xp, wp, M, N, K);
+  const char* sm = getenv("KUKEON_SK2_SERIAL");
+  if (sm && MT == 4 && splitk == 1) {
+    const int mask = atoi(sm);
+    if (mask == 1)
+      skinny2_kernel<4, false, 1><<<grid, 256, 0, stream>>>(
+          op, nullptr, xp, wp, M, N, K);
+    else if (mask == 2)
+      skinny2_kernel<4, false, 2><<<grid, 256, 0, stream>>>(
+          op, nullptr, xp, wp, M, N, K);
+    else if (mask == 4)
+      skinny2_kernel<4, false, 4><<<grid, 256, 0, stream>>>(
+          op, nullptr, xp, wp, M, N, K);
+    else
+      skinny2_kernel<4, false, 7><<<grid, 256, 0, stream>>>(
+          op, nullptr, xp, wp, M, N, K);
     HIP_CHECK_KERNEL();
     return;
   }
