@@ -388,7 +388,7 @@ __global__ __launch_bounds__(256) void skinny2_kernel(
   for (long ks = ks0; ks < K; ks += kadv, cur ^= 1) {
     // glds(s+1) is older than w0(s+1): <=8 outstanding => glds landed
     asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    __syncthreads();
+    if (!(SMASK & 16)) __syncthreads();
     const unsigned short* xb = xbuf[cur];
     const long ksn = ks + kadv;
     issue_w8(w1, p1);
@@ -396,8 +396,9 @@ __global__ __launch_bounds__(256) void skinny2_kernel(
     // its landing deadline is the next loop-top vmcnt(8), a whole
     // 2-tile consume away. Always 8 DMAs (clamped to slice 0 past the
     // end) so every wait constant holds on every iteration.
-    glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
-                 wid, lane);
+    if (!(SMASK & 8))
+      glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
+                   wid, lane);
     if (SMASK & 1) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     consume8<MT, 16>(w0, xb, acc0, row16, kgrp);
     const unsigned short* p0n = (ksn < K) ? p0 + kadv : p0;
@@ -462,12 +463,12 @@ void skinny_gemm2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   const long total = (long)M * N;
 #define SK2_LAUNCH(MT_)                                                      \
   if (splitk == 1) {                                                         \
-    skinny2_kernel<MT_, false><<<grid, 256, 0, stream>>>(                    \
+    skinny2_kernel<MT_, false, 2><<<grid, 256, 0, stream>>>(                 \
         op, nullptr, xp, wp, M, N, K);                                       \
   } else {                                                                   \
     float* wsp = ws.data_ptr<float>();                                       \
     TORCH_CHECK(ws.numel() >= total * splitk, "sk2 workspace too small");    \
-    skinny2_kernel<MT_, true><<<grid, 256, 0, stream>>>(                     \
+    skinny2_kernel<MT_, true, 2><<<grid, 256, 0, stream>>>(                  \
         nullptr, wsp, xp, wp, M, N, K);                                      \
     skinny_reduce_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256,   \
                            0, stream>>>(op, wsp, total, splitk);             \
@@ -475,18 +476,22 @@ void skinny_gemm2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   const char* sm = getenv("KUKEON_SK2_SERIAL");
   if (sm && MT == 4 && splitk == 1) {
     const int mask = atoi(sm);
-    if (mask == 1)
-      skinny2_kernel<4, false, 1><<<grid, 256, 0, stream>>>(
-          op, nullptr, xp, wp, M, N, K);
-    else if (mask == 2)
-      skinny2_kernel<4, false, 2><<<grid, 256, 0, stream>>>(
-          op, nullptr, xp, wp, M, N, K);
-    else if (mask == 4)
-      skinny2_kernel<4, false, 4><<<grid, 256, 0, stream>>>(
-          op, nullptr, xp, wp, M, N, K);
-    else
-      skinny2_kernel<4, false, 7><<<grid, 256, 0, stream>>>(
-          op, nullptr, xp, wp, M, N, K);
+    switch (mask) {
+      case 1: skinny2_kernel<4, false, 1><<<grid, 256, 0, stream>>>(
+                  op, nullptr, xp, wp, M, N, K); break;
+      case 2: skinny2_kernel<4, false, 2><<<grid, 256, 0, stream>>>(
+                  op, nullptr, xp, wp, M, N, K); break;
+      case 4: skinny2_kernel<4, false, 4><<<grid, 256, 0, stream>>>(
+                  op, nullptr, xp, wp, M, N, K); break;
+      case 8: skinny2_kernel<4, false, 8><<<grid, 256, 0, stream>>>(
+                  op, nullptr, xp, wp, M, N, K); break;
+      case 16: skinny2_kernel<4, false, 16><<<grid, 256, 0, stream>>>(
+                  op, nullptr, xp, wp, M, N, K); break;
+      case 24: skinny2_kernel<4, false, 24><<<grid, 256, 0, stream>>>(
+                  op, nullptr, xp, wp, M, N, K); break;
+      default: skinny2_kernel<4, false, 7><<<grid, 256, 0, stream>>>(
+                  op, nullptr, xp, wp, M, N, K); break;
+    }
     HIP_CHECK_KERNEL();
     return;
   }
