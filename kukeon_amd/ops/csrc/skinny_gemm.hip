@@ -400,11 +400,11 @@ __global__ __launch_bounds__(256) void skinny2_kernel(
       glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
                    wid, lane);
     if (SMASK & 1) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    consume8<MT, 16>(w0, xb, acc0, row16, kgrp);
+    if (!(SMASK & 32)) consume8<MT, 16>(w0, xb, acc0, row16, kgrp);
     const unsigned short* p0n = (ksn < K) ? p0 + kadv : p0;
     issue_w8(w0, p0n);                      // tile0 of the NEXT slice
     if (SMASK & 4) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    consume8<MT, 16>(w1, xb, acc1, row16, kgrp);
+    if (!(SMASK & 32)) consume8<MT, 16>(w1, xb, acc1, row16, kgrp);
     if (SMASK & 2) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     p0 = p0n;
     if (ksn < K) p1 += kadv;
@@ -488,6 +488,12 @@ void skinny_gemm2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
       case 16: skinny2_kernel<4, false, 16><<<grid, 256, 0, stream>>>(
                   op, nullptr, xp, wp, M, N, K); break;
       case 24: skinny2_kernel<4, false, 24><<<grid, 256, 0, stream>>>(
+                  op, nullptr, xp, wp, M, N, K); break;
+      case 34: skinny2_kernel<4, false, 34><<<grid, 256, 0, stream>>>(
+                  op, nullptr, xp, wp, M, N, K); break;
+      case 42: skinny2_kernel<4, false, 42><<<grid, 256, 0, stream>>>(
+                  op, nullptr, xp, wp, M, N, K); break;
+      case 10: skinny2_kernel<4, false, 10><<<grid, 256, 0, stream>>>(
                   op, nullptr, xp, wp, M, N, K); break;
       default: skinny2_kernel<4, false, 7><<<grid, 256, 0, stream>>>(
                   op, nullptr, xp, wp, M, N, K); break;
