@@ -375,43 +375,45 @@ __global__ __launch_bounds__(256) void skinny2_kernel(
   const unsigned short* p0 = w + (long)(n0t0 + row16) * K + ks0 + 8 * kgrp;
   const unsigned short* p1 = w + (long)(n0t1 + row16) * K + ks0 + 8 * kgrp;
 
-  // prologue: tile0 W issued FIRST so its destination registers are
-  // live when the glds source addresses are computed — the allocator
-  // must then keep them disjoint (a load landing into a register the
-  // DMA engine still holds as a source address was the round-2 wild-
-  // write fault). One full drain before the loop keeps the loop-top
-  // wait constant.
-  issue_w8(w0, p0);
+  // Stream discipline (v3, measured): draining the W stream to zero at
+  // a slice boundary costs the full LADEN memory latency (~2-4us under
+  // load) per slice — the ablation showed the consume path is entirely
+  // hidden and the drain dominates. So 16 W loads stay in flight across
+  // every barrier, and the only forced retirement is the x-stage DMA
+  // (vmcnt(16) at the slice top). Issue order inside a slice mirrors the
+  // validated v1 pattern: glds mid-slice, then the next slice's W loads
+  // (crossing loads YOUNGER than the DMA — the reverse order faulted).
+  // prologue order = the loop's invariant order: glds oldest, then the
+  // two W tile sets — the loop-top vmcnt(16) retires exactly the glds
+  // while the 16 W loads keep flying
   glds_stage_x(xbuf[0], x, K, ks0, KSLICE, M, wid, lane);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  issue_w8(w0, p0);
+  issue_w8(w1, p1);
   int cur = 0;
   for (long ks = ks0; ks < K; ks += kadv, cur ^= 1) {
-    // glds(s+1) is older than w0(s+1): <=8 outstanding => glds landed
-    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    if (!(SMASK & 16)) __syncthreads();
+    // invariant on entry: glds(s) retired or retiring, w0(s)+w1(s) in
+    // flight (16 loads). vmcnt(16) forces glds(s) landed (it is older
+    // than the 16 W loads issued after it — except on the first
+    // iteration, handled by the prologue wait above).
+    asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+    __syncthreads();
     const unsigned short* xb = xbuf[cur];
     const long ksn = ks + kadv;
-    issue_w8(w1, p1);
-    // stage slice s+1 with BOTH W register sets live (disjoint regs);
-    // its landing deadline is the next loop-top vmcnt(8), a whole
-    // 2-tile consume away. Always 8 DMAs (clamped to slice 0 past the
-    // end) so every wait constant holds on every iteration.
-    if (!(SMASK & 8))
-      glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
-                   wid, lane);
-    if (SMASK & 1) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    if (!(SMASK & 32)) consume8<MT, 16>(w0, xb, acc0, row16, kgrp);
+    consume8<MT, 8>(w0, xb, acc0, row16, kgrp);      // vmcnt(15-u)
+    // stage slice s+1, then its W loads (younger than the DMA)
+    glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
+                 wid, lane);
     const unsigned short* p0n = (ksn < K) ? p0 + kadv : p0;
-    issue_w8(w0, p0n);                      // tile0 of the NEXT slice
-    if (SMASK & 4) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    if (!(SMASK & 32)) consume8<MT, 16>(w1, xb, acc1, row16, kgrp);
-    if (SMASK & 2) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    issue_w8(w0, p0n);
+    consume8<MT, 16>(w1, xb, acc1, row16, kgrp);     // vmcnt(23-u)
+    const unsigned short* p1n = (ksn < K) ? p1 + kadv : p1;
+    issue_w8(w1, p1n);
     p0 = p0n;
-    if (ksn < K) p1 += kadv;
+    p1 = p1n;
   }
-  // drain the dangling tile0-next loads (hipcc cannot see them, so it
-  // emits no vmcnt(0) before s_endpgm; a load landing after the wave
-  // slot is re-issued to the NEXT kernel's waves corrupts their VGPRs)
+  // drain the cross-boundary loads (hipcc cannot see them, so it emits
+  // no vmcnt(0) before s_endpgm; a load landing after the wave slot is
+  // re-issued to the NEXT kernel's waves corrupts their VGPRs)
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 
   const int ncol0 = n0t0 + 4 * kgrp;
