@@ -30,6 +30,8 @@ void mfma_probe16(torch::Tensor out, torch::Tensor a, torch::Tensor b);
 void mfma_probe16k(torch::Tensor out, torch::Tensor a, torch::Tensor b);
 void skinny_gemm2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                   torch::Tensor ws);
+void skinny_gemm4(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                  torch::Tensor ws);
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws);
 void glds_probe(torch::Tensor out, torch::Tensor src);
@@ -64,6 +66,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "16x16x32 bf16 MFMA fragment-layout probe");
   m.def("mfma_probe16k", &kukeon::mfma_probe16k,
         "16x16x16 (K=16) bf16 MFMA fragment-layout probe");
+  m.def("skinny_gemm4", &kukeon::skinny_gemm4,
+        "G=4 deep-pipeline weight-streaming GEMM");
   m.def("skinny_gemm2", &kukeon::skinny_gemm2,
         "G-walk hand-counted weight-streaming GEMM (v2)");
   m.def("skinny_gemm", &kukeon::skinny_gemm,

@@ -438,6 +438,139 @@ __global__ __launch_bounds__(256) void skinny2_kernel(
   }
 }
 
+// G=4 variant: four N-tiles per block, four static W register sets, 32
+// loads in flight per wave (128 KB per CU) — sized so the laden memory
+// latency under full load (~3 us) still sustains the per-CU share of HBM
+// bandwidth by Little's law. Same never-drain discipline as skinny2 v3.
+template <int MT, bool SPLIT>
+__global__ __launch_bounds__(256) void skinny4_kernel(
+    unsigned short* __restrict__ out, float* __restrict__ ws,
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ w, int M, int N, long K) {
+  __shared__ __align__(16) unsigned short xbuf[2][64 * KSLICE];
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int row16 = lane & 15;
+  const int kgrp = lane >> 4;
+  f32x4_t acc0[MT], acc1[MT], acc2[MT], acc3[MT];
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    acc0[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+    acc1[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+    acc2[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+    acc3[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+  }
+  const long kadv = (long)gridDim.y * KSLICE;
+  const long ks0 = (long)blockIdx.y * KSLICE;
+  if (ks0 >= K) return;
+  u32x4_t w0[8], w1[8], w2[8], w3[8];
+  const int nb = blockIdx.x * 4;
+  const unsigned short* p0 =
+      w + (long)((nb + 0) * 64 + wid * 16 + row16) * K + ks0 + 8 * kgrp;
+  const unsigned short* p1 =
+      w + (long)((nb + 1) * 64 + wid * 16 + row16) * K + ks0 + 8 * kgrp;
+  const unsigned short* p2 =
+      w + (long)((nb + 2) * 64 + wid * 16 + row16) * K + ks0 + 8 * kgrp;
+  const unsigned short* p3 =
+      w + (long)((nb + 3) * 64 + wid * 16 + row16) * K + ks0 + 8 * kgrp;
+
+  glds_stage_x(xbuf[0], x, K, ks0, KSLICE, M, wid, lane);
+  issue_w8(w0, p0);
+  issue_w8(w1, p1);
+  issue_w8(w2, p2);
+  issue_w8(w3, p3);
+  int cur = 0;
+  for (long ks = ks0; ks < K; ks += kadv, cur ^= 1) {
+    // entry: glds(s) oldest, t0..t3 (32 loads) younger
+    asm volatile("s_waitcnt vmcnt(32)" ::: "memory");
+    __syncthreads();
+    const unsigned short* xb = xbuf[cur];
+    const long ksn = ks + kadv;
+    consume8<MT, 24>(w0, xb, acc0, row16, kgrp);   // vmcnt(31-u)
+    glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
+                 wid, lane);
+    const unsigned short* p0n = (ksn < K) ? p0 + kadv : p0;
+    issue_w8(w0, p0n);
+    consume8<MT, 32>(w1, xb, acc1, row16, kgrp);   // vmcnt(39-u)
+    const unsigned short* p1n = (ksn < K) ? p1 + kadv : p1;
+    issue_w8(w1, p1n);
+    consume8<MT, 32>(w2, xb, acc2, row16, kgrp);
+    const unsigned short* p2n = (ksn < K) ? p2 + kadv : p2;
+    issue_w8(w2, p2n);
+    consume8<MT, 32>(w3, xb, acc3, row16, kgrp);
+    const unsigned short* p3n = (ksn < K) ? p3 + kadv : p3;
+    issue_w8(w3, p3n);
+    p0 = p0n; p1 = p1n; p2 = p2n; p3 = p3n;
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  float* slab = SPLIT ? ws + (long)blockIdx.y * M * N : nullptr;
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    const int mrow = m * 16 + row16;
+    if (mrow >= M) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long o0 = (long)mrow * N + (nb + 0) * 64 + wid * 16 + 4 * kgrp + r;
+      const long o1 = (long)mrow * N + (nb + 1) * 64 + wid * 16 + 4 * kgrp + r;
+      const long o2 = (long)mrow * N + (nb + 2) * 64 + wid * 16 + 4 * kgrp + r;
+      const long o3 = (long)mrow * N + (nb + 3) * 64 + wid * 16 + 4 * kgrp + r;
+      if (SPLIT) {
+        slab[o0] = acc0[m][r]; slab[o1] = acc1[m][r];
+        slab[o2] = acc2[m][r]; slab[o3] = acc3[m][r];
+      } else {
+        out[o0] = f2us(acc0[m][r]); out[o1] = f2us(acc1[m][r]);
+        out[o2] = f2us(acc2[m][r]); out[o3] = f2us(acc3[m][r]);
+      }
+    }
+  }
+}
+
+void skinny_gemm4(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                  torch::Tensor ws) {
+  const int M = x.size(0);
+  const long K = x.size(1);
+  const int N = w.size(0);
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 256 == 0 && K % KSLICE == 0);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int ngroups = N / 256;
+  const int nslices = (int)(K / KSLICE);
+  int splitk = 1;
+  if (ngroups < 256) splitk = min(nslices, (256 + ngroups - 1) / ngroups);
+  if (const char* ov = getenv("KUKEON_SK4_SPLITK")) {
+    const int v = atoi(ov);
+    if (v > 0) splitk = min(nslices, v);
+  }
+  const int MT = (M + 15) / 16;
+  dim3 grid(ngroups, splitk);
+  auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
+  auto* xp = reinterpret_cast<const unsigned short*>(x.data_ptr());
+  auto* wp = reinterpret_cast<const unsigned short*>(w.data_ptr());
+  const long total = (long)M * N;
+#define SK4_LAUNCH(MT_)                                                      \
+  if (splitk == 1) {                                                         \
+    skinny4_kernel<MT_, false><<<grid, 256, 0, stream>>>(                    \
+        op, nullptr, xp, wp, M, N, K);                                       \
+  } else {                                                                   \
+    float* wsp = ws.data_ptr<float>();                                       \
+    TORCH_CHECK(ws.numel() >= total * splitk, "sk4 workspace too small");    \
+    skinny4_kernel<MT_, true><<<grid, 256, 0, stream>>>(                     \
+        nullptr, wsp, xp, wp, M, N, K);                                      \
+    skinny_reduce_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256,   \
+                           0, stream>>>(op, wsp, total, splitk);             \
+  }
+  switch (MT) {
+    case 1: SK4_LAUNCH(1); break;
+    case 2: SK4_LAUNCH(2); break;
+    case 3: SK4_LAUNCH(3); break;
+    default: SK4_LAUNCH(4); break;
+  }
+#undef SK4_LAUNCH
+  HIP_CHECK_KERNEL();
+}
+
 void skinny_gemm2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                   torch::Tensor ws) {
   const int M = x.size(0);
