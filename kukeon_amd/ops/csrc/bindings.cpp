@@ -32,6 +32,8 @@ void skinny_gemm2(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                   torch::Tensor ws);
 void skinny_gemm4(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                   torch::Tensor ws);
+void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                  torch::Tensor ws);
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws);
 void glds_probe(torch::Tensor out, torch::Tensor src);
@@ -66,6 +68,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "16x16x32 bf16 MFMA fragment-layout probe");
   m.def("mfma_probe16k", &kukeon::mfma_probe16k,
         "16x16x16 (K=16) bf16 MFMA fragment-layout probe");
+  m.def("skinny_gemm5", &kukeon::skinny_gemm5,
+        "full-line W stream via wave-private LDS image (v5)");
   m.def("skinny_gemm4", &kukeon::skinny_gemm4,
         "G=4 deep-pipeline weight-streaming GEMM");
   m.def("skinny_gemm2", &kukeon::skinny_gemm2,

@@ -526,6 +526,233 @@ __global__ __launch_bounds__(256) void skinny4_kernel(
   }
 }
 
+// ===========================================================================
+// skinny v5 — full-line W stream through a wave-private LDS image.
+//
+// The request-granularity probe (scripts/pattern_probe3.hip) measured the
+// fragment-shaped W read (16 rows x 64 B per instruction, half a 128-B
+// line per row) at 5.7 TB/s vs 7.3 TB/s for an 8-rows x 128-B contiguous
+// shape at the same geometry — the memory system is request-bound, not
+// byte-bound. v5 therefore loads W contiguously (each instruction = 8
+// rows x 128 B = 8 full-line requests), lands it in registers, and
+// redistributes to MFMA fragment shape through a WAVE-PRIVATE LDS image
+// (no barrier: only the issuing wave reads it), XOR-swizzled so the
+// 16-row fragment reads are bank-conflict-free. The never-drain pipeline
+// holds glds(x) + 16 W loads in flight across every slice barrier with
+// every wait a constant vmcnt(16); the W image for slice s is written at
+// the END of slice s-1 from loads issued a full slice earlier.
+// ===========================================================================
+
+// 8 contiguous nt loads for one tile's 16-row x 512-B slice: instruction
+// (rb, u) covers rows rb..rb+7 at bytes [u*128, u*128+128). p_lo/p_hi are
+// per-lane base pointers for rows (l>>3) and (8 + l>>3).
+DEV_INLINE void issue_wc8(u32x4_t (&reg)[8], const unsigned short* p_lo,
+                          const unsigned short* p_hi) {
+#define SK5_LD(i, P, off)                                                 \
+  asm volatile("global_load_dwordx4 %0, %1, off offset:" #off " nt"      \
+               : "=&v"(reg[i]) : "v"(P) : "memory")
+  SK5_LD(0, p_lo, 0); SK5_LD(1, p_lo, 128);
+  SK5_LD(2, p_lo, 256); SK5_LD(3, p_lo, 384);
+  SK5_LD(4, p_hi, 0); SK5_LD(5, p_hi, 128);
+  SK5_LD(6, p_hi, 256); SK5_LD(7, p_hi, 384);
+#undef SK5_LD
+}
+
+// W-image byte offset inside one 8 KiB tile slot: row-major [16][512 B]
+// with the 16-B column slot XOR-swizzled by row (same scheme as xswz).
+DEV_INLINE int wimg_off(int row, int byte_in_row) {
+  return row * 512 + (byte_in_row ^ ((row & 15) << 4));
+}
+
+template <int MT>
+DEV_INLINE void consume_img(const unsigned short* wimg,
+                            const unsigned short* xb,
+                            f32x4_t (&acc)[MT], int row16, int kgrp) {
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    const int kc = u * 32 + 8 * kgrp;
+    const uint4 av = *reinterpret_cast<const uint4*>(
+        reinterpret_cast<const char*>(wimg) + wimg_off(row16, kc * 2));
+    const bf16x8_t af = frag_of(av);
+#pragma unroll
+    for (int m = 0; m < MT; ++m) {
+      const int xr = m * 16 + row16;
+      const uint4 xv = *reinterpret_cast<const uint4*>(
+          reinterpret_cast<const char*>(xb) + xswz(xr, kc * 2));
+      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af, frag_of(xv), acc[m], 0, 0, 0);
+    }
+  }
+}
+
+// write one tile's 16x512B register set into its LDS image slot
+DEV_INLINE void wimg_write(unsigned short* wimg, u32x4_t (&reg)[8],
+                           int lane) {
+  const int seg16 = (lane & 7) * 16;  // byte offset of this lane's 16 B
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    const int row = (i & 3) * 0 + (i < 4 ? (lane >> 3) : 8 + (lane >> 3));
+    const int u = i & 3;
+    *reinterpret_cast<uint4*>(
+        reinterpret_cast<char*>(wimg) +
+        wimg_off(row, u * 128 + seg16)) = __builtin_bit_cast(uint4, reg[i]);
+  }
+}
+
+template <int MT, bool SPLIT>
+__global__ __launch_bounds__(256) void skinny5_kernel(
+    unsigned short* __restrict__ out, float* __restrict__ ws,
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ w, int M, int N, long K) {
+  // x: 2 x 32 KiB shared double buffer; W: per-wave 2-tile image (16 KiB
+  // x 4 waves) — 128 KiB total, 1 block/CU.
+  __shared__ __align__(16) unsigned short xbuf[2][64 * KSLICE];
+  __shared__ __align__(16) unsigned short wimg_all[4][2][16 * 256];
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int row16 = lane & 15;
+  const int kgrp = lane >> 4;
+  f32x4_t acc0[MT], acc1[MT];
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    acc0[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+    acc1[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+  }
+  const long kadv = (long)gridDim.y * KSLICE;
+  const long ks0 = (long)blockIdx.y * KSLICE;
+  if (ks0 >= K) return;
+  u32x4_t w0[8], w1[8];
+  const int n0t0 = (blockIdx.x * 2 + 0) * 64 + wid * 16;
+  const int n0t1 = (blockIdx.x * 2 + 1) * 64 + wid * 16;
+  const int rlo = lane >> 3, seg = (lane & 7) * 8;  // elems, not bytes
+  const unsigned short* p0l = w + (long)(n0t0 + rlo) * K + ks0 + seg;
+  const unsigned short* p0h = w + (long)(n0t0 + 8 + rlo) * K + ks0 + seg;
+  const unsigned short* p1l = w + (long)(n0t1 + rlo) * K + ks0 + seg;
+  const unsigned short* p1h = w + (long)(n0t1 + 8 + rlo) * K + ks0 + seg;
+  unsigned short* img0 = wimg_all[wid][0];
+  unsigned short* img1 = wimg_all[wid][1];
+
+  // prologue: land slice-0 W, build its image, then put slice-1 loads +
+  // slice-0 x stage in flight (invariant order: glds oldest)
+  issue_wc8(w0, p0l, p0h);
+  issue_wc8(w1, p1l, p1h);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  wimg_write(img0, w0, lane);
+  wimg_write(img1, w1, lane);
+  glds_stage_x(xbuf[0], x, K, ks0, KSLICE, M, wid, lane);
+  {
+    const long ks1 = ks0 + kadv;
+    const unsigned short* q0l = (ks1 < K) ? p0l + kadv : p0l;
+    const unsigned short* q0h = (ks1 < K) ? p0h + kadv : p0h;
+    const unsigned short* q1l = (ks1 < K) ? p1l + kadv : p1l;
+    const unsigned short* q1h = (ks1 < K) ? p1h + kadv : p1h;
+    issue_wc8(w0, q0l, q0h);
+    issue_wc8(w1, q1l, q1h);
+    p0l = q0l; p0h = q0h; p1l = q1l; p1h = q1h;
+  }
+  int cur = 0;
+  for (long ks = ks0; ks < K; ks += kadv, cur ^= 1) {
+    // entry: glds(s) oldest, then t0 loads(s+1), t1 loads(s+1)
+    asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+    __syncthreads();
+    const unsigned short* xb = xbuf[cur];
+    const long ksn = ks + kadv;
+    glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
+                 wid, lane);
+    consume_img<MT>(img0, xb, acc0, row16, kgrp);   // LDS-only
+    // t0 loads(s+1) landed long ago: younger = t1L(8) + glds(8)
+    asm volatile("s_waitcnt vmcnt(16)"
+                 : "+v"(w0[0]), "+v"(w0[1]), "+v"(w0[2]), "+v"(w0[3]),
+                   "+v"(w0[4]), "+v"(w0[5]), "+v"(w0[6]), "+v"(w0[7]));
+    wimg_write(img0, w0, lane);       // after img0 frag reads (in-order)
+    const long ks2 = ksn + kadv;
+    {
+      const unsigned short* q0l = (ks2 < K) ? p0l + kadv : p0l;
+      const unsigned short* q0h = (ks2 < K) ? p0h + kadv : p0h;
+      issue_wc8(w0, q0l, q0h);
+      p0l = q0l; p0h = q0h;
+    }
+    consume_img<MT>(img1, xb, acc1, row16, kgrp);
+    asm volatile("s_waitcnt vmcnt(16)"
+                 : "+v"(w1[0]), "+v"(w1[1]), "+v"(w1[2]), "+v"(w1[3]),
+                   "+v"(w1[4]), "+v"(w1[5]), "+v"(w1[6]), "+v"(w1[7]));
+    wimg_write(img1, w1, lane);
+    {
+      const unsigned short* q1l = (ks2 < K) ? p1l + kadv : p1l;
+      const unsigned short* q1h = (ks2 < K) ? p1h + kadv : p1h;
+      issue_wc8(w1, q1l, q1h);
+      p1l = q1l; p1h = q1h;
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  const int ncol0 = n0t0 + 4 * kgrp;
+  const int ncol1 = n0t1 + 4 * kgrp;
+  float* slab = SPLIT ? ws + (long)blockIdx.y * M * N : nullptr;
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    const int mrow = m * 16 + row16;
+    if (mrow >= M) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long off0 = (long)mrow * N + ncol0 + r;
+      const long off1 = (long)mrow * N + ncol1 + r;
+      if (SPLIT) {
+        slab[off0] = acc0[m][r];
+        slab[off1] = acc1[m][r];
+      } else {
+        out[off0] = f2us(acc0[m][r]);
+        out[off1] = f2us(acc1[m][r]);
+      }
+    }
+  }
+}
+
+void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                  torch::Tensor ws) {
+  const int M = x.size(0);
+  const long K = x.size(1);
+  const int N = w.size(0);
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 128 == 0 && K % KSLICE == 0);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int ngroups = N / 128;
+  const int nslices = (int)(K / KSLICE);
+  int splitk = 1;
+  if (ngroups < 256) splitk = min(nslices, (256 + ngroups - 1) / ngroups);
+  if (const char* ov = getenv("KUKEON_SK5_SPLITK")) {
+    const int v = atoi(ov);
+    if (v > 0) splitk = min(nslices, v);
+  }
+  const int MT = (M + 15) / 16;
+  dim3 grid(ngroups, splitk);
+  auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
+  auto* xp = reinterpret_cast<const unsigned short*>(x.data_ptr());
+  auto* wp = reinterpret_cast<const unsigned short*>(w.data_ptr());
+  const long total = (long)M * N;
+#define SK5_LAUNCH(MT_)                                                      \
+  if (splitk == 1) {                                                         \
+    skinny5_kernel<MT_, false><<<grid, 256, 0, stream>>>(                    \
+        op, nullptr, xp, wp, M, N, K);                                       \
+  } else {                                                                   \
+    float* wsp = ws.data_ptr<float>();                                       \
+    TORCH_CHECK(ws.numel() >= total * splitk, "sk5 workspace too small");    \
+    skinny5_kernel<MT_, true><<<grid, 256, 0, stream>>>(                     \
+        nullptr, wsp, xp, wp, M, N, K);                                      \
+    skinny_reduce_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256,   \
+                           0, stream>>>(op, wsp, total, splitk);             \
+  }
+  switch (MT) {
+    case 1: SK5_LAUNCH(1); break;
+    case 2: SK5_LAUNCH(2); break;
+    case 3: SK5_LAUNCH(3); break;
+    default: SK5_LAUNCH(4); break;
+  }
+#undef SK5_LAUNCH
+  HIP_CHECK_KERNEL();
+}
+
 void skinny_gemm4(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                   torch::Tensor ws) {
   const int M = x.size(0);
