@@ -599,13 +599,33 @@ DEV_INLINE void wimg_write(unsigned short* wimg, u32x4_t (&reg)[8],
   }
 }
 
+// write one wave's 16 x rows (contiguous-loaded) into the SHARED xswz
+// image; published to the other waves by the slice barrier
+DEV_INLINE void ximg_write(unsigned short* xb, u32x4_t (&reg)[8], int wid,
+                           int lane) {
+  const int seg16 = (lane & 7) * 16;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    const int row = wid * 16 + (i < 4 ? (lane >> 3) : 8 + (lane >> 3));
+    const int u = i & 3;
+    *reinterpret_cast<uint4*>(
+        reinterpret_cast<char*>(xb) +
+        xswz(row, u * 128 + seg16)) = __builtin_bit_cast(uint4, reg[i]);
+  }
+}
+
 template <int MT, bool SPLIT>
 __global__ __launch_bounds__(256) void skinny5_kernel(
     unsigned short* __restrict__ out, float* __restrict__ ws,
     const unsigned short* __restrict__ x,
     const unsigned short* __restrict__ w, int M, int N, long K) {
-  // x: 2 x 32 KiB shared double buffer; W: per-wave 2-tile image (16 KiB
-  // x 4 waves) — 128 KiB total, 1 block/CU.
+  // x: 2 x 32 KiB shared double buffer (cooperatively written, barrier-
+  // published); W: per-wave 2-tile image (16 KiB x 4 waves). 128 KiB,
+  // 1 block/CU. NO LDS-DMA anywhere: glds completions do not stay
+  // in-order with plain loads on the vmcnt counter (mixed-class counted
+  // waits released early -> the round-2 wild-write fault), so both
+  // operands stage through registers + ds_write and every wait is a
+  // constant in-order vmcnt(16).
   __shared__ __align__(16) unsigned short xbuf[2][64 * KSLICE];
   __shared__ __align__(16) unsigned short wimg_all[4][2][16 * 256];
   const int wid = threadIdx.x / WAVE;
@@ -621,68 +641,73 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
   const long kadv = (long)gridDim.y * KSLICE;
   const long ks0 = (long)blockIdx.y * KSLICE;
   if (ks0 >= K) return;
-  u32x4_t w0[8], w1[8];
+  u32x4_t w0[8], w1[8], xr8[8];
   const int n0t0 = (blockIdx.x * 2 + 0) * 64 + wid * 16;
   const int n0t1 = (blockIdx.x * 2 + 1) * 64 + wid * 16;
-  const int rlo = lane >> 3, seg = (lane & 7) * 8;  // elems, not bytes
+  const int rlo = lane >> 3, seg = (lane & 7) * 8;  // elems
   const unsigned short* p0l = w + (long)(n0t0 + rlo) * K + ks0 + seg;
   const unsigned short* p0h = w + (long)(n0t0 + 8 + rlo) * K + ks0 + seg;
   const unsigned short* p1l = w + (long)(n0t1 + rlo) * K + ks0 + seg;
   const unsigned short* p1h = w + (long)(n0t1 + 8 + rlo) * K + ks0 + seg;
+  // this wave's 16 x rows (clamped for M < 64: junk rows are staged but
+  // their accumulators are discarded by the epilogue's mrow guard)
+  const int xrow_l = wid * 16 + rlo;
+  const int xrow_h = wid * 16 + 8 + rlo;
+  const unsigned short* pxl = x + (long)min(M - 1, xrow_l) * K + ks0 + seg;
+  const unsigned short* pxh = x + (long)min(M - 1, xrow_h) * K + ks0 + seg;
   unsigned short* img0 = wimg_all[wid][0];
   unsigned short* img1 = wimg_all[wid][1];
 
-  // prologue: land slice-0 W, build its image, then put slice-1 loads +
-  // slice-0 x stage in flight (invariant order: glds oldest)
+  // prologue: land slice 0, build the images, put slice 1 in flight
   issue_wc8(w0, p0l, p0h);
   issue_wc8(w1, p1l, p1h);
+  issue_wc8(xr8, pxl, pxh);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   wimg_write(img0, w0, lane);
   wimg_write(img1, w1, lane);
-  glds_stage_x(xbuf[0], x, K, ks0, KSLICE, M, wid, lane);
+  ximg_write(xbuf[0], xr8, wid, lane);
   {
     const long ks1 = ks0 + kadv;
-    const unsigned short* q0l = (ks1 < K) ? p0l + kadv : p0l;
-    const unsigned short* q0h = (ks1 < K) ? p0h + kadv : p0h;
-    const unsigned short* q1l = (ks1 < K) ? p1l + kadv : p1l;
-    const unsigned short* q1h = (ks1 < K) ? p1h + kadv : p1h;
-    issue_wc8(w0, q0l, q0h);
-    issue_wc8(w1, q1l, q1h);
-    p0l = q0l; p0h = q0h; p1l = q1l; p1h = q1h;
+    if (ks1 < K) {
+      p0l += kadv; p0h += kadv; p1l += kadv; p1h += kadv;
+      pxl += kadv; pxh += kadv;
+    }
+    issue_wc8(w0, p0l, p0h);
+    issue_wc8(w1, p1l, p1h);
+    issue_wc8(xr8, pxl, pxh);
   }
   int cur = 0;
   for (long ks = ks0; ks < K; ks += kadv, cur ^= 1) {
-    // entry: glds(s) oldest, then t0 loads(s+1), t1 loads(s+1)
-    asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+    // entry: t0L(s+1), t1L(s+1), xL(s+1) in flight (24 plain loads, in
+    // that order); images(s) + xbuf[cur] ready
     __syncthreads();
     const unsigned short* xb = xbuf[cur];
     const long ksn = ks + kadv;
-    glds_stage_x(xbuf[cur ^ 1], x, K, (ksn < K ? ksn : 0), KSLICE, M,
-                 wid, lane);
+    const bool adv = (ksn + kadv) < K;
     consume_img<MT>(img0, xb, acc0, row16, kgrp);   // LDS-only
-    // t0 loads(s+1) landed long ago: younger = t1L(8) + glds(8)
+    // t0L(s+1) landed: younger = t1L(8) + xL(8); all one class,
+    // strictly in-order
     asm volatile("s_waitcnt vmcnt(16)"
                  : "+v"(w0[0]), "+v"(w0[1]), "+v"(w0[2]), "+v"(w0[3]),
                    "+v"(w0[4]), "+v"(w0[5]), "+v"(w0[6]), "+v"(w0[7]));
     wimg_write(img0, w0, lane);       // after img0 frag reads (in-order)
-    const long ks2 = ksn + kadv;
-    {
-      const unsigned short* q0l = (ks2 < K) ? p0l + kadv : p0l;
-      const unsigned short* q0h = (ks2 < K) ? p0h + kadv : p0h;
-      issue_wc8(w0, q0l, q0h);
-      p0l = q0l; p0h = q0h;
-    }
+    if (adv) { p0l += kadv; p0h += kadv; }
+    issue_wc8(w0, p0l, p0h);
     consume_img<MT>(img1, xb, acc1, row16, kgrp);
     asm volatile("s_waitcnt vmcnt(16)"
                  : "+v"(w1[0]), "+v"(w1[1]), "+v"(w1[2]), "+v"(w1[3]),
                    "+v"(w1[4]), "+v"(w1[5]), "+v"(w1[6]), "+v"(w1[7]));
     wimg_write(img1, w1, lane);
-    {
-      const unsigned short* q1l = (ks2 < K) ? p1l + kadv : p1l;
-      const unsigned short* q1h = (ks2 < K) ? p1h + kadv : p1h;
-      issue_wc8(w1, q1l, q1h);
-      p1l = q1l; p1h = q1h;
-    }
+    if (adv) { p1l += kadv; p1h += kadv; }
+    issue_wc8(w1, p1l, p1h);
+    // xL(s+1) landed: younger = t0L(s+2) + t1L(s+2)
+    asm volatile("s_waitcnt vmcnt(16)"
+                 : "+v"(xr8[0]), "+v"(xr8[1]), "+v"(xr8[2]), "+v"(xr8[3]),
+                   "+v"(xr8[4]), "+v"(xr8[5]), "+v"(xr8[6]), "+v"(xr8[7]));
+    ximg_write(xbuf[cur ^ 1], xr8, wid, lane);
+    if (adv) { pxl += kadv; pxh += kadv; }
+    issue_wc8(xr8, pxl, pxh);
+    // exit: t0L(s+2), t1L(s+2), xL(s+2) in flight — invariant restored
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 
