@@ -13,7 +13,8 @@ if which == "blas":
 else:
     from kukeon_amd import _C
     wrk = torch.empty(16 * 64 * N, dtype=torch.float32, device="cuda")
+    fn = {"v2": _C.skinny_gemm2, "v5": _C.skinny_gemm5}[which]
     for i in range(30):
-        _C.skinny_gemm2(out, x, ws_list[i % 3], wrk)
+        fn(out, x, ws_list[i % 3], wrk)
 torch.cuda.synchronize()
 print("done")
