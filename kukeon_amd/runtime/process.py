@@ -57,11 +57,13 @@ class ContainerProbe:
 class Runtime:
     """Interface the runner programs against (ProcessRuntime / FakeRuntime)."""
 
-    def start_root(self, cell_dir: Path, cgroup_rel: str) -> int:
+    def start_root(self, cell_dir: Path, cgroup_rel: str,
+                   ns: Optional[Dict] = None) -> int:
         raise NotImplementedError
 
     def start_container(self, cdir: Path, spec: api.ContainerSpec,
-                        env: List[str], cgroup_rel: str) -> int:
+                        env: List[str], cgroup_rel: str,
+                        ns: Optional[Dict] = None) -> int:
         raise NotImplementedError
 
     def probe(self, cdir: Path) -> ContainerProbe:
@@ -104,10 +106,24 @@ class ProcessRuntime(Runtime):
             self.cgroups.attach(cgroup_rel, p.pid)
         return p.pid
 
-    def start_root(self, cell_dir: Path, cgroup_rel: str) -> int:
+    def start_root(self, cell_dir: Path, cgroup_rel: str,
+                   ns: Optional[Dict] = None) -> int:
         cdir = cell_dir / ROOT_CONTAINER
         cdir.mkdir(parents=True, exist_ok=True)
-        pid = self._spawn_shim(cdir, ["--pause"], cgroup_rel)
+        args = ["--pause"]
+        if ns and ns.get("unshare"):
+            args += ["--unshare", ",".join(ns["unshare"]),
+                     "--ns-record", str(cdir / "ns.json")]
+            if ns.get("hostname"):
+                args += ["--hostname", ns["hostname"]]
+        pid = self._spawn_shim(cdir, args, cgroup_rel)
+        if ns and ns.get("unshare"):
+            # wait for the pause to record its namespace outcome so the
+            # caller can plumb the netns before peers join
+            for _ in range(300):
+                if (cdir / "ns.json").exists() or not proc.alive(pid):
+                    break
+                time.sleep(0.01)
         st = proc.proc_starttime(pid)
         _write_json(cdir / "runtime.json", {
             "shimPid": pid, "shimStarttime": st, "workloadPid": pid,
@@ -115,7 +131,8 @@ class ProcessRuntime(Runtime):
         return pid
 
     def start_container(self, cdir: Path, spec: api.ContainerSpec,
-                        env: List[str], cgroup_rel: str) -> int:
+                        env: List[str], cgroup_rel: str,
+                        ns: Optional[Dict] = None) -> int:
         cdir.mkdir(parents=True, exist_ok=True)
         argv = self._argv(spec)
         spawn = {
@@ -128,6 +145,7 @@ class ProcessRuntime(Runtime):
                       for r in (spec.repos or [])],
             "git": ({"name": spec.git.name, "email": spec.git.email}
                     if spec.git else {}),
+            "ns": ns or {},
         }
         # clear stale exit state from a previous run
         with contextlib.suppress(FileNotFoundError):
@@ -211,11 +229,11 @@ class FakeRuntime(Runtime):
     def _key(self, cdir: Path) -> str:
         return str(cdir)
 
-    def start_root(self, cell_dir: Path, cgroup_rel: str) -> int:
+    def start_root(self, cell_dir: Path, cgroup_rel: str, ns=None) -> int:
         return self.start_container(cell_dir / ROOT_CONTAINER,
                                     api.ContainerSpec(), [], cgroup_rel)
 
-    def start_container(self, cdir, spec, env, cgroup_rel) -> int:
+    def start_container(self, cdir, spec, env, cgroup_rel, ns=None) -> int:
         key = self._key(cdir)
         if key in self.fail_on:
             raise self.fail_on[key]

@@ -47,8 +47,41 @@ def write_json(path: Path, data: dict) -> None:
     os.replace(tmp, path)
 
 
-def run_pause() -> int:
-    """Root-container pause: reap children, exit 0 on TERM/INT."""
+def run_pause(unshare_kinds: str = "", hostname: str = "",
+              ns_record: str = "") -> int:
+    """Root-container pause: owns the cell's namespaces (net/uts/ipc —
+    the pod model of reference internal/ctr/spec.go:38), reaps children,
+    exits 0 on TERM/INT. Namespace setup degrades gracefully on hosts
+    without CAP_SYS_ADMIN and records the outcome for the runner."""
+    held = []
+    if unshare_kinds:
+        from kukeon_amd.runtime import namespaces as nsmod
+        for kind in unshare_kinds.split(","):
+            fl = nsmod.NS_FLAGS.get(kind)
+            if not fl or kind in ("mnt", "pid"):
+                continue
+            try:
+                nsmod.unshare(fl)
+                held.append(kind)
+            except OSError as e:
+                sys.stderr.write(f"unshare {kind}: {e}\n")
+        if "uts" in held and hostname:
+            try:
+                nsmod.sethostname(hostname)
+            except OSError:
+                pass
+        if "net" in held:
+            # fresh netns: bring loopback up so in-cell localhost works
+            try:
+                from kukeon_amd.runtime.netlink import Rtnl
+                with Rtnl() as nl:
+                    nl.set_link_up_by_index(1)
+            except OSError as e:
+                sys.stderr.write(f"netns lo up: {e}\n")
+    if ns_record:
+        write_json(Path(ns_record),
+                   {"held": held, "degraded": bool(unshare_kinds) and
+                    set(unshare_kinds.split(",")) - set(held) != set()})
     stop = {"flag": False}
 
     def on_term(signum, frame):
@@ -135,6 +168,73 @@ class Shim:
                 st["error"] = str(e)[-500:]
             states.append(st)
         write_json(self.dir / "setup.json", {"repos": states})
+
+    def enter_namespaces(self) -> None:
+        """Join the cell root's net/uts/ipc namespaces and take a
+        private mount namespace with rendered /etc files (and optional
+        image rootfs). Runs in the shim BEFORE any fork so the PTY and
+        workload inherit everything. Degrades per-kind; the outcome is
+        recorded to ns.json for status derivation."""
+        ns = self.spec.get("ns") or {}
+        if not ns:
+            return
+        from kukeon_amd.runtime import namespaces as nsmod
+        held: list = []
+        degraded = False
+        join_pid = ns.get("joinPid") or 0
+        if join_pid and ns.get("join"):
+            held += nsmod.join(join_pid, list(ns["join"]))
+            if set(ns["join"]) - set(held):
+                degraded = True
+        if ns.get("mountNs"):
+            try:
+                nsmod.unshare(nsmod.CLONE_NEWNS)
+                nsmod.make_mounts_private()
+                held.append("mnt")
+                rootfs = ns.get("rootfs") or ""
+                if rootfs:
+                    try:
+                        self._enter_rootfs(nsmod, ns, rootfs)
+                        held.append("rootfs")
+                    except OSError as e:
+                        degraded = True
+                        sys.stderr.write(f"rootfs: {e}\n")
+                elif ns.get("hostname"):
+                    try:
+                        nsmod.setup_etc(ns["hostname"],
+                                        ns.get("hosts") or {},
+                                        str(self.dir / "etc"))
+                    except OSError as e:
+                        sys.stderr.write(f"etc render: {e}\n")
+            except OSError as e:
+                degraded = True
+                sys.stderr.write(f"mount ns: {e}\n")
+        write_json(self.dir / "ns.json",
+                   {"held": held, "degraded": degraded})
+
+    def _enter_rootfs(self, nsmod, ns: dict, rootfs: str) -> None:
+        """chroot into an image rootfs keeping the shim functional: the
+        container state dir, /proc and /dev are bind-mounted inside at
+        their host paths so runtime/status files and the PTY keep
+        working; /etc/hostname + hosts are rendered INTO the layer."""
+        import kukeon_amd.runtime.proc  # noqa: F401  pre-import (chroot)
+        for sub, src in ((str(self.dir).lstrip("/"), str(self.dir)),
+                         ("proc", "/proc"), ("dev", "/dev"),
+                         ("tmp", "/tmp")):
+            tgt = os.path.join(rootfs, sub)
+            os.makedirs(tgt, exist_ok=True)
+            nsmod.mount(src, tgt, "", nsmod.MS_BIND | nsmod.MS_REC)
+        etc = os.path.join(rootfs, "etc")
+        os.makedirs(etc, exist_ok=True)
+        hostname = ns.get("hostname") or "cell"
+        with open(os.path.join(etc, "hostname"), "w") as f:
+            f.write(hostname + "\n")
+        with open(os.path.join(etc, "hosts"), "w") as f:
+            f.write("127.0.0.1\tlocalhost\n")
+            f.write(f"127.0.1.1\t{hostname}\n")
+            for name, ip in sorted((ns.get("hosts") or {}).items()):
+                f.write(f"{ip}\t{name}\n")
+        nsmod.enter_rootfs(rootfs)
 
     def child_env(self) -> dict:
         env = dict(os.environ)
@@ -314,10 +414,14 @@ def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--dir")
     ap.add_argument("--pause", action="store_true")
+    ap.add_argument("--unshare", default="")
+    ap.add_argument("--hostname", default="")
+    ap.add_argument("--ns-record", default="")
     args = ap.parse_args()
     if args.pause:
-        return run_pause()
+        return run_pause(args.unshare, args.hostname, args.ns_record)
     shim = Shim(Path(args.dir))
+    shim.enter_namespaces()
 
     def on_term(signum, frame):
         # forward to the workload; the wait loop observes the exit

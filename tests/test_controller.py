@@ -592,7 +592,7 @@ class RaceCheckRuntime(FakeRuntime):
         self._mu = threading.Lock()
         self._spawning = set()
 
-    def start_container(self, cdir, spec, env, cgroup_rel):
+    def start_container(self, cdir, spec, env, cgroup_rel, ns=None):
         import time as _t
         key = self._key(cdir)
         with self._mu:
@@ -602,7 +602,7 @@ class RaceCheckRuntime(FakeRuntime):
             self._spawning.add(key)
         _t.sleep(0.002)  # widen the probe->spawn window
         try:
-            return super().start_container(cdir, spec, env, cgroup_rel)
+            return super().start_container(cdir, spec, env, cgroup_rel, ns)
         finally:
             with self._mu:
                 self._spawning.discard(key)
