@@ -658,17 +658,34 @@ def image_prune(ctx):
 
 
 @cli.command()
-@click.argument("name")
-@click.option("-f", "--file", "file_", required=True,
-              type=click.Path(exists=True),
-              help="image spec YAML (entrypoint/env/capabilities)")
+@click.argument("context", type=click.Path(exists=True), required=False)
+@click.option("-t", "--tag", "tag", required=True, help="image name")
+@click.option("-f", "--file", "file_", type=click.Path(exists=True),
+              help="Kukefile (default <context>/Kukefile); a YAML file "
+                   "falls back to profile registration")
 @pass_ctx
-def build(ctx, name, file_):
-    """Register an image from a build spec (kukebuild analog; OCI builds
-    are delegated to an external engine when one exists)."""
-    spec = yaml.safe_load(Path(file_).read_text()) or {}
-    ctx._controller().register_image(name, spec)
-    click.echo(f"image {name} registered")
+def build(ctx, context, tag, file_):
+    """Build a layered rootfs image from a Kukefile (kukebuild analog:
+    FROM/COPY/RUN/ENV/CMD/WORKDIR/LABEL executed natively with
+    overlayfs layers — reference cmd/kukebuild/main.go:17-49). With a
+    YAML -f and no context, registers a runtime profile instead."""
+    from kukeon_amd.images import Builder, ImageStore
+    ctl = ctx._controller()
+    if context is None:
+        if not file_:
+            raise click.UsageError("need a build context or -f profile")
+        spec = yaml.safe_load(Path(file_).read_text()) or {}
+        ctl.register_image(tag, spec)
+        click.echo(f"image {tag} registered (profile)")
+        return
+    kf = Path(file_) if file_ else Path(context) / "Kukefile"
+    if not kf.exists():
+        raise click.UsageError(f"no Kukefile at {kf}")
+    store = ImageStore(str(ctl.run_path))
+    man = Builder(store).build(Path(context), kf.read_text(), tag,
+                               log=lambda m: click.echo(m))
+    click.echo(f"image {tag}: {len(man['layers'])} layer(s), "
+               f"{man['sizeBytes']} bytes")
 
 
 @cli.group()

@@ -605,6 +605,29 @@ class Controller:
             if net_info and net_info.get("ip"):
                 hosts[doc.metadata.name] = net_info["ip"]
             cfg["hosts"] = hosts
+            # built layered image -> overlay rootfs mounted by the shim
+            # in ITS mount namespace (reference: the OCI rootfs the
+            # containerd snapshotter provides)
+            if c.image and c.image not in ("none", "host"):
+                from kukeon_amd.images import ImageStore
+                istore = ImageStore(str(self.run_path))
+                if istore.exists(c.image):
+                    man = istore.get(c.image)
+                    st = cell_dir / (c.id or "main") / "rootfs-state"
+                    cfg["rootfs"] = {
+                        "layers": [str(p) for p in
+                                   istore.layer_paths(c.image)],
+                        "upper": str(st / "up"),
+                        "work": str(st / "w"),
+                        "mnt": str(st / "m"),
+                    }
+                    ic = man.get("config", {})
+                    if ic.get("cmd"):
+                        cfg["imageCmd"] = ic["cmd"]
+                    if ic.get("workdir"):
+                        cfg["imageWorkdir"] = ic["workdir"]
+                    if ic.get("env"):
+                        cfg["imageEnv"] = list(ic["env"])
         return cfg or None
 
     def _spec_drifted(self, doc: api.CellDoc) -> bool:

@@ -212,12 +212,23 @@ class Shim:
         write_json(self.dir / "ns.json",
                    {"held": held, "degraded": degraded})
 
-    def _enter_rootfs(self, nsmod, ns: dict, rootfs: str) -> None:
+    def _enter_rootfs(self, nsmod, ns: dict, rootfs) -> None:
         """chroot into an image rootfs keeping the shim functional: the
         container state dir, /proc and /dev are bind-mounted inside at
         their host paths so runtime/status files and the PTY keep
-        working; /etc/hostname + hosts are rendered INTO the layer."""
+        working; /etc/hostname + hosts are rendered INTO the layer.
+        `rootfs` is either a direct path or an overlay config
+        {layers, upper, work, mnt} mounted here so it dies with this
+        mount namespace."""
         import kukeon_amd.runtime.proc  # noqa: F401  pre-import (chroot)
+        if isinstance(rootfs, dict):
+            lowers = ":".join(reversed(rootfs["layers"]))
+            for k in ("upper", "work", "mnt"):
+                os.makedirs(rootfs[k], exist_ok=True)
+            nsmod.mount("overlay", rootfs["mnt"], "overlay", 0,
+                        f"lowerdir={lowers},upperdir={rootfs['upper']},"
+                        f"workdir={rootfs['work']}")
+            rootfs = rootfs["mnt"]
         for sub, src in ((str(self.dir).lstrip("/"), str(self.dir)),
                          ("proc", "/proc"), ("dev", "/dev"),
                          ("tmp", "/tmp")):
@@ -238,6 +249,9 @@ class Shim:
 
     def child_env(self) -> dict:
         env = dict(os.environ)
+        for kv in (self.spec.get("ns") or {}).get("imageEnv", []):
+            k, _, v = kv.partition("=")
+            env[k] = v
         for kv in self.spec.get("env", []):
             k, _, v = kv.partition("=")
             env[k] = v
@@ -258,12 +272,20 @@ class Shim:
 
     def _exec_child(self):
         spec = self.spec
+        ns = spec.get("ns") or {}
         if spec.get("cwd"):
             try:
                 os.chdir(spec["cwd"])
             except OSError:
                 pass
+        if ns.get("imageWorkdir"):
+            try:
+                os.chdir(ns["imageWorkdir"])
+            except OSError:
+                pass
         argv = spec["argv"]
+        if argv == ["/bin/sh"] and ns.get("imageCmd"):
+            argv = ["/bin/sh", "-c", ns["imageCmd"]]
         try:
             os.execvpe(argv[0], argv, self.child_env())
         except OSError as e:

@@ -1,0 +1,188 @@
+"""Image build + rootfs e2e (kukebuild analog).
+
+Builds a layered image with the native Kukefile builder (COPY + RUN via
+overlayfs), runs a real process cell chroot'ed into it, and verifies the
+layer's files are visible inside the cell and absent for host-rootfs
+cells. The in-image shell is a dash + libc closure copied from the host
+into the build context (no base images exist in this offline env).
+"""
+import shutil
+import subprocess
+import time
+import uuid
+from pathlib import Path
+
+import pytest
+
+from kukeon_amd.api import v1beta1 as api
+from kukeon_amd.controller.core import Controller
+from kukeon_amd.images import Builder, BuildError, ImageStore, \
+    overlay_supported
+from kukeon_amd.runtime import namespaces as nsmod
+
+HAVE_MNT = nsmod.can_unshare(nsmod.CLONE_NEWNS)
+HAVE_OVL = HAVE_MNT and overlay_supported()
+
+
+def ldd_closure(binary: str):
+    """Paths needed to run `binary` in a bare chroot."""
+    out = subprocess.run(["ldd", binary], capture_output=True, text=True)
+    paths = [binary]
+    for line in out.stdout.splitlines():
+        parts = line.split()
+        for p in parts:
+            if p.startswith("/") and Path(p).exists():
+                paths.append(p)
+    return paths
+
+
+def make_shell_context(ctx: Path):
+    """Build context containing sh/cat/ls + their library closures."""
+    for tool in ("/usr/bin/dash", "/usr/bin/cat", "/usr/bin/ls"):
+        for src in ldd_closure(tool):
+            rel = src.lstrip("/")
+            dst = ctx / "shellfs" / rel
+            dst.parent.mkdir(parents=True, exist_ok=True)
+            if not dst.exists():
+                shutil.copy2(src, dst)
+    bind = ctx / "shellfs" / "bin"
+    bind.mkdir(parents=True, exist_ok=True)
+    shutil.copy2("/usr/bin/dash", bind / "sh")
+    shutil.copy2("/usr/bin/cat", bind / "cat")
+    shutil.copy2("/usr/bin/ls", bind / "ls")
+
+
+def wait_for(pred, timeout=15.0):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        if pred():
+            return True
+        time.sleep(0.05)
+    return False
+
+
+@pytest.mark.skipif(not HAVE_OVL, reason="no overlayfs/mount ns")
+def test_build_layers_run_and_prune(tmp_path):
+    store = ImageStore(str(tmp_path / "run"))
+    ctx = tmp_path / "ctx"
+    ctx.mkdir()
+    make_shell_context(ctx)
+    (ctx / "hello.txt").write_text("from-the-layer\n")
+    kukefile = """
+FROM scratch
+COPY shellfs /
+COPY hello.txt /data/
+RUN echo built-at-build-time > /data/built.txt
+ENV KUKE_IMG=1
+CMD cat /data/hello.txt
+"""
+    logs = []
+    man = Builder(store).build(ctx, kukefile, "test/base", logs.append)
+    assert len(man["layers"]) == 3  # shellfs, hello, RUN
+    assert man["config"]["cmd"] == "cat /data/hello.txt"
+    # derived image stacking on the base
+    ctx2 = tmp_path / "ctx2"
+    ctx2.mkdir()
+    (ctx2 / "extra.txt").write_text("layer-two\n")
+    man2 = Builder(store).build(
+        ctx2, "FROM test/base\nCOPY extra.txt /data/\n", "test/child")
+    assert len(man2["layers"]) == 4
+    assert man2["layers"][:3] == man["layers"]
+    # prune: deleting the child leaves base layers alive
+    store.delete("test/child")
+    dropped = store.prune_layers()
+    assert len(dropped) == 1  # only the child's unique layer
+    # RUN failure surfaces stderr
+    with pytest.raises(BuildError):
+        Builder(store).build(ctx2, "FROM test/base\nRUN exit 3\n", "bad")
+
+
+@pytest.mark.skipif(not HAVE_OVL, reason="no overlayfs/mount ns")
+def test_cell_runs_on_image_rootfs(tmp_path):
+    ctl = Controller(str(tmp_path / "run"), gpu_devices=[])
+    ctl.bootstrap()
+    store = ImageStore(str(ctl.run_path))
+    ctx = tmp_path / "ctx"
+    ctx.mkdir()
+    make_shell_context(ctx)
+    (ctx / "hello.txt").write_text("from-the-layer\n")
+    Builder(store).build(
+        ctx,
+        "FROM scratch\nCOPY shellfs /\nCOPY hello.txt /data/\n"
+        "RUN echo built > /data/built.txt\n",
+        "test/rootfs")
+    name = f"img-{uuid.uuid4().hex[:6]}"
+    cdir = ctl.store.cell_dir("default", "default", "default", name) / "main"
+    doc = api.CellDoc(
+        metadata=api.Metadata(name=name),
+        spec=api.CellSpec(
+            realm_id="default", space_id="default", stack_id="default",
+            containers=[api.ContainerSpec(
+                id="main", image="test/rootfs", command="/bin/sh",
+                args=["-c",
+                      f"cat /data/hello.txt > {cdir}/got.txt; "
+                      f"cat /data/built.txt >> {cdir}/got.txt; "
+                      f"ls /etc/hostname >> {cdir}/got.txt 2>&1; "
+                      "sleep 30"])]))
+    ctl.create_cell(doc)
+    ctl.start_cell("default", "default", "default", name)
+    assert wait_for(lambda: (cdir / "got.txt").exists())
+    time.sleep(0.2)
+    got = (cdir / "got.txt").read_text()
+    assert "from-the-layer" in got     # COPY layer visible inside
+    assert "built" in got              # RUN layer visible inside
+    ns = ctl.store.read(cdir / "ns.json") or {}
+    assert "rootfs" in ns.get("held", []), ns
+    # writable upper: the cell wrote through the overlay, layers untouched
+    base_layer = store.layer_paths("test/rootfs")[1] / "data" / "hello.txt"
+    assert base_layer.read_text() == "from-the-layer\n"
+    ctl.delete_cell("default", "default", "default", name, force=True)
+
+    # a host-rootfs cell does NOT see the layer's files
+    name2 = f"host-{uuid.uuid4().hex[:6]}"
+    c2dir = ctl.store.cell_dir("default", "default", "default",
+                               name2) / "main"
+    doc2 = api.CellDoc(
+        metadata=api.Metadata(name=name2),
+        spec=api.CellSpec(
+            realm_id="default", space_id="default", stack_id="default",
+            containers=[api.ContainerSpec(
+                id="main", image="none", command="sh",
+                args=["-c", "ls /data/hello.txt > seen.txt 2>&1; "
+                            "sleep 30"])]))
+    ctl.create_cell(doc2)
+    ctl.start_cell("default", "default", "default", name2)
+    assert wait_for(lambda: (c2dir / "seen.txt").exists())
+    time.sleep(0.1)
+    assert "No such file" in (c2dir / "seen.txt").read_text()
+    ctl.delete_cell("default", "default", "default", name2, force=True)
+
+
+@pytest.mark.skipif(not HAVE_OVL, reason="no overlayfs/mount ns")
+def test_image_cmd_and_env_defaults(tmp_path):
+    """A container with no command runs the image CMD with image ENV."""
+    ctl = Controller(str(tmp_path / "run"), gpu_devices=[])
+    ctl.bootstrap()
+    store = ImageStore(str(ctl.run_path))
+    ctx = tmp_path / "ctx"
+    ctx.mkdir()
+    make_shell_context(ctx)
+    name = f"cmd-{uuid.uuid4().hex[:6]}"
+    cdir = ctl.store.cell_dir("default", "default", "default", name) / "main"
+    Builder(store).build(
+        ctx,
+        "FROM scratch\nCOPY shellfs /\nENV GREETING=hello-img\n"
+        f"CMD echo $GREETING > {cdir}/cmd.txt; sleep 30\n",
+        "test/cmdimg")
+    doc = api.CellDoc(
+        metadata=api.Metadata(name=name),
+        spec=api.CellSpec(
+            realm_id="default", space_id="default", stack_id="default",
+            containers=[api.ContainerSpec(id="main",
+                                          image="test/cmdimg")]))
+    ctl.create_cell(doc)
+    ctl.start_cell("default", "default", "default", name)
+    assert wait_for(lambda: (cdir / "cmd.txt").exists())
+    time.sleep(0.1)
+    assert (cdir / "cmd.txt").read_text().strip() == "hello-img"
+    ctl.delete_cell("default", "default", "default", name, force=True)
