@@ -17,8 +17,11 @@ POOL_PREFIX = "10.88"
 
 
 class SubnetAllocator:
-    def __init__(self, store: Store):
+    def __init__(self, store: Store, honor_host_routes: bool = True,
+                 route_file: str = "/proc/net/route"):
         self.store = store
+        self.honor_host_routes = honor_host_routes
+        self.route_file = route_file
         self._mu = threading.Lock()
 
     def _network_path(self, realm: str, space: str):
@@ -39,8 +42,10 @@ class SubnetAllocator:
         # also honor LIVE host routes in the pool: another instance (or a
         # crashed daemon whose store was wiped) may still hold a bridge —
         # colliding subnets blackhole reply traffic
+        if not self.honor_host_routes:
+            return used
         try:
-            with open("/proc/net/route") as f:
+            with open(self.route_file) as f:
                 for line in f.read().splitlines()[1:]:
                     parts = line.split()
                     if len(parts) < 8:

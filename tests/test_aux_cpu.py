@@ -207,13 +207,14 @@ def test_subnet_allocator_persistence_release_exhaustion(tmp_path):
     from kukeon_amd.controller.subnet import SubnetAllocator
 
     store = Store(str(tmp_path / "run"))
-    a = SubnetAllocator(store)
+    a = SubnetAllocator(store, honor_host_routes=False)
     s1 = a.allocate("default", "alpha")
     s2 = a.allocate("default", "beta")
     assert s1 != s2 and s1.endswith(".0/24")
     assert a.allocate("default", "alpha") == s1   # idempotent
     # persisted: a fresh allocator over the same tree sees it
-    b = SubnetAllocator(Store(str(tmp_path / "run")))
+    b = SubnetAllocator(Store(str(tmp_path / "run")),
+                        honor_host_routes=False)
     assert b.lookup("default", "alpha") == s1
     b.release("default", "alpha")
     assert b.lookup("default", "alpha") is None
@@ -224,6 +225,31 @@ def test_subnet_allocator_persistence_release_exhaustion(tmp_path):
         b.allocate("default", f"sp{i}")
     with pytest.raises(errors.KukeonError, match="exhausted"):
         b.allocate("default", "overflow")
+
+
+def test_subnet_allocator_skips_live_host_routes(tmp_path):
+    """Octets with live host routes (e.g. a crashed daemon's bridge)
+    must not be re-allocated: colliding subnets blackhole replies."""
+    from kukeon_amd.state.store import Store
+    from kukeon_amd.controller.subnet import SubnetAllocator
+
+    rt = tmp_path / "route"
+    # fake /proc/net/route with 10.88.0.0/24 and 10.88.1.0/24 held
+    rt.write_text(
+        "Iface Dest Gateway Flags RefCnt Use Metric Mask MTU Win IRTT\n"
+        "k-dead 0058580A 00000000 0001 0 0 0 00FFFFFF 0 0 0\n"
+        .replace("0058580A", "0058580A") +
+        "k-dead2 0158580A 00000000 0001 0 0 0 00FFFFFF 0 0 0\n")
+    # hex little-endian: 0A58_5800 ... craft via real pack below instead
+    import socket as _s, struct as _st
+    def hx(ip):
+        return "%08X" % _st.unpack("<I", _s.inet_aton(ip))[0]
+    rt.write_text(
+        "Iface Dest Gateway Flags RefCnt Use Metric Mask MTU Win IRTT\n"
+        f"k-dead {hx('10.88.0.0')} 00000000 0001 0 0 0 00FFFFFF 0 0 0\n"
+        f"k-dead2 {hx('10.88.1.0')} 00000000 0001 0 0 0 00FFFFFF 0 0 0\n")
+    a = SubnetAllocator(Store(str(tmp_path / "run")), route_file=str(rt))
+    assert a.allocate("default", "alpha") == "10.88.2.0/24"
 
 
 def test_gpu_allocator_top_up_after_spec_growth(tmp_path):
