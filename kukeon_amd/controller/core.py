@@ -605,6 +605,27 @@ class Controller:
             if net_info and net_info.get("ip"):
                 hosts[doc.metadata.name] = net_info["ip"]
             cfg["hosts"] = hosts
+            # REAL volume mounts: with a private mount namespace each
+            # volume binds at its declared target (host-wide state is
+            # untouched); the env/symlink surface stays for tooling
+            binds = []
+            for vm in c.volumes:
+                if not vm.target:
+                    continue
+                try:
+                    if vm.source:
+                        src = str(Path(vm.source))
+                    else:
+                        vol = self.get_volume(doc.spec.realm_id,
+                                              doc.spec.space_id,
+                                              vm.name or "")
+                        src = str(vol.status.path)
+                    binds.append({"src": src, "dst": vm.target})
+                except errors.NotFound:
+                    log.warning("cell %s: volume %r not found; skipped",
+                                doc.metadata.name, vm.name)
+            if binds:
+                cfg["binds"] = binds
             # built layered image -> overlay rootfs mounted by the shim
             # in ITS mount namespace (reference: the OCI rootfs the
             # containerd snapshotter provides)

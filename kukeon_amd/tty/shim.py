@@ -25,6 +25,7 @@ stages.go:62-88, kukepause PID-1 semantics cmd/kukepause/main.go:17-60 via
 from __future__ import annotations
 
 import argparse
+import contextlib
 import fcntl
 import json
 import os
@@ -206,11 +207,36 @@ class Shim:
                                         str(self.dir / "etc"))
                     except OSError as e:
                         sys.stderr.write(f"etc render: {e}\n")
+                if not rootfs:
+                    self._bind_volumes(nsmod, ns)
             except OSError as e:
                 degraded = True
                 sys.stderr.write(f"mount ns: {e}\n")
         write_json(self.dir / "ns.json",
                    {"held": held, "degraded": degraded})
+
+    def _bind_volumes(self, nsmod, ns: dict, prefix: str = "") -> None:
+        """Bind-mount declared volumes at their targets inside this
+        container's private mount namespace (real mounts — reference
+        internal/ctr/spec.go volume mounts; the host tree is
+        unaffected)."""
+        for b in ns.get("binds") or []:
+            src, dst = b.get("src"), b.get("dst")
+            if not src or not dst:
+                continue
+            if prefix:
+                dst = os.path.join(prefix, dst.lstrip("/"))
+            try:
+                if os.path.isdir(src):
+                    os.makedirs(dst, exist_ok=True)
+                else:
+                    os.makedirs(os.path.dirname(dst) or "/",
+                                exist_ok=True)
+                    if not os.path.exists(dst):
+                        open(dst, "a").close()
+                nsmod.mount(src, dst, "", nsmod.MS_BIND | nsmod.MS_REC)
+            except OSError as e:
+                sys.stderr.write(f"bind {src} -> {dst}: {e}\n")
 
     def _enter_rootfs(self, nsmod, ns: dict, rootfs) -> None:
         """chroot into an image rootfs keeping the shim functional: the
@@ -245,6 +271,9 @@ class Shim:
             f.write(f"127.0.1.1\t{hostname}\n")
             for name, ip in sorted((ns.get("hosts") or {}).items()):
                 f.write(f"{ip}\t{name}\n")
+        # volumes bind BEFORE the chroot (host sources are unreachable
+        # after), at rootfs-prefixed targets
+        self._bind_volumes(nsmod, ns, prefix=rootfs)
         nsmod.enter_rootfs(rootfs)
 
     def child_env(self) -> dict:
@@ -339,12 +368,21 @@ class Shim:
         activity = self.dir / "activity"
         activity.touch()
         clients = []
+        # SCM_RIGHTS fast path (reference cmd/kuketty/main.go:17-30): one
+        # client may hold the PTY master fd directly, making its byte
+        # pump CLI<->kernel with zero shim copies. While an fd-holder is
+        # live the shim stays out of the master (reads would steal bytes
+        # from the holder), so capture pauses for that span — a
+        # documented deviation; it resumes on detach.
+        self._fd_holder = None
         rc = 0
         fcntl.fcntl(master, fcntl.F_SETFL,
                     fcntl.fcntl(master, fcntl.F_GETFL) | os.O_NONBLOCK)
         try:
             while True:
-                rl = [srv, master] + clients
+                rl = [srv] + clients
+                if self._fd_holder is None:
+                    rl.append(master)
                 try:
                     rd, _, _ = select.select(rl, [], [], 0.5)
                 except InterruptedError:
@@ -382,6 +420,8 @@ class Shim:
                             data = b""
                         if not data:
                             clients.remove(r)
+                            if self._fd_holder is r:
+                                self._fd_holder = None  # resume capture
                             r.close()
                             continue
                         activity.touch()
@@ -411,7 +451,21 @@ class Shim:
             pass
 
     def _handle_client_data(self, c, master: int, data: bytes) -> None:
-        # in-band resize escape: \x00R{"rows":..,"cols":..}\n ; raw otherwise
+        # control escapes: \x00F\n requests the PTY master via SCM_RIGHTS;
+        # \x00R{"rows":..,"cols":..}\n resizes; raw bytes otherwise
+        if data.startswith(b"\x00F"):
+            nl = data.find(b"\n")
+            if nl >= 0:
+                data = data[nl + 1:]
+            if self._fd_holder is None:
+                try:
+                    socket.send_fds(c, [b"\x00FDOK\n"], [master])
+                    self._fd_holder = c
+                except OSError:
+                    pass
+            else:
+                with contextlib.suppress(OSError):
+                    c.sendall(b"\x00FNO\n")
         if data.startswith(b"\x00R"):
             nl = data.find(b"\n")
             if nl > 0:

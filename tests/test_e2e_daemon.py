@@ -353,3 +353,79 @@ def test_container_repo_setup(harness, tmp_path):
     assert repos and repos[0]["state"] == "cloned", repos
     client.KillCell(realm="default", space="default", stack="default",
                     name="repocell")
+
+
+def test_scm_rights_fd_attach(harness):
+    """SCM_RIGHTS fast path (reference cmd/kuketty/main.go:17-30): the
+    first attacher gets the PTY master fd and pumps bytes with no shim
+    relay; a concurrent second attacher falls back to the relay path;
+    capture resumes once the fd holder detaches."""
+    import socket as sk
+
+    ctl, srv, client = harness
+    doc = api.CellDoc(
+        metadata=api.Metadata(name="fdcell"),
+        spec=api.CellSpec(
+            realm_id="default", space_id="default", stack_id="default",
+            containers=[api.ContainerSpec(
+                id="term", image="busybox", command="/bin/sh",
+                args=["-i"], attachable=True)]))
+    client.CreateCell(doc=doc.to_dict())
+    client.StartCell(realm="default", space="default", stack="default",
+                     name="fdcell")
+    res = client.AttachContainer(realm="default", space="default",
+                                 stack="default", name="fdcell")
+    path = res["hostSocketPath"]
+    deadline = time.monotonic() + 5
+    while not attach_mod.ping(path) and time.monotonic() < deadline:
+        time.sleep(0.1)
+
+    s1 = sk.socket(sk.AF_UNIX, sk.SOCK_STREAM)
+    s1.connect(path)
+    s1.settimeout(5.0)
+    assert b"kukeon-tty/1" in s1.recv(4096)
+    mfd = attach_mod._request_fd(s1)
+    assert mfd is not None  # got the PTY master via SCM_RIGHTS
+    # the DIRECT pump: write to the master fd, read the echo back
+    os.write(mfd, b"echo fd-$((40+2))\n")
+    buf = b""
+    deadline = time.monotonic() + 5
+    while b"fd-42" not in buf and time.monotonic() < deadline:
+        import select as _sel
+        rd, _, _ = _sel.select([mfd], [], [], 0.5)
+        if rd:
+            buf += os.read(mfd, 4096)
+    assert b"fd-42" in buf
+
+    # second concurrent attacher is refused the fd (falls back to relay)
+    s2 = sk.socket(sk.AF_UNIX, sk.SOCK_STREAM)
+    s2.connect(path)
+    s2.settimeout(5.0)
+    s2.recv(4096)
+    assert attach_mod._request_fd(s2) is None
+    s2.close()
+
+    # detach the fd holder; the shim resumes the relay + capture
+    os.close(mfd)
+    s1.close()
+    time.sleep(0.3)
+    s3 = sk.socket(sk.AF_UNIX, sk.SOCK_STREAM)
+    s3.connect(path)
+    s3.settimeout(5.0)
+    s3.recv(4096)
+    s3.sendall(b"echo post-$((40+3))\n")
+    buf = b""
+    deadline = time.monotonic() + 5
+    while b"post-43" not in buf and time.monotonic() < deadline:
+        try:
+            buf += s3.recv(4096)
+        except OSError:
+            break
+    assert b"post-43" in buf
+    s3.close()
+    # capture holds the post-detach output (recorded by the shim again)
+    res = client.LogPath(realm="default", space="default", stack="default",
+                         name="fdcell")
+    assert b"post-43" in Path(res["path"]).read_bytes()
+    client.KillCell(realm="default", space="default", stack="default",
+                    name="fdcell")

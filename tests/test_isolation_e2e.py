@@ -7,6 +7,7 @@ asserted by observing the packets land (or not) on a live listener.
 Hosts without CAP_SYS_ADMIN/CAP_NET_ADMIN skip (the controller records
 the same degradation at runtime).
 """
+import contextlib
 import socket
 import threading
 import time
@@ -230,3 +231,41 @@ def test_default_allow_space_routes_everything(ctl):
     assert any(line.split()[1] == "00000000"
                for line in routes.splitlines()[1:] if line.strip()), routes
     ctl.delete_cell("default", space, "default", name, force=True)
+
+
+@pytest.mark.skipif(not nsmod.can_unshare(nsmod.CLONE_NEWNS),
+                    reason="no mount ns")
+def test_volumes_are_real_bind_mounts(ctl, tmp_path):
+    """A declared volume appears AT ITS TARGET inside the container's
+    mount namespace as a real mount; the host tree is untouched."""
+    vol = tmp_path / "shared"
+    vol.mkdir()
+    (vol / "payload.txt").write_text("mounted-for-real\n")
+    name = f"vol-{uuid.uuid4().hex[:6]}"
+    doc = api.CellDoc(
+        metadata=api.Metadata(name=name),
+        spec=api.CellSpec(
+            realm_id="default", space_id="default", stack_id="default",
+            containers=[api.ContainerSpec(
+                id="main", image="none", command="sh",
+                args=["-c", "cat /mnt/kuke-vol/payload.txt > got.txt "
+                            "2>&1; echo inner > /mnt/kuke-vol/back.txt; "
+                            "sleep 30"],
+                volumes=[api.VolumeMount(name="shared",
+                                         source=str(vol),
+                                         target="/mnt/kuke-vol")])]))
+    ctl.create_cell(doc)
+    ctl.start_cell("default", "default", "default", name)
+    cdir = ctl.store.cell_dir("default", "default", "default", name) / "main"
+    assert wait_for(lambda: (cdir / "got.txt").exists())
+    time.sleep(0.1)
+    assert "mounted-for-real" in (cdir / "got.txt").read_text()
+    # writes through the mount land in the volume source
+    assert wait_for(lambda: (vol / "back.txt").exists())
+    # the MOUNT is private: the host sees only the empty mountpoint
+    # directory (created on the shared fs), never the volume's content
+    host_view = Path("/mnt/kuke-vol")
+    assert not (host_view / "payload.txt").exists()
+    ctl.delete_cell("default", "default", "default", name, force=True)
+    with contextlib.suppress(OSError):
+        host_view.rmdir()
