@@ -75,7 +75,14 @@ class LLMEngine:
         if nblocks <= 0:
             if self.is_cuda:
                 free, _total = torch.cuda.mem_get_info(self.device)
-                budget = int(free * ecfg.kv_mem_fraction) - (2 << 30)
+                # oversubscribed rehearsals (more local ranks than GPUs,
+                # e.g. a world-2 torchrun on a 1-GPU box) share the HBM:
+                # split the KV budget so the second rank isn't starved
+                import os as _os
+                lw = int(_os.environ.get("LOCAL_WORLD_SIZE", "1") or 1)
+                ndev = max(1, torch.cuda.device_count())
+                share = max(1, -(-lw // ndev))
+                budget = int(free * ecfg.kv_mem_fraction) // share - (2 << 30)
                 nblocks = PagedKVCache.blocks_from_bytes(
                     max(budget, 1 << 28), cfg.num_layers, self.hk,
                     ecfg.block_size, cfg.head_dim, dtype_bytes=kv_bytes)

@@ -37,6 +37,43 @@ class AttnMeta:
     tmp_ml: Optional[torch.Tensor] = None
 
 
+def split_prefill_meta(meta: AttnMeta, parts: int = 2):
+    """Split a packed prefill batch into sequence groups (token-balanced)
+    for the pipelined TP path. Returns [(row0, row1, AttnMeta), ...]."""
+    qs = meta.q_starts  # [nseq, 2] (ctx_start, packed row offset)
+    nseq = int(qs.shape[0])
+    T = int(meta.positions.shape[0])
+    row_offs = [int(qs[i, 1]) for i in range(nseq)] + [T]
+    # choose the sequence boundary closest to half the tokens
+    target = T // 2
+    split = max(1, min(nseq - 1,
+                       min(range(1, nseq),
+                           key=lambda i: abs(row_offs[i] - target))))
+    groups = []
+    for (s0, s1) in ((0, split), (split, nseq)):
+        r0, r1 = row_offs[s0], row_offs[s1]
+        g_qs = qs[s0:s1].clone()
+        g_qs[:, 1] -= r0
+        qb_seq, qb_start = [], []
+        for s in range(s0, s1):
+            chunk = row_offs[s + 1] - row_offs[s]
+            for qb in range(0, chunk, 32):
+                qb_seq.append(s - s0)
+                qb_start.append(qb)
+        dev = meta.positions.device
+        gm = AttnMeta(
+            mode="prefill",
+            positions=meta.positions[r0:r1],
+            slot_mapping=meta.slot_mapping[r0:r1],
+            block_table=meta.block_table[s0:s1],
+            seq_lens=meta.seq_lens[s0:s1],
+            q_starts=g_qs,
+            qb_seq=torch.tensor(qb_seq, dtype=torch.int32, device=dev),
+            qb_start=torch.tensor(qb_start, dtype=torch.int32, device=dev))
+        groups.append((r0, r1, gm))
+    return groups
+
+
 def _init_weight(shape, device, std=0.02, seed=None):
     w = torch.empty(shape, dtype=torch.bfloat16, device=device)
     w.normal_(0.0, std)
@@ -145,6 +182,30 @@ class LlamaLayer(nn.Module):
         h = self.mlp.forward(h)
         return h, residual
 
+    # -- pipelined TP prefill phases (comm/GEMM overlap): the o-proj and
+    # down-proj partials are returned UN-reduced; the caller overlaps
+    # their all-reduce (comm stream) with the other sequence group's
+    # compute (parallel.tp_all_reduce_async) --
+    def part_attn(self, x, residual, cos_sin, kc, vc, meta):
+        if residual is None:
+            residual = x
+            h = torch.empty_like(x)
+            ops.rmsnorm(h, x, self.input_norm, self.eps)
+        else:
+            ops.fused_add_rmsnorm(x, residual, self.input_norm, self.eps)
+            h = x
+        attn_out = self.attn.forward_pre_o(h, cos_sin, kc, vc, meta)
+        return ops.linear(attn_out, self.attn.o_w), residual
+
+    def part_mlp(self, o_reduced, residual):
+        ops.fused_add_rmsnorm(o_reduced, residual, self.post_norm, self.eps)
+        h = o_reduced
+        gu = ops.linear(h, self.mlp.gate_up_w)
+        act = torch.empty(h.shape[0], self.mlp.inter, dtype=h.dtype,
+                          device=h.device)
+        ops.silu_mul(act, gu)
+        return ops.linear(act, self.mlp.down_w)
+
 
 class LlamaModel(nn.Module):
     """Random-init Llama decoder for the serving engine (bench contract:
@@ -178,6 +239,13 @@ class LlamaModel(nn.Module):
 
     def forward(self, input_ids: torch.Tensor, kv_k: torch.Tensor,
                 kv_v: torch.Tensor, meta: AttnMeta) -> torch.Tensor:
+        import os as _os
+        if (parallel.tp_size() > 1 and meta.mode == "prefill"
+                and meta.q_starts is not None
+                and int(meta.q_starts.shape[0]) >= 2
+                and _os.environ.get("KUKEON_TP_OVERLAP", "1") != "0"):
+            return self._forward_prefill_overlap(input_ids, kv_k, kv_v,
+                                                 meta)
         x = F.embedding(input_ids.long(), self.embed)
         residual = None
         for i, layer in enumerate(self.layers):
@@ -185,6 +253,45 @@ class LlamaModel(nn.Module):
                                         kv_v[i], meta)
         ops.fused_add_rmsnorm(x, residual, self.final_norm, self.cfg.rms_eps)
         return x
+
+    _overlap_runs = 0  # test hook: counts pipelined-path invocations
+
+    def _forward_prefill_overlap(self, input_ids, kv_k, kv_v,
+                                 meta: AttnMeta) -> torch.Tensor:
+        """TP prefill with comm/GEMM overlap: sequences split into two
+        groups whose layer phases interleave, so each group's o-proj and
+        down-proj all-reduce (on the comm stream, RCCL over xGMI) flies
+        under the OTHER group's attention / MLP GEMMs. Numerically
+        identical to the plain path (group split is by whole sequences;
+        reduction order within each all-reduce is unchanged) — the
+        gloo world-2 equivalence test pins that."""
+        LlamaModel._overlap_runs += 1
+        groups = split_prefill_meta(meta)
+        xs, res, ar = [], [], []
+        for (r0, r1, gm) in groups:
+            xs.append(F.embedding(input_ids[r0:r1].long(), self.embed))
+            res.append(None)
+            ar.append(None)
+        os_ = [None] * len(groups)
+        for i, layer in enumerate(self.layers):
+            for g, (r0, r1, gm) in enumerate(groups):
+                parallel.wait_comm(ar[g])  # down-proj AR of layer i-1
+                os_[g], res[g] = layer.part_attn(xs[g], res[g],
+                                                 self.cos_sin, kv_k[i],
+                                                 kv_v[i], gm)
+                ar[g] = parallel.tp_all_reduce_async(os_[g])
+            for g, (r0, r1, gm) in enumerate(groups):
+                parallel.wait_comm(ar[g])  # o-proj AR
+                xs[g] = layer.part_mlp(os_[g], res[g])
+                ar[g] = parallel.tp_all_reduce_async(xs[g])
+        out = torch.empty(input_ids.shape[0], self.cfg.hidden_size,
+                          dtype=xs[0].dtype, device=xs[0].device)
+        for g, (r0, r1, gm) in enumerate(groups):
+            parallel.wait_comm(ar[g])
+            ops.fused_add_rmsnorm(xs[g], res[g], self.final_norm,
+                                  self.cfg.rms_eps)
+            out[r0:r1] = xs[g]
+        return out
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
         return ops.linear(hidden, self.lm_head)

@@ -77,6 +77,45 @@ def tp_all_reduce(t: torch.Tensor) -> torch.Tensor:
     return t
 
 
+# ---- comm/GEMM overlap (north star: all-reduce on a dedicated HIP
+# stream overlapped with the next GEMM; used by the pipelined TP prefill
+# in models/llama.py) ----
+_COMM_STREAM = None
+
+
+def comm_stream():
+    global _COMM_STREAM
+    if _COMM_STREAM is None and torch.cuda.is_available():
+        _COMM_STREAM = torch.cuda.Stream()
+    return _COMM_STREAM
+
+
+def tp_all_reduce_async(t: torch.Tensor):
+    """Launch the TP all-reduce on the comm stream; returns an event the
+    consumer stream must wait on (None when already complete — CPU/gloo
+    or TP=1). RCCL ties the collective to the stream current at call
+    time, so compute on the default stream overlaps the transfer."""
+    if _TP_SIZE <= 1:
+        return None
+    if not t.is_cuda:
+        dist.all_reduce(t, group=_TP_GROUP)
+        return None
+    ready = torch.cuda.Event()
+    ready.record()
+    cs = comm_stream()
+    with torch.cuda.stream(cs):
+        cs.wait_event(ready)
+        dist.all_reduce(t, group=_TP_GROUP)
+        done = torch.cuda.Event()
+        done.record()
+    return done
+
+
+def wait_comm(ev) -> None:
+    if ev is not None:
+        torch.cuda.current_stream().wait_event(ev)
+
+
 def barrier() -> None:
     if dist.is_initialized():
         dist.barrier()

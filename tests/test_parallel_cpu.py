@@ -72,7 +72,48 @@ def _tp_engine_worker(rank, world, port, result_dir):
     dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("worker", [_tp_linear_worker, _tp_engine_worker])
+def _tp_overlap_worker(rank, world, port, result_dir):
+    """Pipelined TP prefill (comm/GEMM overlap, 2 sequence groups) must
+    produce exactly the tokens of the plain TP path."""
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from kukeon_amd import parallel
+    parallel.init_tensor_parallel(world)
+    from kukeon_amd.engine.config import (EngineConfig, SamplingParams,
+                                          tiny_llama)
+    from kukeon_amd.engine.engine import LLMEngine
+    from kukeon_amd.engine.kv_cache import SequenceKV
+    from kukeon_amd.models.llama import LlamaModel
+
+    cfg = tiny_llama()
+    prompts = [[3, 1, 4, 1, 5, 9, 2, 6], [2, 7, 1, 8, 2, 8],
+               [1, 1, 2, 3, 5, 8, 13, 21, 34, 55]]
+    runs = {}
+    for mode in ("0", "1"):
+        os.environ["KUKEON_TP_OVERLAP"] = mode
+        ecfg = EngineConfig(max_model_len=128, max_sessions=4,
+                            num_kv_blocks=96, use_graphs=False,
+                            tp_size=world)
+        model = LlamaModel(cfg, device="cpu")
+        engine = LLMEngine(model, cfg, ecfg, device="cpu")
+        for p in prompts:  # one multi-sequence prefill batch
+            engine.add_request(SequenceKV(ecfg.block_size), list(p),
+                               SamplingParams(temperature=0.0,
+                                              max_new_tokens=4))
+        toks = []
+        while engine.has_work():
+            for o in engine.step():
+                toks.append((o.req_id, tuple(o.new_tokens)))
+        runs[mode] = sorted(toks)
+    assert runs["0"] == runs["1"], runs
+    assert LlamaModel._overlap_runs > 0  # the pipelined path really ran
+    del os.environ["KUKEON_TP_OVERLAP"]
+    dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("worker", [_tp_linear_worker, _tp_engine_worker,
+                                    _tp_overlap_worker])
 def test_tp_world2(worker, tmp_path):
     port = 29600 + (os.getpid() + (0 if worker is _tp_linear_worker else 7)) % 500
     mp.spawn(worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
