@@ -395,9 +395,21 @@ def kill(ctx, name):
 
 @cli.command()
 @click.argument("name")
+@click.option("-c", "--container", default="",
+              help="restart only this container of the cell")
 @pass_ctx
-def restart(ctx, name):
-    """Restart a cell."""
+def restart(ctx, name, container):
+    """Restart a cell (or one container of it with -c)."""
+    if container:
+        try:
+            cell = ctx.client.call("RestartContainer", realm=ctx.realm,
+                                   space=ctx.space, stack=ctx.stack,
+                                   name=name, container=container)
+            click.echo(f"cell {name}/{container}: "
+                       f"{cell['status']['state']}")
+        except errors.KukeonError as e:
+            _die(e)
+        return
     _lifecycle(ctx, "RestartCell", name)
 
 

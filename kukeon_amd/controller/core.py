@@ -698,6 +698,58 @@ class Controller:
         self._persist_cell(doc)
         return doc
 
+    @locked_cell
+    def stop_container(self, realm, space, stack, name,
+                       container_id: str) -> api.CellDoc:
+        """Stop ONE container of a cell (reference runner single-
+        container lifecycle, start.go:1239); cell status re-derives on
+        the next reconcile."""
+        doc = self.get_cell(realm, space, stack, name)
+        if not any((c.id or "main") == container_id
+                   for c in doc.spec.containers):
+            raise errors.ContainerNotFound(f"{name}/{container_id}")
+        cdir = self.store.cell_dir(realm, space, stack, name) / container_id
+        self.runtime.stop(cdir)
+        return self.reconcile_cell(realm, space, stack, name)
+
+    @locked_cell
+    def start_container(self, realm, space, stack, name,
+                        container_id: str) -> api.CellDoc:
+        """(Re)start ONE container of a running cell with a freshly
+        rendered env (GPU pinning and volume binds preserved)."""
+        doc = self.get_cell(realm, space, stack, name)
+        spec = None
+        for c in doc.spec.containers:
+            if (c.id or "main") == container_id:
+                spec = c
+                break
+        if spec is None:
+            raise errors.ContainerNotFound(f"{name}/{container_id}")
+        cell_dir = self.store.cell_dir(realm, space, stack, name)
+        cdir = cell_dir / container_id
+        probe = self.runtime.probe(cdir)
+        if not probe.running:
+            gpu_ids = self.gpus.assignments.get(self._gpu_owner(doc), [])
+            cursor = 0
+            mine: List[int] = []
+            for c in doc.spec.containers:
+                take = gpu_ids[cursor:cursor + c.gpus]
+                cursor += c.gpus
+                if (c.id or "main") == container_id:
+                    mine = take
+            env = self._container_env(doc, spec, mine)
+            env += self._mount_volumes(doc, spec, cdir)
+            net_info = self.store.read(cell_dir / "network.json")
+            cns = self._container_ns_config(doc, spec, cell_dir, net_info)
+            self.runtime.start_container(cdir, spec, env,
+                                         self._cell_cgroup(doc), cns)
+        return self.reconcile_cell(realm, space, stack, name)
+
+    def restart_container(self, realm, space, stack, name,
+                          container_id: str) -> api.CellDoc:
+        self.stop_container(realm, space, stack, name, container_id)
+        return self.start_container(realm, space, stack, name, container_id)
+
     def restart_cell(self, realm, space, stack, name) -> api.CellDoc:
         self.stop_cell(realm, space, stack, name)
         return self.start_cell(realm, space, stack, name)

@@ -165,6 +165,21 @@ class Service:
         return self.ctl.restart_cell(p["realm"], p["space"], p["stack"],
                                      p["name"]).to_dict()
 
+    def StartContainer(self, p):
+        return self.ctl.start_container(
+            p["realm"], p["space"], p["stack"], p["name"],
+            p["container"]).to_dict()
+
+    def StopContainer(self, p):
+        return self.ctl.stop_container(
+            p["realm"], p["space"], p["stack"], p["name"],
+            p["container"]).to_dict()
+
+    def RestartContainer(self, p):
+        return self.ctl.restart_container(
+            p["realm"], p["space"], p["stack"], p["name"],
+            p["container"]).to_dict()
+
     def DeleteCell(self, p):
         self.ctl.delete_cell(p["realm"], p["space"], p["stack"], p["name"],
                              p.get("force", False))
