@@ -104,6 +104,8 @@ class Controller:
         # + reconcile loop); every cell mutation holds its cell's lock —
         # reference runner/runner.go:333-340
         self.cell_locks = ScopeLocks()
+        self._session_cache: Dict[Tuple[str, str, str],
+                                  Optional[api.SessionDoc]] = {}
 
     # ==================================================================
     # bootstrap
@@ -391,14 +393,29 @@ class Controller:
         return [f"{k}={v}" for k, v in env.items()]
 
     def _stack_session(self, realm, space, stack) -> Optional[api.SessionDoc]:
+        # memoized per stack (invalidated on any session mutation): the
+        # per-start directory scan was O(sessions) JSON reads on every
+        # container-env build
+        key = (realm, space, stack)
+        if key in self._session_cache:
+            return self._session_cache[key]
         d = self.run_path / "sessions" / realm / space / stack
-        if not d.is_dir():
-            return None
-        for p in sorted(d.glob("*.json")):
-            data = self.store.read(p)
-            if data and data.get("status", {}).get("state") ==                     api.STATE_RUNNING:
-                return api.SessionDoc.from_dict(data)
-        return None
+        found = None
+        if d.is_dir():
+            for p in sorted(d.glob("*.json")):
+                data = self.store.read(p)
+                if data and data.get("status", {}).get("state") ==                         api.STATE_RUNNING:
+                    found = api.SessionDoc.from_dict(data)
+                    break
+        self._session_cache[key] = found
+        return found
+
+    def _invalidate_session_cache(self, realm=None, space=None,
+                                  stack=None) -> None:
+        if realm is None:
+            self._session_cache.clear()
+        else:
+            self._session_cache.pop((realm, space, stack), None)
 
     @locked_cell
     def start_cell(self, realm, space, stack, name) -> api.CellDoc:
@@ -982,6 +999,7 @@ class Controller:
     # sessions
     # ==================================================================
     def create_session(self, doc: api.SessionDoc) -> api.SessionDoc:
+        self._invalidate_session_cache()
         parser.validate_document(doc)
         realm = doc.spec.realm_id or naming.DEFAULT_REALM
         space = doc.spec.space_id or naming.DEFAULT_SPACE
@@ -1020,6 +1038,7 @@ class Controller:
 
     def close_session(self, realm, space, stack, name,
                       state: str = api.STATE_COMPLETED) -> api.SessionDoc:
+        self._invalidate_session_cache(realm, space, stack)
         doc = self.get_session(realm, space, stack, name)
         if doc.status.state in (api.STATE_COMPLETED, api.STATE_TERMINATED):
             return doc
@@ -1079,6 +1098,7 @@ class Controller:
         return doc
 
     def delete_session(self, realm, space, stack, name) -> None:
+        self._invalidate_session_cache(realm, space, stack)
         self.close_session(realm, space, stack, name, api.STATE_TERMINATED)
         self.store.delete(self.store.session_path(realm, space, stack, name))
 
@@ -1367,6 +1387,7 @@ class Controller:
             doc.status.out_of_sync_error = str(e)
 
     def reconcile_sessions(self) -> int:
+        self._invalidate_session_cache()
         visited = 0
         now = self.now()
         for doc in self.list_sessions():

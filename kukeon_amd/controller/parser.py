@@ -27,13 +27,13 @@ def detect_and_convert(raw: Dict[str, Any], validate: bool = True):
     kind = raw.get("kind")
     if not kind:
         raise errors.ValidationError("document missing kind")
-    cls = api.DOC_TYPES.get(kind)
-    if cls is None:
+    if api.DOC_TYPES.get(kind) is None:
         raise errors.ValidationError(f"unknown kind {kind!r}")
-    av = raw.get("apiVersion", api.API_VERSION)
-    if av not in (api.API_VERSION, f"kukeon.io/{api.API_VERSION}"):
-        raise errors.ValidationError(f"unsupported apiVersion {av!r}")
-    doc = cls.from_dict(raw)
+    # version seam: detect + up-convert + default (api/scheme.py — the
+    # apischeme analog); the controller only ever sees internal-shape
+    # v1beta1 documents
+    from kukeon_amd.api import scheme
+    doc = scheme.normalize_doc(raw)
     if validate:
         validate_document(doc)
     return doc

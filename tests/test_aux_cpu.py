@@ -288,3 +288,86 @@ def test_store_delete_leaves_lock_tombstone(tmp_path):
     ino_before = lockp.stat().st_ino
     store.create_exclusive(doc, {"spec": {}})
     assert lockp.stat().st_ino == ino_before
+
+
+def test_apischeme_version_translation():
+    """api/scheme.py: v1alpha1 wire docs up-convert through the seam
+    (scope keys, flat container shorthand, enum renames) and internal
+    docs export back down with named field loss."""
+    from kukeon_amd.api import scheme
+    from kukeon_amd.api import v1beta1 as api
+
+    alpha_cell = {
+        "apiVersion": "v1alpha1", "kind": "Cell",
+        "metadata": {"name": "old-style"},
+        "spec": {"realm": "prod", "space": "web", "stack": "app",
+                 "autoRemove": True, "image": "tool", "command": "sleep",
+                 "args": ["5"]}}
+    doc = scheme.normalize_doc(alpha_cell)
+    assert doc.api_version == "v1beta1"
+    assert doc.spec.realm_id == "prod" and doc.spec.stack_id == "app"
+    assert doc.spec.auto_delete is True
+    assert len(doc.spec.containers) == 1
+    assert doc.spec.containers[0].command == "sleep"
+    assert doc.spec.containers[0].id == "main"
+
+    alpha_space = {
+        "apiVersion": "kukeon.io/v1alpha1", "kind": "Space",
+        "metadata": {"name": "web"},
+        "spec": {"realm": "prod",
+                 "network": {"egress": {"mode": "deny",
+                                        "allow": [{"cidr": "10.0.0.0/8"}]}}}}
+    sp = scheme.normalize_doc(alpha_space)
+    assert sp.spec.network.egress.default == "deny"
+
+    # defaulting: scope falls back to the default hierarchy
+    bare = scheme.normalize_doc({"kind": "Cell",
+                                 "metadata": {"name": "c"},
+                                 "spec": {"containers": [
+                                     {"image": "x", "command": "sleep"}]}})
+    assert bare.spec.realm_id == "default"
+    assert bare.spec.containers[0].id == "main"
+
+    # downgrade export names what it drops
+    beta = api.CellDoc(
+        metadata=api.Metadata(name="c"),
+        spec=api.CellSpec(realm_id="r", space_id="s", stack_id="t",
+                          containers=[api.ContainerSpec(
+                              id="main", image="x", command="sleep",
+                              restart_policy="on-failure",
+                              repos=[api.ContainerRepo(url="u")])]))
+    wire, lost = scheme.to_wire(beta, "v1alpha1")
+    assert wire["spec"]["realm"] == "r"
+    assert wire["spec"]["containers"][0]["restartPolicy"] == "onFailure"
+    assert any("repos" in f for f in lost)
+    # unsupported version rejected
+    with pytest.raises(errors.KukeonError):
+        scheme.normalize({"apiVersion": "v2", "kind": "Cell",
+                          "metadata": {"name": "x"}, "spec": {}})
+
+
+def test_apply_accepts_v1alpha1_documents(tmp_path):
+    """The whole apply pipeline accepts legacy-version YAML through the
+    scheme seam."""
+    from kukeon_amd.controller.core import Controller
+    from kukeon_amd.runtime.process import FakeRuntime
+
+    ctl = Controller(str(tmp_path / "run"), runtime=FakeRuntime(),
+                     gpu_devices=[])
+    ctl.bootstrap()
+    text = """
+apiVersion: v1alpha1
+kind: Cell
+metadata: {name: legacy}
+spec:
+  realm: default
+  space: default
+  stack: default
+  image: busybox
+  command: sleep
+  args: ["9"]
+"""
+    res = ctl.apply_documents(text)
+    assert res[0].action == "created", (res[0].action, res[0].error)
+    doc = ctl.get_cell("default", "default", "default", "legacy")
+    assert doc.spec.containers[0].args == ["9"]
