@@ -33,6 +33,8 @@ class RoleRef:
     ref: str = ""
     needs: RoleNeeds = field(default_factory=RoleNeeds)
     harnesses: List[str] = field(default_factory=list)
+    # per-role template variables (merged over team vars at render time)
+    vars: dict = field(default_factory=dict)
 
 
 @dataclass
@@ -53,7 +55,9 @@ class ProjectTeam:
     metadata: Metadata = field(default_factory=Metadata)
     source: TeamSource = field(default_factory=TeamSource)
     default_harnesses: List[str] = field(default_factory=list)
+    default_needs: RoleNeeds = field(default_factory=RoleNeeds)
     roles: List[RoleRef] = field(default_factory=list)
+    vars: dict = field(default_factory=dict)
 
 
 @dataclass
@@ -112,7 +116,10 @@ def parse_team_doc(raw: Dict[str, Any]):
             roles.append(RoleRef(ref=r.get("ref", ""),
                                  needs=RoleNeeds(
                                      image=list(needs.get("image", []))),
-                                 harnesses=list(r.get("harnesses", []))))
+                                 harnesses=list(r.get("harnesses", [])),
+                                 vars=dict(r.get("vars", {}))))
+        defaults = spec.get("defaults") or {}
+        dneeds = defaults.get("needs") or {}
         doc = ProjectTeam(
             metadata=md,
             source=TeamSource(repo=src.get("repo", ""),
@@ -120,9 +127,10 @@ def parse_team_doc(raw: Dict[str, Any]):
                               tag=src.get("tag", ""),
                               commit=src.get("commit", ""),
                               path=src.get("path", "")),
-            default_harnesses=list(
-                (spec.get("defaults") or {}).get("harnesses", [])),
-            roles=roles)
+            default_harnesses=list(defaults.get("harnesses", [])),
+            default_needs=RoleNeeds(image=list(dneeds.get("image", []))),
+            roles=roles,
+            vars=dict(spec.get("vars", {})))
         if not doc.metadata.name:
             raise errors.ValidationError("ProjectTeam needs metadata.name")
         if not doc.roles:

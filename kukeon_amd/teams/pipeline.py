@@ -147,7 +147,11 @@ def render_team(team: ProjectTeam, roles: Dict[str, Role],
             h = harnesses.get(hname)
             if h is None:
                 raise errors.ValidationError(f"harness {hname!r} not found")
-            image = select_image(catalog, hname, rr.needs.image)
+            # needs-merge (reference teamrender): team defaults union
+            # role requirements before image selection
+            from kukeon_amd.teams.template import TemplateEngine, merge_needs
+            needs = merge_needs(team.default_needs.image, rr.needs.image)
+            image = select_image(catalog, hname, needs)
             tmpl_path = source_dir / h.template
             if not tmpl_path.exists():
                 raise errors.ValidationError(
@@ -160,8 +164,13 @@ def render_team(team: ProjectTeam, roles: Dict[str, Role],
                 "TEAM_ROOT": str(team_root),
                 "SKILL_PATH": h.skill_path,
                 "ROLE_PROMPT": role.prompt,
+                "NEEDS": needs,
+                "ROLE_ENV": list(role.env),
             }
-            rendered = Template(tmpl_path.read_text()).safe_substitute(ctx)
+            ctx.update(team.vars)
+            ctx.update(rr.vars)
+            engine = TemplateEngine(partials_dir=source_dir / "partials")
+            rendered = engine.render(tmpl_path.read_text(), ctx)
             for raw in yaml.safe_load_all(rendered):
                 if not raw:
                     continue

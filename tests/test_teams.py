@@ -166,3 +166,123 @@ def test_parse_rejects_wrong_apiversion():
     with pytest.raises(errors.ValidationError):
         parse_team_file("apiVersion: v1beta1\nkind: Role\n"
                         "metadata: {name: x}\n")
+
+
+def test_template_engine_partials_conditionals_each(tmp_path):
+    """teamrender engine: partials, {{#if}}/{{else}}, {{#each}} and the
+    ${VAR} layer compose (reference internal/teamrender partials
+    pipeline)."""
+    from kukeon_amd.teams.template import TemplateEngine, merge_needs
+    from kukeon_amd.api import errors as kerrors
+
+    pd = tmp_path / "partials"
+    pd.mkdir()
+    (pd / "resources.tmpl").write_text(
+        "{{#if GPU}}gpus: ${GPU}{{else}}gpus: 0{{/if}}\n")
+    (pd / "env.tmpl").write_text(
+        "env:\n{{#each ROLE_ENV}}  - ${ITEM}\n{{/each}}")
+    eng = TemplateEngine(partials_dir=pd)
+    text = ("name: ${ROLE}\n"
+            "{{> resources}}"
+            "{{> env}}")
+    out = eng.render(text, {"ROLE": "builder", "GPU": "2",
+                            "ROLE_ENV": ["A=1", "B=2"]})
+    assert "name: builder" in out
+    assert "gpus: 2" in out
+    assert "  - A=1" in out and "  - B=2" in out
+    # falsy branch
+    out = eng.render("{{#if GPU}}y{{else}}n{{/if}}", {"GPU": ""})
+    assert out == "n"
+    # nested if inside each
+    out = eng.render(
+        "{{#each XS}}{{#if LOUD}}${ITEM}!{{else}}${ITEM}{{/if}},{{/each}}",
+        {"XS": ["a", "b"], "LOUD": "yes"})
+    assert out == "a!,b!,"
+    # missing partial is a validation error
+    import pytest as _pytest
+    with _pytest.raises(kerrors.ValidationError):
+        eng.render("{{> nope}}", {})
+    # needs-merge unions in order without dupes
+    assert merge_needs(["git"], ["git", "gpu"], None) == ["git", "gpu"]
+
+
+def test_render_team_with_partials_and_needs_merge(tmp_path):
+    """Full pipeline: a harness template using partials + conditionals
+    renders per role with team/role vars and the merged needs drive
+    image selection."""
+    from kukeon_amd.teams import parse_team_doc
+    from kukeon_amd.teams.pipeline import render_team
+    import yaml as _yaml
+
+    src = tmp_path / "agents"
+    (src / "partials").mkdir(parents=True)
+    (src / "partials" / "cellspec.tmpl").write_text(
+        "containers:\n"
+        "  - id: main\n"
+        "    image: ${IMAGE}\n"
+        "    command: sleep\n"
+        "    args: [\"60\"]\n"
+        "{{#if GPU}}    gpus: ${GPU}\n{{/if}}")
+    (src / "agent.tmpl").write_text(
+        "apiVersion: v1beta1\n"
+        "kind: CellBlueprint\n"
+        "metadata: {name: ${TEAM}-${ROLE}-${HARNESS}}\n"
+        "spec:\n"
+        "  namePrefix: ${ROLE}\n"
+        "  template:\n"
+        "    spec:\n"
+        "      stackId: default\n"
+        "      {{> cellspec}}\n")
+    team = parse_team_doc(_yaml.safe_load("""
+apiVersion: kuketeams.io/v1
+kind: ProjectTeam
+metadata: {name: squad}
+spec:
+  source: {path: .}
+  defaults:
+    harnesses: [cc]
+    needs: {image: [git]}
+  vars: {GPU: "1"}
+  roles:
+    - ref: builder
+      needs: {image: [gpu]}
+    - ref: scout
+      vars: {GPU: ""}
+"""))
+    roles = {
+        "builder": parse_team_doc(_yaml.safe_load(
+            "apiVersion: kuketeams.io/v1\nkind: Role\n"
+            "metadata: {name: builder}\nspec: {prompt: build}")),
+        "scout": parse_team_doc(_yaml.safe_load(
+            "apiVersion: kuketeams.io/v1\nkind: Role\n"
+            "metadata: {name: scout}\nspec: {prompt: scout}")),
+    }
+    harness = parse_team_doc(_yaml.safe_load(
+        "apiVersion: kuketeams.io/v1\nkind: Harness\n"
+        "metadata: {name: cc}\nspec: {template: agent.tmpl}"))
+    catalog = parse_team_doc(_yaml.safe_load("""
+apiVersion: kuketeams.io/v1
+kind: ImageCatalog
+metadata: {name: cat}
+spec:
+  images:
+    - ref: base
+      harness: cc
+      capabilities: [git]
+      image: reg/base
+    - ref: gpu
+      harness: cc
+      capabilities: [git, gpu]
+      image: reg/gpu
+"""))
+    docs = render_team(team, roles, {"cc": harness}, catalog, src,
+                       tmp_path, "default", "default")
+    by_name = {d["metadata"]["name"]: d for d in docs}
+    b = by_name["squad-builder-cc"]["spec"]["template"]["spec"]
+    s = by_name["squad-scout-cc"]["spec"]["template"]["spec"]
+    # merged needs [git, gpu] -> the gpu-capable image for builder
+    assert b["containers"][0]["image"] == "reg/gpu"
+    assert s["containers"][0]["image"] == "reg/base"
+    # conditional: builder (GPU=1 team var) gets gpus; scout overrode
+    assert b["containers"][0].get("gpus") == 1
+    assert "gpus" not in s["containers"][0]
