@@ -191,6 +191,30 @@ class LLMEngine:
 
     # ------------------------------------------------------------------
     def step(self) -> List[StepOutput]:
+        import kukeon_amd.parallel as parallel
+        if parallel.ep_size() > 1:
+            return self._step_ep_lockstep()
+        return self._step_local()
+
+    def _step_ep_lockstep(self) -> List[StepOutput]:
+        """EP serving: every MoE forward is a collective, so all EP
+        ranks must make the SAME number of model passes. Per lockstep
+        round each rank contributes exactly ONE pass — a real step
+        (decode micro-batch forced to 1) or a participation pass that
+        only serves this rank's experts. The caller (modelhub engine
+        loop) exchanges the has-work flag so idle ranks keep
+        participating while any rank still serves."""
+        from kukeon_amd.models.mixtral import MixtralModel
+        self.ecfg.decode_microbatch = 1  # one pass per step, always
+        before = MixtralModel.pass_count
+        outs = self._step_local() if self.has_work() else []
+        made = MixtralModel.pass_count - before
+        assert made <= 1, f"EP lockstep violated: {made} passes in a step"
+        if made == 0:
+            self.model.participate()
+        return outs
+
+    def _step_local(self) -> List[StepOutput]:
         if self.waiting:
             batch, blocked = self._admit_prefill()
             # several small idle sessions may need to go before the head

@@ -243,7 +243,14 @@ class MixtralModel(nn.Module):
         # reuse the Llama RoPE table builder
         self.cos_sin = LlamaModel._build_rope_table(self).to(self.device_)
 
+    # EP lockstep accounting: every full forward touches each MoE
+    # layer's collectives exactly once, so EP ranks stay paired as long
+    # as their PASS COUNTS match — a work-free rank contributes a
+    # participation pass (collectives only) instead of a real forward.
+    pass_count = 0
+
     def forward(self, input_ids, kv_k, kv_v, meta: AttnMeta) -> torch.Tensor:
+        MixtralModel.pass_count += 1
         x = F.embedding(input_ids.long(), self.embed)
         residual = None
         for i, layer in enumerate(self.layers):
@@ -251,6 +258,18 @@ class MixtralModel(nn.Module):
                                         kv_v[i], meta)
         ops.fused_add_rmsnorm(x, residual, self.final_norm, self.cfg.rms_eps)
         return x
+
+    def participate(self) -> None:
+        """One collective-only pass: serve this rank's experts for the
+        other EP ranks' tokens without any local batch (VERDICT r01
+        item 9 — the cross-rank lockstep EP serving scheduler). Runs the
+        same all-to-all sequence as a real forward, with zero local
+        tokens."""
+        MixtralModel.pass_count += 1
+        empty = torch.empty(0, self.cfg.hidden_size, dtype=torch.bfloat16,
+                            device=self.device_)
+        for layer in self.layers:
+            layer.moe.forward(empty)
 
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
         return F.linear(hidden, self.lm_head)

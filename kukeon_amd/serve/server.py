@@ -63,19 +63,53 @@ class ModelhubServer:
     # ---- engine loop --------------------------------------------------
     def _engine_loop(self):
         self.engine.capture_all()
-        while not self._stop.is_set():
-            # drain submissions
+        from kukeon_amd import parallel
+        ep = parallel.ep_size() > 1
+        stopping = False
+        while True:
+            if self._stop.is_set():
+                stopping = True
+            if stopping and not ep:
+                return
+            # drain submissions (EP never blocks: the lockstep flag
+            # exchange below must run every iteration so an idle rank
+            # keeps serving its experts for the others)
             try:
                 while True:
                     fn = self._submit.get(
-                        timeout=None if not self.engine.has_work() and
-                        self._submit.empty() else 0.0)
+                        timeout=(0.02 if ep else
+                                 None if not self.engine.has_work() and
+                                 self._submit.empty() else 0.0))
                     if fn is None:
-                        return
+                        stopping = True
+                        if not ep:
+                            return
+                        break
                     fn()
             except queue.Empty:
                 pass
-            if not self.engine.has_work():
+            if ep:
+                # cross-rank lockstep admission + consensus shutdown:
+                # step while ANY rank has work (idle and even STOPPING
+                # ranks keep contributing participation passes so the
+                # others' requests complete); exit only when every rank
+                # voted stop — an EP modelhub drains global work first.
+                import torch as _torch
+                import torch.distributed as _dist
+                has = 1 if (not stopping and self.engine.has_work()) else 0
+                flag = _torch.tensor(
+                    [has, 1 if stopping else 0],
+                    device=self.engine.device if self.engine.is_cuda
+                    else "cpu")
+                _dist.all_reduce(flag, op=_dist.ReduceOp.SUM)
+                if int(flag[1].item()) == _dist.get_world_size():
+                    return
+                if int(flag[0].item()) == 0:
+                    continue
+                if stopping or not self.engine.has_work():
+                    self.engine.model.participate()
+                    continue
+            elif not self.engine.has_work():
                 continue
             import time as _time
             t0 = _time.perf_counter()

@@ -160,3 +160,53 @@ def test_expert_parallel_moe_all_to_all(tmp_path):
     port = 29600 + (os.getpid() + 13) % 500
     mp.spawn(_ep_moe_worker, args=(2, port, str(tmp_path)), nprocs=2,
              join=True)
+
+
+def _ep_serving_worker(rank, world, port, result_dir):
+    """EP SERVING: two modelhub servers, one per EP
+    rank, under ASYMMETRIC load — rank 1 goes idle while rank 0 still
+    serves; the lockstep scheduler keeps rank 1 participating (serving
+    its experts) so rank 0's requests complete."""
+    import uuid
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from kukeon_amd import parallel
+    parallel.init_expert_parallel(world)
+    from kukeon_amd.engine.config import EngineConfig, tiny_mixtral
+    from kukeon_amd.models.mixtral import MixtralModel
+    from kukeon_amd.serve.server import ModelhubClient, ModelhubServer
+
+    cfg = tiny_mixtral()
+    ecfg = EngineConfig(max_model_len=128, max_sessions=4,
+                        num_kv_blocks=64, use_graphs=False)
+    model = MixtralModel(cfg, device="cpu")
+    sock = f"/tmp/mh-ep-{rank}-{uuid.uuid4().hex[:8]}.sock"
+    hub = ModelhubServer(model, cfg, ecfg, sock, device="cpu")
+    hub.start()
+    try:
+        c = ModelhubClient(sock, timeout=180)
+        if rank == 1:
+            # one short request, then idle while rank 0 keeps serving
+            r = c.generate("s1", [5, 6, 7], max_new_tokens=2,
+                           temperature=0.0)
+            assert len(r["tokens"]) == 2
+        else:
+            for turn in range(3):
+                r = c.generate("s0", [1 + turn, 2, 3], max_new_tokens=4,
+                               temperature=0.0)
+                assert len(r["tokens"]) == 4
+        c.close()
+    finally:
+        # no main-thread barrier here: it would collide with the engine
+        # thread's collectives. hub.stop() votes stop through the
+        # lockstep flag; the engine loop exits only when EVERY rank has
+        # voted, draining the other rank's in-flight requests first.
+        hub.stop()
+    dist.destroy_process_group()
+
+
+def test_expert_parallel_serving_lockstep(tmp_path):
+    port = 29700 + (os.getpid() + 29) % 500
+    mp.spawn(_ep_serving_worker, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
