@@ -546,103 +546,144 @@ __global__ __launch_bounds__(256) void skinny4_kernel(
 // 8 contiguous nt loads for one tile's 16-row x 512-B slice: instruction
 // (rb, u) covers rows rb..rb+7 at bytes [u*128, u*128+128). p_lo/p_hi are
 // per-lane base pointers for rows (l>>3) and (8 + l>>3).
-DEV_INLINE void issue_wc8(u32x4_t (&reg)[8], const unsigned short* p_lo,
-                          const unsigned short* p_hi) {
+template <int NI>
+DEV_INLINE void issue_wc(u32x4_t (&reg)[NI], const unsigned short* p_lo,
+                         const unsigned short* p_hi) {
 #define SK5_LD(i, P, off)                                                 \
   asm volatile("global_load_dwordx4 %0, %1, off offset:" #off " nt"      \
                : "=&v"(reg[i]) : "v"(P) : "memory")
-  SK5_LD(0, p_lo, 0); SK5_LD(1, p_lo, 128);
-  SK5_LD(2, p_lo, 256); SK5_LD(3, p_lo, 384);
-  SK5_LD(4, p_hi, 0); SK5_LD(5, p_hi, 128);
-  SK5_LD(6, p_hi, 256); SK5_LD(7, p_hi, 384);
+  if constexpr (NI == 8) {
+    SK5_LD(0, p_lo, 0); SK5_LD(1, p_lo, 128);
+    SK5_LD(2, p_lo, 256); SK5_LD(3, p_lo, 384);
+    SK5_LD(4, p_hi, 0); SK5_LD(5, p_hi, 128);
+    SK5_LD(6, p_hi, 256); SK5_LD(7, p_hi, 384);
+  } else {
+    SK5_LD(0, p_lo, 0); SK5_LD(1, p_lo, 128);
+    SK5_LD(2, p_hi, 0); SK5_LD(3, p_hi, 128);
+  }
 #undef SK5_LD
 }
 
 // same shape for the x operand but WITHOUT nt: x is L2-resident and
 // re-read by every block — evict-first policy on it forces the whole x
 // slice back to HBM for every block (+~50% HBM traffic on gate_up)
-DEV_INLINE void issue_xc8(u32x4_t (&reg)[8], const unsigned short* p_lo,
-                          const unsigned short* p_hi) {
+template <int NI>
+DEV_INLINE void issue_xc(u32x4_t (&reg)[NI], const unsigned short* p_lo,
+                         const unsigned short* p_hi) {
 #define SK5_LDX(i, P, off)                                                \
   asm volatile("global_load_dwordx4 %0, %1, off offset:" #off            \
                : "=&v"(reg[i]) : "v"(P) : "memory")
-  SK5_LDX(0, p_lo, 0); SK5_LDX(1, p_lo, 128);
-  SK5_LDX(2, p_lo, 256); SK5_LDX(3, p_lo, 384);
-  SK5_LDX(4, p_hi, 0); SK5_LDX(5, p_hi, 128);
-  SK5_LDX(6, p_hi, 256); SK5_LDX(7, p_hi, 384);
+  if constexpr (NI == 8) {
+    SK5_LDX(0, p_lo, 0); SK5_LDX(1, p_lo, 128);
+    SK5_LDX(2, p_lo, 256); SK5_LDX(3, p_lo, 384);
+    SK5_LDX(4, p_hi, 0); SK5_LDX(5, p_hi, 128);
+    SK5_LDX(6, p_hi, 256); SK5_LDX(7, p_hi, 384);
+  } else {
+    SK5_LDX(0, p_lo, 0); SK5_LDX(1, p_lo, 128);
+    SK5_LDX(2, p_hi, 0); SK5_LDX(3, p_hi, 128);
+  }
 #undef SK5_LDX
 }
 
-// W-image byte offset inside one 8 KiB tile slot: row-major [16][512 B]
+// W-image byte offset inside one tile slot: row-major [16][KS*2 B]
 // with the 16-B column slot XOR-swizzled by row (same scheme as xswz).
-DEV_INLINE int wimg_off(int row, int byte_in_row) {
-  return row * 512 + (byte_in_row ^ ((row & 15) << 4));
+// counted wait that NAMES every register of the set: consumers (the
+// ds_writes) are data-ordered after it; the count allows the two
+// younger same-class load groups (2*NI) to stay in flight
+template <int NI>
+DEV_INLINE void wait_set(u32x4_t (&r)[NI]) {
+  if constexpr (NI == 8) {
+    asm volatile("s_waitcnt vmcnt(16)"
+                 : "+v"(r[0]), "+v"(r[1]), "+v"(r[2]), "+v"(r[3]),
+                   "+v"(r[4]), "+v"(r[5]), "+v"(r[6]), "+v"(r[7]));
+  } else {
+    asm volatile("s_waitcnt vmcnt(8)"
+                 : "+v"(r[0]), "+v"(r[1]), "+v"(r[2]), "+v"(r[3]));
+  }
 }
 
-template <int MT>
+template <int KS>
+DEV_INLINE int wimg_off(int row, int byte_in_row) {
+  return row * (KS * 2) + (byte_in_row ^ ((row & 15) << 4));
+}
+
+template <int KS>
+DEV_INLINE int xswz_ks(int row, int byte_in_row) {
+  return row * (KS * 2) + (byte_in_row ^ ((row & 15) << 4));
+}
+
+template <int MT, int KS>
 DEV_INLINE void consume_img(const unsigned short* wimg,
                             const unsigned short* xb,
                             f32x4_t (&acc)[MT], int row16, int kgrp) {
 #pragma unroll
-  for (int u = 0; u < 8; ++u) {
+  for (int u = 0; u < KS / 32; ++u) {
     const int kc = u * 32 + 8 * kgrp;
     const uint4 av = *reinterpret_cast<const uint4*>(
-        reinterpret_cast<const char*>(wimg) + wimg_off(row16, kc * 2));
+        reinterpret_cast<const char*>(wimg) + wimg_off<KS>(row16, kc * 2));
     const bf16x8_t af = frag_of(av);
 #pragma unroll
     for (int m = 0; m < MT; ++m) {
       const int xr = m * 16 + row16;
       const uint4 xv = *reinterpret_cast<const uint4*>(
-          reinterpret_cast<const char*>(xb) + xswz(xr, kc * 2));
+          reinterpret_cast<const char*>(xb) + xswz_ks<KS>(xr, kc * 2));
       acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
           af, frag_of(xv), acc[m], 0, 0, 0);
     }
   }
 }
 
-// write one tile's 16x512B register set into its LDS image slot
-DEV_INLINE void wimg_write(unsigned short* wimg, u32x4_t (&reg)[8],
+// write one tile's 16-row register set into its LDS image slot
+template <int KS, int NI = KS / 32>
+DEV_INLINE void wimg_write(unsigned short* wimg, u32x4_t (&reg)[NI],
                            int lane) {
   const int seg16 = (lane & 7) * 16;  // byte offset of this lane's 16 B
+  constexpr int UPH = NI / 2;         // 128-B steps per 8-row half
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    const int row = (i & 3) * 0 + (i < 4 ? (lane >> 3) : 8 + (lane >> 3));
-    const int u = i & 3;
+  for (int i = 0; i < NI; ++i) {
+    const int row = (i < UPH ? (lane >> 3) : 8 + (lane >> 3));
+    const int u = (i < UPH) ? i : i - UPH;
     *reinterpret_cast<uint4*>(
         reinterpret_cast<char*>(wimg) +
-        wimg_off(row, u * 128 + seg16)) = __builtin_bit_cast(uint4, reg[i]);
+        wimg_off<KS>(row, u * 128 + seg16)) =
+        __builtin_bit_cast(uint4, reg[i]);
   }
 }
 
 // write one wave's 16 x rows (contiguous-loaded) into the SHARED xswz
 // image; published to the other waves by the slice barrier
-DEV_INLINE void ximg_write(unsigned short* xb, u32x4_t (&reg)[8], int wid,
+template <int KS, int NI = KS / 32>
+DEV_INLINE void ximg_write(unsigned short* xb, u32x4_t (&reg)[NI], int wid,
                            int lane) {
   const int seg16 = (lane & 7) * 16;
+  constexpr int UPH = NI / 2;
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    const int row = wid * 16 + (i < 4 ? (lane >> 3) : 8 + (lane >> 3));
-    const int u = i & 3;
+  for (int i = 0; i < NI; ++i) {
+    const int row = wid * 16 + (i < UPH ? (lane >> 3) : 8 + (lane >> 3));
+    const int u = (i < UPH) ? i : i - UPH;
     *reinterpret_cast<uint4*>(
         reinterpret_cast<char*>(xb) +
-        xswz(row, u * 128 + seg16)) = __builtin_bit_cast(uint4, reg[i]);
+        xswz_ks<KS>(row, u * 128 + seg16)) =
+        __builtin_bit_cast(uint4, reg[i]);
   }
 }
 
-template <int MT, bool SPLIT>
+template <int MT, bool SPLIT, int KS>
 __global__ __launch_bounds__(256) void skinny5_kernel(
     unsigned short* __restrict__ out, float* __restrict__ ws,
     const unsigned short* __restrict__ x,
     const unsigned short* __restrict__ w, int M, int N, long K) {
-  // x: 2 x 32 KiB shared double buffer (cooperatively written, barrier-
-  // published); W: per-wave 2-tile image (16 KiB x 4 waves). 128 KiB,
-  // 1 block/CU. NO LDS-DMA anywhere: glds completions do not stay
-  // in-order with plain loads on the vmcnt counter (mixed-class counted
-  // waits released early -> the round-2 wild-write fault), so both
-  // operands stage through registers + ds_write and every wait is a
-  // constant in-order vmcnt(16).
-  __shared__ __align__(16) unsigned short xbuf[2][64 * KSLICE];
-  __shared__ __align__(16) unsigned short wimg_all[4][2][16 * 256];
+  // x: 2 shared slice buffers (cooperatively written, barrier-
+  // published); W: per-wave 2-tile image. KS=256: 128 KiB total ->
+  // 1 block/CU; KS=128: 64 KiB -> 2 blocks/CU (2 waves/SIMD hide the
+  // consume's MFMA/LDS dependency stalls). NO LDS-DMA anywhere: glds
+  // completions do not stay in-order with plain loads on the vmcnt
+  // counter (mixed-class counted waits released early -> the round-2
+  // wild-write fault), so both operands stage through registers +
+  // ds_write and every wait is a constant in-order vmcnt.
+  constexpr int NI = KS / 32;            // 16B-loads per tile slice
+  __shared__ __align__(16) unsigned short xbuf[2][64 * KS];
+  __shared__ __align__(16) unsigned short wimg_all[4][2][16 * (KS)];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int row16 = lane & 15;
@@ -653,10 +694,10 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
     acc0[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
     acc1[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
   }
-  const long kadv = (long)gridDim.y * KSLICE;
-  const long ks0 = (long)blockIdx.y * KSLICE;
+  const long kadv = (long)gridDim.y * KS;
+  const long ks0 = (long)blockIdx.y * KS;
   if (ks0 >= K) return;
-  u32x4_t w0[8], w1[8], xr8[8];
+  u32x4_t w0[NI], w1[NI], xr8[NI];
   const int n0t0 = (blockIdx.x * 2 + 0) * 64 + wid * 16;
   const int n0t1 = (blockIdx.x * 2 + 1) * 64 + wid * 16;
   const int rlo = lane >> 3, seg = (lane & 7) * 8;  // elems
@@ -664,8 +705,6 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
   const unsigned short* p0h = w + (long)(n0t0 + 8 + rlo) * K + ks0 + seg;
   const unsigned short* p1l = w + (long)(n0t1 + rlo) * K + ks0 + seg;
   const unsigned short* p1h = w + (long)(n0t1 + 8 + rlo) * K + ks0 + seg;
-  // this wave's 16 x rows (clamped for M < 64: junk rows are staged but
-  // their accumulators are discarded by the epilogue's mrow guard)
   const int xrow_l = wid * 16 + rlo;
   const int xrow_h = wid * 16 + 8 + rlo;
   const unsigned short* pxl = x + (long)min(M - 1, xrow_l) * K + ks0 + seg;
@@ -674,54 +713,47 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
   unsigned short* img1 = wimg_all[wid][1];
 
   // prologue: land slice 0, build the images, put slice 1 in flight
-  issue_wc8(w0, p0l, p0h);
-  issue_wc8(w1, p1l, p1h);
-  issue_xc8(xr8, pxl, pxh);
+  issue_wc<NI>(w0, p0l, p0h);
+  issue_wc<NI>(w1, p1l, p1h);
+  issue_xc<NI>(xr8, pxl, pxh);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  wimg_write(img0, w0, lane);
-  wimg_write(img1, w1, lane);
-  ximg_write(xbuf[0], xr8, wid, lane);
+  wimg_write<KS>(img0, w0, lane);
+  wimg_write<KS>(img1, w1, lane);
+  ximg_write<KS>(xbuf[0], xr8, wid, lane);
   {
     const long ks1 = ks0 + kadv;
     if (ks1 < K) {
       p0l += kadv; p0h += kadv; p1l += kadv; p1h += kadv;
       pxl += kadv; pxh += kadv;
     }
-    issue_wc8(w0, p0l, p0h);
-    issue_wc8(w1, p1l, p1h);
-    issue_xc8(xr8, pxl, pxh);
+    issue_wc<NI>(w0, p0l, p0h);
+    issue_wc<NI>(w1, p1l, p1h);
+    issue_xc<NI>(xr8, pxl, pxh);
   }
   int cur = 0;
   for (long ks = ks0; ks < K; ks += kadv, cur ^= 1) {
-    // entry: t0L(s+1), t1L(s+1), xL(s+1) in flight (24 plain loads, in
-    // that order); images(s) + xbuf[cur] ready
+    // entry: t0L(s+1), t1L(s+1), xL(s+1) in flight (3*NI plain loads,
+    // in that order); images(s) + xbuf[cur] ready
     __syncthreads();
     const unsigned short* xb = xbuf[cur];
     const long ksn = ks + kadv;
     const bool adv = (ksn + kadv) < K;
-    consume_img<MT>(img0, xb, acc0, row16, kgrp);   // LDS-only
-    // t0L(s+1) landed: younger = t1L(8) + xL(8); all one class,
-    // strictly in-order
-    asm volatile("s_waitcnt vmcnt(16)"
-                 : "+v"(w0[0]), "+v"(w0[1]), "+v"(w0[2]), "+v"(w0[3]),
-                   "+v"(w0[4]), "+v"(w0[5]), "+v"(w0[6]), "+v"(w0[7]));
-    wimg_write(img0, w0, lane);       // after img0 frag reads (in-order)
+    consume_img<MT, KS>(img0, xb, acc0, row16, kgrp);   // LDS-only
+    // t0L(s+1) landed: younger = t1L(NI) + xL(NI); one class, in-order
+    wait_set<NI>(w0);
+    wimg_write<KS>(img0, w0, lane);     // after img0 frag reads (in-order)
     if (adv) { p0l += kadv; p0h += kadv; }
-    issue_wc8(w0, p0l, p0h);
-    consume_img<MT>(img1, xb, acc1, row16, kgrp);
-    asm volatile("s_waitcnt vmcnt(16)"
-                 : "+v"(w1[0]), "+v"(w1[1]), "+v"(w1[2]), "+v"(w1[3]),
-                   "+v"(w1[4]), "+v"(w1[5]), "+v"(w1[6]), "+v"(w1[7]));
-    wimg_write(img1, w1, lane);
+    issue_wc<NI>(w0, p0l, p0h);
+    consume_img<MT, KS>(img1, xb, acc1, row16, kgrp);
+    wait_set<NI>(w1);
+    wimg_write<KS>(img1, w1, lane);
     if (adv) { p1l += kadv; p1h += kadv; }
-    issue_wc8(w1, p1l, p1h);
+    issue_wc<NI>(w1, p1l, p1h);
     // xL(s+1) landed: younger = t0L(s+2) + t1L(s+2)
-    asm volatile("s_waitcnt vmcnt(16)"
-                 : "+v"(xr8[0]), "+v"(xr8[1]), "+v"(xr8[2]), "+v"(xr8[3]),
-                   "+v"(xr8[4]), "+v"(xr8[5]), "+v"(xr8[6]), "+v"(xr8[7]));
-    ximg_write(xbuf[cur ^ 1], xr8, wid, lane);
+    wait_set<NI>(xr8);
+    ximg_write<KS>(xbuf[cur ^ 1], xr8, wid, lane);
     if (adv) { pxl += kadv; pxh += kadv; }
-    issue_xc8(xr8, pxl, pxh);
+    issue_xc<NI>(xr8, pxl, pxh);
     // exit: t0L(s+2), t1L(s+2), xL(s+2) in flight — invariant restored
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -753,14 +785,21 @@ void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   const int M = x.size(0);
   const long K = x.size(1);
   const int N = w.size(0);
-  TORCH_CHECK(M >= 1 && M <= 64 && N % 128 == 0 && K % KSLICE == 0);
+  int KS = 128;  // 64 KiB LDS -> 2 blocks/CU (2 waves/SIMD)
+  if (const char* ov = getenv("KUKEON_SK5_KS")) {
+    KS = atoi(ov) == 256 ? 256 : 128;
+  }
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 128 == 0 && K % KS == 0);
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int ngroups = N / 128;
-  const int nslices = (int)(K / KSLICE);
+  const int nslices = (int)(K / KS);
+  // 2 blocks/CU at KS=128: target ~512 blocks before settling for less
+  const int target = (KS == 128) ? 512 : 256;
   int splitk = 1;
-  if (ngroups < 256) splitk = min(nslices, (256 + ngroups - 1) / ngroups);
+  if (ngroups < target)
+    splitk = min(nslices, (target + ngroups - 1) / ngroups);
   if (const char* ov = getenv("KUKEON_SK5_SPLITK")) {
     const int v = atoi(ov);
     if (v > 0) splitk = min(nslices, v);
@@ -771,18 +810,20 @@ void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   auto* xp = reinterpret_cast<const unsigned short*>(x.data_ptr());
   auto* wp = reinterpret_cast<const unsigned short*>(w.data_ptr());
   const long total = (long)M * N;
-#define SK5_LAUNCH(MT_)                                                      \
+#define SK5_LAUNCH_KS(MT_, KS_)                                              \
   if (splitk == 1) {                                                         \
-    skinny5_kernel<MT_, false><<<grid, 256, 0, stream>>>(                    \
+    skinny5_kernel<MT_, false, KS_><<<grid, 256, 0, stream>>>(               \
         op, nullptr, xp, wp, M, N, K);                                       \
   } else {                                                                   \
     float* wsp = ws.data_ptr<float>();                                       \
     TORCH_CHECK(ws.numel() >= total * splitk, "sk5 workspace too small");    \
-    skinny5_kernel<MT_, true><<<grid, 256, 0, stream>>>(                     \
+    skinny5_kernel<MT_, true, KS_><<<grid, 256, 0, stream>>>(                \
         nullptr, wsp, xp, wp, M, N, K);                                      \
     skinny_reduce_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256,   \
                            0, stream>>>(op, wsp, total, splitk);             \
   }
+#define SK5_LAUNCH(MT_)                                                      \
+  if (KS == 128) { SK5_LAUNCH_KS(MT_, 128) } else { SK5_LAUNCH_KS(MT_, 256) }
   switch (MT) {
     case 1: SK5_LAUNCH(1); break;
     case 2: SK5_LAUNCH(2); break;
@@ -790,6 +831,7 @@ void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
     default: SK5_LAUNCH(4); break;
   }
 #undef SK5_LAUNCH
+#undef SK5_LAUNCH_KS
   HIP_CHECK_KERNEL();
 }
 

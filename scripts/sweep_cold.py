@@ -62,14 +62,20 @@ for (M, N, K, tag) in SHAPES:
         us = t_rot(lambda i: _C.skinny_gemm2(out, x, ws_list[i], wrk), nw)
         line += f"  v2k{sk or 'A'}={us:6.1f}"
     os.environ.pop("KUKEON_SK2_SPLITK", None)
-    for sk in (0, 2, 4, 5, 8, 16):
-        if sk:
-            os.environ["KUKEON_SK5_SPLITK"] = str(sk)
-        else:
-            os.environ.pop("KUKEON_SK5_SPLITK", None)
-        us = t_rot(lambda i: _C.skinny_gemm5(out, x, ws_list[i], wrk), nw)
-        line += f"  v5k{sk or 'A'}={us:6.1f}"
-    os.environ.pop("KUKEON_SK5_SPLITK", None)
+    for ks in (128, 256):
+        os.environ["KUKEON_SK5_KS"] = str(ks)
+        for sk in (0, 2, 4, 8, 16, 32):
+            if sk:
+                os.environ["KUKEON_SK5_SPLITK"] = str(sk)
+            else:
+                os.environ.pop("KUKEON_SK5_SPLITK", None)
+            if sk and sk > (K // ks):
+                continue
+            us = t_rot(lambda i: _C.skinny_gemm5(out, x, ws_list[i], wrk),
+                       nw)
+            line += f"  v5-{ks}k{sk or 'A'}={us:6.1f}"
+        os.environ.pop("KUKEON_SK5_SPLITK", None)
+    os.environ.pop("KUKEON_SK5_KS", None)
     if N % 256 == 0:
         for sk in (0, 2, 4, 8, 16):
             if sk:
