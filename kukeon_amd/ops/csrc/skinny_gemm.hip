@@ -587,21 +587,6 @@ DEV_INLINE void issue_xc(u32x4_t (&reg)[NI], const unsigned short* p_lo,
 
 // W-image byte offset inside one tile slot: row-major [16][KS*2 B]
 // with the 16-B column slot XOR-swizzled by row (same scheme as xswz).
-// counted wait that NAMES every register of the set: consumers (the
-// ds_writes) are data-ordered after it; the count allows the two
-// younger same-class load groups (2*NI) to stay in flight
-template <int NI>
-DEV_INLINE void wait_set(u32x4_t (&r)[NI]) {
-  if constexpr (NI == 8) {
-    asm volatile("s_waitcnt vmcnt(16)"
-                 : "+v"(r[0]), "+v"(r[1]), "+v"(r[2]), "+v"(r[3]),
-                   "+v"(r[4]), "+v"(r[5]), "+v"(r[6]), "+v"(r[7]));
-  } else {
-    asm volatile("s_waitcnt vmcnt(8)"
-                 : "+v"(r[0]), "+v"(r[1]), "+v"(r[2]), "+v"(r[3]));
-  }
-}
-
 template <int KS>
 DEV_INLINE int wimg_off(int row, int byte_in_row) {
   return row * (KS * 2) + (byte_in_row ^ ((row & 15) << 4));
@@ -610,6 +595,31 @@ DEV_INLINE int wimg_off(int row, int byte_in_row) {
 template <int KS>
 DEV_INLINE int xswz_ks(int row, int byte_in_row) {
   return row * (KS * 2) + (byte_in_row ^ ((row & 15) << 4));
+}
+
+// contiguous full-line loads for one tile slice: instruction (half, u)
+// covers rows half*8+(lane>>3) at bytes [u*128, +128) — 8 whole 128-B
+// lines per instruction (the request-granularity lever measured in
+// scripts/pattern_probe3.hip). All loads are hipcc-VISIBLE (values in
+// locals): with no LDS-DMA in this kernel the compiler's own counted
+// vmcnt bookkeeping is exact, and register lifetimes are its problem —
+// the invisible-asm variants corrupted under MT=4 register pressure
+// (ledger: compiler reuse of in-flight asm destinations).
+template <int NI, bool NT>
+DEV_INLINE void load_tile(u32x4_t (&reg)[NI], const unsigned short* p_lo,
+                          const unsigned short* p_hi) {
+  constexpr int UPH = NI / 2;
+#pragma unroll
+  for (int i = 0; i < NI; ++i) {
+    const unsigned short* p = (i < UPH ? p_lo : p_hi) +
+                              (i < UPH ? i : i - UPH) * 64;
+    if constexpr (NT) {
+      reg[i] = __builtin_nontemporal_load(
+          reinterpret_cast<const u32x4_t*>(p));
+    } else {
+      reg[i] = *reinterpret_cast<const u32x4_t*>(p);
+    }
+  }
 }
 
 template <int MT, int KS>
@@ -637,8 +647,8 @@ DEV_INLINE void consume_img(const unsigned short* wimg,
 template <int KS, int NI = KS / 32>
 DEV_INLINE void wimg_write(unsigned short* wimg, u32x4_t (&reg)[NI],
                            int lane) {
-  const int seg16 = (lane & 7) * 16;  // byte offset of this lane's 16 B
-  constexpr int UPH = NI / 2;         // 128-B steps per 8-row half
+  const int seg16 = (lane & 7) * 16;
+  constexpr int UPH = NI / 2;
 #pragma unroll
   for (int i = 0; i < NI; ++i) {
     const int row = (i < UPH ? (lane >> 3) : 8 + (lane >> 3));
@@ -650,8 +660,8 @@ DEV_INLINE void wimg_write(unsigned short* wimg, u32x4_t (&reg)[NI],
   }
 }
 
-// write one wave's 16 x rows (contiguous-loaded) into the SHARED xswz
-// image; published to the other waves by the slice barrier
+// write one wave's 16 x rows into the SHARED xswz image; published to
+// the other waves by the slice barrier
 template <int KS, int NI = KS / 32>
 DEV_INLINE void ximg_write(unsigned short* xb, u32x4_t (&reg)[NI], int wid,
                            int lane) {
@@ -673,17 +683,15 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
     unsigned short* __restrict__ out, float* __restrict__ ws,
     const unsigned short* __restrict__ x,
     const unsigned short* __restrict__ w, int M, int N, long K) {
-  // x: 2 shared slice buffers (cooperatively written, barrier-
-  // published); W: per-wave 2-tile image. KS=256: 128 KiB total ->
-  // 1 block/CU; KS=128: 64 KiB -> 2 blocks/CU (2 waves/SIMD hide the
-  // consume's MFMA/LDS dependency stalls). NO LDS-DMA anywhere: glds
-  // completions do not stay in-order with plain loads on the vmcnt
-  // counter (mixed-class counted waits released early -> the round-2
-  // wild-write fault), so both operands stage through registers +
-  // ds_write and every wait is a constant in-order vmcnt.
-  constexpr int NI = KS / 32;            // 16B-loads per tile slice
+  // Never-drain full-line pipeline, pure HIP: slice s+1's W/x loads are
+  // issued BEFORE the slice-s barrier (a full memory fence pins them
+  // there) and ds_written to the images at the end of slice s, so ~3*NI
+  // loads are always in flight and no wait ever sees the laden-queue
+  // round trip. KS=256: 128 KiB LDS -> 1 block/CU; KS=128: 64 KiB ->
+  // 2 blocks/CU (2 waves/SIMD hide the consume's dependency stalls).
+  constexpr int NI = KS / 32;
   __shared__ __align__(16) unsigned short xbuf[2][64 * KS];
-  __shared__ __align__(16) unsigned short wimg_all[4][2][16 * (KS)];
+  __shared__ __align__(16) unsigned short wimg_all[4][2][16 * KS];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int row16 = lane & 15;
@@ -713,10 +721,9 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
   unsigned short* img1 = wimg_all[wid][1];
 
   // prologue: land slice 0, build the images, put slice 1 in flight
-  issue_wc<NI>(w0, p0l, p0h);
-  issue_wc<NI>(w1, p1l, p1h);
-  issue_xc<NI>(xr8, pxl, pxh);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  load_tile<NI, true>(w0, p0l, p0h);
+  load_tile<NI, true>(w1, p1l, p1h);
+  load_tile<NI, false>(xr8, pxl, pxh);
   wimg_write<KS>(img0, w0, lane);
   wimg_write<KS>(img1, w1, lane);
   ximg_write<KS>(xbuf[0], xr8, wid, lane);
@@ -726,37 +733,31 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
       p0l += kadv; p0h += kadv; p1l += kadv; p1h += kadv;
       pxl += kadv; pxh += kadv;
     }
-    issue_wc<NI>(w0, p0l, p0h);
-    issue_wc<NI>(w1, p1l, p1h);
-    issue_xc<NI>(xr8, pxl, pxh);
+    load_tile<NI, true>(w0, p0l, p0h);
+    load_tile<NI, true>(w1, p1l, p1h);
+    load_tile<NI, false>(xr8, pxl, pxh);
   }
   int cur = 0;
   for (long ks = ks0; ks < K; ks += kadv, cur ^= 1) {
-    // entry: t0L(s+1), t1L(s+1), xL(s+1) in flight (3*NI plain loads,
-    // in that order); images(s) + xbuf[cur] ready
+    // entry: slice s+1 loads in flight (w0/w1/xr8 values pending);
+    // images(s) + xbuf[cur] ready
     __syncthreads();
     const unsigned short* xb = xbuf[cur];
     const long ksn = ks + kadv;
     const bool adv = (ksn + kadv) < K;
     consume_img<MT, KS>(img0, xb, acc0, row16, kgrp);   // LDS-only
-    // t0L(s+1) landed: younger = t1L(NI) + xL(NI); one class, in-order
-    wait_set<NI>(w0);
-    wimg_write<KS>(img0, w0, lane);     // after img0 frag reads (in-order)
+    wimg_write<KS>(img0, w0, lane);   // first USE of w0 -> counted wait
     if (adv) { p0l += kadv; p0h += kadv; }
-    issue_wc<NI>(w0, p0l, p0h);
+    load_tile<NI, true>(w0, p0l, p0h);                  // slice s+2
     consume_img<MT, KS>(img1, xb, acc1, row16, kgrp);
-    wait_set<NI>(w1);
     wimg_write<KS>(img1, w1, lane);
     if (adv) { p1l += kadv; p1h += kadv; }
-    issue_wc<NI>(w1, p1l, p1h);
-    // xL(s+1) landed: younger = t0L(s+2) + t1L(s+2)
-    wait_set<NI>(xr8);
+    load_tile<NI, true>(w1, p1l, p1h);
     ximg_write<KS>(xbuf[cur ^ 1], xr8, wid, lane);
     if (adv) { pxl += kadv; pxh += kadv; }
-    issue_xc<NI>(xr8, pxl, pxh);
-    // exit: t0L(s+2), t1L(s+2), xL(s+2) in flight — invariant restored
+    load_tile<NI, false>(xr8, pxl, pxh);
+    // exit: slice s+2 loads in flight — invariant restored
   }
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 
   const int ncol0 = n0t0 + 4 * kgrp;
   const int ncol1 = n0t1 + 4 * kgrp;
