@@ -42,7 +42,8 @@ for (M, N, K, tag) in SHAPES:
                for _ in range(nw)]
     out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
     nslices = (K + 255) // 256
-    wrk = torch.empty(min(16, nslices) * M * N, dtype=torch.float32, device="cuda")
+    wrk = torch.empty(min(48, 2 * nslices) * M * N, dtype=torch.float32,
+                      device="cuda")
     floor = wbytes / 6.3e12 * 1e6
     us_bl = t_rot(lambda i: F.linear(x, ws_list[i]), nw)
     line = (f"{tag:>8} ({nw} copies): blas {us_bl:6.1f}  floor {floor:6.1f}")
