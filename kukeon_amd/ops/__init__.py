@@ -92,11 +92,20 @@ _SKINNY_WS_RETIRED = []
 # 11.9 vs 19.2 at M=16 on 4096x4096) and trails it on qkv/gate_up/down/
 # lm_head, where hipBLASLt is 55-100% of the HBM floor. KUKEON_SKINNY_GEMM=1
 # forces the skinny kernel for every eligible shape (benchmarking).
-_USE_SKINNY = __import__("os").environ.get("KUKEON_SKINNY_GEMM", "0") == "1"
+_USE_SKINNY = __import__("os").environ.get("KUKEON_SKINNY_GEMM", "0")
 
 
 def _skinny_wins(rows: int, N: int, K: int) -> bool:
     return N == K == 4096
+
+
+# v5 (full-line never-drain pipeline, KS=128/2 blocks-per-CU): beats
+# hipBLASLt cold on the down-projection (32.2us vs 39.6 on 4096x14336 at
+# M=64, floor 18.6 — profiles/r02_progress.md); qkv ties, gate_up and
+# lm_head stay on the library. KUKEON_SKINNY_GEMM=5 forces it everywhere
+# for benchmarking.
+def _skinny5_wins(rows: int, N: int, K: int) -> bool:
+    return N == 4096 and K == 14336
 
 
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
@@ -104,9 +113,31 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     shapes where it measured faster than hipBLASLt; hipBLASLt otherwise."""
     rows = x.shape[0]
     if (x.is_cuda and x.dim() == 2 and rows <= 64
+            and w.shape[0] % 128 == 0 and w.shape[1] % 128 == 0
+            and x.dtype == torch.bfloat16
+            and (_USE_SKINNY == "5"
+                 or _skinny5_wins(rows, w.shape[0], w.shape[1]))):
+        N = w.shape[0]
+        K = w.shape[1]
+        out = torch.empty(rows, N, dtype=x.dtype, device=x.device)
+        key = (x.device.index or 0)
+        ws = _SKINNY_WS.get(key)
+        ngroups = N // 128
+        nslices = -(-K // 128)
+        splitk = 1 if ngroups >= 256 else min(nslices, -(-256 // ngroups))
+        need = max(1, splitk) * 64 * N
+        if ws is None or ws.numel() < need:
+            if ws is not None:
+                _SKINNY_WS_RETIRED.append(ws)
+            ws = torch.empty(need, dtype=torch.float32, device=x.device)
+            _SKINNY_WS[key] = ws
+        _native().skinny_gemm5(out, x, w, ws)
+        return out
+    if (x.is_cuda and x.dim() == 2 and rows <= 64
             and w.shape[0] % 64 == 0 and w.shape[1] % 32 == 0
             and x.dtype == torch.bfloat16
-            and (_USE_SKINNY or _skinny_wins(rows, w.shape[0], w.shape[1]))):
+            and (_USE_SKINNY == "1"
+                 or _skinny_wins(rows, w.shape[0], w.shape[1]))):
         N = w.shape[0]
         K = w.shape[1]
         out = torch.empty(rows, N, dtype=x.dtype, device=x.device)

@@ -796,11 +796,12 @@ void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int ngroups = N / 128;
   const int nslices = (int)(K / KS);
-  // 2 blocks/CU at KS=128: target ~512 blocks before settling for less
-  const int target = (KS == 128) ? 512 : 256;
+  // ~256 blocks: the sweep's knee for every shape — beyond it the
+  // extra split-K slab traffic outweighs occupancy (down: sk8/256
+  // blocks 32.2us vs sk16/512 blocks 40.6us cold)
   int splitk = 1;
-  if (ngroups < target)
-    splitk = min(nslices, (target + ngroups - 1) / ngroups);
+  if (ngroups < 256)
+    splitk = min(nslices, (256 + ngroups - 1) / ngroups);
   if (const char* ov = getenv("KUKEON_SK5_SPLITK")) {
     const int v = atoi(ov);
     if (v > 0) splitk = min(nslices, v);
