@@ -172,9 +172,28 @@ def linear_add_rmsnorm(x: torch.Tensor, w: torch.Tensor,
     rows = x.shape[0]
     N, K = w.shape
     if (x.is_cuda and parallel.tp_size() == 1 and x.dim() == 2
+            and rows <= 64 and N % 2048 == 0 and N <= 8192 and K % 128 == 0
+            and x.dtype == torch.bfloat16
+            and (_USE_SKINNY == "5" or _skinny5_wins(rows, N, K))):
+        key = (x.device.index or 0)
+        ws = _SKINNY_WS.get(key)
+        ngroups = N // 128
+        nslices = -(-K // 128)
+        splitk = min(nslices, -(-256 // ngroups))
+        need = max(1, splitk) * 64 * N
+        if ws is None or ws.numel() < need:
+            if ws is not None:
+                _SKINNY_WS_RETIRED.append(ws)
+            ws = torch.empty(need, dtype=torch.float32, device=x.device)
+            _SKINNY_WS[key] = ws
+        normed = torch.empty(rows, N, dtype=x.dtype, device=x.device)
+        _native().skinny_gemm5_fused_norm(normed, x, w, ws, residual,
+                                          norm_weight, eps)
+        return normed
+    if (x.is_cuda and parallel.tp_size() == 1 and x.dim() == 2
             and rows <= 64 and N % 2048 == 0 and N <= 8192 and K % 32 == 0
             and x.dtype == torch.bfloat16
-            and (_USE_SKINNY or _skinny_wins(rows, N, K))):
+            and (_USE_SKINNY == "1" or _skinny_wins(rows, N, K))):
         key = (x.device.index or 0)
         ws = _SKINNY_WS.get(key)
         ntiles = N // 64

@@ -37,6 +37,10 @@ void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws);
 void glds_probe(torch::Tensor out, torch::Tensor src);
+void skinny_gemm5_fused_norm(torch::Tensor normed, torch::Tensor x,
+                             torch::Tensor w, torch::Tensor ws,
+                             torch::Tensor residual, torch::Tensor nw,
+                             double eps);
 void skinny_gemm_fused_norm(torch::Tensor normed, torch::Tensor x,
                             torch::Tensor w, torch::Tensor ws,
                             torch::Tensor residual, torch::Tensor nw,
@@ -78,6 +82,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "weight-streaming decode GEMM (M<=64)");
   m.def("glds_probe", &kukeon::glds_probe,
         "asm global_lds round-trip validator");
+  m.def("skinny_gemm5_fused_norm", &kukeon::skinny_gemm5_fused_norm,
+        "skinny5 + fused split-K reduce + residual add + RMSNorm");
   m.def("skinny_gemm_fused_norm", &kukeon::skinny_gemm_fused_norm,
         "skinny GEMM + split-K reduce + residual add + RMSNorm");
   m.def("sample", &kukeon::sample, "top-k/top-p/temperature sampling");

@@ -314,6 +314,11 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 __global__ void skinny_reduce_kernel(unsigned short* __restrict__ out,
                                      const float* __restrict__ ws, long n,
                                      int splitk);
+__global__ __launch_bounds__(256) void skinny_reduce_add_rmsnorm_kernel(
+    unsigned short* __restrict__ normed,
+    unsigned short* __restrict__ residual, const float* __restrict__ ws,
+    const unsigned short* __restrict__ nw, int N, long total, int splitk,
+    float eps);
 
 DEV_INLINE void issue_w8(u32x4_t (&reg)[8], const unsigned short* p) {
   // 8 nt 16B loads at 64 B stride (one 256-k slice of one W row per lane
@@ -1068,6 +1073,55 @@ void skinny_gemm_fused_norm(torch::Tensor normed, torch::Tensor x,
     default: SKF_LAUNCH(4); break;
   }
 #undef SKF_LAUNCH
+  HIP_CHECK_KERNEL();
+  skinny_reduce_add_rmsnorm_kernel<<<dim3((unsigned)M), 256, 0, stream>>>(
+      reinterpret_cast<unsigned short*>(normed.data_ptr()),
+      reinterpret_cast<unsigned short*>(residual.data_ptr()), wsp,
+      reinterpret_cast<const unsigned short*>(nw.data_ptr()), N, total,
+      splitk, (float)eps);
+  HIP_CHECK_KERNEL();
+}
+
+void skinny_gemm5_fused_norm(torch::Tensor normed, torch::Tensor x,
+                             torch::Tensor w, torch::Tensor ws,
+                             torch::Tensor residual, torch::Tensor nw,
+                             double eps) {
+  // skinny5 (full-line never-drain pipeline) + the split-K reduce fused
+  // with residual add + RMSNorm: the decode down-projection epilogue in
+  // ONE kernel instead of reduce + norm (two ~4.5us in-graph launches).
+  const int M = x.size(0);
+  const long K = x.size(1);
+  const int N = w.size(0);
+  constexpr int KS = 128;
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 2048 == 0 && N <= 8192 &&
+              K % KS == 0);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() &&
+              residual.is_contiguous() && normed.is_contiguous());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int ngroups = N / 128;
+  const int nslices = (int)(K / KS);
+  int splitk = min(nslices, (256 + ngroups - 1) / ngroups);
+  if (const char* ov = getenv("KUKEON_SK5_SPLITK")) {
+    const int v = atoi(ov);
+    if (v > 0) splitk = min(nslices, v);
+  }
+  const int MT = (M + 15) / 16;
+  dim3 grid(ngroups, splitk);
+  auto* xp = reinterpret_cast<const unsigned short*>(x.data_ptr());
+  auto* wp = reinterpret_cast<const unsigned short*>(w.data_ptr());
+  const long total = (long)M * N;
+  float* wsp = ws.data_ptr<float>();
+  TORCH_CHECK(ws.numel() >= total * splitk, "workspace too small");
+#define SK5F_LAUNCH(MT_)                                                    \
+  skinny5_kernel<MT_, true, KS><<<grid, 256, 0, stream>>>(                  \
+      nullptr, wsp, xp, wp, M, N, K)
+  switch (MT) {
+    case 1: SK5F_LAUNCH(1); break;
+    case 2: SK5F_LAUNCH(2); break;
+    case 3: SK5F_LAUNCH(3); break;
+    default: SK5F_LAUNCH(4); break;
+  }
+#undef SK5F_LAUNCH
   HIP_CHECK_KERNEL();
   skinny_reduce_add_rmsnorm_kernel<<<dim3((unsigned)M), 256, 0, stream>>>(
       reinterpret_cast<unsigned short*>(normed.data_ptr()),
