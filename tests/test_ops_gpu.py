@@ -391,6 +391,54 @@ def test_skinny_gemm2(M, N, K, splitk, monkeypatch):
                                atol=3e-2)
 
 
+@pytest.mark.parametrize("M,N,K,ks,sk", [
+    (64, 4096, 14336, 128, 0), (64, 28672, 4096, 128, 0),
+    (64, 6144, 4096, 128, 4), (17, 6144, 4096, 128, 3),
+    (64, 4096, 14336, 256, 8), (64, 128256, 4096, 128, 0)])
+def test_skinny_gemm5(M, N, K, ks, sk, monkeypatch):
+    """v6 full-line never-drain pipeline (pure HIP, visible loads) vs
+    fp32 reference, both LDS geometries."""
+    from kukeon_amd import _C
+    monkeypatch.setenv("KUKEON_SK5_KS", str(ks))
+    if sk:
+        monkeypatch.setenv("KUKEON_SK5_SPLITK", str(sk))
+    torch.manual_seed(13)
+    x = (torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.5)
+    w = (torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05)
+    out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+    ws = torch.empty(48 * 64 * N, dtype=torch.float32, device=DEV)
+    _C.skinny_gemm5(out, x, w, ws)
+    ref = (x.float() @ w.float().T)
+    torch.testing.assert_close(out.float().cpu(), ref.cpu(), rtol=3e-2,
+                               atol=6e-2)
+
+
+def test_skinny_gemm5_fused_norm():
+    """skinny5 + fused split-K reduce + residual add + RMSNorm (the
+    decode down epilogue) vs the unfused reference chain."""
+    from kukeon_amd import _C
+    torch.manual_seed(7)
+    M, N, K = 64, 4096, 14336
+    x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.3
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05
+    resid = torch.randn(M, N, dtype=torch.bfloat16, device=DEV)
+    nw = torch.rand(N, dtype=torch.bfloat16, device=DEV) + 0.5
+    eps = 1e-5
+    resid_ref = resid.clone()
+    ws = torch.empty(32 * 64 * N, dtype=torch.float32, device=DEV)
+    normed = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+    _C.skinny_gemm5_fused_norm(normed, x, w, ws, resid, nw, eps)
+    # reference: fp32 gemm -> bf16 add -> rmsnorm
+    y = (x.float() @ w.float().T)
+    r2 = (resid_ref.float() + y).to(torch.bfloat16).float()
+    ref = (r2 * torch.rsqrt(r2.pow(2).mean(-1, keepdim=True) + eps) *
+           nw.float())
+    torch.testing.assert_close(normed.float().cpu(), ref.cpu(), rtol=4e-2,
+                               atol=6e-2)
+    torch.testing.assert_close(resid.float().cpu(), r2.cpu(), rtol=3e-2,
+                               atol=5e-2)
+
+
 @pytest.mark.parametrize("kv_dtype", ["bf16", "fp8"])
 def test_decode_graphs_match_eager(kv_dtype):
     """Self-advancing graph decode must produce the same tokens as the
