@@ -90,11 +90,18 @@ def _die(e: Exception):
 @cli.command()
 @pass_ctx
 def init(ctx):
-    """Bootstrap the run path, hierarchies and cgroups."""
+    """Bootstrap the run path, hierarchies, cgroups and the kukeon
+    system group (reference kuke init: sysuser.EnsureUserGroup +
+    ownership fixups — non-root hosts degrade to single-user)."""
     ctl = ctx._controller()
     ctl.bootstrap()
     from kukeon_amd.daemon.server import verify_or_write_instance
     verify_or_write_instance(Path(ctx.run_path))
+    from kukeon_amd.runtime import sysuser
+    gid = sysuser.ensure_group()
+    if gid is not None:
+        changed = sysuser.chown_tree(Path(ctx.run_path), gid)
+        click.echo(f"group kukeon (gid {gid}): {changed} entries owned")
     click.echo(f"initialized kukeon at {ctx.run_path} "
                f"(cgroups: {ctl.cgroups.mode}, "
                f"gpus: {ctl.gpus.devices or 'none'})")
@@ -538,6 +545,26 @@ def doctor(ctx, what):
     row("git", bool(sh.which("git")),
         "present" if sh.which("git") else
         "absent (container repo setup unavailable)")
+    from kukeon_amd.runtime import namespaces as nsmod
+    from kukeon_amd.runtime import netlink as nlmod
+    from kukeon_amd.runtime import sysuser
+    from kukeon_amd.images import overlay_supported
+    ns_ok = nsmod.can_unshare(nsmod.CLONE_NEWUTS | nsmod.CLONE_NEWIPC)
+    row("namespaces", ns_ok,
+        "uts/ipc/mount available" if ns_ok else
+        "denied (cells run in host namespaces)")
+    net_ok = nsmod.can_unshare(nsmod.CLONE_NEWNET) and nlmod.available()
+    row("netns+rtnetlink", net_ok,
+        "per-space bridges available" if net_ok else
+        "denied (space networks degrade to host net)")
+    ovl = overlay_supported()
+    row("overlayfs", ovl,
+        "image rootfs + RUN builds available" if ovl else
+        "absent (COPY-only builds, direct-chroot rootfs)")
+    gid = sysuser.lookup_group()
+    row("kukeon group", gid is not None,
+        f"gid {gid}" if gid is not None else
+        "absent (kuke init creates it; single-user until then)")
     run = Path(ctx.run_path)
     writable = os.access(run if run.exists() else run.parent, os.W_OK)
     row("run path", writable, str(run), required=True)
@@ -728,7 +755,9 @@ def daemon_serve(ctx, reconcile_interval, configuration, foreground):
     ctl = ctx._controller(server_cfg)
     ctl.bootstrap()
     from kukeon_amd.daemon.server import Server
-    srv = Server(ctl, ctx.socket_path, reconcile_interval)
+    from kukeon_amd.runtime import sysuser
+    srv = Server(ctl, ctx.socket_path, reconcile_interval,
+                 socket_gid=sysuser.lookup_group())
     srv.start()
     if foreground:
         srv.wait()

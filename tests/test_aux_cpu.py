@@ -371,3 +371,38 @@ spec:
     assert res[0].action == "created", (res[0].action, res[0].error)
     doc = ctl.get_cell("default", "default", "default", "legacy")
     assert doc.spec.containers[0].args == ["9"]
+
+
+def test_sysuser_group_and_ownership(tmp_path):
+    """sysuser: ensure_group + chown_tree + socket perms (reference
+    internal/sysuser EnsureUserGroup/ChownTreeAndChmodSkip). Uses a
+    throwaway group name when running as root; asserts graceful None
+    degrade otherwise."""
+    import os
+    from kukeon_amd.runtime import sysuser
+
+    (tmp_path / "data").mkdir()
+    f = tmp_path / "data" / "doc.json"
+    f.write_text("{}")
+    (tmp_path / "data" / "doc.json.lock").write_text("")
+    if os.geteuid() != 0:
+        assert sysuser.ensure_group("kukeon-test-nope") is None
+        return
+    import subprocess
+    name = f"kuketest{os.getpid() % 10000}"
+    try:
+        gid = sysuser.ensure_group(name)
+        assert gid is not None
+        assert sysuser.ensure_group(name) == gid  # idempotent
+        changed = sysuser.chown_tree(tmp_path, gid)
+        assert changed >= 2
+        assert f.stat().st_gid == gid
+        # lock tombstones skipped
+        assert (tmp_path / "data" / "doc.json.lock").stat().st_gid != gid \
+            or gid == 0
+        # dirs got setgid+group-rwx
+        assert (tmp_path / "data").stat().st_mode & 0o2070 == 0o2070
+        sysuser.apply_socket_group(str(f), gid)
+        assert f.stat().st_mode & 0o777 == 0o660
+    finally:
+        subprocess.run(["groupdel", name], capture_output=True)
