@@ -105,7 +105,10 @@ def _skinny_wins(rows: int, N: int, K: int) -> bool:
 # lm_head stay on the library. KUKEON_SKINNY_GEMM=5 forces it everywhere
 # for benchmarking.
 def _skinny5_wins(rows: int, N: int, K: int) -> bool:
-    return N == 4096 and K == 14336
+    # llama-3-8b down (32.2us vs blas 39.6 cold) and llama-3-70b down
+    # (79.3 vs 83.1 at M=16, 92.6 vs 139.3 at M=64) — measured on
+    # MI355X, profiles/r02_progress.md
+    return (N, K) in ((4096, 14336), (8192, 28672))
 
 
 def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:

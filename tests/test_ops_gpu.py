@@ -394,7 +394,8 @@ def test_skinny_gemm2(M, N, K, splitk, monkeypatch):
 @pytest.mark.parametrize("M,N,K,ks,sk", [
     (64, 4096, 14336, 128, 0), (64, 28672, 4096, 128, 0),
     (64, 6144, 4096, 128, 4), (17, 6144, 4096, 128, 3),
-    (64, 4096, 14336, 256, 8), (64, 128256, 4096, 128, 0)])
+    (64, 4096, 14336, 256, 8), (64, 128256, 4096, 128, 0),
+    (16, 8192, 28672, 128, 0), (64, 8192, 28672, 128, 0)])
 def test_skinny_gemm5(M, N, K, ks, sk, monkeypatch):
     """v6 full-line never-drain pipeline (pure HIP, visible loads) vs
     fp32 reference, both LDS geometries."""
