@@ -34,6 +34,12 @@ void skinny_gemm4(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                   torch::Tensor ws);
 void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                   torch::Tensor ws);
+void skinny_gemm6(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                  torch::Tensor ws);
+void skinny_gemm6_fused_norm(torch::Tensor normed, torch::Tensor x,
+                             torch::Tensor w, torch::Tensor ws,
+                             torch::Tensor residual, torch::Tensor nw,
+                             double eps);
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws);
 void glds_probe(torch::Tensor out, torch::Tensor src);
@@ -72,6 +78,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "16x16x32 bf16 MFMA fragment-layout probe");
   m.def("mfma_probe16k", &kukeon::mfma_probe16k,
         "16x16x16 (K=16) bf16 MFMA fragment-layout probe");
+  m.def("skinny_gemm6", &kukeon::skinny_gemm6,
+        "decode GEMM v6 (barrier-free register-x pipeline)");
+  m.def("skinny_gemm6_fused_norm", &kukeon::skinny_gemm6_fused_norm,
+        "v6 + fused split-K reduce + residual add + RMSNorm");
   m.def("skinny_gemm5", &kukeon::skinny_gemm5,
         "full-line W stream via wave-private LDS image (v5)");
   m.def("skinny_gemm4", &kukeon::skinny_gemm4,

@@ -1131,6 +1131,263 @@ void skinny_gemm5_fused_norm(torch::Tensor normed, torch::Tensor x,
   HIP_CHECK_KERNEL();
 }
 
+
+// ---------------------------------------------------------------------------
+// v6: barrier-free register-x pipeline.
+//
+// v5's residual gap (chip ~4 TB/s vs 6.1-6.75 for every ingredient probe,
+// ledger profiles/r02_progress.md item 4) correlates with its one structural
+// cost: the shared-LDS x image forces a per-slice __syncthreads(), and at
+// ~256 blocks on 256 CUs there is ONE wave per SIMD — so every slice, all
+// four waves convoy on the barrier and the slowest wave's memory jitter is
+// paid by the whole block, every slice.
+//
+// v6 removes the barrier entirely: the MFMA x operand (B fragment) is 16
+// contiguous bytes per lane at x[row][kc..kc+8], so it can be loaded
+// DIRECTLY from global memory (x is L2-resident) into registers one slice
+// ahead — no LDS staging, no cross-wave publication, no barrier. W keeps
+// the v5 full-line nt stream + per-wave XOR-swizzled LDS image. Each wave
+// is fully self-paced; the only cross-lane structure left is the wave
+// itself. LDS drops to the W images (KS*256 B per block).
+// ---------------------------------------------------------------------------
+template <int MT, int XNI>
+DEV_INLINE void load_xfrags(u32x4_t (&xr)[MT][XNI],
+                            const unsigned short* const (&xq)[MT]) {
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+#pragma unroll
+    for (int u = 0; u < XNI; ++u) {
+      xr[m][u] = *reinterpret_cast<const u32x4_t*>(xq[m] + u * 32);
+    }
+  }
+}
+
+template <int MT, int KS>
+DEV_INLINE void consume_reg(const unsigned short* wimg,
+                            const u32x4_t (&xr)[MT][KS / 32],
+                            f32x4_t (&acc)[MT], int row16, int kgrp) {
+#pragma unroll
+  for (int u = 0; u < KS / 32; ++u) {
+    const int kc = u * 32 + 8 * kgrp;
+    const uint4 av = *reinterpret_cast<const uint4*>(
+        reinterpret_cast<const char*>(wimg) + wimg_off<KS>(row16, kc * 2));
+    const bf16x8_t af = frag_of(av);
+#pragma unroll
+    for (int m = 0; m < MT; ++m) {
+      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af, frag_of(xr[m][u]), acc[m], 0, 0, 0);
+    }
+  }
+}
+
+template <int MT, bool SPLIT, int KS>
+__global__ __launch_bounds__(256) void skinny6_kernel(
+    unsigned short* __restrict__ out, float* __restrict__ ws,
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ w, int M, int N, long K) {
+  constexpr int NI = KS / 32;
+  constexpr int XNI = KS / 32;
+  __shared__ __align__(16) unsigned short wimg_all[4][2][16 * KS];
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int row16 = lane & 15;
+  const int kgrp = lane >> 4;
+  f32x4_t acc0[MT], acc1[MT];
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    acc0[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+    acc1[m] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+  }
+  const long kadv = (long)gridDim.y * KS;
+  const long ks0 = (long)blockIdx.y * KS;
+  if (ks0 >= K) return;
+  u32x4_t w0[NI], w1[NI];
+  u32x4_t xa[MT][XNI], xb[MT][XNI];
+  const int n0t0 = (blockIdx.x * 2 + 0) * 64 + wid * 16;
+  const int n0t1 = (blockIdx.x * 2 + 1) * 64 + wid * 16;
+  const int rlo = lane >> 3, seg = (lane & 7) * 8;  // elems
+  const unsigned short* p0l = w + (long)(n0t0 + rlo) * K + ks0 + seg;
+  const unsigned short* p0h = w + (long)(n0t0 + 8 + rlo) * K + ks0 + seg;
+  const unsigned short* p1l = w + (long)(n0t1 + rlo) * K + ks0 + seg;
+  const unsigned short* p1h = w + (long)(n0t1 + 8 + rlo) * K + ks0 + seg;
+  const unsigned short* xq[MT];
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    xq[m] = x + (long)min(M - 1, m * 16 + row16) * K + ks0 + 8 * kgrp;
+  }
+  unsigned short* img0 = wimg_all[wid][0];
+  unsigned short* img1 = wimg_all[wid][1];
+
+  // prologue: land slice 0, build the W images, put slice 1 in flight
+  load_tile<NI, true>(w0, p0l, p0h);
+  load_tile<NI, true>(w1, p1l, p1h);
+  load_xfrags<MT, XNI>(xa, xq);
+  wimg_write<KS>(img0, w0, lane);
+  wimg_write<KS>(img1, w1, lane);
+  {
+    const long ks1 = ks0 + kadv;
+    if (ks1 < K) {
+      p0l += kadv; p0h += kadv; p1l += kadv; p1h += kadv;
+#pragma unroll
+      for (int m = 0; m < MT; ++m) xq[m] += kadv;
+    }
+    load_tile<NI, true>(w0, p0l, p0h);
+    load_tile<NI, true>(w1, p1l, p1h);
+    load_xfrags<MT, XNI>(xb, xq);
+  }
+  long ks = ks0;
+  // per-slice body: consume slice `ks` from the images + xcur, keep the
+  // never-drain invariant (slice ks+2's W and x loads in flight on exit).
+  // No __syncthreads anywhere: images and x fragments are wave-private,
+  // and a wave's LDS reads/writes to the same image are ordered by the
+  // per-wave in-order LDS pipe (the same guarantee v5's consume->rewrite
+  // sequence already relies on).
+  auto step6 = [&](u32x4_t (&xcur)[MT][XNI]) {
+    const bool adv = (ks + 2 * kadv) < K;
+    consume_reg<MT, KS>(img0, xcur, acc0, row16, kgrp);
+    wimg_write<KS>(img0, w0, lane);  // first USE of w0 -> counted wait
+    if (adv) { p0l += kadv; p0h += kadv; }
+    load_tile<NI, true>(w0, p0l, p0h);
+    consume_reg<MT, KS>(img1, xcur, acc1, row16, kgrp);
+    wimg_write<KS>(img1, w1, lane);
+    if (adv) { p1l += kadv; p1h += kadv; }
+    load_tile<NI, true>(w1, p1l, p1h);
+    if (adv) {
+#pragma unroll
+      for (int m = 0; m < MT; ++m) xq[m] += kadv;
+    }
+    load_xfrags<MT, XNI>(xcur, xq);
+  };
+  for (;;) {
+    step6(xa);
+    ks += kadv;
+    if (ks >= K) break;
+    step6(xb);
+    ks += kadv;
+    if (ks >= K) break;
+  }
+
+  const int ncol0 = n0t0 + 4 * kgrp;
+  const int ncol1 = n0t1 + 4 * kgrp;
+  float* slab = SPLIT ? ws + (long)blockIdx.y * M * N : nullptr;
+#pragma unroll
+  for (int m = 0; m < MT; ++m) {
+    const int mrow = m * 16 + row16;
+    if (mrow >= M) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long off0 = (long)mrow * N + ncol0 + r;
+      const long off1 = (long)mrow * N + ncol1 + r;
+      if (SPLIT) {
+        slab[off0] = acc0[m][r];
+        slab[off1] = acc1[m][r];
+      } else {
+        out[off0] = f2us(acc0[m][r]);
+        out[off1] = f2us(acc1[m][r]);
+      }
+    }
+  }
+}
+
+void skinny_gemm6(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                  torch::Tensor ws) {
+  const int M = x.size(0);
+  const long K = x.size(1);
+  const int N = w.size(0);
+  int KS = 128;
+  if (const char* ov = getenv("KUKEON_SK6_KS")) {
+    KS = atoi(ov) == 256 ? 256 : 128;
+  }
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 128 == 0 && K % KS == 0);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int ngroups = N / 128;
+  const int nslices = (int)(K / KS);
+  int splitk = 1;
+  if (ngroups < 256)
+    splitk = min(nslices, (256 + ngroups - 1) / ngroups);
+  if (const char* ov = getenv("KUKEON_SK6_SPLITK")) {
+    const int v = atoi(ov);
+    if (v > 0) splitk = min(nslices, v);
+  }
+  const int MT = (M + 15) / 16;
+  dim3 grid(ngroups, splitk);
+  auto* op = reinterpret_cast<unsigned short*>(out.data_ptr());
+  auto* xp = reinterpret_cast<const unsigned short*>(x.data_ptr());
+  auto* wp = reinterpret_cast<const unsigned short*>(w.data_ptr());
+  const long total = (long)M * N;
+#define SK6_LAUNCH_KS(MT_, KS_)                                              \
+  if (splitk == 1) {                                                         \
+    skinny6_kernel<MT_, false, KS_><<<grid, 256, 0, stream>>>(               \
+        op, nullptr, xp, wp, M, N, K);                                       \
+  } else {                                                                   \
+    float* wsp = ws.data_ptr<float>();                                       \
+    TORCH_CHECK(ws.numel() >= total * splitk, "sk6 workspace too small");    \
+    skinny6_kernel<MT_, true, KS_><<<grid, 256, 0, stream>>>(                \
+        nullptr, wsp, xp, wp, M, N, K);                                      \
+    skinny_reduce_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256,   \
+                           0, stream>>>(op, wsp, total, splitk);             \
+  }
+#define SK6_LAUNCH(MT_)                                                      \
+  if (KS == 128) { SK6_LAUNCH_KS(MT_, 128) } else { SK6_LAUNCH_KS(MT_, 256) }
+  switch (MT) {
+    case 1: SK6_LAUNCH(1); break;
+    case 2: SK6_LAUNCH(2); break;
+    case 3: SK6_LAUNCH(3); break;
+    default: SK6_LAUNCH(4); break;
+  }
+#undef SK6_LAUNCH
+#undef SK6_LAUNCH_KS
+  HIP_CHECK_KERNEL();
+}
+
+void skinny_gemm6_fused_norm(torch::Tensor normed, torch::Tensor x,
+                             torch::Tensor w, torch::Tensor ws,
+                             torch::Tensor residual, torch::Tensor nw,
+                             double eps) {
+  const int M = x.size(0);
+  const long K = x.size(1);
+  const int N = w.size(0);
+  constexpr int KS = 128;
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 2048 == 0 && N <= 8192 &&
+              K % KS == 0);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() &&
+              residual.is_contiguous() && normed.is_contiguous());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int ngroups = N / 128;
+  const int nslices = (int)(K / KS);
+  int splitk = min(nslices, (256 + ngroups - 1) / ngroups);
+  if (const char* ov = getenv("KUKEON_SK6_SPLITK")) {
+    const int v = atoi(ov);
+    if (v > 0) splitk = min(nslices, v);
+  }
+  const int MT = (M + 15) / 16;
+  dim3 grid(ngroups, splitk);
+  auto* xp = reinterpret_cast<const unsigned short*>(x.data_ptr());
+  auto* wp = reinterpret_cast<const unsigned short*>(w.data_ptr());
+  const long total = (long)M * N;
+  float* wsp = ws.data_ptr<float>();
+  TORCH_CHECK(ws.numel() >= total * splitk, "workspace too small");
+#define SK6F_LAUNCH(MT_)                                                    \
+  skinny6_kernel<MT_, true, KS><<<grid, 256, 0, stream>>>(                  \
+      nullptr, wsp, xp, wp, M, N, K)
+  switch (MT) {
+    case 1: SK6F_LAUNCH(1); break;
+    case 2: SK6F_LAUNCH(2); break;
+    case 3: SK6F_LAUNCH(3); break;
+    default: SK6F_LAUNCH(4); break;
+  }
+#undef SK6F_LAUNCH
+  HIP_CHECK_KERNEL();
+  skinny_reduce_add_rmsnorm_kernel<<<dim3((unsigned)M), 256, 0, stream>>>(
+      reinterpret_cast<unsigned short*>(normed.data_ptr()),
+      reinterpret_cast<unsigned short*>(residual.data_ptr()), wsp,
+      reinterpret_cast<const unsigned short*>(nw.data_ptr()), N, total,
+      splitk, (float)eps);
+  HIP_CHECK_KERNEL();
+}
+
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws) {
   const int M = x.size(0);
