@@ -500,6 +500,27 @@ def status(ctx):
             checks.append(("daemon rpc", False, str(e)))
     data = Path(ctx.run_path) / "data"
     checks.append(("state tree", data.is_dir(), str(data)))
+    if data.is_dir():
+        # per-realm storage stats (reference ctr/client.go:169-181 exposes
+        # per-namespace storage from boltdb; here the state tree + per-run
+        # image layers are the storage domains)
+        def _du(p: Path) -> int:
+            total = 0
+            for f in p.rglob("*"):
+                try:
+                    if f.is_file() and not f.is_symlink():
+                        total += f.stat().st_size
+                except OSError:
+                    continue
+            return total
+
+        for realm in sorted(d.name for d in data.iterdir() if d.is_dir()):
+            checks.append((f"storage:{realm}", True,
+                           f"{_du(data / realm) / 1024:.0f} KiB"))
+        layers = Path(ctx.run_path) / "layers"
+        if layers.is_dir():
+            checks.append(("storage:layers", True,
+                           f"{_du(layers) / (1 << 20):.1f} MiB"))
     ok = True
     for name, good, detail in checks:
         mark = "ok" if good else "FAIL"
