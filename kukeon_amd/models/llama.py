@@ -193,11 +193,8 @@ class LlamaLayer(nn.Module):
         h2 = ops.linear_add_rmsnorm(attn_out, self.attn.o_w, residual,
                                     self.post_norm, self.eps)
         gu = ops.linear(h2, self.mlp.gate_up_w)
-        act = torch.empty(h2.shape[0], self.mlp.inter, dtype=h2.dtype,
-                          device=h2.device)
-        ops.silu_mul(act, gu)
-        h_next = ops.linear_add_rmsnorm(act, self.mlp.down_w, residual,
-                                        next_norm_w, eps)
+        h_next = ops.mlp_down_fused(gu, self.mlp.down_w, residual,
+                                    next_norm_w, eps)
         return h_next, residual
 
     # -- pipelined TP prefill phases (comm/GEMM overlap): the o-proj and

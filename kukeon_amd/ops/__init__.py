@@ -217,6 +217,45 @@ def linear_add_rmsnorm(x: torch.Tensor, w: torch.Tensor,
     return h
 
 
+
+def mlp_down_fused(gu: torch.Tensor, w: torch.Tensor,
+                   residual: torch.Tensor, norm_weight: torch.Tensor,
+                   eps: float) -> torch.Tensor:
+    """The decode MLP tail — silu(gate)*up -> down-projection ->
+    TP all-reduce -> residual add -> RMSNorm — with the activation fused
+    into the down GEMM's x staging on the skinny decode shapes (the
+    standalone silu_mul launch and its act-tensor HBM round trip
+    disappear; profiles/r02_progress.md). `gu` is the fused gate_up GEMM
+    output [rows, 2K]. Equivalent to silu_mul + linear_add_rmsnorm.
+    """
+    from kukeon_amd import parallel
+    rows = gu.shape[0]
+    N, K = w.shape
+    if (gu.is_cuda and parallel.tp_size() == 1 and gu.dim() == 2
+            and rows <= 64 and gu.shape[1] == 2 * K
+            and N % 2048 == 0 and N <= 8192 and K % 128 == 0
+            and gu.dtype == torch.bfloat16
+            and (_USE_SKINNY == "5" or _skinny5_wins(rows, N, K))):
+        key = (gu.device.index or 0)
+        ws = _SKINNY_WS.get(key)
+        ngroups = N // 128
+        nslices = -(-K // 128)
+        splitk = min(nslices, -(-256 // ngroups))
+        need = max(1, splitk) * 64 * N
+        if ws is None or ws.numel() < need:
+            if ws is not None:
+                _SKINNY_WS_RETIRED.append(ws)
+            ws = torch.empty(need, dtype=torch.float32, device=gu.device)
+            _SKINNY_WS[key] = ws
+        normed = torch.empty(rows, N, dtype=gu.dtype, device=gu.device)
+        _native().skinny_gemm5_silu_fused_norm(normed, gu, w, ws,
+                                               residual, norm_weight, eps)
+        return normed
+    act = torch.empty(rows, K, dtype=gu.dtype, device=gu.device)
+    silu_mul(act, gu)
+    return linear_add_rmsnorm(act, w, residual, norm_weight, eps)
+
+
 def moe_gather_tokens(out, input, row_map) -> None:
     _impl(input).moe_gather_tokens(out, input, row_map)
 

@@ -40,6 +40,10 @@ void skinny_gemm6_fused_norm(torch::Tensor normed, torch::Tensor x,
                              torch::Tensor w, torch::Tensor ws,
                              torch::Tensor residual, torch::Tensor nw,
                              double eps);
+void skinny_gemm5_silu_fused_norm(torch::Tensor normed, torch::Tensor gu,
+                                  torch::Tensor w, torch::Tensor ws,
+                                  torch::Tensor residual, torch::Tensor nw,
+                                  double eps);
 void skinny_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                  torch::Tensor ws);
 void glds_probe(torch::Tensor out, torch::Tensor src);
@@ -82,6 +86,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "decode GEMM v6 (barrier-free register-x pipeline)");
   m.def("skinny_gemm6_fused_norm", &kukeon::skinny_gemm6_fused_norm,
         "v6 + fused split-K reduce + residual add + RMSNorm");
+  m.def("skinny_gemm5_silu_fused_norm",
+        &kukeon::skinny_gemm5_silu_fused_norm,
+        "silu(gate)*up + down GEMM + reduce + add + RMSNorm fused");
   m.def("skinny_gemm5", &kukeon::skinny_gemm5,
         "full-line W stream via wave-private LDS image (v5)");
   m.def("skinny_gemm4", &kukeon::skinny_gemm4,

@@ -683,7 +683,26 @@ DEV_INLINE void ximg_write(unsigned short* xb, u32x4_t (&reg)[NI], int wid,
   }
 }
 
-template <int MT, bool SPLIT, int KS>
+// fused SwiGLU on a 16-B register fragment (8 bf16): same math as
+// activation.hip silu_mul_kernel so the fused path matches the two-kernel
+// path bit-for-bit through f2us round-to-nearest-even.
+DEV_INLINE u32x4_t silu_mul8(u32x4_t g, u32x4_t u) {
+  u32x4_t r;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const unsigned int gg = g[i], uu = u[i];
+    const float g0 = us2f((unsigned short)(gg & 0xffffu));
+    const float g1 = us2f((unsigned short)(gg >> 16));
+    const float u0 = us2f((unsigned short)(uu & 0xffffu));
+    const float u1 = us2f((unsigned short)(uu >> 16));
+    const float r0 = g0 / (1.f + __expf(-g0)) * u0;
+    const float r1 = g1 / (1.f + __expf(-g1)) * u1;
+    r[i] = (unsigned int)f2us(r0) | ((unsigned int)f2us(r1) << 16);
+  }
+  return r;
+}
+
+template <int MT, bool SPLIT, int KS, bool SILU = false>
 __global__ __launch_bounds__(256) void skinny5_kernel(
     unsigned short* __restrict__ out, float* __restrict__ ws,
     const unsigned short* __restrict__ x,
@@ -720,8 +739,18 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
   const unsigned short* p1h = w + (long)(n0t1 + 8 + rlo) * K + ks0 + seg;
   const int xrow_l = wid * 16 + rlo;
   const int xrow_h = wid * 16 + 8 + rlo;
-  const unsigned short* pxl = x + (long)min(M - 1, xrow_l) * K + ks0 + seg;
-  const unsigned short* pxh = x + (long)min(M - 1, xrow_h) * K + ks0 + seg;
+  // SILU: x is the fused gate_up GEMM output [M, 2K] — gate in the first
+  // K columns, up in the second; the activation happens in registers on
+  // the way into the LDS image and the standalone silu_mul kernel (plus
+  // its act-tensor HBM round trip) disappears.
+  const long xstride = SILU ? 2 * K : K;
+  const unsigned short* pxl =
+      x + (long)min(M - 1, xrow_l) * xstride + ks0 + seg;
+  const unsigned short* pxh =
+      x + (long)min(M - 1, xrow_h) * xstride + ks0 + seg;
+  const unsigned short* pul = SILU ? pxl + K : nullptr;
+  const unsigned short* puh = SILU ? pxh + K : nullptr;
+  u32x4_t ur8[SILU ? NI : 1];
   unsigned short* img0 = wimg_all[wid][0];
   unsigned short* img1 = wimg_all[wid][1];
 
@@ -729,6 +758,11 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
   load_tile<NI, true>(w0, p0l, p0h);
   load_tile<NI, true>(w1, p1l, p1h);
   load_tile<NI, false>(xr8, pxl, pxh);
+  if constexpr (SILU) {
+    load_tile<NI, false>(ur8, pul, puh);
+#pragma unroll
+    for (int i = 0; i < NI; ++i) xr8[i] = silu_mul8(xr8[i], ur8[i]);
+  }
   wimg_write<KS>(img0, w0, lane);
   wimg_write<KS>(img1, w1, lane);
   ximg_write<KS>(xbuf[0], xr8, wid, lane);
@@ -737,10 +771,12 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
     if (ks1 < K) {
       p0l += kadv; p0h += kadv; p1l += kadv; p1h += kadv;
       pxl += kadv; pxh += kadv;
+      if constexpr (SILU) { pul += kadv; puh += kadv; }
     }
     load_tile<NI, true>(w0, p0l, p0h);
     load_tile<NI, true>(w1, p1l, p1h);
     load_tile<NI, false>(xr8, pxl, pxh);
+    if constexpr (SILU) load_tile<NI, false>(ur8, pul, puh);
   }
   int cur = 0;
   for (long ks = ks0; ks < K; ks += kadv, cur ^= 1) {
@@ -758,9 +794,17 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
     wimg_write<KS>(img1, w1, lane);
     if (adv) { p1l += kadv; p1h += kadv; }
     load_tile<NI, true>(w1, p1l, p1h);
+    if constexpr (SILU) {
+#pragma unroll
+      for (int i = 0; i < NI; ++i) xr8[i] = silu_mul8(xr8[i], ur8[i]);
+    }
     ximg_write<KS>(xbuf[cur ^ 1], xr8, wid, lane);
-    if (adv) { pxl += kadv; pxh += kadv; }
+    if (adv) {
+      pxl += kadv; pxh += kadv;
+      if constexpr (SILU) { pul += kadv; puh += kadv; }
+    }
     load_tile<NI, false>(xr8, pxl, pxh);
+    if constexpr (SILU) load_tile<NI, false>(ur8, pul, puh);
     // exit: slice s+2 loads in flight — invariant restored
   }
 
@@ -1287,6 +1331,59 @@ __global__ __launch_bounds__(256) void skinny6_kernel(
       }
     }
   }
+}
+
+void skinny_gemm5_silu_fused_norm(torch::Tensor normed, torch::Tensor gu,
+                                  torch::Tensor w, torch::Tensor ws,
+                                  torch::Tensor residual, torch::Tensor nw,
+                                  double eps) {
+  // The whole decode MLP tail in two kernels: silu(gate)*up happens in
+  // registers inside the down-projection's x staging (SILU template
+  // path), and the split-K reduce fuses with the next layer's residual
+  // add + RMSNorm. Replaces silu_mul + skinny_gemm5 + reduce_add_norm —
+  // one fewer launch and no act-tensor HBM round trip per layer.
+  const int M = gu.size(0);
+  const int N = w.size(0);
+  const long K = w.size(1);
+  constexpr int KS = 128;
+  TORCH_CHECK(M >= 1 && M <= 64 && N % 2048 == 0 && N <= 8192 &&
+              K % KS == 0);
+  TORCH_CHECK(gu.size(1) == 2 * K, "gate_up output must be [M, 2K]");
+  TORCH_CHECK(gu.is_contiguous() && w.is_contiguous() &&
+              residual.is_contiguous() && normed.is_contiguous());
+  TORCH_CHECK(gu.scalar_type() == torch::kBFloat16);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int ngroups = N / 128;
+  const int nslices = (int)(K / KS);
+  int splitk = min(nslices, (256 + ngroups - 1) / ngroups);
+  if (const char* ov = getenv("KUKEON_SK5_SPLITK")) {
+    const int v = atoi(ov);
+    if (v > 0) splitk = min(nslices, v);
+  }
+  const int MT = (M + 15) / 16;
+  dim3 grid(ngroups, splitk);
+  auto* xp = reinterpret_cast<const unsigned short*>(gu.data_ptr());
+  auto* wp = reinterpret_cast<const unsigned short*>(w.data_ptr());
+  const long total = (long)M * N;
+  float* wsp = ws.data_ptr<float>();
+  TORCH_CHECK(ws.numel() >= total * splitk, "workspace too small");
+#define SK5S_LAUNCH(MT_)                                                    \
+  skinny5_kernel<MT_, true, KS, true><<<grid, 256, 0, stream>>>(            \
+      nullptr, wsp, xp, wp, M, N, K)
+  switch (MT) {
+    case 1: SK5S_LAUNCH(1); break;
+    case 2: SK5S_LAUNCH(2); break;
+    case 3: SK5S_LAUNCH(3); break;
+    default: SK5S_LAUNCH(4); break;
+  }
+#undef SK5S_LAUNCH
+  HIP_CHECK_KERNEL();
+  skinny_reduce_add_rmsnorm_kernel<<<dim3((unsigned)M), 256, 0, stream>>>(
+      reinterpret_cast<unsigned short*>(normed.data_ptr()),
+      reinterpret_cast<unsigned short*>(residual.data_ptr()), wsp,
+      reinterpret_cast<const unsigned short*>(nw.data_ptr()), N, total,
+      splitk, (float)eps);
+  HIP_CHECK_KERNEL();
 }
 
 void skinny_gemm6(torch::Tensor out, torch::Tensor x, torch::Tensor w,

@@ -440,6 +440,36 @@ def test_skinny_gemm5_fused_norm():
                                atol=5e-2)
 
 
+@pytest.mark.parametrize("M", [1, 16, 64])
+def test_skinny_gemm5_silu_fused_norm(M):
+    """Whole decode MLP tail fused — silu(gate)*up folded into the down
+    GEMM's x staging + reduce/add/RMSNorm epilogue — vs the unfused
+    fp32 reference chain."""
+    from kukeon_amd import _C
+    torch.manual_seed(11)
+    N, K = 4096, 14336
+    gu = torch.randn(M, 2 * K, dtype=torch.bfloat16, device=DEV) * 0.4
+    w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05
+    resid = torch.randn(M, N, dtype=torch.bfloat16, device=DEV)
+    nw = torch.rand(N, dtype=torch.bfloat16, device=DEV) + 0.5
+    eps = 1e-5
+    resid_ref = resid.clone()
+    ws = torch.empty(32 * 64 * N, dtype=torch.float32, device=DEV)
+    normed = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+    _C.skinny_gemm5_silu_fused_norm(normed, gu, w, ws, resid, nw, eps)
+    g = gu[:, :K].float()
+    u = gu[:, K:].float()
+    act = (torch.nn.functional.silu(g) * u).to(torch.bfloat16)
+    y = act.float() @ w.float().T
+    r2 = (resid_ref.float() + y).to(torch.bfloat16).float()
+    ref = (r2 * torch.rsqrt(r2.pow(2).mean(-1, keepdim=True) + eps) *
+           nw.float())
+    torch.testing.assert_close(normed.float().cpu(), ref.cpu(), rtol=4e-2,
+                               atol=6e-2)
+    torch.testing.assert_close(resid.float().cpu(), r2.cpu(), rtol=3e-2,
+                               atol=5e-2)
+
+
 @pytest.mark.parametrize("kv_dtype", ["bf16", "fp8"])
 def test_decode_graphs_match_eager(kv_dtype):
     """Self-advancing graph decode must produce the same tokens as the
