@@ -93,6 +93,7 @@ _SKINNY_WS_RETIRED = []
 # lm_head, where hipBLASLt is 55-100% of the HBM floor. KUKEON_SKINNY_GEMM=1
 # forces the skinny kernel for every eligible shape (benchmarking).
 _USE_SKINNY = __import__("os").environ.get("KUKEON_SKINNY_GEMM", "0")
+_FUSE_SILU = __import__("os").environ.get("KUKEON_FUSE_SILU", "1") != "0"
 
 
 def _skinny_wins(rows: int, N: int, K: int) -> bool:
@@ -231,7 +232,8 @@ def mlp_down_fused(gu: torch.Tensor, w: torch.Tensor,
     from kukeon_amd import parallel
     rows = gu.shape[0]
     N, K = w.shape
-    if (gu.is_cuda and parallel.tp_size() == 1 and gu.dim() == 2
+    if (_FUSE_SILU and gu.is_cuda and parallel.tp_size() == 1
+            and gu.dim() == 2
             and rows <= 64 and gu.shape[1] == 2 * K
             and N % 2048 == 0 and N <= 8192 and K % 128 == 0
             and gu.dtype == torch.bfloat16
