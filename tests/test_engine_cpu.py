@@ -371,3 +371,26 @@ def test_multi_victim_eviction_unblocks_admission():
     assert len(got) == 8
     # at least two idle sessions were evicted
     assert sum(1 for kv in idle if not kv.blocks) >= 2
+
+
+def test_mlp_down_fused_matches_unfused_chain():
+    """ops.mlp_down_fused (CPU fallback path) must equal
+    silu_mul + linear + residual add + RMSNorm exactly."""
+    import torch
+    from kukeon_amd import ops
+
+    torch.manual_seed(5)
+    M, N, K = 8, 64, 96
+    gu = torch.randn(M, 2 * K, dtype=torch.bfloat16) * 0.4
+    w = torch.randn(N, K, dtype=torch.bfloat16) * 0.05
+    resid = torch.randn(M, N, dtype=torch.bfloat16)
+    nw = torch.rand(N, dtype=torch.bfloat16) + 0.5
+    r1 = resid.clone()
+    out = ops.mlp_down_fused(gu, w, r1, nw, 1e-5)
+
+    r2 = resid.clone()
+    act = torch.empty(M, K, dtype=torch.bfloat16)
+    ops.silu_mul(act, gu)
+    ref = ops.linear_add_rmsnorm(act, w, r2, nw, 1e-5)
+    torch.testing.assert_close(out, ref, rtol=0, atol=0)
+    torch.testing.assert_close(r1, r2, rtol=0, atol=0)
