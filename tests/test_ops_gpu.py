@@ -397,7 +397,7 @@ def test_skinny_gemm2(M, N, K, splitk, monkeypatch):
     (64, 4096, 14336, 256, 8), (64, 128256, 4096, 128, 0),
     (16, 8192, 28672, 128, 0), (64, 8192, 28672, 128, 0)])
 def test_skinny_gemm5(M, N, K, ks, sk, monkeypatch):
-    """v6 full-line never-drain pipeline (pure HIP, visible loads) vs
+    """v5 full-line never-drain pipeline (pure HIP, visible loads) vs
     fp32 reference, both LDS geometries."""
     from kukeon_amd import _C
     monkeypatch.setenv("KUKEON_SK5_KS", str(ks))
@@ -409,6 +409,26 @@ def test_skinny_gemm5(M, N, K, ks, sk, monkeypatch):
     out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
     ws = torch.empty(48 * 64 * N, dtype=torch.float32, device=DEV)
     _C.skinny_gemm5(out, x, w, ws)
+    ref = (x.float() @ w.float().T)
+    torch.testing.assert_close(out.float().cpu(), ref.cpu(), rtol=3e-2,
+                               atol=6e-2)
+
+
+@pytest.mark.parametrize("M,N,K,sk", [
+    (64, 4096, 14336, 0), (16, 4096, 4096, 4), (33, 6144, 4096, 0)])
+def test_skinny_gemm6(M, N, K, sk, monkeypatch):
+    """v6 barrier-free register-x pipeline vs fp32 reference (kept
+    in-tree as the documented negative result — slower than v5, see
+    profiles/r02_progress.md — but it must stay correct)."""
+    from kukeon_amd import _C
+    if sk:
+        monkeypatch.setenv("KUKEON_SK6_SPLITK", str(sk))
+    torch.manual_seed(17)
+    x = (torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.5)
+    w = (torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05)
+    out = torch.empty(M, N, dtype=torch.bfloat16, device=DEV)
+    ws = torch.empty(48 * 64 * N, dtype=torch.float32, device=DEV)
+    _C.skinny_gemm6(out, x, w, ws)
     ref = (x.float() @ w.float().T)
     torch.testing.assert_close(out.float().cpu(), ref.cpu(), rtol=3e-2,
                                atol=6e-2)
