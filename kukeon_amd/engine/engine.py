@@ -272,10 +272,19 @@ class LLMEngine:
         free = self.kv.allocator.num_free
         avail_rows = len(self._free_rows)
         budget = self.ecfg.max_prefill_tokens
+        # soft tail: a request whose remainder only slightly overflows the
+        # budget is admitted whole rather than leaving a tiny (<1/16
+        # budget) chunk for a full extra model pass next step — at 64
+        # sessions x 257 tokens vs a 16384 budget the hard cutoff cost an
+        # entire 64-token full-layer sweep (~a decode step) per turn.
+        slack = budget // 16
         while self.waiting and avail_rows > 0 and tokens < budget:
             req = self.waiting[0]
             remaining = len(req.prompt_tokens) - req.prefill_pos
             chunk = min(remaining, budget - tokens)
+            if 0 < remaining - chunk <= slack and tokens + remaining <= \
+                    budget + slack:
+                chunk = remaining
             if chunk <= 0:
                 break
             finishing = (req.prefill_pos + chunk) == len(req.prompt_tokens)

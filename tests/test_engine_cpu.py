@@ -394,3 +394,35 @@ def test_mlp_down_fused_matches_unfused_chain():
     ref = ops.linear_add_rmsnorm(act, w, r2, nw, 1e-5)
     torch.testing.assert_close(out, ref, rtol=0, atol=0)
     torch.testing.assert_close(r1, r2, rtol=0, atol=0)
+
+
+def test_prefill_soft_tail_admission():
+    """A request whose remainder barely overflows the prefill budget is
+    admitted whole instead of leaving a tiny tail chunk that costs a
+    full extra model pass."""
+    import torch
+    from kukeon_amd.engine.config import (EngineConfig, SamplingParams,
+                                          tiny_llama)
+    from kukeon_amd.engine.engine import LLMEngine
+    from kukeon_amd.models.llama import LlamaModel
+
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=512, max_sessions=8,
+                        num_kv_blocks=256, max_prefill_tokens=64,
+                        use_graphs=False)
+    torch.manual_seed(0)
+    model = LlamaModel(cfg, device="cpu")
+    eng = LLMEngine(model, cfg, ecfg, device="cpu")
+    from kukeon_amd.engine.kv_cache import SequenceKV
+    # 66 tokens vs budget 64, slack 4: admitted in ONE chunk
+    eng.add_request(SequenceKV(ecfg.block_size), list(range(66)),
+                    SamplingParams(max_new_tokens=2))
+    batch, blocked = eng._admit_prefill()
+    assert not blocked
+    assert [c for _, c in batch] == [66]
+    # 80 tokens vs budget 64: overflow > slack -> still chunked
+    eng2 = LLMEngine(LlamaModel(cfg, device="cpu"), cfg, ecfg, device="cpu")
+    eng2.add_request(SequenceKV(ecfg.block_size), list(range(80)),
+                     SamplingParams(max_new_tokens=2))
+    batch2, _ = eng2._admit_prefill()
+    assert [c for _, c in batch2] == [64]
