@@ -1340,8 +1340,11 @@ void skinny_gemm5_silu_fused_norm(torch::Tensor normed, torch::Tensor gu,
   // The whole decode MLP tail in two kernels: silu(gate)*up happens in
   // registers inside the down-projection's x staging (SILU template
   // path), and the split-K reduce fuses with the next layer's residual
-  // add + RMSNorm. Replaces silu_mul + skinny_gemm5 + reduce_add_norm —
-  // one fewer launch and no act-tensor HBM round trip per layer.
+  // add + RMSNorm. Measured SLOWER than silu_mul + skinny_gemm5 in situ
+  // (54.2 vs 42.9 us cold — doubling the x-staging bytes beats the
+  // saved launch; ledger profiles/r02_progress.md) so dispatch is off
+  // by default (KUKEON_FUSE_SILU=1 re-enables); kept as the documented
+  // negative result.
   const int M = gu.size(0);
   const int N = w.size(0);
   const long K = w.size(1);
