@@ -1132,11 +1132,18 @@ class Controller:
         kind, name = doc.kind, doc.metadata.name
         if kind == api.KIND_REALM:
             try:
-                self.get_realm(name)
-                return ResourceResult(kind, name, "unchanged")
+                cur = self.get_realm(name)
             except errors.RealmNotFound:
                 self.create_realm(doc)
                 return ResourceResult(kind, name, "created")
+            if diffmod.diff_realm(doc, cur).change_type == \
+                    diffmod.ChangeType.NONE:
+                return ResourceResult(kind, name, "unchanged")
+            cur.metadata.labels = dict(cur.metadata.labels or {},
+                                       **(doc.metadata.labels or {}))
+            self.store.write_cas(
+                self.store.realm_dir(name) / METADATA_FILE, cur.to_dict())
+            return ResourceResult(kind, name, "updated")
         if kind == api.KIND_SPACE:
             try:
                 cur = self.get_space(doc.spec.realm_id, name)
@@ -1154,11 +1161,20 @@ class Controller:
             return ResourceResult(kind, name, "updated")
         if kind == api.KIND_STACK:
             try:
-                self.get_stack(doc.spec.realm_id, doc.spec.space_id, name)
-                return ResourceResult(kind, name, "unchanged")
+                cur = self.get_stack(doc.spec.realm_id, doc.spec.space_id,
+                                     name)
             except errors.StackNotFound:
                 self.create_stack(doc)
                 return ResourceResult(kind, name, "created")
+            if diffmod.diff_stack(doc, cur).change_type == \
+                    diffmod.ChangeType.NONE:
+                return ResourceResult(kind, name, "unchanged")
+            cur.metadata.labels = dict(cur.metadata.labels or {},
+                                       **(doc.metadata.labels or {}))
+            self.store.write_cas(
+                self.store.stack_dir(doc.spec.realm_id, doc.spec.space_id,
+                                     name) / METADATA_FILE, cur.to_dict())
+            return ResourceResult(kind, name, "updated")
         if kind == api.KIND_SECRET:
             self.put_secret(doc)
             return ResourceResult(kind, name, "updated")
@@ -1197,12 +1213,30 @@ class Controller:
                     self.recreate_cell(doc)
                     return ResourceResult(kind, name, "recreated")
                 cur.spec = doc.spec
+                if doc.metadata.labels:
+                    cur.metadata.labels = dict(cur.metadata.labels or {},
+                                               **doc.metadata.labels)
                 # spec update: bump generation, guarded against a racing
                 # writer between our read and this write
                 self.store.write_cas(
                     self._cell_path(doc.spec.realm_id, doc.spec.space_id,
                                     doc.spec.stack_id, name), cur.to_dict(),
                     expected_generation=cur.metadata.generation or None)
+                # converge a RUNNING cell now: the spec-hash start path
+                # respawns exactly the drifted containers (and spawns
+                # additive ones) under the existing cell namespaces —
+                # without this, a Compatible apply would not land until
+                # the next manual start (reference UpdateCell child
+                # recreate dance, apply/diff.go classification notes)
+                if d.respawn and (cur.status and cur.status.state in
+                                  (api.STATE_READY, api.STATE_DEGRADED)):
+                    try:
+                        self.start_cell(doc.spec.realm_id,
+                                        doc.spec.space_id,
+                                        doc.spec.stack_id, name)
+                    except errors.KukeonError as e:
+                        return ResourceResult(kind, name, "updated",
+                                              error=f"converge: {e}")
                 return ResourceResult(kind, name, "updated")
         raise errors.ValidationError(f"kind {kind} is not applyable")
 

@@ -1,8 +1,30 @@
-"""Desired-vs-actual diff with change classification.
+"""Desired-vs-actual diff with per-field change classification.
 
-Reference contract (internal/controller/apply/diff.go): ChangeType in
-{None, Additive, Compatible, Breaking}; Breaking changes force a cell
-recreate; annotations and provenance are deliberately not compared.
+Reference contract (internal/controller/apply/diff.go, 1.2k LoC):
+ChangeType in {None, Additive, Compatible, Breaking}; Breaking changes
+force a cell recreate; Compatible changes land in place; annotations and
+provenance are deliberately not compared.
+
+The per-field classification here is grounded in THIS runtime's
+semantics, not a transcription of the reference's containerd rules:
+
+* A cell's uts/ipc/net namespaces, its veth/IP and its GPU reservation
+  are created once by the root pause shim / start path and cannot be
+  re-stamped onto a live cell -> fields baked into them are BREAKING
+  (`host_network`, `host_pid`, `gpus`, scope identity).
+* Everything a container shim renders into its spawn spec at exec time
+  (image rootfs, command/args/env/cwd, user, devices, volume binds,
+  secrets, repos, git identity, tty/attachable) is COMPATIBLE: the
+  spec-hash respawn path (controller start, reference start.go:867+
+  analog) recreates exactly the drifted containers under the existing
+  cell namespaces. These are reported in DiffResult.respawn so apply
+  can converge a running cell immediately instead of waiting for the
+  next manual start.
+* Restart-policy knobs and labels are consulted by the reconciler from
+  metadata on every pass -> COMPATIBLE with no respawn.
+* A new container id is ADDITIVE (start spawns it); a removed id is
+  BREAKING (this runtime has no in-place child-removal path — recreate
+  is the safe converge).
 """
 from __future__ import annotations
 
@@ -26,6 +48,9 @@ _SEVERITY = {ChangeType.NONE: 0, ChangeType.ADDITIVE: 1,
 class DiffResult:
     def __init__(self):
         self.changes: List[Tuple[str, ChangeType]] = []
+        # container ids whose spawn spec drifted: a running cell is
+        # converged by respawning exactly these (spec-hash path)
+        self.respawn: List[str] = []
 
     def add(self, path: str, ct: ChangeType):
         self.changes.append((path, ct))
@@ -42,37 +67,67 @@ class DiffResult:
     def paths(self) -> List[str]:
         return [p for p, _ in self.changes]
 
+    @property
+    def breaking_paths(self) -> List[str]:
+        return [p for p, ct in self.changes if ct == ChangeType.BREAKING]
 
-# container fields whose change requires recreating the process
-_BREAKING_CONTAINER_FIELDS = [
+
+# Baked into cell-level state at bring-up (pause-shim namespaces, GPU
+# allocation): cannot be re-stamped in place.
+_BREAKING_CONTAINER_FIELDS = ["host_network", "host_pid", "gpus"]
+
+# Rendered into the container's spawn spec at exec: a child respawn under
+# the existing cell namespaces lands the change (spec-hash drift path).
+_RESPAWN_CONTAINER_FIELDS = [
     "image", "command", "args", "working_dir", "env", "user", "privileged",
-    "host_network", "host_pid", "devices", "gpus", "attachable", "volumes",
-    "secrets", "repos", "git", "tty",
-]
-_COMPATIBLE_CONTAINER_FIELDS = [
-    "restart_policy", "restart_backoff_seconds", "restart_max_retries",
+    "devices", "attachable", "volumes", "secrets", "repos", "git", "tty",
     "resources",
 ]
+
+# Consulted from metadata by the reconciler every pass: no respawn.
+_METADATA_CONTAINER_FIELDS = [
+    "restart_policy", "restart_backoff_seconds", "restart_max_retries",
+]
+
+
+def _labels(doc) -> dict:
+    return dict(getattr(doc.metadata, "labels", None) or {})
 
 
 def diff_cell(desired: api.CellDoc, actual: api.CellDoc) -> DiffResult:
     r = DiffResult()
+    if _scope(desired) != _scope(actual):
+        # identity/scope change: a different cell entirely
+        r.add("spec.scope", ChangeType.BREAKING)
     if desired.spec.auto_delete != actual.spec.auto_delete:
+        # the reaper reads AutoDelete from metadata on the next pass
         r.add("spec.autoDelete", ChangeType.COMPATIBLE)
+    if _labels(desired) != _labels(actual):
+        r.add("metadata.labels", ChangeType.COMPATIBLE)
     des = {_cname(c): c for c in desired.spec.containers}
     act = {_cname(c): c for c in actual.spec.containers}
     for name, c in des.items():
         if name not in act:
             r.add(f"spec.containers[{name}]", ChangeType.ADDITIVE)
+            r.respawn.append(name)
             continue
         a = act[name]
+        drifted = False
         for fname in _BREAKING_CONTAINER_FIELDS:
             if getattr(c, fname) != getattr(a, fname):
-                r.add(f"spec.containers[{name}].{fname}", ChangeType.BREAKING)
-        for fname in _COMPATIBLE_CONTAINER_FIELDS:
+                r.add(f"spec.containers[{name}].{fname}",
+                      ChangeType.BREAKING)
+        for fname in _RESPAWN_CONTAINER_FIELDS:
             if getattr(c, fname) != getattr(a, fname):
                 r.add(f"spec.containers[{name}].{fname}",
                       ChangeType.COMPATIBLE)
+                drifted = True
+        for fname in _METADATA_CONTAINER_FIELDS:
+            if getattr(c, fname) != getattr(a, fname):
+                r.add(f"spec.containers[{name}].{fname}",
+                      ChangeType.COMPATIBLE)
+        if drifted:
+            r.respawn.append(name)
     for name in act:
         if name not in des:
             r.add(f"spec.containers[{name}]", ChangeType.BREAKING)
@@ -84,12 +139,36 @@ def diff_space(desired: api.SpaceDoc, actual: api.SpaceDoc) -> DiffResult:
     dn = desired.spec.network.to_dict() if desired.spec.network else {}
     an = actual.spec.network.to_dict() if actual.spec.network else {}
     if dn != an:
+        # egress chains + routes are re-asserted by the space-network
+        # reconcile pass; existing cells keep their veth/IP
         r.add("spec.network", ChangeType.COMPATIBLE)
     dd = desired.spec.defaults.to_dict() if desired.spec.defaults else {}
     ad = actual.spec.defaults.to_dict() if actual.spec.defaults else {}
     if dd != ad:
+        # inheritance is applied at cell materialization; existing cells
+        # are not restamped (reference: Compatible)
         r.add("spec.defaults", ChangeType.COMPATIBLE)
+    if _labels(desired) != _labels(actual):
+        r.add("metadata.labels", ChangeType.COMPATIBLE)
     return r
+
+
+def diff_realm(desired: api.RealmDoc, actual: api.RealmDoc) -> DiffResult:
+    r = DiffResult()
+    if _labels(desired) != _labels(actual):
+        r.add("metadata.labels", ChangeType.COMPATIBLE)
+    return r
+
+
+def diff_stack(desired: api.StackDoc, actual: api.StackDoc) -> DiffResult:
+    r = DiffResult()
+    if _labels(desired) != _labels(actual):
+        r.add("metadata.labels", ChangeType.COMPATIBLE)
+    return r
+
+
+def _scope(d: api.CellDoc):
+    return (d.spec.realm_id, d.spec.space_id, d.spec.stack_id)
 
 
 def _cname(c: api.ContainerSpec) -> str:

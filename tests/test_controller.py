@@ -184,9 +184,17 @@ spec:
     results = ctl.apply_documents(yaml_text)
     assert all(r.action == "unchanged" for r in results), [
         (r.kind, r.action, r.error) for r in results]
-    # breaking change to the cell -> recreate
+    # args change on a child container -> Compatible (spec-hash respawn
+    # lands it under the existing cell namespaces); spec persisted
     results = ctl.apply_documents(yaml_text.replace('args: ["60"]',
                                                     'args: ["120"]'))
+    cellres = [r for r in results if r.kind == "Cell"][0]
+    assert cellres.action == "updated"
+    got = ctl.get_cell("prod", "web", "app", "worker")
+    assert got.spec.containers[0].args == ["120"]
+    # gpus change is baked into cell bring-up -> Breaking -> recreate
+    results = ctl.apply_documents(yaml_text.replace(
+        'args: ["60"]', 'args: ["120"]\n      gpus: 1'))
     cellres = [r for r in results if r.kind == "Cell"][0]
     assert cellres.action == "recreated"
 
@@ -235,7 +243,8 @@ def test_blueprint_run_and_outofsync(ctl):
     rec = ctl.reconcile_cell("default", "default", "default",
                              doc.metadata.name)
     assert rec.status.out_of_sync
-    assert "Breaking" in rec.status.out_of_sync_reason
+    assert rec.status.out_of_sync_reason == \
+        "Compatible: spec.containers[main].args"
 
 
 def test_session_wallclock_deadline(ctl):

@@ -142,9 +142,11 @@ def test_apply_matrix_document_kinds(harness):
 
 
 def test_apply_compatible_update_preserves_process(harness):
-    """A COMPATIBLE spec change (restart policy knob) updates in place —
-    the running container keeps its PID; a BREAKING change (argv)
-    recreates it (spec-hash + diff engine together)."""
+    """Per-field diff classification end-to-end: a metadata-only change
+    (restart policy knob) keeps the running PID; an argv change is
+    Compatible on a child container and the apply itself converges the
+    running cell (spec-hash respawn -> new PID, no manual start); a
+    gpus change is Breaking -> recreate."""
     ctl, srv, client = harness
     client.CreateCell(doc=cell_doc("upd").to_dict())
     client.StartCell(realm="default", space="default", stack="default",
@@ -152,7 +154,7 @@ def test_apply_compatible_update_preserves_process(harness):
     cdir = ctl.store.cell_dir("default", "default", "default",
                               "upd") / "main"
     pid1 = json.loads((cdir / "runtime.json").read_text())["workloadPid"]
-    # compatible: restart policy change
+    # metadata-only: restart policy change -> same process
     d = cell_doc("upd", rp="on-failure").to_dict()
     res = client.ApplyDocuments(yaml=yaml.safe_dump(d))
     assert res[0]["action"] == "updated"
@@ -160,17 +162,24 @@ def test_apply_compatible_update_preserves_process(harness):
                      name="upd")
     pid2 = json.loads((cdir / "runtime.json").read_text())["workloadPid"]
     assert pid2 == pid1  # unchanged spawn spec -> same process
-    # breaking: argv change -> recreate (new pid)
+    # argv change: Compatible, converged by the apply itself
     d = cell_doc("upd", cmd_args=("60",)).to_dict()
     res = client.ApplyDocuments(yaml=yaml.safe_dump(d))
-    assert res[0]["action"] == "recreated"
+    assert res[0]["action"] == "updated", res
     deadline = time.monotonic() + 10
     pid3 = pid2
     while pid3 == pid2 and time.monotonic() < deadline:
         data = ctl.store.read(cdir / "runtime.json")
         pid3 = (data or {}).get("workloadPid", pid2)
         time.sleep(0.1)
-    assert pid3 != pid2
+    assert pid3 != pid2  # respawned with the new argv by apply
+    got = ctl.get_cell("default", "default", "default", "upd")
+    assert got.spec.containers[0].args == ["60"]
+    # hostNetwork flip: baked into the cell namespaces -> Breaking
+    d = cell_doc("upd", cmd_args=("60",)).to_dict()
+    d["spec"]["containers"][0]["hostNetwork"] = True
+    res = client.ApplyDocuments(yaml=yaml.safe_dump(d))
+    assert res[0]["action"] == "recreated", res
     client.KillCell(realm="default", space="default", stack="default",
                     name="upd")
 
