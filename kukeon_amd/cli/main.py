@@ -241,8 +241,13 @@ KIND_ALIASES = {
 @click.argument("name", required=False)
 @click.option("-o", "--output", default="table",
               type=click.Choice(["table", "yaml", "json"]))
+@click.option("--api-version", "api_version", default=None,
+              type=click.Choice(["v1alpha1", "v1beta1"]),
+              help="export documents at a wire version (v1alpha1 is a "
+                   "lossy downgrade; dropped fields are reported on "
+                   "stderr)")
 @pass_ctx
-def get(ctx, kind, name, output):
+def get(ctx, kind, name, output, api_version):
     """Get resources: kuke get cells | kuke get cell NAME -o yaml."""
     k = KIND_ALIASES.get(kind.lower())
     if k is None:
@@ -293,6 +298,18 @@ def get(ctx, kind, name, output):
             docs = []
     except errors.KukeonError as e:
         _die(e)
+    if api_version and api_version != "v1beta1":
+        from kukeon_amd.api import scheme
+        wired = []
+        for d in docs:
+            w, lost = scheme.to_wire(d, api_version)
+            wired.append(w)
+            for f in lost:
+                click.echo(f"warn: {d['metadata']['name']}: dropped in "
+                           f"{api_version}: {f}", err=True)
+        docs = wired
+        if output == "table":
+            output = "yaml"  # a downgraded doc only makes sense as a doc
     if output == "yaml":
         click.echo(yaml.safe_dump_all(docs, sort_keys=False).rstrip())
     elif output == "json":

@@ -170,3 +170,41 @@ def test_cli_doctor(tmp_path):
     for field in ("cgroups", "cgroup:memory", "amdgpu", "kfd", "iptables",
                   "git", "run path", "hip extension"):
         assert field in out, out
+
+
+def test_get_api_version_downgrade(tmp_path):
+    """kuke get --api-version v1alpha1 exports the lossy downgrade
+    (scope keys renamed, dropped beta-only fields reported)."""
+    import yaml as yamlmod
+    from click.testing import CliRunner
+    from kukeon_amd.cli.main import cli
+
+    runner = CliRunner()
+    rp = str(tmp_path / "run")
+    r = runner.invoke(cli, ["--run-path", rp, "--local", "init"])
+    assert r.exit_code == 0, r.output
+    cell = {
+        "apiVersion": "v1beta1", "kind": "Cell",
+        "metadata": {"name": "alpha-export"},
+        "spec": {"realmId": "default", "spaceId": "default",
+                 "stackId": "default",
+                 "containers": [{"id": "main", "command": "sleep",
+                                 "args": ["5"],
+                                 "repos": [{"url": "https://x/y.git"}]}]},
+    }
+    f = tmp_path / "cell.yaml"
+    f.write_text(yamlmod.safe_dump(cell))
+    r = runner.invoke(cli, ["--run-path", rp, "--local", "apply", "-f",
+                            str(f)])
+    assert r.exit_code == 0, r.output
+    r = runner.invoke(cli, ["--run-path", rp, "--local", "get", "cell",
+                            "alpha-export", "-o", "yaml",
+                            "--api-version", "v1alpha1"])
+    assert r.exit_code == 0, r.output
+    docs = list(yamlmod.safe_load_all(
+        "\n".join(l for l in r.output.splitlines()
+                  if not l.startswith("warn:"))))
+    assert docs[0]["apiVersion"] == "v1alpha1"
+    assert docs[0]["spec"]["realm"] == "default"  # scope key renamed
+    assert "realmId" not in docs[0]["spec"]
+    assert "dropped in v1alpha1" in r.output  # repos is beta-only
