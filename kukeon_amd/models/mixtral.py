@@ -144,12 +144,12 @@ class MixtralMoE(nn.Module):
         """
         T = x.shape[0]
         logits = F.linear(x, self.router_w).float()            # [T, E]
-        probs = torch.softmax(logits, dim=-1)
-        topv, topi = probs.topk(self.K, dim=-1)                # [T, K]
-        topv = topv / topv.sum(dim=-1, keepdim=True)
-        wdense = torch.zeros(T, self.E, dtype=torch.float32,
+        # fused router: softmax top-K renorm scatter in ONE launch (the
+        # eager chain was ~5 kernels/layer, ~6% of the Mixtral decode
+        # run — profiles/r02_mixtral_stats.txt)
+        wdense = torch.empty(T, self.E, dtype=torch.float32,
                              device=x.device)
-        wdense.scatter_(1, topi, topv)                         # [T, E]
+        ops.moe_router_weights(wdense, logits, self.K)
         x_rep = x.unsqueeze(0).expand(self.E, T, x.shape[1])
         gu = torch.bmm(x_rep, self.gate_up_w.transpose(1, 2))  # [E,T,2I]
         act = torch.empty(self.E * T, self.inter, dtype=x.dtype,
@@ -157,8 +157,10 @@ class MixtralMoE(nn.Module):
         ops.silu_mul(act, gu.reshape(self.E * T, 2 * self.inter))
         y = torch.bmm(act.view(self.E, T, self.inter),
                       self.down_w.transpose(1, 2))             # [E,T,H]
-        out = (y.float() * wdense.t().unsqueeze(-1)).sum(dim=0)
-        return parallel.tp_all_reduce(out.to(x.dtype))
+        # fused combine: skips zero-weight experts' rows entirely
+        out = torch.empty(T, x.shape[1], dtype=x.dtype, device=x.device)
+        ops.moe_dense_combine(out, y.view(self.E, T, x.shape[1]), wdense)
+        return parallel.tp_all_reduce(out)
 
     def _forward_sparse(self, x: torch.Tensor) -> torch.Tensor:
         T = x.shape[0]

@@ -262,6 +262,14 @@ def mlp_down_fused(gu: torch.Tensor, w: torch.Tensor,
     return linear_add_rmsnorm(act, w, residual, norm_weight, eps)
 
 
+def moe_router_weights(wdense, logits, k: int) -> None:
+    _impl(logits).moe_router_weights(wdense, logits, k)
+
+
+def moe_dense_combine(out, y, wdense) -> None:
+    _impl(y).moe_dense_combine(out, y, wdense)
+
+
 def moe_gather_tokens(out, input, row_map) -> None:
     _impl(input).moe_gather_tokens(out, input, row_map)
 

@@ -58,6 +58,10 @@ void skinny_gemm_fused_norm(torch::Tensor normed, torch::Tensor x,
 void sample(torch::Tensor tokens, torch::Tensor logits, torch::Tensor temps,
             torch::Tensor top_k, torch::Tensor top_p, torch::Tensor seed,
             torch::Tensor workspace);
+void moe_router_weights(torch::Tensor wdense, torch::Tensor logits,
+                        long K);
+void moe_dense_combine(torch::Tensor out, torch::Tensor y,
+                       torch::Tensor wdense);
 void moe_gather_tokens(torch::Tensor out, torch::Tensor input,
                        torch::Tensor row_map);
 void moe_scatter_tokens(torch::Tensor out, torch::Tensor input,
@@ -107,6 +111,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_advance", &kukeon::decode_advance,
         "on-device decode cursor advance (self-advancing graph)");
   m.def("moe_gather_tokens", &kukeon::moe_gather_tokens, "MoE permute");
+  m.def("moe_router_weights", &kukeon::moe_router_weights,
+        "softmax top-K renorm -> dense [T,E] weights, one launch");
+  m.def("moe_dense_combine", &kukeon::moe_dense_combine,
+        "out = sum_e w[t,e]*y[e,t,:], zero-weight experts skipped");
   m.def("moe_scatter_tokens", &kukeon::moe_scatter_tokens,
         "MoE unpermute + weighted combine");
 }

@@ -233,3 +233,17 @@ def moe_scatter_tokens(out: torch.Tensor, input: torch.Tensor,
     acc = (input[inv_map.long()].float() *
            weights.unsqueeze(-1)).sum(dim=1)  # [T, H]
     out.copy_(acc.to(out.dtype))
+
+def moe_router_weights(wdense: torch.Tensor, logits: torch.Tensor,
+                       K: int) -> None:
+    probs = torch.softmax(logits.float(), dim=-1)
+    topv, topi = probs.topk(K, dim=-1)
+    topv = topv / topv.sum(dim=-1, keepdim=True)
+    wdense.zero_()
+    wdense.scatter_(1, topi, topv)
+
+
+def moe_dense_combine(out: torch.Tensor, y: torch.Tensor,
+                      wdense: torch.Tensor) -> None:
+    acc = (y.float() * wdense.t().unsqueeze(-1)).sum(dim=0)
+    out.copy_(acc.to(out.dtype))

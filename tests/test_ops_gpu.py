@@ -490,6 +490,42 @@ def test_skinny_gemm5_silu_fused_norm(M):
                                atol=5e-2)
 
 
+@pytest.mark.parametrize("T,E,K", [(32, 8, 2), (1, 8, 2), (7, 16, 4)])
+def test_moe_router_weights(T, E, K):
+    """Fused softmax/top-K/renorm/scatter router vs the eager chain."""
+    from kukeon_amd import _C
+    torch.manual_seed(23)
+    logits = torch.randn(T, E, dtype=torch.float32, device=DEV) * 2
+    w = torch.empty(T, E, dtype=torch.float32, device=DEV)
+    _C.moe_router_weights(w, logits, K)
+    probs = torch.softmax(logits, dim=-1)
+    topv, topi = probs.topk(K, dim=-1)
+    topv = topv / topv.sum(dim=-1, keepdim=True)
+    ref = torch.zeros(T, E, dtype=torch.float32, device=DEV)
+    ref.scatter_(1, topi, topv)
+    torch.testing.assert_close(w.cpu(), ref.cpu(), rtol=1e-4, atol=1e-5)
+    assert (w != 0).sum().item() == T * K
+
+
+def test_moe_dense_combine():
+    """out = sum_e w[t,e] * y[e,t,:] vs the eager fp32 chain."""
+    from kukeon_amd import _C
+    torch.manual_seed(29)
+    E, T, H = 8, 32, 4096
+    y = torch.randn(E, T, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.zeros(T, E, dtype=torch.float32, device=DEV)
+    idx = torch.stack([torch.randperm(E, device=DEV)[:2]
+                       for _ in range(T)])
+    vals = torch.rand(T, 2, device=DEV) + 0.1
+    vals = vals / vals.sum(-1, keepdim=True)
+    w.scatter_(1, idx, vals)
+    out = torch.empty(T, H, dtype=torch.bfloat16, device=DEV)
+    _C.moe_dense_combine(out, y, w)
+    ref = (y.float() * w.t().unsqueeze(-1)).sum(dim=0).to(torch.bfloat16)
+    torch.testing.assert_close(out.float().cpu(), ref.float().cpu(),
+                               rtol=2e-2, atol=2e-2)
+
+
 @pytest.mark.parametrize("kv_dtype", ["bf16", "fp8"])
 def test_decode_graphs_match_eager(kv_dtype):
     """Self-advancing graph decode must produce the same tokens as the
