@@ -143,8 +143,9 @@ class MixtralMoE(nn.Module):
         hipGraph-capturable (the sparse path's per-expert loop is neither).
         """
         T = x.shape[0]
-        logits = F.linear(x, self.router_w).float()            # [T, E]
-        # fused router: softmax top-K renorm scatter in ONE launch (the
+        logits = F.linear(x, self.router_w)                    # [T, E]
+        # fused router: softmax top-K renorm scatter in ONE launch,
+        # reading the bf16 logits directly — no .float() cast pass (the
         # eager chain was ~5 kernels/layer, ~6% of the Mixtral decode
         # run — profiles/r02_mixtral_stats.txt)
         wdense = torch.empty(T, self.E, dtype=torch.float32,

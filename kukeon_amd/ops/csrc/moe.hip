@@ -47,14 +47,17 @@ __global__ void moe_scatter_kernel(unsigned short* __restrict__ out,   // [T,H]
 // decode step, ~6% of the Mixtral run in launch+tensor overhead).
 // One thread per token row; E <= 32, K <= 4.
 __global__ void moe_router_kernel(float* __restrict__ wdense,   // [T,E]
-                                  const float* __restrict__ logits,
-                                  int T, int E, int K) {
+                                  const void* __restrict__ logits,
+                                  int T, int E, int K, bool bf16_in) {
   const int t = blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= T) return;
   float l[32];
   float mx = -1e30f;
   for (int e = 0; e < E; ++e) {
-    l[e] = logits[(long)t * E + e];
+    l[e] = bf16_in
+        ? us2f(reinterpret_cast<const unsigned short*>(
+              logits)[(long)t * E + e])
+        : reinterpret_cast<const float*>(logits)[(long)t * E + e];
     mx = fmaxf(mx, l[e]);
   }
   float denom = 0.f;
@@ -114,15 +117,16 @@ void moe_router_weights(torch::Tensor wdense, torch::Tensor logits,
                         long K) {
   const int T = logits.size(0);
   const int E = logits.size(1);
+  const bool bf16_in = logits.scalar_type() == torch::kBFloat16;
   TORCH_CHECK(E <= 32 && K <= 4 && K <= E);
-  TORCH_CHECK(logits.scalar_type() == torch::kFloat32 &&
-              wdense.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(bf16_in || logits.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(wdense.scalar_type() == torch::kFloat32);
   TORCH_CHECK(wdense.size(0) == T && wdense.size(1) == E);
   TORCH_CHECK(logits.is_contiguous() && wdense.is_contiguous());
   if (T == 0) return;
   auto stream = c10::hip::getCurrentHIPStream().stream();
   moe_router_kernel<<<dim3((unsigned)((T + 255) / 256)), 256, 0, stream>>>(
-      wdense.data_ptr<float>(), logits.data_ptr<float>(), T, E, (int)K);
+      wdense.data_ptr<float>(), logits.data_ptr(), T, E, (int)K, bf16_in);
   HIP_CHECK_KERNEL();
 }
 

@@ -236,6 +236,7 @@ def moe_scatter_tokens(out: torch.Tensor, input: torch.Tensor,
 
 def moe_router_weights(wdense: torch.Tensor, logits: torch.Tensor,
                        K: int) -> None:
+    # accepts bf16 or fp32 logits, like the HIP kernel
     probs = torch.softmax(logits.float(), dim=-1)
     topv, topi = probs.topk(K, dim=-1)
     topv = topv / topv.sum(dim=-1, keepdim=True)

@@ -498,6 +498,13 @@ def test_moe_router_weights(T, E, K):
     logits = torch.randn(T, E, dtype=torch.float32, device=DEV) * 2
     w = torch.empty(T, E, dtype=torch.float32, device=DEV)
     _C.moe_router_weights(w, logits, K)
+    # bf16 input path must agree with the fp32 path on bf16-rounded data
+    wb = torch.empty(T, E, dtype=torch.float32, device=DEV)
+    lb = logits.to(torch.bfloat16)
+    _C.moe_router_weights(wb, lb, K)
+    _C.moe_router_weights(w, lb.float(), K)
+    torch.testing.assert_close(wb.cpu(), w.cpu(), rtol=1e-4, atol=1e-5)
+    _C.moe_router_weights(w, logits, K)
     probs = torch.softmax(logits, dim=-1)
     topv, topi = probs.topk(K, dim=-1)
     topv = topv / topv.sum(dim=-1, keepdim=True)
