@@ -702,7 +702,7 @@ DEV_INLINE u32x4_t silu_mul8(u32x4_t g, u32x4_t u) {
   return r;
 }
 
-template <int MT, bool SPLIT, int KS, bool SILU = false>
+template <int MT, bool SPLIT, int KS, bool SILU = false, int TILES = 2>
 __global__ __launch_bounds__(256) void skinny5_kernel(
     unsigned short* __restrict__ out, float* __restrict__ ws,
     const unsigned short* __restrict__ x,
@@ -715,7 +715,7 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
   // 2 blocks/CU (2 waves/SIMD hide the consume's dependency stalls).
   constexpr int NI = KS / 32;
   __shared__ __align__(16) unsigned short xbuf[2][64 * KS];
-  __shared__ __align__(16) unsigned short wimg_all[4][2][16 * KS];
+  __shared__ __align__(16) unsigned short wimg_all[4][TILES][16 * KS];
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int row16 = lane & 15;
@@ -730,8 +730,8 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
   const long ks0 = (long)blockIdx.y * KS;
   if (ks0 >= K) return;
   u32x4_t w0[NI], w1[NI], xr8[NI];
-  const int n0t0 = (blockIdx.x * 2 + 0) * 64 + wid * 16;
-  const int n0t1 = (blockIdx.x * 2 + 1) * 64 + wid * 16;
+  const int n0t0 = (blockIdx.x * TILES + 0) * 64 + wid * 16;
+  const int n0t1 = (blockIdx.x * TILES + (TILES - 1)) * 64 + wid * 16;
   const int rlo = lane >> 3, seg = (lane & 7) * 8;  // elems
   const unsigned short* p0l = w + (long)(n0t0 + rlo) * K + ks0 + seg;
   const unsigned short* p0h = w + (long)(n0t0 + 8 + rlo) * K + ks0 + seg;
@@ -752,11 +752,11 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
   const unsigned short* puh = SILU ? pxh + K : nullptr;
   u32x4_t ur8[SILU ? NI : 1];
   unsigned short* img0 = wimg_all[wid][0];
-  unsigned short* img1 = wimg_all[wid][1];
+  unsigned short* img1 = wimg_all[wid][TILES - 1];
 
   // prologue: land slice 0, build the images, put slice 1 in flight
   load_tile<NI, true>(w0, p0l, p0h);
-  load_tile<NI, true>(w1, p1l, p1h);
+  if constexpr (TILES == 2) load_tile<NI, true>(w1, p1l, p1h);
   load_tile<NI, false>(xr8, pxl, pxh);
   if constexpr (SILU) {
     load_tile<NI, false>(ur8, pul, puh);
@@ -764,7 +764,7 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
     for (int i = 0; i < NI; ++i) xr8[i] = silu_mul8(xr8[i], ur8[i]);
   }
   wimg_write<KS>(img0, w0, lane);
-  wimg_write<KS>(img1, w1, lane);
+  if constexpr (TILES == 2) wimg_write<KS>(img1, w1, lane);
   ximg_write<KS>(xbuf[0], xr8, wid, lane);
   {
     const long ks1 = ks0 + kadv;
@@ -774,7 +774,7 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
       if constexpr (SILU) { pul += kadv; puh += kadv; }
     }
     load_tile<NI, true>(w0, p0l, p0h);
-    load_tile<NI, true>(w1, p1l, p1h);
+    if constexpr (TILES == 2) load_tile<NI, true>(w1, p1l, p1h);
     load_tile<NI, false>(xr8, pxl, pxh);
     if constexpr (SILU) load_tile<NI, false>(ur8, pul, puh);
   }
@@ -790,10 +790,12 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
     wimg_write<KS>(img0, w0, lane);   // first USE of w0 -> counted wait
     if (adv) { p0l += kadv; p0h += kadv; }
     load_tile<NI, true>(w0, p0l, p0h);                  // slice s+2
-    consume_img<MT, KS>(img1, xb, acc1, row16, kgrp);
-    wimg_write<KS>(img1, w1, lane);
-    if (adv) { p1l += kadv; p1h += kadv; }
-    load_tile<NI, true>(w1, p1l, p1h);
+    if constexpr (TILES == 2) {
+      consume_img<MT, KS>(img1, xb, acc1, row16, kgrp);
+      wimg_write<KS>(img1, w1, lane);
+      if (adv) { p1l += kadv; p1h += kadv; }
+      load_tile<NI, true>(w1, p1l, p1h);
+    }
     if constexpr (SILU) {
 #pragma unroll
       for (int i = 0; i < NI; ++i) xr8[i] = silu_mul8(xr8[i], ur8[i]);
@@ -821,10 +823,10 @@ __global__ __launch_bounds__(256) void skinny5_kernel(
       const long off1 = (long)mrow * N + ncol1 + r;
       if (SPLIT) {
         slab[off0] = acc0[m][r];
-        slab[off1] = acc1[m][r];
+        if constexpr (TILES == 2) slab[off1] = acc1[m][r];
       } else {
         out[off0] = f2us(acc0[m][r]);
-        out[off1] = f2us(acc1[m][r]);
+        if constexpr (TILES == 2) out[off1] = f2us(acc1[m][r]);
       }
     }
   }
@@ -843,14 +845,20 @@ void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  const int ngroups = N / 128;
+  // TILES=1 (KUKEON_SK5_NSPLIT=1): 64 N-cols per block -> double the
+  // grid at the SAME split-K slab traffic, so two blocks co-reside per
+  // CU and cover each other's slice-barrier bubbles
+  const char* nv = getenv("KUKEON_SK5_NSPLIT");
+  const int tiles = (nv && atoi(nv) == 1) ? 1 : 2;
+  const int ngroups = N / (tiles * 64);
   const int nslices = (int)(K / KS);
-  // ~256 blocks: the sweep's knee for every shape — beyond it the
-  // extra split-K slab traffic outweighs occupancy (down: sk8/256
+  // ~256 blocks (TILES=2): the sweep's knee for every shape — beyond it
+  // the extra split-K slab traffic outweighs occupancy (down: sk8/256
   // blocks 32.2us vs sk16/512 blocks 40.6us cold)
+  const int target = tiles == 1 ? 512 : 256;
   int splitk = 1;
-  if (ngroups < 256)
-    splitk = min(nslices, (256 + ngroups - 1) / ngroups);
+  if (ngroups < target)
+    splitk = min(nslices, (target + ngroups - 1) / ngroups);
   if (const char* ov = getenv("KUKEON_SK5_SPLITK")) {
     const int v = atoi(ov);
     if (v > 0) splitk = min(nslices, v);
@@ -862,14 +870,22 @@ void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   auto* wp = reinterpret_cast<const unsigned short*>(w.data_ptr());
   const long total = (long)M * N;
 #define SK5_LAUNCH_KS(MT_, KS_)                                              \
-  if (splitk == 1) {                                                         \
+  if (splitk == 1 && tiles == 2) {                                           \
     skinny5_kernel<MT_, false, KS_><<<grid, 256, 0, stream>>>(               \
+        op, nullptr, xp, wp, M, N, K);                                       \
+  } else if (splitk == 1) {                                                  \
+    skinny5_kernel<MT_, false, KS_, false, 1><<<grid, 256, 0, stream>>>(     \
         op, nullptr, xp, wp, M, N, K);                                       \
   } else {                                                                   \
     float* wsp = ws.data_ptr<float>();                                       \
     TORCH_CHECK(ws.numel() >= total * splitk, "sk5 workspace too small");    \
-    skinny5_kernel<MT_, true, KS_><<<grid, 256, 0, stream>>>(                \
-        nullptr, wsp, xp, wp, M, N, K);                                      \
+    if (tiles == 2) {                                                        \
+      skinny5_kernel<MT_, true, KS_><<<grid, 256, 0, stream>>>(              \
+          nullptr, wsp, xp, wp, M, N, K);                                    \
+    } else {                                                                 \
+      skinny5_kernel<MT_, true, KS_, false, 1><<<grid, 256, 0, stream>>>(    \
+          nullptr, wsp, xp, wp, M, N, K);                                    \
+    }                                                                        \
     skinny_reduce_kernel<<<dim3((unsigned)((total / 8 + 255) / 256)), 256,   \
                            0, stream>>>(op, wsp, total, splitk);             \
   }
