@@ -845,20 +845,21 @@ void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  // TILES=1 (KUKEON_SK5_NSPLIT=1): 64 N-cols per block -> double the
-  // grid at the SAME split-K slab traffic, so two blocks co-reside per
-  // CU and cover each other's slice-barrier bubbles
+  // TILES=1 (default): 64 N-cols per block at the same ~256-block
+  // split-K budget -> HALF the slab traffic per block count, measured
+  // 3-4% faster on every dispatched shape (down 31.7 vs 32.6 us,
+  // 70b-down 80.2 vs 83.9 — sweep_nsplit.py; the 512-block
+  // co-residency variant was NOT the win, smaller blocks at equal
+  // splitk were). KUKEON_SK5_NSPLIT=2 restores the two-tile geometry.
   const char* nv = getenv("KUKEON_SK5_NSPLIT");
-  const int tiles = (nv && atoi(nv) == 1) ? 1 : 2;
+  const int tiles = (nv && atoi(nv) == 2) ? 2 : 1;
   const int ngroups = N / (tiles * 64);
   const int nslices = (int)(K / KS);
-  // ~256 blocks (TILES=2): the sweep's knee for every shape — beyond it
-  // the extra split-K slab traffic outweighs occupancy (down: sk8/256
-  // blocks 32.2us vs sk16/512 blocks 40.6us cold)
-  const int target = tiles == 1 ? 512 : 256;
+  // ~256 blocks: the sweep's knee for every shape — beyond it the
+  // extra split-K slab traffic outweighs occupancy
   int splitk = 1;
-  if (ngroups < target)
-    splitk = min(nslices, (target + ngroups - 1) / ngroups);
+  if (ngroups < 256)
+    splitk = min(nslices, (256 + ngroups - 1) / ngroups);
   if (const char* ov = getenv("KUKEON_SK5_SPLITK")) {
     const int v = atoi(ov);
     if (v > 0) splitk = min(nslices, v);
@@ -1158,7 +1159,9 @@ void skinny_gemm5_fused_norm(torch::Tensor normed, torch::Tensor x,
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous() &&
               residual.is_contiguous() && normed.is_contiguous());
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  const int ngroups = N / 128;
+  const char* nv = getenv("KUKEON_SK5_NSPLIT");
+  const int tiles = (nv && atoi(nv) == 2) ? 2 : 1;   // see skinny_gemm5
+  const int ngroups = N / (tiles * 64);
   const int nslices = (int)(K / KS);
   int splitk = min(nslices, (256 + ngroups - 1) / ngroups);
   if (const char* ov = getenv("KUKEON_SK5_SPLITK")) {
@@ -1173,8 +1176,13 @@ void skinny_gemm5_fused_norm(torch::Tensor normed, torch::Tensor x,
   float* wsp = ws.data_ptr<float>();
   TORCH_CHECK(ws.numel() >= total * splitk, "workspace too small");
 #define SK5F_LAUNCH(MT_)                                                    \
-  skinny5_kernel<MT_, true, KS><<<grid, 256, 0, stream>>>(                  \
-      nullptr, wsp, xp, wp, M, N, K)
+  if (tiles == 2) {                                                         \
+    skinny5_kernel<MT_, true, KS><<<grid, 256, 0, stream>>>(                \
+        nullptr, wsp, xp, wp, M, N, K);                                     \
+  } else {                                                                  \
+    skinny5_kernel<MT_, true, KS, false, 1><<<grid, 256, 0, stream>>>(      \
+        nullptr, wsp, xp, wp, M, N, K);                                     \
+  }
   switch (MT) {
     case 1: SK5F_LAUNCH(1); break;
     case 2: SK5F_LAUNCH(2); break;
@@ -1372,7 +1380,9 @@ void skinny_gemm5_silu_fused_norm(torch::Tensor normed, torch::Tensor gu,
               residual.is_contiguous() && normed.is_contiguous());
   TORCH_CHECK(gu.scalar_type() == torch::kBFloat16);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  const int ngroups = N / 128;
+  const char* nv = getenv("KUKEON_SK5_NSPLIT");
+  const int tiles = (nv && atoi(nv) == 2) ? 2 : 1;   // see skinny_gemm5
+  const int ngroups = N / (tiles * 64);
   const int nslices = (int)(K / KS);
   int splitk = min(nslices, (256 + ngroups - 1) / ngroups);
   if (const char* ov = getenv("KUKEON_SK5_SPLITK")) {
@@ -1387,8 +1397,13 @@ void skinny_gemm5_silu_fused_norm(torch::Tensor normed, torch::Tensor gu,
   float* wsp = ws.data_ptr<float>();
   TORCH_CHECK(ws.numel() >= total * splitk, "workspace too small");
 #define SK5S_LAUNCH(MT_)                                                    \
-  skinny5_kernel<MT_, true, KS, true><<<grid, 256, 0, stream>>>(            \
-      nullptr, wsp, xp, wp, M, N, K)
+  if (tiles == 2) {                                                         \
+    skinny5_kernel<MT_, true, KS, true><<<grid, 256, 0, stream>>>(          \
+        nullptr, wsp, xp, wp, M, N, K);                                     \
+  } else {                                                                  \
+    skinny5_kernel<MT_, true, KS, true, 1><<<grid, 256, 0, stream>>>(       \
+        nullptr, wsp, xp, wp, M, N, K);                                     \
+  }
   switch (MT) {
     case 1: SK5S_LAUNCH(1); break;
     case 2: SK5S_LAUNCH(2); break;
