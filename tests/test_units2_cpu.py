@@ -230,3 +230,41 @@ def test_scope_locks_refcount_and_exclusion():
         with locks.hold(key):
             assert locks.held_count() == 1
     assert locks.held_count() == 0
+
+
+# ---------------------------------------------------------------------------
+# scheme round-trips
+# ---------------------------------------------------------------------------
+def test_scheme_roundtrip_and_idempotency():
+    from kukeon_amd.api import scheme
+
+    beta = {
+        "apiVersion": "v1beta1", "kind": "Cell",
+        "metadata": {"name": "rt"},
+        "spec": {"realmId": "r1", "spaceId": "s1", "stackId": "st1",
+                 "autoDelete": True,
+                 "containers": [{"id": "main", "image": "img",
+                                 "command": "run", "args": ["-v"],
+                                 "restartPolicy": "on-failure"}]},
+    }
+    # normalize is idempotent
+    n1 = scheme.normalize(beta)
+    assert scheme.normalize(n1) == n1
+    # beta -> alpha -> beta preserves everything non-lossy
+    wire, lost = scheme.to_wire(dict(n1), "v1alpha1")
+    assert lost == []
+    assert wire["spec"]["realm"] == "r1" and wire["spec"]["autoRemove"]
+    assert wire["spec"]["containers"][0]["restartPolicy"] == "onFailure"
+    back = scheme.normalize(wire)
+    assert back["spec"]["realmId"] == "r1"
+    assert back["spec"]["autoDelete"] is True
+    assert back["spec"]["containers"][0]["restartPolicy"] == "on-failure"
+    # alpha flat shorthand up-converts to a containers list with defaults
+    alpha = {"apiVersion": "v1alpha1", "kind": "Cell",
+             "metadata": {"name": "flat"},
+             "spec": {"realm": "r1", "image": "busybox",
+                      "command": "sh"}}
+    n = scheme.normalize(alpha)
+    assert n["spec"]["containers"][0]["image"] == "busybox"
+    assert n["spec"]["containers"][0]["id"] == "main"
+    assert n["spec"]["spaceId"] == "default"  # cross-version defaulting
