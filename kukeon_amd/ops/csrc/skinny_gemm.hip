@@ -837,10 +837,12 @@ void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   const int M = x.size(0);
   const long K = x.size(1);
   const int N = w.size(0);
-  int KS = 128;  // LDS per block: 24 KiB (KS=64) / 48 (128) / 96 (256)
+  int KS = 128;
+  // KS < 128 is unsound: the XOR swizzle offsets (row&15)<<4 span 256
+  // bytes, a full KS=128 row — at KS=64 they cross rows (measured:
+  // garbage output, sweep_ks.py). 128 and 256 are the valid geometries.
   if (const char* ov = getenv("KUKEON_SK5_KS")) {
-    const int v = atoi(ov);
-    KS = (v == 256 || v == 64) ? v : 128;
+    KS = atoi(ov) == 256 ? 256 : 128;
   }
   TORCH_CHECK(M >= 1 && M <= 64 && N % 128 == 0 && K % KS == 0);
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && out.is_contiguous());
@@ -892,9 +894,7 @@ void skinny_gemm5(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                            0, stream>>>(op, wsp, total, splitk);             \
   }
 #define SK5_LAUNCH(MT_)                                                      \
-  if (KS == 128) { SK5_LAUNCH_KS(MT_, 128) }                                 \
-  else if (KS == 64) { SK5_LAUNCH_KS(MT_, 64) }                              \
-  else { SK5_LAUNCH_KS(MT_, 256) }
+  if (KS == 128) { SK5_LAUNCH_KS(MT_, 128) } else { SK5_LAUNCH_KS(MT_, 256) }
   switch (MT) {
     case 1: SK5_LAUNCH(1); break;
     case 2: SK5_LAUNCH(2); break;
