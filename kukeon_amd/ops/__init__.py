@@ -101,6 +101,8 @@ _FUSE_SILU = __import__("os").environ.get("KUKEON_FUSE_SILU", "0") == "1"
 
 
 def _skinny_wins(rows: int, N: int, K: int) -> bool:
+    if _USE_SKINNY == "off":   # A/B: force the library everywhere
+        return False
     return N == K == 4096
 
 
@@ -110,6 +112,8 @@ def _skinny_wins(rows: int, N: int, K: int) -> bool:
 # lm_head stay on the library. KUKEON_SKINNY_GEMM=5 forces it everywhere
 # for benchmarking.
 def _skinny5_wins(rows: int, N: int, K: int) -> bool:
+    if _USE_SKINNY == "off":   # A/B: force the library everywhere
+        return False
     # llama-3-8b down (32.2us vs blas 39.6 cold) and llama-3-70b down
     # (79.3 vs 83.1 at M=16, 92.6 vs 139.3 at M=64) — measured on
     # MI355X, profiles/r02_progress.md
