@@ -19,6 +19,7 @@ interleave across clients — the engine continuously batches them):
 from __future__ import annotations
 
 import argparse
+import collections
 import json
 import logging
 import os
@@ -26,6 +27,7 @@ import queue
 import socket
 import socketserver
 import threading
+import time
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 
@@ -43,6 +45,7 @@ class _Pending:
     reply: "queue.Queue"
     tokens: List[int]
     stream: bool = False
+    t0: float = 0.0
 
 
 class ModelhubServer:
@@ -59,6 +62,9 @@ class ModelhubServer:
         self._threads = []
         self.metrics = {"tokens_generated": 0, "turns_completed": 0,
                         "engine_steps": 0, "busy_seconds": 0.0}
+        # last-N turn latencies (seconds) for the stats percentiles —
+        # the SURVEY aux-subsystem requirement (turn/latency metrics)
+        self._turn_latency = collections.deque(maxlen=512)
 
     # ---- engine loop --------------------------------------------------
     def _engine_loop(self):
@@ -148,6 +154,9 @@ class ModelhubServer:
                         p.reply.put({"delta": list(o.new_tokens)})
                     if o.finished:
                         self.metrics["turns_completed"] += 1
+                        if p.t0:
+                            self._turn_latency.append(
+                                _time.perf_counter() - p.t0)
                         del self._pending[o.req_id]
                         if p.stream:
                             p.reply.put({"delta": list(o.new_tokens),
@@ -164,12 +173,23 @@ class ModelhubServer:
             return {"ok": True, "pid": os.getpid()}
         if method == "stats":
             kv = self.engine.kv
+            lat = sorted(self._turn_latency)
+
+            def pct(q: float) -> float:
+                if not lat:
+                    return 0.0
+                return round(lat[min(len(lat) - 1,
+                                     int(q * len(lat)))] * 1000, 1)
+
             return {
                 "sessions": len(self.sessions),
                 "running": self.engine.num_running,
                 "waiting": len(self.engine.waiting),
                 "kv_blocks_total": kv.num_blocks,
                 "kv_blocks_free": kv.allocator.num_free,
+                "turn_latency_ms": {"p50": pct(0.50), "p95": pct(0.95),
+                                    "p99": pct(0.99),
+                                    "n": len(lat)},
                 **self.metrics,
             }
         if method == "release":
@@ -224,7 +244,8 @@ class ModelhubServer:
                 reply.put({"error": str(e)})
                 return
             with self._lock:
-                self._pending[rid] = _Pending(rid, reply, [], stream=stream)
+                self._pending[rid] = _Pending(rid, reply, [], stream=stream,
+                                              t0=time.perf_counter())
         self._submit.put(_do_submit)
         while True:
             out = reply.get(timeout=600)

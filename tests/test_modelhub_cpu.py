@@ -260,3 +260,17 @@ def test_engine_step_error_fails_pending_and_keeps_serving(hub):
     r = c.generate("alive", [4, 5, 6], max_new_tokens=2, temperature=0.0)
     assert len(r["tokens"]) == 2
     c.close()
+
+
+def test_stats_turn_latency_percentiles(hub):
+    """The stats verb reports p50/p95/p99 turn latency over the last
+    N completed turns (SURVEY aux: turn/latency metrics)."""
+    h, sock = hub
+    c = ModelhubClient(sock, timeout=120)
+    c.generate("lat-s1", [5, 6, 7], max_new_tokens=4)
+    c.generate("lat-s1", [8], max_new_tokens=4)
+    lat = c.call("stats")["turn_latency_ms"]
+    assert lat["n"] >= 2
+    assert lat["p50"] > 0
+    assert lat["p50"] <= lat["p95"] <= lat["p99"]
+    c.close()
