@@ -250,8 +250,10 @@ class Builder:
         (work / "up").mkdir()
         (work / "ovl").mkdir()
         (work / "w").mkdir()
+        (work / "lo0").mkdir()  # FROM scratch + RUN-first: overlayfs
+        # refuses upperdir as its own lowerdir, so give it an empty lower
         lowers = ":".join(str(self.store.layer_root(l))
-                          for l in reversed(layers)) or str(work / "up")
+                          for l in reversed(layers)) or str(work / "lo0")
         rc_r, rc_w = os.pipe()
         pid = os.fork()
         if pid == 0:

@@ -248,15 +248,42 @@ def team_init(controller, team_file: str, realm: str = "default",
 
     built: List[str] = []
     if build_images and catalog:
+        from kukeon_amd.images import (Builder, BuildError, ImageStore,
+                                       overlay_supported)
+        can_build = overlay_supported()
+        store = ImageStore(str(controller.run_path)) if can_build else None
         for ref in build_order(catalog):
             entry = next(e for e in catalog.images if e.ref == ref)
-            controller.register_image(
-                name=f"kukeon.internal/{ref}",
-                spec={"harness": entry.harness,
-                      "capabilities": entry.capabilities,
-                      "buildContext": entry.build_context,
-                      "dockerfile": entry.build_dockerfile,
-                      "source": str(source_dir)})
+            tag = f"kukeon.internal/{ref}"
+            kf = (source_dir / entry.build_dockerfile
+                  if entry.build_dockerfile else None)
+            ctx = (source_dir / entry.build_context
+                   if entry.build_context else source_dir)
+            if can_build and kf is not None and kf.is_file():
+                # real layered build (kukebuild analog): catalog entries
+                # with a Kukefile become rootfs images cells can pivot
+                # into; FROM kukeon.internal/<base> layers on the parent
+                # built earlier in the topo order. The built manifest IS
+                # the registration (same catalog path) — registering on
+                # top would clobber its layer list.
+                try:
+                    man = Builder(store).build(ctx, kf.read_text(), tag)
+                except BuildError as e:
+                    raise errors.ValidationError(
+                        f"catalog image {ref!r} build failed: {e}")
+                man["labels"].update({"harness": entry.harness,
+                                      "capabilities": entry.capabilities,
+                                      "team-source": str(source_dir)})
+                store.put_manifest(tag, man["layers"], man["config"],
+                                   man["labels"])
+            else:
+                controller.register_image(
+                    name=tag,
+                    spec={"harness": entry.harness,
+                          "capabilities": entry.capabilities,
+                          "buildContext": entry.build_context,
+                          "dockerfile": entry.build_dockerfile,
+                          "source": str(source_dir)})
             built.append(ref)
 
     rendered = render_team(team, roles, harnesses, catalog, source_dir,

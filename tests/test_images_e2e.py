@@ -186,3 +186,93 @@ def test_image_cmd_and_env_defaults(tmp_path):
     time.sleep(0.1)
     assert (cdir / "cmd.txt").read_text().strip() == "hello-img"
     ctl.delete_cell("default", "default", "default", name, force=True)
+
+
+@pytest.mark.skipif(not HAVE_OVL, reason="no overlayfs/mount ns")
+def test_team_catalog_kukefile_builds_real_image(tmp_path):
+    """A team ImageCatalog entry carrying a Kukefile produces a real
+    layered image at kukeon.internal/<ref> (teambuild -> kukebuild
+    analog), FROM-ordered so children layer on parents."""
+    from kukeon_amd.teams.pipeline import team_init
+    from kukeon_amd.runtime.process import FakeRuntime
+
+    # agents source tree with a catalog + trivial role/harness
+    src = tmp_path / "agents"
+    (src / "harnesses").mkdir(parents=True)
+    (src / "img").mkdir()
+    make_shell_context(src / "img")
+    (src / "img" / "Kukefile.base").write_text(
+        "FROM scratch\nCOPY shellfs /\n"
+        "RUN echo base > /base.txt\n")
+    (src / "img" / "Kukefile.child").write_text(
+        "FROM kukeon.internal/base\n"
+        "RUN echo child > /child.txt\n")
+    (src / "harnesses" / "cat.yaml").write_text("""
+apiVersion: kuketeams.io/v1
+kind: ImageCatalog
+metadata: {name: cat}
+spec:
+  images:
+    - ref: base
+      harness: h1
+      build: {context: img, dockerfile: img/Kukefile.base}
+    - ref: child
+      harness: h1
+      base: base
+      build: {context: img, dockerfile: img/Kukefile.child}
+""")
+    (src / "harnesses" / "h1.yaml").write_text("""
+apiVersion: kuketeams.io/v1
+kind: Harness
+metadata: {name: h1}
+spec:
+  template: harnesses/h1.tmpl.yaml
+""")
+    (src / "harnesses" / "h1.tmpl.yaml").write_text("""
+kind: CellBlueprint
+metadata: {name: ${TEAM}-${ROLE}-${HARNESS}}
+spec:
+  namePrefix: ${ROLE}
+  template:
+    spec:
+      realmId: default
+      spaceId: default
+      stackId: default
+      containers:
+        - id: agent
+          image: "${IMAGE}"
+          command: sleep
+          args: ["5"]
+""")
+    (src / "roles").mkdir()
+    (src / "roles" / "dev.yaml").write_text("""
+apiVersion: kuketeams.io/v1
+kind: Role
+metadata: {name: dev}
+spec: {description: dev agent}
+""")
+    team_file = tmp_path / "kuketeam.yaml"
+    team_file.write_text(f"""
+apiVersion: kuketeams.io/v1
+kind: ProjectTeam
+metadata: {{name: t1}}
+spec:
+  source: {{path: {src}}}
+  defaults: {{harnesses: [h1]}}
+  roles:
+    - ref: dev
+""")
+    run = tmp_path / "run"
+    ctl = Controller(str(run), runtime=FakeRuntime())
+    ctl.bootstrap()
+    res = team_init(ctl, str(team_file), teams_root=str(tmp_path / "th"))
+    assert res["built"] == ["base", "child"]
+    store = ImageStore(str(run))
+    base = store.get("kukeon.internal/base")
+    child = store.get("kukeon.internal/child")
+    assert (store.layer_root(base["layers"][-1]) /
+            "base.txt").read_text().strip() == "base"
+    # child layers on base: parent layers first, child's delta on top
+    assert child["layers"][:len(base["layers"])] == base["layers"]
+    assert (store.layer_root(child["layers"][-1]) /
+            "child.txt").read_text().strip() == "child"
