@@ -106,3 +106,7 @@ class SamplingParams:
     top_k: int = 50
     top_p: float = 0.9
     max_new_tokens: int = 128
+    # generation ends when one of these ids is sampled (the id itself is
+    # the turn's last token); agent protocols stop at eos/tool markers
+    # long before max_new_tokens
+    stop_token_ids: tuple = ()

@@ -405,7 +405,8 @@ class LLMEngine:
 
     def _append_token(self, req: Request, tok: int) -> StepOutput:
         req.output_tokens.append(tok)
-        if len(req.output_tokens) >= req.sampling.max_new_tokens:
+        if (len(req.output_tokens) >= req.sampling.max_new_tokens
+                or tok in (req.sampling.stop_token_ids or ())):
             req.finished = True
             req.kv.pending_token = tok
             self._release_row(req)
@@ -548,9 +549,23 @@ class LLMEngine:
             r.kv.history.extend([prev] + toks[:-1])
             r.kv.num_tokens += k
             fin = False
-            for t in toks:
+            emitted = []
+            for i, t in enumerate(toks):
+                emitted.append(t)
                 fin = self._append_token(r, t).finished
-            outs.append(StepOutput(r.req_id, toks, fin))
+                if fin and i + 1 < k:
+                    # stop token mid-microbatch: the device already
+                    # advanced KV/history for the discarded tail — roll
+                    # the host counters back so the next turn's context
+                    # ends at the stop token (staging re-derives
+                    # pos/seq_lens from kv.num_tokens)
+                    extra = k - (i + 1)
+                    r.kv.num_tokens -= extra
+                    del r.kv.history[len(r.kv.history) - extra:]
+                    break
+                if fin:
+                    break
+            outs.append(StepOutput(r.req_id, emitted, fin))
         return outs
 
     def _decode_forward(self, B: int) -> None:

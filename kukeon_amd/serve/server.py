@@ -230,7 +230,8 @@ class ModelhubServer:
             temperature=float(params.get("temperature", 0.7)),
             top_k=int(params.get("top_k", 50)),
             top_p=float(params.get("top_p", 0.9)),
-            max_new_tokens=int(params.get("max_new_tokens", 128)))
+            max_new_tokens=int(params.get("max_new_tokens", 128)),
+            stop_token_ids=tuple(params.get("stop_token_ids", ()) or ()))
         reply: "queue.Queue" = queue.Queue()
 
         def _do_submit():

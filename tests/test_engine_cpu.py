@@ -426,3 +426,52 @@ def test_prefill_soft_tail_admission():
                      SamplingParams(max_new_tokens=2))
     batch2, _ = eng2._admit_prefill()
     assert [c for _, c in batch2] == [64]
+
+
+def test_stop_token_ends_generation_and_rolls_back_context():
+    """A sampled stop token finishes the turn mid-microbatch; the
+    discarded tail's KV/history advances are rolled back so the next
+    turn's context ends exactly at the stop token."""
+    import torch
+    from kukeon_amd.engine.config import SamplingParams
+    from kukeon_amd.engine.kv_cache import SequenceKV
+
+    torch.manual_seed(0)
+    engine, cfg, ecfg = make_engine()
+    prompt = [7, 3, 99, 140, 11, 42]
+    sp = SamplingParams(temperature=0.0, max_new_tokens=8)
+    kv = SequenceKV(ecfg.block_size)
+    engine.add_request(kv, prompt, sp)
+    ref = []
+    while engine.has_work():
+        for o in engine.step():
+            ref.extend(o.new_tokens)
+    assert len(ref) == 8
+    # invariant: context tokens = prompt + emitted - 1 (the last emitted
+    # token is pending); history mirrors the context for evict-recompute
+    assert kv.num_tokens == len(prompt) + len(ref) - 1
+    assert len(kv.history) == kv.num_tokens
+    stop = ref[0]  # greedy is deterministic: this WILL be sampled first
+
+    torch.manual_seed(0)
+    engine2, _, _ = make_engine()
+    kv2 = SequenceKV(ecfg.block_size)
+    sp2 = SamplingParams(temperature=0.0, max_new_tokens=8,
+                         stop_token_ids=(stop,))
+    engine2.add_request(kv2, prompt, sp2)
+    got = []
+    while engine2.has_work():
+        for o in engine2.step():
+            got.extend(o.new_tokens)
+    assert got == ref[:1]           # truncated at the stop token
+    assert kv2.pending_token == stop
+    # the microbatch tail past the stop was rolled back: same invariant
+    assert kv2.num_tokens == len(prompt) + len(got) - 1
+    assert len(kv2.history) == kv2.num_tokens
+    # the session keeps working after a stop-token turn
+    engine2.add_request(kv2, [5], sp2)
+    more = []
+    while engine2.has_work():
+        for o in engine2.step():
+            more.extend(o.new_tokens)
+    assert 1 <= len(more) <= 8
