@@ -695,6 +695,61 @@ def test_engine_gpu_preemption_recompute():
         assert len(kv.history) + kv.num_tokens >= expect[i] - 1
 
 
+def test_engine_gpu_stop_token_rollback():
+    """Stop-token finish mid-microbatch on the graphed HIP decode path:
+    the device ring already advanced past the stop — the host rollback
+    must leave the context ending at the stop token, and the session
+    must keep serving."""
+    from kukeon_amd.engine.config import (EngineConfig, SamplingParams,
+                                          tiny_llama)
+    from kukeon_amd.engine.engine import LLMEngine
+    from kukeon_amd.engine.kv_cache import SequenceKV
+    from kukeon_amd.models.llama import LlamaModel
+
+    torch.manual_seed(0)
+    cfg = tiny_llama()
+    ecfg = EngineConfig(max_model_len=512, max_sessions=2,
+                        num_kv_blocks=256, use_graphs=True,
+                        decode_microbatch=8, graph_buckets=(2,))
+    model = LlamaModel(cfg, device=DEV)
+    engine = LLMEngine(model, cfg, ecfg, device=DEV)
+    prompt = [7, 3, 99, 140, 11, 42]
+    kv = SequenceKV(ecfg.block_size)
+    engine.add_request(kv, prompt,
+                       SamplingParams(temperature=0.0, max_new_tokens=16))
+    ref = []
+    while engine.has_work():
+        for o in engine.step():
+            ref.extend(o.new_tokens)
+    assert len(ref) == 16
+    stop = ref[2]
+    cut = ref.index(stop)  # first occurrence ends the stopped run
+
+    torch.manual_seed(0)
+    model2 = LlamaModel(cfg, device=DEV)
+    engine2 = LLMEngine(model2, cfg, ecfg, device=DEV)
+    kv2 = SequenceKV(ecfg.block_size)
+    engine2.add_request(kv2, prompt,
+                        SamplingParams(temperature=0.0, max_new_tokens=16,
+                                       stop_token_ids=(stop,)))
+    got = []
+    while engine2.has_work():
+        for o in engine2.step():
+            got.extend(o.new_tokens)
+    assert got == ref[:cut + 1]
+    assert kv2.pending_token == stop
+    assert kv2.num_tokens == len(prompt) + len(got) - 1
+    assert len(kv2.history) == kv2.num_tokens
+    # session continues cleanly after the rollback
+    engine2.add_request(kv2, [5],
+                        SamplingParams(temperature=0.0, max_new_tokens=4))
+    more = []
+    while engine2.has_work():
+        for o in engine2.step():
+            more.extend(o.new_tokens)
+    assert len(more) == 4
+
+
 def test_mfma_probe16k_layout():
     from kukeon_amd import _C
     torch.manual_seed(0)
