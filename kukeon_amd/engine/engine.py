@@ -403,6 +403,28 @@ class LLMEngine:
             req.row = -1
             self.num_running -= 1
 
+    def cancel(self, rid: int) -> bool:
+        """Abort a request at the next step boundary (client went away /
+        explicit cancel verb). Must run on the engine thread (the server
+        routes it through the submit queue). The session's KV keeps the
+        already-computed prefix — the turn just ends early, exactly like
+        a stop token at the last emitted position."""
+        for i, r in enumerate(self.waiting):
+            if r.req_id == rid:
+                self.waiting.pop(i)
+                r.finished = True
+                if r.output_tokens:
+                    r.kv.pending_token = r.output_tokens[-1]
+                return True
+        for r in self._rows:
+            if r is not None and r.req_id == rid:
+                r.finished = True
+                if r.output_tokens:
+                    r.kv.pending_token = r.output_tokens[-1]
+                self._release_row(r)
+                return True
+        return False
+
     def _append_token(self, req: Request, tok: int) -> StepOutput:
         req.output_tokens.append(tok)
         if (len(req.output_tokens) >= req.sampling.max_new_tokens

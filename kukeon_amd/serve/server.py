@@ -46,6 +46,7 @@ class _Pending:
     tokens: List[int]
     stream: bool = False
     t0: float = 0.0
+    session: str = ""
 
 
 class ModelhubServer:
@@ -192,6 +193,26 @@ class ModelhubServer:
                                     "n": len(lat)},
                 **self.metrics,
             }
+        if method == "cancel":
+            name = params["session"]
+            done: "queue.Queue" = queue.Queue()
+
+            def _do_cancel():
+                with self._lock:
+                    rids = [rid for rid, p in self._pending.items()
+                            if p.session == name]
+                n = 0
+                for rid in rids:
+                    if self.engine.cancel(rid):
+                        n += 1
+                    with self._lock:
+                        p = self._pending.pop(rid, None)
+                    if p is not None:
+                        p.reply.put({"canceled": True,
+                                     "tokens": p.tokens})
+                done.put({"canceled": n})
+            self._submit.put(_do_cancel)
+            return done.get(timeout=60)
         if method == "release":
             name = params["session"]
             done: "queue.Queue" = queue.Queue()
@@ -246,7 +267,8 @@ class ModelhubServer:
                 return
             with self._lock:
                 self._pending[rid] = _Pending(rid, reply, [], stream=stream,
-                                              t0=time.perf_counter())
+                                              t0=time.perf_counter(),
+                                              session=name)
         self._submit.put(_do_submit)
         while True:
             out = reply.get(timeout=600)
@@ -254,7 +276,14 @@ class ModelhubServer:
                 raise ValueError(out["error"])
             if "delta" in out:
                 if not out.get("_final_stub"):
-                    emit({"delta": out["delta"]})
+                    try:
+                        emit({"delta": out["delta"]})
+                    except OSError:
+                        # streaming client went away: stop decoding for
+                        # it at the next step boundary instead of
+                        # burning GPU to max_new_tokens
+                        self.handle("cancel", {"session": name})
+                        raise
                 continue
             break
         kv = self.sessions.get(name)

@@ -274,3 +274,40 @@ def test_stats_turn_latency_percentiles(hub):
     assert lat["p50"] > 0
     assert lat["p50"] <= lat["p95"] <= lat["p99"]
     c.close()
+
+
+def test_cancel_aborts_inflight_request(hub):
+    """`cancel` ends a session's in-flight generate at the next step
+    boundary: the blocked generate returns what was produced so far with
+    canceled=True, and the session stays usable."""
+    import threading
+
+    h, sock = hub
+    c = ModelhubClient(sock, timeout=120)
+    got = {}
+
+    def long_gen():
+        c2 = ModelhubClient(sock, timeout=120)
+        got["r"] = c2.call("generate", session="c1",
+                           tokens=[1, 2, 3], max_new_tokens=200,
+                           temperature=0.0)
+        c2.close()
+
+    t = threading.Thread(target=long_gen)
+    t.start()
+    # wait until the request is actually running
+    for _ in range(200):
+        if c.call("stats")["running"] or c.call("stats")["waiting"]:
+            break
+        import time as _t
+        _t.sleep(0.02)
+    r = c.call("cancel", session="c1")
+    assert r["canceled"] == 1
+    t.join(30)
+    assert not t.is_alive()
+    assert got["r"].get("canceled") is True
+    assert len(got["r"]["tokens"]) < 200
+    # session still serves
+    r2 = c.generate("c1", [9], max_new_tokens=2, temperature=0.0)
+    assert len(r2["tokens"]) == 2
+    c.close()
