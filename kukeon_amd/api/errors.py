@@ -63,6 +63,10 @@ class AlreadyExists(KukeonError):
     code = "ErrAlreadyExists"
 
 
+class Conflict(KukeonError):
+    code = "ErrConflict"
+
+
 class InvalidArgument(KukeonError):
     code = "ErrInvalidArgument"
 

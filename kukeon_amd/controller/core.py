@@ -1483,13 +1483,7 @@ class Controller:
                     out.append(data)
         return out
 
-    def delete_image(self, name: str) -> None:
-        safe = name.replace("/", "_")
-        if not self.store.delete(self.run_path / "images" / f"{safe}.json"):
-            raise errors.NotFound(f"image {name}")
-
-    def prune_images(self) -> List[str]:
-        """Remove images not referenced by any cell spec."""
+    def _images_in_use(self) -> set:
         used = set()
         for realm in self.store.list_children(self.store.data_root):
             for space in self.store.list_children(self.store.realm_dir(realm)):
@@ -1501,10 +1495,26 @@ class Controller:
                             doc = self.get_cell(realm, space, stack, cell)
                             for c in doc.spec.containers:
                                 used.add(c.image)
+        return used
+
+    def delete_image(self, name: str, force: bool = False) -> None:
+        if not force and name in self._images_in_use():
+            # a cell may be running on this image's overlay lowerdirs;
+            # yanking the layers under a live mount corrupts its rootfs
+            raise errors.Conflict(
+                f"image {name} is referenced by a cell spec; delete the "
+                "cell first (or --force)")
+        safe = name.replace("/", "_")
+        if not self.store.delete(self.run_path / "images" / f"{safe}.json"):
+            raise errors.NotFound(f"image {name}")
+
+    def prune_images(self) -> List[str]:
+        """Remove images not referenced by any cell spec."""
+        used = self._images_in_use()
         removed = []
         for img in self.list_images():
             if img["name"] not in used:
-                self.delete_image(img["name"])
+                self.delete_image(img["name"], force=True)
                 removed.append(img["name"])
         return removed
 

@@ -268,3 +268,26 @@ def test_scheme_roundtrip_and_idempotency():
     assert n["spec"]["containers"][0]["image"] == "busybox"
     assert n["spec"]["containers"][0]["id"] == "main"
     assert n["spec"]["spaceId"] == "default"  # cross-version defaulting
+
+
+def test_delete_image_refuses_while_referenced(tmp_path):
+    from kukeon_amd.api import errors, v1beta1 as api
+    from kukeon_amd.controller.core import Controller
+    from kukeon_amd.runtime.process import FakeRuntime
+
+    ctl = Controller(str(tmp_path / "run"), runtime=FakeRuntime())
+    ctl.bootstrap()
+    ctl.register_image("tool:v1", spec={})
+    doc = api.CellDoc(
+        metadata=api.Metadata(name="imguser"),
+        spec=api.CellSpec(realm_id="default", space_id="default",
+                          stack_id="default",
+                          containers=[api.ContainerSpec(
+                              id="main", image="tool:v1",
+                              command="sleep", args=["5"])]))
+    ctl.create_cell(doc)
+    with pytest.raises(errors.Conflict):
+        ctl.delete_image("tool:v1")
+    ctl.delete_image("tool:v1", force=True)   # explicit override works
+    ctl.register_image("tool:v2", spec={})
+    ctl.delete_image("tool:v2")               # unreferenced: fine
