@@ -718,10 +718,12 @@ def image_get(ctx, name):
 
 @image.command("delete")
 @click.argument("name")
+@click.option("--force", is_flag=True,
+              help="delete even while a cell spec references the image")
 @pass_ctx
-def image_delete(ctx, name):
+def image_delete(ctx, name, force):
     try:
-        ctx._controller().delete_image(name)
+        ctx._controller().delete_image(name, force=force)
         click.echo(f"image {name} deleted")
     except errors.KukeonError as e:
         _die(e)
