@@ -104,6 +104,7 @@ def run_pause(unshare_kinds: str = "", hostname: str = "",
 
 
 class Shim:
+    CAP_LIMIT = 32 << 20   # capture.log rotation threshold
     def __init__(self, cdir: Path):
         self.dir = cdir
         self.spec = json.loads((cdir / "spawn.json").read_text())
@@ -364,7 +365,20 @@ class Shim:
 
         self.child_pid = pid
         self.record_runtime(pid)
-        capture = open(self.dir / "capture.log", "ab", buffering=0)
+        # capture with size-capped rotation: a chatty agent must not
+        # fill the state volume; one .1 generation keeps recent history
+        # for `kuke log` (32 MiB cap per generation)
+        cap_path = self.dir / "capture.log"
+        capture = open(cap_path, "ab", buffering=0)
+        cap_limit = self.CAP_LIMIT
+
+        def cap_write(data: bytes):
+            nonlocal capture
+            capture.write(data)
+            if capture.tell() >= cap_limit:
+                capture.close()
+                os.replace(cap_path, str(cap_path) + ".1")
+                capture = open(cap_path, "ab", buffering=0)
         activity = self.dir / "activity"
         activity.touch()
         clients = []
@@ -403,7 +417,7 @@ class Shim:
                             data = b""
                         if not data:
                             raise EOFError
-                        capture.write(data)
+                        cap_write(data)
                         dead = []
                         for c in clients:
                             try:

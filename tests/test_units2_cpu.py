@@ -291,3 +291,34 @@ def test_delete_image_refuses_while_referenced(tmp_path):
     ctl.delete_image("tool:v1", force=True)   # explicit override works
     ctl.register_image("tool:v2", spec={})
     ctl.delete_image("tool:v2")               # unreferenced: fine
+
+
+def test_capture_log_rotation(tmp_path):
+    """The shim's capture file rotates at its size cap instead of
+    growing unbounded (one .1 generation kept for kuke log)."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    d = tmp_path / "c"
+    (d / "tty").mkdir(parents=True)
+    spec = {"id": "noisy",
+            "argv": ["/bin/sh", "-c",
+                     "i=0; while [ $i -lt 64 ]; do "
+                     "head -c 4096 /dev/zero | tr '\\0' x; i=$((i+1)); "
+                     "done"],
+            "env": [], "attachable": True, "capture": True}
+    (d / "spawn.json").write_text(json.dumps(spec))
+    code = (
+        "import sys; sys.path.insert(0, %r);"
+        "from kukeon_amd.tty import shim as m;"
+        "m.Shim.CAP_LIMIT = 64 * 1024;"
+        "s = m.Shim(__import__('pathlib').Path(%r));"
+        "sys.exit(s.run_attachable())" % (os.getcwd(), str(d)))
+    r = subprocess.run([sys.executable, "-c", code], timeout=60,
+                       capture_output=True, text=True)
+    rot = d / "capture.log.1"
+    assert rot.exists(), (r.returncode, r.stdout[-300:], r.stderr[-300:])
+    assert rot.stat().st_size >= 64 * 1024
+    assert (d / "capture.log").stat().st_size < 1 << 20
