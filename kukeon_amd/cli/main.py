@@ -478,7 +478,11 @@ def log(ctx, name, container, follow, lines):
     path = Path(res["path"])
     if not path.exists():
         _die(errors.NotFound(f"no log at {path}"))
-    data = path.read_bytes().splitlines()[-lines:]
+    raw = path.read_bytes()
+    rot = Path(str(path) + ".1")   # shim rotation keeps one generation
+    if len(raw.splitlines()) < lines and rot.exists():
+        raw = rot.read_bytes() + raw
+    data = raw.splitlines()[-lines:]
     for line in data:
         click.echo(line.decode("utf-8", "replace"))
     if follow:
