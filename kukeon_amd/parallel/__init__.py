@@ -147,4 +147,11 @@ def ep_rank() -> int:
 
 def ep_all_to_all(out: torch.Tensor, inp: torch.Tensor,
                   out_splits, in_splits) -> None:
+    if out.is_cuda and dist.get_backend() == "gloo":
+        # oversubscribed rehearsal (more ranks than GPUs falls back to
+        # gloo, which cannot move CUDA tensors): stage through host
+        h_out = torch.empty_like(out, device="cpu")
+        dist.all_to_all_single(h_out, inp.cpu(), out_splits, in_splits)
+        out.copy_(h_out)
+        return
     dist.all_to_all_single(out, inp, out_splits, in_splits)
