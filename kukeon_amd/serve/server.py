@@ -336,7 +336,14 @@ class ModelhubServer:
         finally:
             os.chdir(cwd)
         self._srv.daemon_threads = True
-        os.chmod(sp, 0o666)
+        # same ownership model as the kukeond socket: kukeon-group 0660
+        # when the group exists, 0666 single-user fallback
+        from kukeon_amd.runtime import sysuser
+        gid = sysuser.lookup_group()
+        if gid is not None:
+            sysuser.apply_socket_group(sp, gid)
+        else:
+            os.chmod(sp, 0o666)
         t = threading.Thread(target=self._srv.serve_forever, daemon=True)
         t.start()
         self._threads.append(t)
