@@ -633,7 +633,7 @@ def test_concurrent_verb_fuzz_no_double_spawn(tmp_path):
         rng = random.Random(seed)
         for _ in range(25):
             verb = rng.choice(["start", "start", "stop", "reconcile",
-                               "kill", "apply"])
+                               "kill", "apply", "restart_c", "stop_c"])
             try:
                 if verb == "start":
                     ctl.start_cell("default", "default", "default", "racy")
@@ -647,6 +647,12 @@ def test_concurrent_verb_fuzz_no_double_spawn(tmp_path):
                                        "racy")
                 elif verb == "apply":
                     ctl.apply_documents(cell_yaml)
+                elif verb == "restart_c":
+                    ctl.restart_container("default", "default", "default",
+                                          "racy", "main")
+                elif verb == "stop_c":
+                    ctl.stop_container("default", "default", "default",
+                                       "racy", "main")
             except errors.KukeonError:
                 pass
             except Exception as e:  # noqa: BLE001
