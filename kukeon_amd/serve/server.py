@@ -11,7 +11,9 @@ interleave across clients — the engine continuously batches them):
       "tokens": [...], "max_new_tokens": 128, "temperature": 0.7,
       "top_k": 50, "top_p": 0.9}}
   -> {"id": N, "result": {"tokens": [...], "context_len": M}}
-  other methods: ping, stats, release (free a session's KV). Passing
+  other methods: ping, stats (KV occupancy + turn-latency
+  percentiles), cancel (abort a session's in-flight turn), release
+  (free a session's KV). Passing
   "stream": true on generate delivers incremental
   {"id": N, "delta": [tok, ...]} frames (one per decode micro-batch)
   ahead of the final result frame — time-to-first-token for agents.
