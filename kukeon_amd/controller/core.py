@@ -449,7 +449,7 @@ class Controller:
                 cursor += c.gpus
                 env = self._container_env(doc, c, mine)
                 env += self._mount_volumes(doc, c, cdir)
-                h = spec_hash(c)
+                h = spec_hash(c, self._image_layers(c.image))
                 prev = self.store.read(cdir / METADATA_FILE) or {}
                 prev_hash = prev.get("metadata", {}).get(
                     "labels", {}).get(SPEC_HASH_LABEL, "")
@@ -679,7 +679,8 @@ class Controller:
                 cell_dir / (c.id or "main") / METADATA_FILE) or {}
             stored = prev.get("metadata", {}).get(
                 "labels", {}).get(SPEC_HASH_LABEL, "")
-            if stored and stored != spec_hash(c):
+            if stored and stored != spec_hash(c,
+                                              self._image_layers(c.image)):
                 return True
         return False
 
@@ -1482,6 +1483,19 @@ class Controller:
                 if data:
                     out.append(data)
         return out
+
+    def _image_layers(self, image: str):
+        """Layer ids when `image` resolves to a built store manifest
+        (content identity for the spawn hash); None for registered-only
+        or external image names."""
+        if not image:
+            return None
+        try:
+            from kukeon_amd.images import ImageStore
+            man = ImageStore(str(self.run_path)).get(image)
+            return man.get("layers") or None
+        except Exception:  # noqa: BLE001  (NotFound / malformed manifest)
+            return None
 
     def _images_in_use(self) -> set:
         used = set()

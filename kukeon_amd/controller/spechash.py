@@ -22,9 +22,14 @@ _NON_SPAWN_FIELDS = (
 )
 
 
-def spec_hash(c: api.ContainerSpec) -> str:
+def spec_hash(c: api.ContainerSpec, image_layers=None) -> str:
     payload = c.to_dict()
     for k in _NON_SPAWN_FIELDS:
         payload.pop(k, None)
+    if image_layers:
+        # image CONTENT drift (reference ctr chainID drift detection,
+        # client.go:142-161): a rebuilt tag changes its layer ids, so a
+        # live container on the old rootfs respawns at the next start
+        payload["_imageLayers"] = list(image_layers)
     blob = json.dumps(payload, sort_keys=True).encode()
     return hashlib.sha256(blob).hexdigest()[:16]

@@ -718,3 +718,41 @@ def test_spec_hash_reuse_and_recreate(ctl):
     key = str(ctl.store.cell_dir("default", "default", "default",
                                  "hashy") / "main")
     assert key in rt.killed
+
+
+def test_image_rebuild_respawns_container(tmp_path):
+    """A rebuilt image tag (new layer ids, same name) drifts the spawn
+    hash: the next start respawns the container onto the new rootfs
+    (reference ctr chainID drift detection)."""
+    from kukeon_amd.images import ImageStore
+
+    rt = FakeRuntime()
+    ctl = Controller(str(tmp_path / "run"), runtime=rt, gpu_devices=[])
+    ctl.bootstrap()
+    store = ImageStore(str(tmp_path / "run"))
+    src = tmp_path / "layer"
+    src.mkdir()
+    (src / "f").write_text("v1")
+    l1 = store.add_layer_from_dir(src)
+    store.put_manifest("app:latest", [l1])
+    doc = make_cell("imgdrift")
+    doc.spec.containers[0].image = "app:latest"
+    ctl.create_cell(doc)
+    ctl.start_cell("default", "default", "default", "imgdrift")
+    cdir = str(ctl.store.cell_dir("default", "default", "default",
+                                  "imgdrift") / "main")
+    spawns1 = rt.started.count(cdir)
+    assert spawns1 == 1
+    # same spec, same image content: start is a no-op (no respawn)
+    ctl.start_cell("default", "default", "default", "imgdrift")
+    assert rt.started.count(cdir) == spawns1
+    assert rt.killed.count(cdir) == 0
+    # rebuild the tag with different content -> new layer id -> respawn
+    (src / "f").write_text("v2")
+    l2 = store.add_layer_from_dir(src)
+    assert l2 != l1
+    store.put_manifest("app:latest", [l2])
+    ctl.start_cell("default", "default", "default", "imgdrift")
+    assert rt.killed.count(cdir) == 1    # old process bounced
+    assert rt.started.count(cdir) == spawns1 + 1
+    ctl.kill_cell("default", "default", "default", "imgdrift")
