@@ -817,6 +817,26 @@ def daemon_stop(ctx):
         _die(e)
 
 
+@daemon.command("recreate")
+@click.option("--model", default="llama-3-8b")
+@click.option("--gpus", type=int, default=1)
+@pass_ctx
+def daemon_recreate(ctx, model, gpus):
+    """Re-provision the system modelhub cell (reference `kuke daemon
+    recreate` re-provisions the containerized kukeond cell via
+    ProvisionKukeondCell, controller.go:253-280 — here the system cell
+    is the inference server)."""
+    from kukeon_amd.controller import naming
+    ctl = ctx._controller()
+    try:
+        ctl.delete_cell(naming.SYSTEM_REALM, naming.SYSTEM_SPACE,
+                        naming.SYSTEM_STACK, "modelhub", force=True)
+    except errors.NotFound:
+        pass
+    doc = ctl.provision_modelhub_cell(model=model, gpus=gpus)
+    click.echo(f"modelhub cell recreated: {doc.status.state}")
+
+
 @daemon.command("status")
 @pass_ctx
 def daemon_status(ctx):

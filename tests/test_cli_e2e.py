@@ -208,3 +208,21 @@ def test_get_api_version_downgrade(tmp_path):
     assert docs[0]["spec"]["realm"] == "default"  # scope key renamed
     assert "realmId" not in docs[0]["spec"]
     assert "dropped in v1alpha1" in r.output  # repos is beta-only
+
+
+def test_daemon_recreate_reprovisions_modelhub(tmp_path):
+    from click.testing import CliRunner
+    from kukeon_amd.cli.main import cli
+
+    runner = CliRunner()
+    rp = str(tmp_path / "run")
+    assert runner.invoke(cli, ["--run-path", rp, "--local",
+                               "init"]).exit_code == 0
+    r = runner.invoke(cli, ["--run-path", rp, "--local", "daemon",
+                            "recreate", "--gpus", "0"])
+    assert r.exit_code == 0, r.output
+    assert "modelhub cell recreated" in r.output
+    # idempotent: a second recreate deletes and re-provisions
+    r2 = runner.invoke(cli, ["--run-path", rp, "--local", "daemon",
+                             "recreate", "--gpus", "0"])
+    assert r2.exit_code == 0, r2.output
