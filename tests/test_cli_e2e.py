@@ -226,3 +226,14 @@ def test_daemon_recreate_reprovisions_modelhub(tmp_path):
     r2 = runner.invoke(cli, ["--run-path", rp, "--local", "daemon",
                              "recreate", "--gpus", "0"])
     assert r2.exit_code == 0, r2.output
+
+
+def test_version_and_autocomplete_smoke():
+    from click.testing import CliRunner
+    from kukeon_amd.cli.main import cli
+
+    runner = CliRunner()
+    r = runner.invoke(cli, ["version"])
+    assert r.exit_code == 0 and "kuke" in r.output
+    r = runner.invoke(cli, ["autocomplete"])
+    assert r.exit_code == 0 and "complete" in r.output.lower()
