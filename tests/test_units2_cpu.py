@@ -322,3 +322,24 @@ def test_capture_log_rotation(tmp_path):
     assert rot.exists(), (r.returncode, r.stdout[-300:], r.stderr[-300:])
     assert rot.stat().st_size >= 64 * 1024
     assert (d / "capture.log").stat().st_size < 1 << 20
+
+
+def test_spec_hash_semantics():
+    """Restart knobs and scope ids never bounce a healthy process;
+    spawn-feeding fields and image layers do."""
+    from kukeon_amd.api import v1beta1 as api
+    from kukeon_amd.controller.spechash import spec_hash
+
+    base = api.ContainerSpec(id="main", image="img", command="run",
+                             args=["-v"], restart_policy="always")
+    h = spec_hash(base)
+    knobs = api.ContainerSpec(id="main", image="img", command="run",
+                              args=["-v"], restart_policy="on-failure",
+                              restart_max_retries=9)
+    assert spec_hash(knobs) == h          # metadata-only knobs
+    argv = api.ContainerSpec(id="main", image="img", command="run",
+                             args=["-x"], restart_policy="always")
+    assert spec_hash(argv) != h           # spawn-feeding field
+    assert spec_hash(base, image_layers=["l1"]) != h
+    assert spec_hash(base, image_layers=["l1"]) != \
+        spec_hash(base, image_layers=["l2"])   # content drift
