@@ -343,3 +343,25 @@ def test_spec_hash_semantics():
     assert spec_hash(base, image_layers=["l1"]) != h
     assert spec_hash(base, image_layers=["l1"]) != \
         spec_hash(base, image_layers=["l2"])   # content drift
+
+
+def test_cell_ipam_allocate_release(tmp_path):
+    """Per-cell host-local IPAM inside the space /24 (ungated twin of
+    the root-only isolation e2e): idempotent allocation, distinct IPs,
+    release-reuse, .1 reserved for the gateway."""
+    from kukeon_amd.controller.subnet import SubnetAllocator
+    from kukeon_amd.state.store import Store
+
+    alloc = SubnetAllocator(Store(str(tmp_path / "run")),
+                            honor_host_routes=False)
+    alloc.allocate("default", "web")
+    gw = alloc.gateway("default", "web")
+    assert gw and gw.endswith(".1")
+    a = alloc.allocate_ip("default", "web", "cell-a")
+    assert alloc.allocate_ip("default", "web", "cell-a") == a  # idempotent
+    b = alloc.allocate_ip("default", "web", "cell-b")
+    assert a != b and a != gw and b != gw
+    assert a.rsplit(".", 1)[0] == gw.rsplit(".", 1)[0]  # same /24
+    alloc.release_ip("default", "web", "cell-a")
+    c = alloc.allocate_ip("default", "web", "cell-c")
+    assert c == a  # released address is reusable
