@@ -365,3 +365,29 @@ def test_cell_ipam_allocate_release(tmp_path):
     alloc.release_ip("default", "web", "cell-a")
     c = alloc.allocate_ip("default", "web", "cell-c")
     assert c == a  # released address is reusable
+
+
+def test_log_formatter_and_naming():
+    """ReformatFormatter emits `ts LEVEL "msg" k=v` lines (reference
+    slog handler contract); naming validates and generates ids."""
+    import logging
+
+    from kukeon_amd.utils.logging import ReformatFormatter
+    from kukeon_amd.controller import naming
+    from kukeon_amd.api import errors as kerrors
+
+    rec = logging.LogRecord("kukeon.cell", logging.INFO, __file__, 1,
+                            'cell "started"', (), None)
+    rec.kv = {"cell": "dev-a1b2c3", "pid": 412}
+    line = ReformatFormatter().format(rec)
+    assert "INFO  " in line and "cell 'started'" in line
+    assert "cell=dev-a1b2c3 pid=412" in line
+    assert line.endswith("logger=kukeon.cell")
+    assert line[4] == "-" and line.split(" ")[0].endswith("Z")
+
+    assert naming.validate_name("web-1") == "web-1"
+    with pytest.raises(kerrors.ValidationError):
+        naming.validate_name("Bad/Name")
+    assert naming.root_container_id("s", "st", "c") == "s-st-c"
+    n = naming.generate_cell_name("dev", taken=set())
+    assert n.startswith("dev-") and len(n) == len("dev-") + 6
