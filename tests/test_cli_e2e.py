@@ -237,3 +237,49 @@ def test_version_and_autocomplete_smoke():
     assert r.exit_code == 0 and "kuke" in r.output
     r = runner.invoke(cli, ["autocomplete"])
     assert r.exit_code == 0 and "complete" in r.output.lower()
+
+
+def test_client_configuration_defaults(tmp_path, monkeypatch):
+    """~/.kuke/kuke.yaml (ClientConfiguration) supplies default scope:
+    a cell created in a non-default realm/space is found by `kuke get`
+    without --realm/--space flags."""
+    import yaml as yamlmod
+    from click.testing import CliRunner
+    from kukeon_amd.cli.main import cli
+
+    runner = CliRunner()
+    rp = str(tmp_path / "run")
+    assert runner.invoke(cli, ["--run-path", rp, "--local",
+                               "init"]).exit_code == 0
+    docs = [
+        {"apiVersion": "v1beta1", "kind": "Realm",
+         "metadata": {"name": "prod"}},
+        {"apiVersion": "v1beta1", "kind": "Space",
+         "metadata": {"name": "web"}, "spec": {"realmId": "prod"}},
+        {"apiVersion": "v1beta1", "kind": "Stack",
+         "metadata": {"name": "app"},
+         "spec": {"realmId": "prod", "spaceId": "web"}},
+        {"apiVersion": "v1beta1", "kind": "Cell",
+         "metadata": {"name": "scoped"},
+         "spec": {"realmId": "prod", "spaceId": "web", "stackId": "app",
+                  "containers": [{"id": "main", "command": "sleep",
+                                  "args": ["5"]}]}},
+    ]
+    f = tmp_path / "m.yaml"
+    f.write_text(yamlmod.safe_dump_all(docs))
+    assert runner.invoke(cli, ["--run-path", rp, "--local", "apply",
+                               "-f", str(f)]).exit_code == 0
+    cfgp = tmp_path / "kuke.yaml"
+    cfgp.write_text(yamlmod.safe_dump({
+        "apiVersion": "v1beta1", "kind": "ClientConfiguration",
+        "spec": {"defaultRealm": "prod", "defaultSpace": "web",
+                 "defaultStack": "app"}}))
+    monkeypatch.setenv("KUKE_CONFIG", str(cfgp))
+    r = runner.invoke(cli, ["--run-path", rp, "--local", "get", "cell",
+                            "scoped"])
+    assert r.exit_code == 0 and "scoped" in r.output, r.output
+    # without the config the default scope misses it
+    monkeypatch.setenv("KUKE_CONFIG", str(tmp_path / "none.yaml"))
+    r2 = runner.invoke(cli, ["--run-path", rp, "--local", "get", "cell",
+                             "scoped"])
+    assert r2.exit_code != 0
