@@ -391,3 +391,13 @@ def test_log_formatter_and_naming():
     assert naming.root_container_id("s", "st", "c") == "s-st-c"
     n = naming.generate_cell_name("dev", taken=set())
     assert n.startswith("dev-") and len(n) == len("dev-") + 6
+
+
+def test_scheme_session_has_no_alpha_representation():
+    from kukeon_amd.api import scheme
+
+    sess = {"apiVersion": "v1beta1", "kind": "Session",
+            "metadata": {"name": "s1"},
+            "spec": {"realmId": "r", "spaceId": "s", "stackId": "st"}}
+    wire, lost = scheme.to_wire(sess, "v1alpha1")
+    assert any("Session" in f for f in lost)
