@@ -401,3 +401,15 @@ def test_scheme_session_has_no_alpha_representation():
             "spec": {"realmId": "r", "spaceId": "s", "stackId": "st"}}
     wire, lost = scheme.to_wire(sess, "v1alpha1")
     assert any("Session" in f for f in lost)
+
+
+def test_instance_pinning_mismatch_fails_fast(tmp_path):
+    """`.kukeon-instance.json` pins daemon identity: a restart with
+    different knobs refuses to start (reference internal/instance)."""
+    from kukeon_amd.api import errors as kerrors
+    from kukeon_amd.daemon.server import verify_or_write_instance
+
+    verify_or_write_instance(tmp_path, cgroup_root="kukeon")
+    verify_or_write_instance(tmp_path, cgroup_root="kukeon")  # same: ok
+    with pytest.raises(kerrors.InvalidArgument):
+        verify_or_write_instance(tmp_path, cgroup_root="other")
