@@ -756,3 +756,17 @@ def test_image_rebuild_respawns_container(tmp_path):
     assert rt.killed.count(cdir) == 1    # old process bounced
     assert rt.started.count(cdir) == spawns1 + 1
     ctl.kill_cell("default", "default", "default", "imgdrift")
+
+
+def test_refresh_all_rederives_statuses(ctl):
+    """refresh_all walks every cell and re-derives status from live
+    probes (the `kuke refresh` backing verb)."""
+    ctl.create_cell(make_cell("fresh1"))
+    ctl.create_cell(make_cell("fresh2"))
+    ctl.start_cell("default", "default", "default", "fresh1")
+    n = ctl.refresh_all()
+    assert n["cells"] >= 2
+    d1 = ctl.get_cell("default", "default", "default", "fresh1")
+    d2 = ctl.get_cell("default", "default", "default", "fresh2")
+    assert d1.status.state == api.STATE_READY
+    assert d2.status.state in (api.STATE_PENDING, "")
