@@ -283,3 +283,20 @@ def test_client_configuration_defaults(tmp_path, monkeypatch):
     r2 = runner.invoke(cli, ["--run-path", rp, "--local", "get", "cell",
                              "scoped"])
     assert r2.exit_code != 0
+
+
+def test_uninstall_removes_state(tmp_path):
+    """kuke uninstall: cascade-deletes realms and removes the run tree
+    (reference internal/controller/uninstall.go)."""
+    from click.testing import CliRunner
+    from kukeon_amd.cli.main import cli
+
+    runner = CliRunner()
+    rp = str(tmp_path / "run")
+    assert runner.invoke(cli, ["--run-path", rp, "--local",
+                               "init"]).exit_code == 0
+    assert (Path(rp) / "data").is_dir()
+    r = runner.invoke(cli, ["--run-path", rp, "--local", "uninstall",
+                            "--yes"])
+    assert r.exit_code == 0, r.output
+    assert not (Path(rp) / "data").exists()
