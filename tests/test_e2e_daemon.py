@@ -429,3 +429,39 @@ def test_scm_rights_fd_attach(harness):
     assert b"post-43" in Path(res["path"]).read_bytes()
     client.KillCell(realm="default", space="default", stack="default",
                     name="fdcell")
+
+
+def test_run_from_blueprint_rpc(harness):
+    """RunFromBlueprint over the wire: materialize + start + lineage
+    labels (the `kuke run -b` backing verb)."""
+    ctl, srv, client = harness
+    bp = {
+        "apiVersion": "v1beta1", "kind": "CellBlueprint",
+        "metadata": {"name": "bp-sleep"},
+        "spec": {
+            "realmId": "default", "spaceId": "default",
+            "namePrefix": "bp",
+            "params": [{"name": "SECS", "default": "30"}],
+            "template": {
+                "kind": "Cell",
+                "spec": {"realmId": "default", "spaceId": "default",
+                         "stackId": "default",
+                         "containers": [{"id": "main",
+                                         "command": "sleep",
+                                         "args": ["${SECS}"]}]}},
+        },
+    }
+    client.PutBlueprint(doc=bp)
+    cell = client.RunFromBlueprint(realm="default", space="default",
+                                   stack="default", blueprint="bp-sleep",
+                                   params={"SECS": "60"}, env=[], name="")
+    name = cell["metadata"]["name"]
+    assert name.startswith("bp-")
+    assert cell["status"]["state"] == "Ready"
+    got = client.GetCell(realm="default", space="default",
+                         stack="default", name=name)
+    assert got["spec"]["containers"][0]["args"] == ["60"]
+    assert got["metadata"]["labels"].get("kukeon.io/blueprint") == \
+        "bp-sleep"
+    client.KillCell(realm="default", space="default", stack="default",
+                    name=name)
