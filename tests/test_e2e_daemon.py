@@ -465,3 +465,41 @@ def test_run_from_blueprint_rpc(harness):
         "bp-sleep"
     client.KillCell(realm="default", space="default", stack="default",
                     name=name)
+
+
+def test_run_from_config_rpc(harness):
+    """RunFromConfig over the wire: blueprint binding + values overlay +
+    provenance stamping (the `kuke run -c` backing verb)."""
+    ctl, srv, client = harness
+    client.PutBlueprint(doc={
+        "apiVersion": "v1beta1", "kind": "CellBlueprint",
+        "metadata": {"name": "bp-c"},
+        "spec": {"realmId": "default", "spaceId": "default",
+                 "namePrefix": "cfged",
+                 "params": [{"name": "SECS", "default": "30"}],
+                 "template": {"kind": "Cell",
+                              "spec": {"realmId": "default",
+                                       "spaceId": "default",
+                                       "stackId": "default",
+                                       "containers": [
+                                           {"id": "main",
+                                            "command": "sleep",
+                                            "args": ["${SECS}"]}]}}}})
+    client.PutConfig(doc={
+        "apiVersion": "v1beta1", "kind": "CellConfig",
+        "metadata": {"name": "cfg-fast"},
+        "spec": {"realmId": "default", "spaceId": "default",
+                 "blueprint": "bp-c", "values": {"SECS": "45"}}})
+    cell = client.RunFromConfig(realm="default", space="default",
+                                stack="default", config="cfg-fast",
+                                params={}, name="")
+    name = cell["metadata"]["name"]
+    got = client.GetCell(realm="default", space="default",
+                         stack="default", name=name)
+    assert got["spec"]["containers"][0]["args"] == ["45"]
+    prov = got["spec"].get("provenance", {})
+    assert (prov.get("bindingKind"), prov.get("bindingRef")) == \
+        ("config", "cfg-fast")
+    assert prov.get("params", {}).get("SECS") == "45"
+    client.KillCell(realm="default", space="default", stack="default",
+                    name=name)
