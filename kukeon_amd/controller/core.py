@@ -18,6 +18,7 @@ from __future__ import annotations
 import base64
 import contextlib
 import logging
+import os
 import shutil
 import time
 from dataclasses import dataclass
@@ -641,6 +642,27 @@ class Controller:
                 except errors.NotFound:
                     log.warning("cell %s: volume %r not found; skipped",
                                 doc.metadata.name, vm.name)
+            # file-channel secrets (ContainerSecret.path — reference
+            # ctr secret file injection): material is written 0600 under
+            # the container dir and bind-mounted read-only at the
+            # declared path inside the container's mount namespace
+            for sref in c.secrets:
+                if not sref.path:
+                    continue
+                try:
+                    sec = self.get_secret(doc.spec.realm_id,
+                                          doc.spec.space_id, sref.name)
+                except errors.NotFound:
+                    log.warning("cell %s: secret %r not found; skipped",
+                                doc.metadata.name, sref.name)
+                    continue
+                sdir = cell_dir / (c.id or "main") / "secrets"
+                sdir.mkdir(parents=True, exist_ok=True)
+                sfile = sdir / sref.name
+                vals = [_b64maybe(v) for v in sec.spec.data.values()]
+                sfile.write_text("\n".join(vals))
+                os.chmod(sfile, 0o600)
+                binds.append({"src": str(sfile), "dst": sref.path})
             if binds:
                 cfg["binds"] = binds
             # built layered image -> overlay rootfs mounted by the shim

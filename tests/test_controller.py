@@ -770,3 +770,29 @@ def test_refresh_all_rederives_statuses(ctl):
     d2 = ctl.get_cell("default", "default", "default", "fresh2")
     assert d1.status.state == api.STATE_READY
     assert d2.status.state in (api.STATE_PENDING, "")
+
+
+def test_secret_file_channel_binds(ctl):
+    """ContainerSecret.path: material lands 0600 under the container
+    dir and is bind-listed at the declared in-container path
+    (reference ctr secret file injection)."""
+    import os as _os
+
+    ctl.put_secret(api.SecretDoc(
+        metadata=api.Metadata(name="tok"),
+        spec=api.SecretSpec(realm_id="default", space_id="default",
+                            data={"value": "s3cr3t"})))
+    doc = make_cell("filesec")
+    doc.spec.containers[0].secrets = [
+        api.ContainerSecret(name="tok", path="/run/secrets/tok")]
+    ctl.create_cell(doc)
+    cell_dir = ctl.store.cell_dir("default", "default", "default",
+                                  "filesec")
+    cfg = ctl._container_ns_config(doc, doc.spec.containers[0],
+                                   cell_dir, None)
+    binds = (cfg or {}).get("binds", [])
+    entry = [b for b in binds if b["dst"] == "/run/secrets/tok"]
+    assert entry, cfg
+    sfile = cell_dir / "main" / "secrets" / "tok"
+    assert sfile.read_text() == "s3cr3t"
+    assert _os.stat(sfile).st_mode & 0o777 == 0o600
