@@ -665,6 +665,8 @@ class Controller:
                 binds.append({"src": str(sfile), "dst": sref.path})
             if binds:
                 cfg["binds"] = binds
+            if c.read_only_root_filesystem:
+                cfg["readOnlyRootfs"] = True
             # built layered image -> overlay rootfs mounted by the shim
             # in ITS mount namespace (reference: the OCI rootfs the
             # containerd snapshotter provides)

@@ -28,6 +28,8 @@ CLONE_NEWPID = 0x20000000
 
 MS_BIND = 4096
 MS_REC = 16384
+MS_RDONLY = 1
+MS_REMOUNT = 32
 MS_PRIVATE = 1 << 18
 
 _libc = ctypes.CDLL(ctypes.util.find_library("c") or "libc.so.6",
