@@ -145,6 +145,7 @@ class ProcessRuntime(Runtime):
                       for r in (spec.repos or [])],
             "git": ({"name": spec.git.name, "email": spec.git.email}
                     if spec.git else {}),
+            "user": spec.user or "",
             "ns": ns or {},
         }
         # clear stale exit state from a previous run

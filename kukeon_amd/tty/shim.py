@@ -309,6 +309,39 @@ class Shim:
         self.record_status(rc if rc >= 0 else 128 - rc)
         return 0
 
+    @staticmethod
+    def _drop_privileges(user: str) -> None:
+        """ContainerSpec.user (reference OCI Process.user): numeric
+        "uid[:gid]" or a passwd name; the workload execs with the
+        dropped identity. Root-only — a non-root shim (or unknown user)
+        records the degrade on stderr and continues."""
+        if not user or os.geteuid() != 0:
+            return
+        try:
+            import pwd as _pwd
+            uid_s, _, gid_s = user.partition(":")
+            if uid_s.isdigit():
+                uid = int(uid_s)
+                gid = int(gid_s) if gid_s.isdigit() else uid
+                try:
+                    name = _pwd.getpwuid(uid).pw_name
+                except KeyError:
+                    name = None
+            else:
+                ent = _pwd.getpwnam(uid_s)
+                uid, gid, name = ent.pw_uid, ent.pw_gid, ent.pw_name
+                if gid_s.isdigit():
+                    gid = int(gid_s)
+            if name is not None:
+                import grp as _grp  # noqa: F401  (ensure module loads)
+                os.initgroups(name, gid)
+            else:
+                os.setgroups([gid])
+            os.setgid(gid)
+            os.setuid(uid)
+        except (OSError, KeyError) as e:
+            sys.stderr.write(f"user {user!r}: {e}; running as root\n")
+
     def _exec_child(self):
         spec = self.spec
         ns = spec.get("ns") or {}
@@ -325,6 +358,7 @@ class Shim:
         argv = spec["argv"]
         if argv == ["/bin/sh"] and ns.get("imageCmd"):
             argv = ["/bin/sh", "-c", ns["imageCmd"]]
+        self._drop_privileges(spec.get("user") or "")
         try:
             os.execvpe(argv[0], argv, self.child_env())
         except OSError as e:

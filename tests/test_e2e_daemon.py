@@ -503,3 +503,37 @@ def test_run_from_config_rpc(harness):
     assert prov.get("params", {}).get("SECS") == "45"
     client.KillCell(realm="default", space="default", stack="default",
                     name=name)
+
+
+def test_container_user_privilege_drop(harness):
+    """ContainerSpec.user: the workload execs with the dropped identity
+    (root-only enforcement; reference OCI Process.user)."""
+    import os as _os
+
+    if _os.geteuid() != 0:
+        pytest.skip("needs root to drop privileges")
+    ctl, srv, client = harness
+    import tempfile
+    out = Path(tempfile.mkdtemp(prefix="kuke-uid-"))
+    _os.chmod(out, 0o1777)   # the dropped user must be able to write
+    doc = {
+        "apiVersion": "v1beta1", "kind": "Cell",
+        "metadata": {"name": "dropped"},
+        "spec": {"realmId": "default", "spaceId": "default",
+                 "stackId": "default",
+                 "containers": [{"id": "main", "command": "sh",
+                                 "user": "65534",   # nobody
+                                 "args": ["-c",
+                                          f"id -u > {out}/uid.txt; "
+                                          "sleep 30"]}]},
+    }
+    client.CreateCell(doc=doc)
+    client.StartCell(realm="default", space="default", stack="default",
+                     name="dropped")
+    deadline = time.monotonic() + 10
+    while not (out / "uid.txt").exists() and time.monotonic() < deadline:
+        time.sleep(0.05)
+    time.sleep(0.1)
+    assert (out / "uid.txt").read_text().strip() == "65534"
+    client.KillCell(realm="default", space="default", stack="default",
+                    name="dropped")
