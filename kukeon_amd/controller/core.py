@@ -467,6 +467,22 @@ class Controller:
                     self.runtime.kill(cdir)
                     probe = self.runtime.probe(cdir)
                 if not probe.running:
+                    # space-defaults inheritance (reference space.go:83-107
+                    # container-isolation defaults): fields the container
+                    # leaves unset inherit from the space before the spawn
+                    space = self.get_space(doc.spec.realm_id,
+                                           doc.spec.space_id)
+                    sd = (space.spec.defaults.container
+                          if space.spec.defaults else None)
+                    if sd:
+                        if sd.user and not c.user:
+                            c = api.ContainerSpec.from_dict(
+                                dict(c.to_dict(), user=sd.user))
+                        if (sd.read_only_root_filesystem
+                                and not c.read_only_root_filesystem):
+                            c = api.ContainerSpec.from_dict(dict(
+                                c.to_dict(),
+                                readOnlyRootFilesystem=True))
                     cns = self._container_ns_config(doc, c, cell_dir,
                                                     net_info)
                     self.runtime.start_container(cdir, c, env, cg, cns)

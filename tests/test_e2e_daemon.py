@@ -537,3 +537,49 @@ def test_container_user_privilege_drop(harness):
     assert (out / "uid.txt").read_text().strip() == "65534"
     client.KillCell(realm="default", space="default", stack="default",
                     name="dropped")
+
+
+def test_space_defaults_user_inheritance(harness):
+    """SpaceDefaults.container.user applies to containers that leave
+    `user` unset (reference space.go:83-107 isolation inheritance)."""
+    import os as _os
+
+    if _os.geteuid() != 0:
+        pytest.skip("needs root to drop privileges")
+    ctl, srv, client = harness
+    client.ApplyDocuments(yaml="""
+apiVersion: v1beta1
+kind: Space
+metadata: {name: lowpriv}
+spec:
+  realmId: default
+  defaults:
+    container: {user: "65534"}
+""")
+    client.ApplyDocuments(yaml="""
+apiVersion: v1beta1
+kind: Stack
+metadata: {name: s}
+spec: {realmId: default, spaceId: lowpriv}
+""")
+    import tempfile
+    out = Path(tempfile.mkdtemp(prefix="kuke-sdu-"))
+    _os.chmod(out, 0o1777)
+    client.CreateCell(doc={
+        "apiVersion": "v1beta1", "kind": "Cell",
+        "metadata": {"name": "inh"},
+        "spec": {"realmId": "default", "spaceId": "lowpriv",
+                 "stackId": "s",
+                 "containers": [{"id": "main", "command": "sh",
+                                 "args": ["-c",
+                                          f"id -u > {out}/uid.txt; "
+                                          "sleep 30"]}]}})
+    client.StartCell(realm="default", space="lowpriv", stack="s",
+                     name="inh")
+    deadline = time.monotonic() + 10
+    while not (out / "uid.txt").exists() and time.monotonic() < deadline:
+        time.sleep(0.05)
+    time.sleep(0.1)
+    assert (out / "uid.txt").read_text().strip() == "65534"
+    client.KillCell(realm="default", space="lowpriv", stack="s",
+                    name="inh")
