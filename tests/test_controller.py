@@ -796,3 +796,40 @@ def test_secret_file_channel_binds(ctl):
     sfile = cell_dir / "main" / "secrets" / "tok"
     assert sfile.read_text() == "s3cr3t"
     assert _os.stat(sfile).st_mode & 0o777 == 0o600
+
+
+def test_session_idle_timeout_enforced(tmp_path):
+    """Session idleTimeout: no shim activity past the window closes
+    the session (frozen clock; activity mtime is the signal)."""
+    import os as _os
+
+    rt = FakeRuntime()
+    ctl = Controller(str(tmp_path / "run"), runtime=rt, gpu_devices=[])
+    ctl.bootstrap()
+    now = [1_000_000.0]
+    ctl.now = lambda: now[0]
+    ctl.create_stack(api.StackDoc(
+        metadata=api.Metadata(name="sess-idle"),
+        spec=api.StackSpec(realm_id="default", space_id="default")))
+    sdoc = api.SessionDoc(
+        metadata=api.Metadata(name="idler"),
+        spec=api.SessionSpec(realm_id="default", space_id="default",
+                             stack_id="sess-idle",
+                             lifetime=api.SessionLifetime(
+                                 idle_timeout="5m")))
+    ctl.create_session(sdoc)
+    # fresh activity inside the stack keeps it alive
+    adir = (ctl.store.stack_dir("default", "default", "sess-idle") /
+            "cellx" / "main")
+    adir.mkdir(parents=True)
+    act = adir / "activity"
+    act.touch()
+    _os.utime(act, (now[0] - 60, now[0] - 60))
+    ctl.reconcile_sessions()
+    got = ctl.get_session("default", "default", "sess-idle", "idler")
+    assert got.status.state == api.STATE_RUNNING
+    # 10 minutes later with no new activity -> terminated
+    now[0] += 600
+    ctl.reconcile_sessions()
+    got = ctl.get_session("default", "default", "sess-idle", "idler")
+    assert got.status.state == api.STATE_TERMINATED
